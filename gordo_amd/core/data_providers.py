@@ -1,0 +1,148 @@
+"""
+Data providers — sources of raw per-tag timeseries.
+
+Spec: gordo_core data providers as used by the reference
+(``RandomDataProvider`` in tests/conftest.py; provider configured via
+``dataset.data_provider.type``). Providers here are synthetic (there is
+no external data store in this environment): random walks and sine
+waves, deterministic per (tag, seed).
+"""
+from __future__ import annotations
+
+import hashlib
+from typing import Iterable, List, Optional, Dict, Any
+
+import numpy as np
+import pandas as pd
+
+from .sensor_tag import SensorTag, normalize_sensor_tag
+from .import_utils import import_location
+
+
+class DataProvider:
+    """Base data provider: serves one pd.Series per SensorTag."""
+
+    def __init__(self, **kwargs):
+        self._params = dict(kwargs)
+
+    @classmethod
+    def from_dict(cls, config: Dict[str, Any]) -> "DataProvider":
+        config = dict(config or {})
+        kind = config.pop("type", None)
+        if kind is None:
+            return RandomDataProvider(**config)
+        if "." in kind:
+            provider_cls = import_location(kind)
+        else:
+            provider_cls = _PROVIDER_REGISTRY.get(kind)
+            if provider_cls is None:
+                from .exceptions import NoSuitableDataProviderError
+
+                raise NoSuitableDataProviderError(
+                    f"No data provider named {kind!r}"
+                )
+        return provider_cls(**config)
+
+    def to_dict(self) -> Dict[str, Any]:
+        d = dict(self._params)
+        d["type"] = type(self).__name__
+        return d
+
+    def can_handle_tag(self, tag: SensorTag) -> bool:
+        return True
+
+    def load_series(
+        self,
+        train_start_date: pd.Timestamp,
+        train_end_date: pd.Timestamp,
+        tag_list: Iterable[SensorTag],
+        resolution: str = "10T",
+    ) -> Iterable[pd.Series]:
+        raise NotImplementedError()
+
+    def __repr__(self):
+        return f"{type(self).__name__}({self._params!r})"
+
+
+def _tag_seed(tag: SensorTag, salt: int) -> int:
+    h = hashlib.sha256(f"{tag.name}:{salt}".encode()).digest()
+    return int.from_bytes(h[:4], "little")
+
+
+def _time_index(start, end, resolution) -> pd.DatetimeIndex:
+    return pd.date_range(start=start, end=end, freq=resolution, inclusive="left")
+
+
+class RandomDataProvider(DataProvider):
+    """Deterministic random-walk series per tag (test/synthetic data)."""
+
+    def __init__(self, min_size: int = 100, max_size: int = 300, **kwargs):
+        super().__init__(min_size=min_size, max_size=max_size, **kwargs)
+        self.min_size = min_size
+        self.max_size = max_size
+
+    def load_series(self, train_start_date, train_end_date, tag_list, resolution="10T"):
+        index = _time_index(train_start_date, train_end_date, resolution)
+        for tag in tag_list:
+            rng = np.random.default_rng(_tag_seed(tag, 0))
+            values = rng.standard_normal(len(index)).cumsum() * 0.1 + rng.uniform(
+                -1.0, 1.0
+            )
+            yield pd.Series(values, index=index, name=tag.name)
+
+
+class SineWaveDataProvider(DataProvider):
+    """
+    Synthetic sine-wave tags (the benchmark data source — BASELINE.md:
+    "synthetic sine-wave tag data"). Each tag gets a deterministic
+    frequency/phase/noise from its name.
+    """
+
+    def __init__(self, noise: float = 0.02, **kwargs):
+        super().__init__(noise=noise, **kwargs)
+        self.noise = noise
+
+    def load_series(self, train_start_date, train_end_date, tag_list, resolution="10T"):
+        index = _time_index(train_start_date, train_end_date, resolution)
+        t = np.arange(len(index), dtype=np.float64)
+        for tag in tag_list:
+            rng = np.random.default_rng(_tag_seed(tag, 1))
+            freq = rng.uniform(0.002, 0.05)
+            phase = rng.uniform(0, 2 * np.pi)
+            amp = rng.uniform(0.5, 2.0)
+            offset = rng.uniform(-1.0, 1.0)
+            values = (
+                amp * np.sin(2 * np.pi * freq * t + phase)
+                + offset
+                + rng.standard_normal(len(index)) * self.noise
+            )
+            yield pd.Series(values, index=index, name=tag.name)
+
+
+class InfluxDataProvider(DataProvider):
+    """
+    Placeholder for the reference's InfluxDB-backed provider. There is
+    no network in this environment; constructing it succeeds (configs
+    parse) but loading data raises.
+    """
+
+    def load_series(self, train_start_date, train_end_date, tag_list, resolution="10T"):
+        raise RuntimeError(
+            "InfluxDataProvider has no reachable InfluxDB in this environment"
+        )
+
+
+_PROVIDER_REGISTRY: Dict[str, type] = {
+    "RandomDataProvider": RandomDataProvider,
+    "SineWaveDataProvider": SineWaveDataProvider,
+    "InfluxDataProvider": InfluxDataProvider,
+    "DataLakeProvider": RandomDataProvider,  # alias: no data lake here
+}
+
+
+def load_data_provider(config: Optional[Dict[str, Any]]) -> DataProvider:
+    if config is None:
+        return RandomDataProvider()
+    if isinstance(config, DataProvider):
+        return config
+    return DataProvider.from_dict(config)

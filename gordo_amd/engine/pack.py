@@ -1,0 +1,500 @@
+"""
+Packed many-model trainers.
+
+The MI355X-native answer to the reference's pod-per-model fan-out
+(SURVEY.md §2.4): G same-architecture models train in lockstep as ONE
+set of grouped tensors — weights ``[G, in, out]``, activations
+``[G, B, F]`` — so every layer of every model in the pack is a single
+grouped MFMA GEMM launch instead of thousands of tiny ones. Per-model
+data, per-model losses, per-model Adam state; one flat fp32 parameter
+buffer (+ bf16 compute mirror on GPU) per pack so the optimizer is one
+fused kernel over the whole fleet shard.
+
+Training math matches Keras defaults (glorot-uniform kernels,
+orthogonal LSTM recurrent kernels, unit forget-gate bias, Adam with
+bias correction, MSE loss) — reference factories:
+gordo/machine/model/factories/feedforward_autoencoder.py,
+lstm_autoencoder.py.
+"""
+from __future__ import annotations
+
+import logging
+import math
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+
+from .. import ops
+from .spec import LayerSpec, ModelSpec
+
+logger = logging.getLogger(__name__)
+
+
+def _compute_dtype(device: torch.device) -> torch.dtype:
+    return torch.bfloat16 if device.type == "cuda" else torch.float32
+
+
+def _glorot_uniform(shape, gen):
+    fan_in, fan_out = shape[-2], shape[-1]
+    limit = math.sqrt(6.0 / (fan_in + fan_out))
+    return (torch.rand(shape, generator=gen, dtype=torch.float32) * 2 - 1) * limit
+
+
+def _orthogonal(shape, gen):
+    rows, cols = shape[-2], shape[-1]
+    a = torch.randn((max(rows, cols), min(rows, cols)), generator=gen)
+    q, r = torch.linalg.qr(a)
+    q = q * torch.sign(torch.diagonal(r))
+    if rows < cols:
+        q = q.T
+    return q[:rows, :cols].contiguous()
+
+
+class _ParamStore:
+    """One flat fp32 master buffer + grads + Adam state (+ bf16 compute
+    mirror on GPU), carved into named [G, ...] views."""
+
+    def __init__(self, device: torch.device):
+        self.device = device
+        self.compute_dtype = _compute_dtype(device)
+        self._shapes: List[Tuple[str, Tuple[int, ...]]] = []
+        self._offsets: Dict[str, Tuple[int, int]] = {}
+        self._total = 0
+        self.views: Dict[str, torch.Tensor] = {}
+        self.cviews: Dict[str, torch.Tensor] = {}
+        self.gviews: Dict[str, torch.Tensor] = {}
+        self.step_count = 0
+
+    def declare(self, name: str, shape: Tuple[int, ...]):
+        n = int(np.prod(shape))
+        self._offsets[name] = (self._total, n)
+        self._shapes.append((name, shape))
+        self._total += n
+
+    def allocate(self):
+        dev = self.device
+        self.p32 = torch.zeros(self._total, dtype=torch.float32, device=dev)
+        self.g32 = torch.zeros(self._total, dtype=torch.float32, device=dev)
+        self.m = torch.zeros(self._total, dtype=torch.float32, device=dev)
+        self.v = torch.zeros(self._total, dtype=torch.float32, device=dev)
+        self.plp = (
+            torch.zeros(self._total, dtype=self.compute_dtype, device=dev)
+            if self.compute_dtype != torch.float32
+            else None
+        )
+        for name, shape in self._shapes:
+            off, n = self._offsets[name]
+            self.views[name] = self.p32[off : off + n].view(*shape)
+            self.gviews[name] = self.g32[off : off + n].view(*shape)
+            self.cviews[name] = (
+                self.plp[off : off + n].view(*shape)
+                if self.plp is not None
+                else self.views[name]
+            )
+
+    def sync_lp(self):
+        if self.plp is not None:
+            self.plp.copy_(self.p32)
+
+    def zero_grad(self):
+        self.g32.zero_()
+
+    def adam_step(self, lr, beta1, beta2, eps):
+        self.step_count += 1
+        ops.adam_step(
+            self.p32, self.g32, self.m, self.v,
+            lr, beta1, beta2, eps, self.step_count, self.plp,
+        )
+
+    def reset_adam(self):
+        self.m.zero_()
+        self.v.zero_()
+        self.step_count = 0
+
+
+class BasePack:
+    def __init__(self, spec: ModelSpec, G: int, device=None, seeds=None):
+        self.spec = spec
+        self.G = G
+        self.device = torch.device(
+            device if device is not None else
+            ("cuda" if torch.cuda.is_available() else "cpu")
+        )
+        self.store = _ParamStore(self.device)
+        self.seeds = list(seeds) if seeds is not None else list(range(G))
+        assert len(self.seeds) == G
+        self._declare_params()
+        self.store.allocate()
+        self._init_weights()
+        self.store.sync_lp()
+
+    # -- interface -------------------------------------------------------
+    def _declare_params(self):
+        raise NotImplementedError
+
+    def _init_weights(self):
+        raise NotImplementedError
+
+    def predict(self, X: torch.Tensor) -> torch.Tensor:
+        raise NotImplementedError
+
+    def train_batch(self, Xb, Tb) -> torch.Tensor:
+        raise NotImplementedError
+
+    # -- common ----------------------------------------------------------
+    @property
+    def compute_dtype(self):
+        return self.store.compute_dtype
+
+    def _to_compute(self, t: torch.Tensor) -> torch.Tensor:
+        return t.to(device=self.device, dtype=self.compute_dtype)
+
+    def param_names(self) -> List[str]:
+        return [name for name, _ in self.store._shapes]
+
+    def state_for_model(self, g: int) -> Dict[str, np.ndarray]:
+        return {
+            name: self.store.views[name][g].detach().cpu().numpy().copy()
+            for name in self.param_names()
+        }
+
+    def load_model_state(self, g: int, state: Dict[str, np.ndarray]):
+        for name, arr in state.items():
+            self.store.views[name][g].copy_(torch.from_numpy(np.asarray(arr)))
+        self.store.sync_lp()
+
+    def optimizer_state_for_model(self, g: int) -> Dict[str, np.ndarray]:
+        out = {}
+        for name in self.param_names():
+            off, n = self.store._offsets[name]
+            shape = self.store.views[name].shape
+            out["m:" + name] = (
+                self.store.m[off : off + n].view(*shape)[g].cpu().numpy().copy()
+            )
+            out["v:" + name] = (
+                self.store.v[off : off + n].view(*shape)[g].cpu().numpy().copy()
+            )
+        return out
+
+    # ---- the epoch loop ------------------------------------------------
+    def fit(
+        self,
+        X: torch.Tensor,
+        Y: torch.Tensor,
+        epochs: int = 1,
+        batch_size: int = 256,
+        shuffle: bool = True,
+        verbose: int = 0,
+        **_,
+    ) -> Dict[str, list]:
+        """
+        Train the whole pack. X: [G, N, F_in] (already on device),
+        Y: [G, N, F_out]. Returns a Keras-style history dict with
+        per-model per-epoch values: {"loss": [[...G...] per epoch],
+        "accuracy": ...}.
+        """
+        G, N = X.shape[0], X.shape[1]
+        adam = self.spec.adam_params
+        history: Dict[str, list] = {"loss": [], "accuracy": []}
+        n_batches = max(1, math.ceil(self._n_samples(N) / batch_size))
+        gens = [torch.Generator().manual_seed(int(s) & 0x7FFFFFFF) for s in self.seeds]
+        for epoch in range(epochs):
+            if shuffle:
+                perm = torch.stack(
+                    [torch.randperm(self._n_samples(N), generator=g) for g in gens]
+                ).to(self.device)
+            else:
+                perm = (
+                    torch.arange(self._n_samples(N), device=self.device)
+                    .unsqueeze(0)
+                    .expand(G, -1)
+                )
+            epoch_loss = torch.zeros(G, dtype=torch.float32, device=self.device)
+            samples_seen = 0
+            for b in range(n_batches):
+                idx = perm[:, b * batch_size : (b + 1) * batch_size]
+                if idx.shape[1] == 0:
+                    continue
+                Xb, Tb = self._gather_batch(X, Y, idx)
+                loss = self.train_batch(Xb, Tb)
+                epoch_loss += loss * idx.shape[1]
+                samples_seen += idx.shape[1]
+            epoch_loss = (epoch_loss / max(samples_seen, 1)).cpu().tolist()
+            history["loss"].append(epoch_loss)
+            history["accuracy"].append([0.0] * G)
+            if verbose:
+                logger.info(
+                    "epoch %d/%d mean-loss=%.6g", epoch + 1, epochs,
+                    float(np.mean(epoch_loss)),
+                )
+        return history
+
+    def _n_samples(self, N: int) -> int:
+        return N
+
+    def _gather_batch(self, X, Y, idx):
+        G, F = X.shape[0], X.shape[2]
+        Fo = Y.shape[2]
+        Xb = X.gather(1, idx.unsqueeze(-1).expand(G, idx.shape[1], F))
+        Tb = Y.gather(1, idx.unsqueeze(-1).expand(G, idx.shape[1], Fo))
+        return Xb, Tb
+
+
+class DensePack(BasePack):
+    """Grouped feedforward autoencoder trainer (kernels K1-K4 of
+    SURVEY.md §2.3)."""
+
+    def _declare_params(self):
+        dims = self.spec.dense_dims()
+        self.layer_meta = dims
+        for i, (fin, fout, _act, _l1) in enumerate(dims):
+            self.store.declare(f"W{i}", (self.G, fin, fout))
+            self.store.declare(f"b{i}", (self.G, fout))
+
+    def _init_weights(self):
+        for g in range(self.G):
+            gen = torch.Generator().manual_seed(int(self.seeds[g]) & 0x7FFFFFFF)
+            for i, (fin, fout, _act, _l1) in enumerate(self.layer_meta):
+                self.store.views[f"W{i}"][g].copy_(_glorot_uniform((fin, fout), gen))
+                # biases stay zero
+
+    def forward(self, X: torch.Tensor) -> List[torch.Tensor]:
+        acts = [X]
+        a = X
+        for i, (_fin, _fout, act, _l1) in enumerate(self.layer_meta):
+            a = ops.grouped_linear_fwd(
+                a, self.store.cviews[f"W{i}"],
+                self.store.views[f"b{i}"], act,
+            )
+            acts.append(a)
+        return acts
+
+    def predict(self, X: torch.Tensor) -> torch.Tensor:
+        return self.forward(self._to_compute(X))[-1]
+
+    def train_batch(self, Xb, Tb) -> torch.Tensor:
+        acts = self.forward(Xb)
+        loss, dA = ops.mse_bwd(acts[-1], Tb.to(acts[-1].dtype))
+        for i in range(len(self.layer_meta) - 1, -1, -1):
+            _fin, _fout, act, l1 = self.layer_meta[i]
+            dZ = ops.act_l1_bwd(dA, acts[i + 1], act, l1)
+            dW, db = ops.grouped_linear_wgrad(acts[i], dZ)
+            self.store.gviews[f"W{i}"].copy_(dW)
+            self.store.gviews[f"b{i}"].copy_(db)
+            if i > 0:
+                dA = ops.grouped_linear_bwd_data(dZ, self.store.cviews[f"W{i}"])
+        a = self.spec.adam_params
+        self.store.adam_step(a["lr"], a["beta1"], a["beta2"], a["eps"])
+        return loss
+
+
+class LSTMPack(BasePack):
+    """Grouped stacked-LSTM autoencoder/forecast trainer (kernels K5-K7).
+
+    The sliding-window featurizer is zero-copy: windows are gathered
+    straight out of the resident [G, N, F] series tensor (the device
+    analog of create_keras_timeseriesgenerator, reference
+    models.py:713-793)."""
+
+    def _declare_params(self):
+        spec = self.spec
+        self.lstm_meta = []  # (fin, H, return_sequences)
+        fin = spec.n_features
+        lstm_layers = [l for l in spec.layers if l.kind == "lstm"]
+        dense_layers = [l for l in spec.layers if l.kind == "dense"]
+        assert len(dense_layers) == 1, "LSTM spec needs exactly one output dense layer"
+        for i, layer in enumerate(lstm_layers):
+            H = layer.units
+            self.store.declare(f"Wx{i}", (self.G, fin, 4 * H))
+            self.store.declare(f"Wh{i}", (self.G, H, 4 * H))
+            self.store.declare(f"bl{i}", (self.G, 4 * H))
+            self.lstm_meta.append((fin, H, layer.return_sequences))
+            fin = H
+        out_layer = dense_layers[0]
+        self.dense_meta = (fin, out_layer.units, out_layer.activation)
+        self.store.declare("Wd", (self.G, fin, out_layer.units))
+        self.store.declare("bd", (self.G, out_layer.units))
+
+    def _init_weights(self):
+        for g in range(self.G):
+            gen = torch.Generator().manual_seed(int(self.seeds[g]) & 0x7FFFFFFF)
+            for i, (fin, H, _rs) in enumerate(self.lstm_meta):
+                self.store.views[f"Wx{i}"][g].copy_(_glorot_uniform((fin, 4 * H), gen))
+                wh = torch.cat([_orthogonal((H, H), gen) for _ in range(4)], dim=1)
+                self.store.views[f"Wh{i}"][g].copy_(wh)
+                b = torch.zeros(4 * H)
+                b[H : 2 * H] = 1.0  # unit forget-gate bias (Keras default)
+                self.store.views[f"bl{i}"][g].copy_(b)
+            fin, fout, _act = self.dense_meta
+            self.store.views["Wd"][g].copy_(_glorot_uniform((fin, fout), gen))
+
+    # ---- sequence forward/backward ------------------------------------
+    def _forward_seq(self, Xw: torch.Tensor, keep: bool):
+        """Xw: [G, B, T, F]. Returns (y, cache)."""
+        G, B, T, _ = Xw.shape
+        seq = Xw
+        cache = []
+        for li, (fin, H, rs) in enumerate(self.lstm_meta):
+            Wx = self.store.cviews[f"Wx{li}"]
+            Wh = self.store.cviews[f"Wh{li}"]
+            b = self.store.views[f"bl{li}"]
+            flat_in = seq.reshape(G, B * T, fin)
+            gates_all = ops.grouped_linear_fwd(flat_in, Wx, b, "linear").view(
+                G, B, T, 4 * H
+            )
+            h = torch.zeros(G, B, H, dtype=self.compute_dtype, device=self.device)
+            c = torch.zeros(G, B, H, dtype=torch.float32, device=self.device)
+            hs = torch.empty(
+                G, B, T, H, dtype=self.compute_dtype, device=self.device
+            )
+            cs = (
+                torch.empty(G, B, T, H, dtype=torch.float32, device=self.device)
+                if keep
+                else None
+            )
+            gacts = (
+                torch.empty(
+                    G, B, T, 4 * H, dtype=self.compute_dtype, device=self.device
+                )
+                if keep
+                else None
+            )
+            for t in range(T):
+                gates = gates_all[:, :, t].contiguous()
+                ops.grouped_gemm_acc(h, Wh, gates)
+                h, c, gact = ops.lstm_pointwise_fwd(gates, c)
+                hs[:, :, t] = h
+                if keep:
+                    cs[:, :, t] = c
+                    gacts[:, :, t] = gact
+            cache.append(
+                dict(seq_in=seq if keep else None, hs=hs, cs=cs, gacts=gacts)
+            )
+            seq = hs
+        h_last = seq[:, :, -1].contiguous()
+        fin, fout, act = self.dense_meta
+        y = ops.grouped_linear_fwd(
+            h_last, self.store.cviews["Wd"], self.store.views["bd"], act
+        )
+        if keep:
+            cache.append(dict(h_last=h_last, y=y))
+        return y, cache
+
+    def predict_windows(self, Xw: torch.Tensor) -> torch.Tensor:
+        y, _ = self._forward_seq(self._to_compute(Xw), keep=False)
+        return y
+
+    def train_batch(self, Xw, Tb) -> torch.Tensor:
+        G, B, T, _ = Xw.shape
+        y, cache = self._forward_seq(Xw, keep=True)
+        loss, dY = ops.mse_bwd(y, Tb.to(y.dtype))
+
+        # output dense layer backward
+        fin, fout, act = self.dense_meta
+        head = cache[-1]
+        dZ = ops.act_l1_bwd(dY, head["y"], act, 0.0)
+        dWd, dbd = ops.grouped_linear_wgrad(head["h_last"], dZ)
+        self.store.gviews["Wd"].copy_(dWd)
+        self.store.gviews["bd"].copy_(dbd)
+        dh_last = ops.grouped_linear_bwd_data(dZ, self.store.cviews["Wd"])
+
+        # BPTT through the LSTM stack
+        dSeq: Optional[torch.Tensor] = None  # grad on layer output sequence
+        for li in range(len(self.lstm_meta) - 1, -1, -1):
+            fin, H, rs = self.lstm_meta[li]
+            lc = cache[li]
+            hs, cs, gacts = lc["hs"], lc["cs"], lc["gacts"]
+            Wh = self.store.cviews[f"Wh{li}"]
+            Wx = self.store.cviews[f"Wx{li}"]
+            dG = torch.empty(
+                G, B, T, 4 * H, dtype=self.compute_dtype, device=self.device
+            )
+            dh_carry = torch.zeros(
+                G, B, H, dtype=self.compute_dtype, device=self.device
+            )
+            dc_carry = torch.zeros(G, B, H, dtype=torch.float32, device=self.device)
+            for t in range(T - 1, -1, -1):
+                dh_t = dh_carry
+                if dSeq is not None:
+                    dh_t = dh_t + dSeq[:, :, t]
+                elif li == len(self.lstm_meta) - 1 and t == T - 1:
+                    dh_t = dh_t + dh_last
+                c_prev = (
+                    cs[:, :, t - 1]
+                    if t > 0
+                    else torch.zeros_like(dc_carry)
+                )
+                dgates, dc_carry = ops.lstm_pointwise_bwd(
+                    dh_t, dc_carry, gacts[:, :, t], cs[:, :, t], c_prev
+                )
+                dG[:, :, t] = dgates
+                dh_carry = ops.grouped_linear_bwd_data(dgates, Wh)
+            # batched weight grads over all (B, T) rows
+            h_prev_all = torch.cat(
+                [torch.zeros_like(hs[:, :, :1]), hs[:, :, :-1]], dim=2
+            )
+            dG_flat = dG.view(G, B * T, 4 * H)
+            dWx, dbl = ops.grouped_linear_wgrad(
+                lc["seq_in"].reshape(G, B * T, fin), dG_flat
+            )
+            dWh, _ = ops.grouped_linear_wgrad(
+                h_prev_all.reshape(G, B * T, H), dG_flat
+            )
+            self.store.gviews[f"Wx{li}"].copy_(dWx)
+            self.store.gviews[f"Wh{li}"].copy_(dWh)
+            self.store.gviews[f"bl{li}"].copy_(dbl)
+            if li > 0:
+                dSeq = ops.grouped_linear_bwd_data(dG_flat, Wx).view(G, B, T, fin)
+
+        a = self.spec.adam_params
+        self.store.adam_step(a["lr"], a["beta1"], a["beta2"], a["eps"])
+        return loss
+
+    # ---- windowed fit/predict over a series ---------------------------
+    @property
+    def lookback(self) -> int:
+        return self.spec.lookback_window
+
+    @property
+    def lookahead(self) -> int:
+        return self.spec.lookahead
+
+    def n_windows(self, N: int) -> int:
+        return N - self.lookback + 1 - self.lookahead
+
+    _n_samples = n_windows
+
+    def _gather_batch(self, X, Y, idx):
+        """idx: [G, B] window start indices. Builds [G,B,T,F] window
+        tensors and [G,B,Fo] targets straight from the series."""
+        G, N, F = X.shape
+        Fo = Y.shape[2]
+        B = idx.shape[1]
+        T = self.lookback
+        rows = idx.unsqueeze(-1) + torch.arange(T, device=X.device)  # [G,B,T]
+        Xw = X.gather(
+            1, rows.reshape(G, B * T, 1).expand(G, B * T, F)
+        ).view(G, B, T, F)
+        trow = idx + (T - 1 + self.lookahead)
+        Tb = Y.gather(1, trow.unsqueeze(-1).expand(G, B, Fo))
+        return Xw, Tb
+
+    def predict(self, X: torch.Tensor, batch_size: int = 4096) -> torch.Tensor:
+        """X: [G, N, F] series → [G, n_windows, Fo] predictions
+        (offset-aligned like the reference: output starts at row
+        lookback-1+lookahead)."""
+        X = self._to_compute(X)
+        G, N, F = X.shape
+        nw = self.n_windows(N)
+        outs = []
+        dummyY = X.new_zeros(G, N, self.dense_meta[1])
+        for s in range(0, nw, batch_size):
+            idx = (
+                torch.arange(s, min(s + batch_size, nw), device=self.device)
+                .unsqueeze(0)
+                .expand(G, -1)
+            )
+            Xw, _ = self._gather_batch(X, dummyY, idx)
+            outs.append(self.predict_windows(Xw))
+        return torch.cat(outs, dim=1)

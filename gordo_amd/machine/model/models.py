@@ -1,0 +1,525 @@
+"""
+sklearn-API estimators backed by the MI355X device engine.
+
+These keep the reference's class names and config surface
+(gordo/machine/model/models.py: KerasAutoEncoder :360, KerasLSTMAutoEncoder
+/ KerasLSTMForecast :463-710, KerasRawModelRegressor :401) so reference
+YAML configs are drop-in, but the compute path is the grouped HIP/MFMA
+engine (engine/pack.py) — there is no TensorFlow anywhere.
+
+An estimator trains as a pack-of-1 when fitted standalone; the fleet
+build scheduler (parallel/) fits many same-architecture estimators as
+one pack (see ``build_pack_spec`` / ``adopt_pack_result``).
+"""
+from __future__ import annotations
+
+import abc
+import logging
+from typing import Any, Callable, Dict, List, Optional, Tuple, Union
+
+import numpy as np
+import torch
+from sklearn.base import BaseEstimator, TransformerMixin
+from sklearn.metrics import explained_variance_score
+
+from .base import GordoBase
+from .register import register_model_builder
+from ...core.import_utils import import_location
+from ...engine.pack import DensePack, LSTMPack
+from ...engine.spec import LayerSpec, ModelSpec
+
+logger = logging.getLogger(__name__)
+
+FLOAT_DTYPE = np.float32
+
+
+def _as_2d_array(X) -> np.ndarray:
+    arr = np.asarray(X.values if hasattr(X, "values") else X, dtype=FLOAT_DTYPE)
+    if arr.ndim == 1:
+        arr = arr.reshape(len(arr), 1)
+    return arr
+
+
+def _default_device() -> str:
+    return "cuda" if torch.cuda.is_available() else "cpu"
+
+
+class KerasBaseEstimator(BaseEstimator, GordoBase):
+    """Base sklearn-compatible estimator; ``kind`` names a registered
+    model-factory (or an import path / callable)."""
+
+    supported_fit_args = [
+        "batch_size",
+        "epochs",
+        "verbose",
+        "callbacks",
+        "validation_split",
+        "shuffle",
+        "class_weight",
+        "initial_epoch",
+        "steps_per_epoch",
+        "validation_batch_size",
+        "max_queue_size",
+        "workers",
+        "use_multiprocessing",
+    ]
+
+    # subclass pack type
+    _pack_cls = DensePack
+
+    def __init__(self, kind: Union[str, Callable], **kwargs) -> None:
+        self.kind = self.load_kind(kind)
+        self.kwargs: Dict[str, Any] = kwargs
+        self._history: Optional[Dict[str, list]] = None
+        self._spec: Optional[ModelSpec] = None
+        self._weights: Optional[Dict[str, np.ndarray]] = None
+        self._pack = None
+        self.n_features: Optional[int] = None
+        self.n_features_out: Optional[int] = None
+
+    # ---- kind handling -------------------------------------------------
+    @staticmethod
+    def parse_module_path(module_path: str) -> Tuple[Optional[str], str]:
+        parts = module_path.split(".")
+        if len(parts) == 1:
+            return None, parts[0]
+        return ".".join(parts[:-1]), parts[-1]
+
+    def load_kind(self, kind) -> str:
+        if callable(kind):
+            register_model_builder(type=self.__class__.__name__)(kind)
+            return kind.__name__
+        module_name, func_name = self.parse_module_path(kind)
+        if module_name is None:
+            factories = register_model_builder.factories.get(
+                self.__class__.__name__, {}
+            )
+            if func_name not in factories:
+                raise ValueError(
+                    f"kind: {kind} is not an available model for type: "
+                    f"{self.__class__.__name__}!"
+                )
+        else:
+            import_location(kind)  # raises if unimportable
+        return kind
+
+    def _resolve_builder(self) -> Callable:
+        module_name, func_name = self.parse_module_path(self.kind)
+        if module_name is None:
+            return register_model_builder.factories[self.__class__.__name__][
+                func_name
+            ]
+        return import_location(self.kind)
+
+    # ---- sklearn plumbing ----------------------------------------------
+    def get_params(self, deep=True) -> Dict[str, Any]:
+        params = {"kind": self.kind}
+        params.update(self.kwargs)
+        return params
+
+    def set_params(self, **params):
+        if "kind" in params:
+            self.kind = self.load_kind(params.pop("kind"))
+        self.kwargs.update(params)
+        return self
+
+    # ---- fit machinery -------------------------------------------------
+    def extract_supported_fit_args(self, kwargs: Dict[str, Any]) -> Dict[str, Any]:
+        return {k: v for k, v in kwargs.items() if k in self.supported_fit_args}
+
+    def _factory_kwargs(self) -> Dict[str, Any]:
+        skip = set(self.supported_fit_args)
+        return {k: v for k, v in self.kwargs.items() if k not in skip}
+
+    def build_pack_spec(
+        self, n_features: int, n_features_out: Optional[int] = None
+    ) -> ModelSpec:
+        """Build the engine ModelSpec for given data shapes."""
+        builder = self._resolve_builder()
+        spec = builder(
+            n_features=n_features,
+            n_features_out=n_features_out,
+            **self._factory_kwargs(),
+        )
+        if not isinstance(spec, ModelSpec):
+            raise TypeError(
+                f"Model factory {self.kind} returned {type(spec)}; expected "
+                "ModelSpec"
+            )
+        return spec
+
+    def fit_args(self) -> Dict[str, Any]:
+        args = self.extract_supported_fit_args(self.kwargs)
+        args.setdefault("epochs", 1)
+        args.setdefault("batch_size", 32)
+        args.setdefault("shuffle", True)
+        return args
+
+    def _make_pack(self, spec: ModelSpec, device=None, seed=None):
+        if seed is None:
+            seed = int(np.random.randint(0, 2 ** 31 - 1))
+        return self._pack_cls(
+            spec, G=1, device=device or _default_device(), seeds=[seed]
+        )
+
+    def fit(self, X, y=None, **kwargs):
+        X = _as_2d_array(X)
+        y = X if y is None else _as_2d_array(y)
+        self.n_features = X.shape[1]
+        self.n_features_out = y.shape[1]
+        spec = self.build_pack_spec(self.n_features, self.n_features_out)
+        pack = self._make_pack(spec)
+        Xd = torch.from_numpy(X).unsqueeze(0).to(pack.device, pack.compute_dtype)
+        Yd = torch.from_numpy(y).unsqueeze(0).to(pack.device, pack.compute_dtype)
+        fit_args = dict(self.fit_args())
+        fit_args.update(self.extract_supported_fit_args(kwargs))
+        history = pack.fit(Xd, Yd, **_clean_fit_args(fit_args))
+        self.adopt_pack_result(
+            spec, pack.state_for_model(0), _history_for_model(history, 0)
+        )
+        self._pack = pack
+        return self
+
+    def adopt_pack_result(
+        self,
+        spec: ModelSpec,
+        weights: Dict[str, np.ndarray],
+        history: Dict[str, list],
+        n_features: Optional[int] = None,
+        n_features_out: Optional[int] = None,
+    ):
+        """Mark this estimator fitted with weights trained in a
+        (possibly shared, G>1) pack."""
+        self._spec = spec
+        self._weights = weights
+        self._history = history
+        self._pack = None
+        if n_features is not None:
+            self.n_features = n_features
+        if n_features_out is not None:
+            self.n_features_out = n_features_out
+        return self
+
+    def _ensure_pack(self):
+        if self._pack is not None:
+            return self._pack
+        if self._spec is None or self._weights is None:
+            raise ValueError(
+                f"This {self.__class__.__name__} has not been fitted yet."
+            )
+        pack = self._pack_cls(self._spec, G=1, device=_default_device(), seeds=[0])
+        pack.load_model_state(0, self._weights)
+        self._pack = pack
+        return pack
+
+    def predict(self, X, **kwargs) -> np.ndarray:
+        X = _as_2d_array(X)
+        pack = self._ensure_pack()
+        with torch.no_grad():
+            out = pack.predict(torch.from_numpy(X).unsqueeze(0))
+        return out[0].float().cpu().numpy()
+
+    # transform == predict so the estimator can sit mid-pipeline
+    def transform(self, X) -> np.ndarray:
+        return self.predict(X)
+
+    def score(self, X, y=None, sample_weight=None) -> float:
+        X = _as_2d_array(X)
+        y = X if y is None else _as_2d_array(y)
+        out = self.predict(X)
+        return explained_variance_score(y[-len(out):], out)
+
+    # ---- metadata / persistence ---------------------------------------
+    def get_metadata(self) -> Dict[str, Any]:
+        if self._history is None:
+            return {}
+        params = dict(self.fit_args())
+        params.pop("shuffle", None)
+        return {
+            "history": {
+                "params": params,
+                **self._history,
+            }
+        }
+
+    @property
+    def history(self):
+        return self._history
+
+    def __getstate__(self):
+        state = self.__dict__.copy()
+        if self._pack is not None and self._weights is None:
+            state["_weights"] = self._pack.state_for_model(0)
+        state["_pack"] = None
+        return state
+
+    def __setstate__(self, state):
+        self.__dict__.update(state)
+        self._pack = None
+
+    def __call__(self):
+        # factory-style call used by some sklearn clone paths
+        return self
+
+
+def _clean_fit_args(args: Dict[str, Any]) -> Dict[str, Any]:
+    out = dict(args)
+    for k in (
+        "callbacks",
+        "validation_split",
+        "class_weight",
+        "initial_epoch",
+        "steps_per_epoch",
+        "validation_batch_size",
+        "max_queue_size",
+        "workers",
+        "use_multiprocessing",
+    ):
+        out.pop(k, None)
+    return out
+
+
+def _history_for_model(history: Dict[str, list], g: int) -> Dict[str, list]:
+    return {
+        key: [float(epoch_vals[g]) for epoch_vals in values]
+        for key, values in history.items()
+    }
+
+
+class KerasAutoEncoder(KerasBaseEstimator, TransformerMixin):
+    """Feedforward autoencoder: fit X→y reconstruction; score =
+    explained variance (reference models.py:360-398)."""
+
+
+class KerasRawModelRegressor(KerasAutoEncoder):
+    """Build a model from a raw ``spec:`` definition — a Sequential of
+    Dense layers with a ``compile:`` section (reference
+    models.py:401-460; only the Dense subset is meaningful here, which
+    is what the reference's own example uses)."""
+
+    def __init__(self, kind: Union[dict, str, Callable] = "raw", **kwargs):
+        if isinstance(kind, dict):
+            kwargs.setdefault("spec", kind)
+            kind = "raw"
+        self._raw_kind = kind
+        super().__init__(kind="feedforward_model", **kwargs)
+        self.kind = "raw"
+
+    def load_kind(self, kind):
+        return "feedforward_model"
+
+    def get_params(self, deep=True):
+        params = {"kind": self._raw_kind}
+        params.update(self.kwargs)
+        return params
+
+    def build_pack_spec(self, n_features, n_features_out=None):
+        spec_def = self.kwargs.get("spec")
+        if not spec_def:
+            raise ValueError("KerasRawModelRegressor requires a 'spec'")
+        return _parse_raw_spec(spec_def, n_features, n_features_out or n_features)
+
+
+def _parse_raw_spec(
+    spec_def: dict, n_features: int, n_features_out: int
+) -> ModelSpec:
+    """Parse a keras-like raw Sequential spec into a dense ModelSpec."""
+    if not isinstance(spec_def, dict) or len(spec_def) != 1:
+        raise ValueError("raw spec must be a single-key dict (Sequential)")
+    seq_key = next(iter(spec_def))
+    if "Sequential" not in seq_key:
+        raise ValueError(f"Unsupported raw model root {seq_key!r}")
+    body = spec_def[seq_key] or {}
+    layers_def = body.get("layers", [])
+    layers: List[LayerSpec] = []
+    for layer_def in layers_def:
+        if isinstance(layer_def, dict) and len(layer_def) == 1:
+            lk = next(iter(layer_def))
+            lkw = layer_def[lk] or {}
+            if "Dense" not in lk:
+                raise ValueError(
+                    f"Unsupported layer {lk!r} in raw spec (Dense only)"
+                )
+            layers.append(
+                LayerSpec(
+                    kind="dense",
+                    units=int(lkw.get("units", n_features_out)),
+                    activation=lkw.get("activation", "linear") or "linear",
+                )
+            )
+        else:
+            raise ValueError(f"Unparsable layer definition {layer_def!r}")
+    if not layers:
+        raise ValueError("raw spec contains no layers")
+    compile_def = {}
+    if "compile" in (spec_def.get(seq_key) or {}):
+        compile_def = spec_def[seq_key]["compile"] or {}
+    return ModelSpec(
+        model_type="feedforward",
+        n_features=n_features,
+        n_features_out=layers[-1].units,
+        layers=layers,
+        loss=str(compile_def.get("loss", "mse")),
+        optimizer=str(compile_def.get("optimizer", "adam")),
+        optimizer_kwargs={},
+    )
+
+
+class KerasLSTMBaseEstimator(KerasBaseEstimator, TransformerMixin, metaclass=abc.ABCMeta):
+    """Lookback-window models over implicit sliding windows (reference
+    models.py:463-710). The windower itself is zero-copy on device —
+    windows are gathered from the resident series (engine/pack.LSTMPack);
+    ``create_keras_timeseriesgenerator`` below reproduces the exact
+    window/target alignment for host-side use and tests."""
+
+    _pack_cls = LSTMPack
+
+    def __init__(
+        self,
+        kind: Union[Callable, str],
+        lookback_window: int = 1,
+        batch_size: int = 32,
+        **kwargs,
+    ) -> None:
+        self.lookback_window = lookback_window
+        kwargs["lookback_window"] = lookback_window
+        kwargs["batch_size"] = batch_size
+        super().__init__(kind=kind, **kwargs)
+
+    @property
+    @abc.abstractmethod
+    def lookahead(self) -> int:
+        """Steps ahead in y the model targets."""
+        ...
+
+    def get_metadata(self):
+        metadata = super().get_metadata()
+        metadata.update({"forecast_steps": self.lookahead})
+        return metadata
+
+    def _validate_and_fix_size_of_X(self, X: np.ndarray) -> np.ndarray:
+        if X.ndim == 1:
+            X = X.reshape(len(X), 1)
+        if self.lookback_window >= X.shape[0]:
+            raise ValueError(
+                "For KerasLSTMForecast lookback_window must be < size of X"
+            )
+        return X
+
+    def _factory_kwargs(self):
+        kw = super()._factory_kwargs()
+        kw.pop("batch_size", None)
+        return kw
+
+    def build_pack_spec(self, n_features, n_features_out=None):
+        spec = super().build_pack_spec(n_features, n_features_out)
+        spec.lookahead = self.lookahead
+        spec.lookback_window = self.lookback_window
+        return spec
+
+    def fit(self, X, y=None, **kwargs):
+        X = self._validate_and_fix_size_of_X(_as_2d_array(X))
+        y = X if y is None else _as_2d_array(y)
+        self.n_features = X.shape[1]
+        self.n_features_out = y.shape[1]
+        spec = self.build_pack_spec(self.n_features, self.n_features_out)
+        pack = self._make_pack(spec)
+        Xd = torch.from_numpy(X).unsqueeze(0).to(pack.device, pack.compute_dtype)
+        Yd = torch.from_numpy(y).unsqueeze(0).to(pack.device, pack.compute_dtype)
+        fit_args = dict(self.fit_args())
+        fit_args.update(self.extract_supported_fit_args(kwargs))
+        history = pack.fit(Xd, Yd, **_clean_fit_args(fit_args))
+        self.adopt_pack_result(
+            spec, pack.state_for_model(0), _history_for_model(history, 0)
+        )
+        self._pack = pack
+        return self
+
+    def predict(self, X, **kwargs) -> np.ndarray:
+        """Output has ``lookback_window - 1 + lookahead`` fewer rows
+        than X (the window offset, reference models.py:618-660)."""
+        X = self._validate_and_fix_size_of_X(_as_2d_array(X))
+        pack = self._ensure_pack()
+        with torch.no_grad():
+            out = pack.predict(torch.from_numpy(X).unsqueeze(0))
+        return out[0].float().cpu().numpy()
+
+    def score(self, X, y=None, sample_weight=None) -> float:
+        X = _as_2d_array(X)
+        y = X if y is None else _as_2d_array(y)
+        out = self.predict(X)
+        return explained_variance_score(y[-len(out):], out)
+
+
+class KerasLSTMForecast(KerasLSTMBaseEstimator):
+    @property
+    def lookahead(self) -> int:
+        return 1
+
+
+class KerasLSTMAutoEncoder(KerasLSTMBaseEstimator):
+    @property
+    def lookahead(self) -> int:
+        return 0
+
+
+class TimeseriesWindows:
+    """Materialized sliding windows with the reference's batch/shape
+    semantics (stand-in for keras TimeseriesGenerator)."""
+
+    def __init__(self, X: np.ndarray, y: np.ndarray, batch_size: int, length: int, lookahead: int):
+        self.X, self.y = X, y
+        self.batch_size = batch_size
+        self.length = length
+        self.lookahead = lookahead
+        self.n_samples = max(0, len(X) - length + 1 - lookahead)
+
+    def __len__(self):
+        return int(np.ceil(self.n_samples / self.batch_size))
+
+    def __getitem__(self, i):
+        start = i * self.batch_size
+        stop = min(start + self.batch_size, self.n_samples)
+        bx = np.stack([self.X[j : j + self.length] for j in range(start, stop)])
+        by = np.stack(
+            [self.y[j + self.length - 1 + self.lookahead] for j in range(start, stop)]
+        )
+        return bx, by
+
+
+def create_keras_timeseriesgenerator(
+    X: np.ndarray,
+    y: Optional[np.ndarray],
+    batch_size: int,
+    lookback_window: int,
+    lookahead: int,
+) -> TimeseriesWindows:
+    """
+    Sliding windows of ``lookback_window`` rows with the target shifted
+    ``lookahead`` steps past the window end. Exactly reproduces the
+    reference's pad_sequences + TimeseriesGenerator alignment
+    (models.py:713-793): for lookahead==0 the sample's last row aligns
+    with its target; lookahead==1 targets one step ahead.
+
+    >>> import numpy as np
+    >>> X, y = np.random.rand(100, 2), np.random.rand(100, 2)
+    >>> gen = create_keras_timeseriesgenerator(X, y, batch_size=10,
+    ...                                        lookback_window=20, lookahead=0)
+    >>> len(gen)  # 9 = ceil((100-20+1)/10)
+    9
+    >>> len(gen[0])  # batchX and batchY
+    2
+    >>> len(gen[0][0])  # batch_size=10
+    10
+    >>> len(gen[0][0][0])  # a single sample, lookback_window=20
+    20
+    >>> len(gen[0][0][0][0])  # n_features=2
+    2
+    """
+    if lookahead < 0:
+        raise ValueError(f"Value of `lookahead` can not be negative, is {lookahead}")
+    if y is None:
+        y = X
+    return TimeseriesWindows(
+        np.asarray(X), np.asarray(y), batch_size, lookback_window, lookahead
+    )

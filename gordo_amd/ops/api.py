@@ -1,0 +1,118 @@
+"""
+Device-op dispatch.
+
+On GPU (ROCm/MI355X) tensors every op runs in the in-tree HIP extension
+(``gordo_amd.ops._gordo_hip``, compiled for gfx950 from ``csrc/``); if
+the extension is missing on a CUDA-capable host the call FAILS loudly —
+there is deliberately no silent eager fallback on GPU. On CPU tensors
+the fp32 reference implementations run (the test oracle / CPU lane).
+"""
+from __future__ import annotations
+
+import os
+from typing import Optional, Tuple
+
+import torch
+
+from . import reference as ref
+from .reference import (  # re-export
+    ACT_LINEAR,
+    ACT_TANH,
+    ACT_RELU,
+    ACT_SIGMOID,
+    act_code,
+)
+
+_hip = None
+_hip_err: Optional[str] = None
+
+
+def _load_hip():
+    global _hip, _hip_err
+    if _hip is not None or _hip_err is not None:
+        return _hip
+    try:
+        from . import _gordo_hip  # built in-tree by setup.py / __graft_entry__.build()
+
+        _hip = _gordo_hip
+    except ImportError as e:
+        _hip_err = str(e)
+    return _hip
+
+
+def hip_available() -> bool:
+    return _load_hip() is not None
+
+
+def _require_hip():
+    mod = _load_hip()
+    if mod is None:
+        raise RuntimeError(
+            "gordo_amd HIP extension (_gordo_hip) is not built but a GPU "
+            f"tensor was passed. Build it with `python setup.py build_ext "
+            f"--inplace` (PYTORCH_ROCM_ARCH=gfx950). Import error: {_hip_err}"
+        )
+    return mod
+
+
+def _on_gpu(t: torch.Tensor) -> bool:
+    return t.is_cuda
+
+
+def grouped_linear_fwd(X, W, b, act) -> torch.Tensor:
+    act = act_code(act)
+    if _on_gpu(X):
+        return _require_hip().grouped_linear_fwd(X, W, b, act)
+    return ref.grouped_linear_fwd(X, W, b, act)
+
+
+def act_l1_bwd(dA, Y, act, l1: float) -> torch.Tensor:
+    act = act_code(act)
+    if _on_gpu(dA):
+        return _require_hip().act_l1_bwd(dA, Y, act, float(l1))
+    return ref.act_l1_bwd(dA, Y, act, l1)
+
+
+def grouped_linear_bwd_data(dZ, W) -> torch.Tensor:
+    if _on_gpu(dZ):
+        return _require_hip().grouped_linear_bwd_data(dZ, W)
+    return ref.grouped_linear_bwd_data(dZ, W)
+
+
+def grouped_linear_wgrad(X, dZ) -> Tuple[torch.Tensor, torch.Tensor]:
+    if _on_gpu(X):
+        return _require_hip().grouped_linear_wgrad(X, dZ)
+    return ref.grouped_linear_wgrad(X, dZ)
+
+
+def grouped_gemm_acc(A, B, C):
+    if _on_gpu(A):
+        return _require_hip().grouped_gemm_acc(A, B, C)
+    return ref.grouped_gemm_acc(A, B, C)
+
+
+def mse_bwd(Y, T) -> Tuple[torch.Tensor, torch.Tensor]:
+    if _on_gpu(Y):
+        return _require_hip().mse_bwd(Y, T)
+    return ref.mse_bwd(Y, T)
+
+
+def adam_step(p, g, m, v, lr, beta1, beta2, eps, step, p_lp=None):
+    if _on_gpu(p):
+        return _require_hip().adam_step(
+            p, g, m, v, float(lr), float(beta1), float(beta2), float(eps),
+            int(step), p_lp,
+        )
+    return ref.adam_step(p, g, m, v, lr, beta1, beta2, eps, step, p_lp)
+
+
+def lstm_pointwise_fwd(gates, c_prev):
+    if _on_gpu(gates):
+        return _require_hip().lstm_pointwise_fwd(gates, c_prev)
+    return ref.lstm_pointwise_fwd(gates, c_prev)
+
+
+def lstm_pointwise_bwd(dh, dc_next, gact, c, c_prev):
+    if _on_gpu(dh):
+        return _require_hip().lstm_pointwise_bwd(dh, dc_next, gact, c, c_prev)
+    return ref.lstm_pointwise_bwd(dh, dc_next, gact, c, c_prev)
