@@ -1,0 +1,3 @@
+from .config_elements.normalized_config import NormalizedConfig
+
+__all__ = ["NormalizedConfig"]
