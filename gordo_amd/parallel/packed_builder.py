@@ -1,0 +1,514 @@
+"""
+Packed fleet builder — the MI355X replacement for the reference's
+pod-per-model Argo fan-out.
+
+Where the reference trains each Machine in its own Kubernetes pod
+(argo-workflow.yml.template:709-776), this builder classifies every
+Machine's model definition, groups machines whose device architecture
+matches (same layer dims/activations, same row count, same fit args),
+and trains each group as ONE ``engine.pack`` — every layer of every
+machine in the group is a single grouped MFMA GEMM launch. Cross-
+validation (TimeSeriesSplit / KFold), DiffBased threshold calculation,
+metadata assembly and the on-disk model layout reproduce
+``ModelBuilder`` semantics exactly; machines whose models aren't
+packable (arbitrary sklearn pipelines, config #1) fall back to the
+per-machine ModelBuilder on CPU.
+"""
+from __future__ import annotations
+
+import concurrent.futures
+import datetime
+import hashlib
+import json
+import logging
+import os
+import time
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional, Tuple
+
+import numpy as np
+import pandas as pd
+import torch
+from sklearn.base import clone as sk_clone
+from sklearn.pipeline import Pipeline
+
+import gordo_amd
+from .. import serializer
+from ..builder.build_model import ModelBuilder
+from ..core.base import GordoBaseDataset
+from ..engine.pack import DensePack, LSTMPack
+from ..engine.spec import ModelSpec
+from ..machine import Machine
+from ..machine.metadata import (
+    BuildMetadata,
+    CrossValidationMetaData,
+    DatasetBuildMetadata,
+    ModelBuildMetadata,
+)
+from ..machine.model.anomaly.diff import DiffBasedAnomalyDetector
+from ..machine.model.models import KerasBaseEstimator, KerasLSTMBaseEstimator
+from ..util import disk_registry
+
+logger = logging.getLogger(__name__)
+
+
+# ---------------------------------------------------------------------------
+@dataclass
+class MachinePlan:
+    machine: Machine
+    model: Any = None            # instantiated estimator graph
+    detector: Optional[DiffBasedAnomalyDetector] = None
+    pre_steps: Optional[List[Tuple[str, Any]]] = None
+    keras_est: Optional[KerasBaseEstimator] = None
+    X: Optional[pd.DataFrame] = None
+    y: Optional[pd.DataFrame] = None
+    dataset_meta: Dict[str, Any] = field(default_factory=dict)
+    query_duration: float = 0.0
+    packable: bool = False
+    error: Optional[BaseException] = None
+    # filled during build
+    scores: Dict[str, Any] = field(default_factory=dict)
+    splits: Dict[str, Any] = field(default_factory=dict)
+    spec: Optional[ModelSpec] = None
+    seed: int = 0
+
+
+def _classify(model) -> Tuple[Optional[DiffBasedAnomalyDetector], Optional[list], Optional[KerasBaseEstimator]]:
+    """Recognize the packable shapes:
+    [DiffBased*Detector(base_estimator=)] [Pipeline(pre..., )] KerasEst."""
+    detector = None
+    inner = model
+    if isinstance(model, DiffBasedAnomalyDetector):
+        detector = model
+        inner = model.base_estimator
+    pre_steps: List[Tuple[str, Any]] = []
+    if isinstance(inner, Pipeline):
+        *pre, last = inner.steps
+        pre_steps = list(pre)
+        inner = last[1]
+    if isinstance(inner, KerasBaseEstimator):
+        return detector, pre_steps, inner
+    return None, None, None
+
+
+def _machine_seed(name: str, eval_seed: int) -> int:
+    h = hashlib.sha256(f"{name}:{eval_seed}".encode()).digest()
+    return int.from_bytes(h[:4], "little") & 0x7FFFFFFF
+
+
+class PackedFleetBuilder:
+    """Build many Machines on one device, packing same-architecture
+    models into grouped trainers."""
+
+    def __init__(
+        self,
+        machines: List[Machine],
+        output_dir: Optional[str] = None,
+        model_register_dir: Optional[str] = None,
+        device: Optional[str] = None,
+        data_workers: int = 8,
+        save_models: bool = True,
+        replace_cache: bool = False,
+    ):
+        self.machines = machines
+        self.output_dir = output_dir
+        self.model_register_dir = model_register_dir
+        self.device = device or (
+            "cuda" if torch.cuda.is_available() else "cpu"
+        )
+        self.data_workers = data_workers
+        self.save_models = save_models
+        self.replace_cache = replace_cache
+
+    # ---- public ----------------------------------------------------------
+    def build_all(self) -> List[Tuple[str, Any]]:
+        """Build every machine. Returns [(name, Machine-with-metadata |
+        exception)]. Models are saved under output_dir/<name>/ when
+        save_models is set."""
+        plans = [MachinePlan(machine=m) for m in self.machines]
+        self._fetch_data(plans)
+        self._instantiate_models(plans)
+
+        packable = [p for p in plans if p.packable and p.error is None]
+        fallback = [p for p in plans if not p.packable and p.error is None]
+
+        for group in self._group(packable):
+            try:
+                self._build_group(group)
+            except Exception as e:  # isolate group failures
+                logger.exception("Pack group build failed")
+                for p in group:
+                    p.error = e
+
+        for p in fallback:
+            try:
+                logger.info("Fallback per-machine build: %s", p.machine.name)
+                builder = ModelBuilder(p.machine)
+                model, machine_out = builder.build(
+                    output_dir=(
+                        os.path.join(self.output_dir, p.machine.name)
+                        if self.output_dir and self.save_models
+                        else None
+                    ),
+                    model_register_dir=self.model_register_dir,
+                    replace_cache=self.replace_cache,
+                )
+                p.machine = machine_out
+                p.model = model
+            except Exception as e:
+                logger.exception("Fallback build failed: %s", p.machine.name)
+                p.error = e
+
+        results: List[Tuple[str, Any]] = []
+        for p in plans:
+            results.append(
+                (p.machine.name, p.error if p.error is not None else p.machine)
+            )
+        return results
+
+    # ---- stages ----------------------------------------------------------
+    def _fetch_data(self, plans: List[MachinePlan]):
+        def fetch(p: MachinePlan):
+            try:
+                start = time.time()
+                dataset = GordoBaseDataset.from_dict(p.machine.dataset.to_dict())
+                p.X, p.y = dataset.get_data()
+                p.query_duration = time.time() - start
+                p.dataset_meta = dataset.get_metadata()
+            except BaseException as e:
+                p.error = e
+
+        with concurrent.futures.ThreadPoolExecutor(self.data_workers) as ex:
+            list(ex.map(fetch, plans))
+
+    def _instantiate_models(self, plans: List[MachinePlan]):
+        for p in plans:
+            if p.error is not None:
+                continue
+            try:
+                p.model = serializer.from_definition(p.machine.model)
+            except Exception as e:
+                p.error = e
+                continue
+            detector, pre_steps, keras_est = _classify(p.model)
+            if keras_est is not None:
+                p.detector = detector
+                p.pre_steps = pre_steps
+                p.keras_est = keras_est
+                p.packable = True
+            p.seed = _machine_seed(
+                p.machine.name, int(p.machine.evaluation.get("seed", 0))
+            )
+
+    def _group(self, plans: List[MachinePlan]) -> List[List[MachinePlan]]:
+        groups: Dict[Any, List[MachinePlan]] = {}
+        for p in plans:
+            est = p.keras_est
+            n_features = p.X.shape[1]
+            n_features_out = p.y.shape[1]
+            try:
+                spec = est.build_pack_spec(n_features, n_features_out)
+            except Exception as e:
+                p.error = e
+                continue
+            p.spec = spec
+            key = (
+                spec.arch_key(),
+                len(p.X),
+                json.dumps(est.fit_args(), sort_keys=True, default=str),
+                json.dumps(p.machine.evaluation, sort_keys=True, default=str),
+                tuple(type(s[1]).__name__ for s in p.pre_steps),
+                p.detector is not None,
+            )
+            groups.setdefault(key, []).append(p)
+        return list(groups.values())
+
+    # ---- the packed group build -----------------------------------------
+    def _pack_cls(self, spec: ModelSpec):
+        return LSTMPack if spec.model_type == "lstm" else DensePack
+
+    def _make_pack(self, spec: ModelSpec, group: List[MachinePlan]):
+        return self._pack_cls(spec)(
+            spec,
+            G=len(group),
+            device=self.device,
+            seeds=[p.seed for p in group],
+        )
+
+    def _stack(self, arrays: List[np.ndarray], pack) -> torch.Tensor:
+        t = torch.from_numpy(np.stack(arrays).astype(np.float32))
+        return t.to(pack.device, pack.compute_dtype)
+
+    def _build_group(self, group: List[MachinePlan]):
+        spec = group[0].spec
+        evaluation = group[0].machine.evaluation
+        fit_args = group[0].keras_est.fit_args()
+        logger.info(
+            "Packed build: %d machines, arch=%s/%s, device=%s",
+            len(group), spec.model_type,
+            [l.units for l in spec.layers], self.device,
+        )
+
+        # cache probe: skip machines already registered
+        to_build: List[MachinePlan] = []
+        for p in group:
+            if self.model_register_dir and not self.replace_cache:
+                key = ModelBuilder(p.machine).cache_key
+                cached = ModelBuilder.check_cache(self.model_register_dir, key)
+                if cached:
+                    p.machine = Machine.from_dict(serializer.load_metadata(cached))
+                    p.model = serializer.load(cached)
+                    continue
+            to_build.append(p)
+        if not to_build:
+            return
+        group = to_build
+
+        # pre-transform (CPU sklearn scalers etc., fitted per machine)
+        Xt_list, y_list = [], []
+        for p in group:
+            Xt = p.X.values.astype(np.float32)
+            for _, step in p.pre_steps:
+                Xt = step.fit_transform(Xt)
+            Xt_list.append(np.asarray(Xt, dtype=np.float32))
+            y_list.append(p.y.values.astype(np.float32))
+
+        cv_mode = str(evaluation.get("cv_mode", "full_build")).lower()
+        cv_duration = None
+        t0_all = time.time()
+        if cv_mode in ("cross_val_only", "full_build"):
+            t0 = time.time()
+            self._cross_validate_group(group, Xt_list, y_list, spec, fit_args)
+            cv_duration = time.time() - t0
+            if cv_mode == "cross_val_only":
+                for p in group:
+                    self._finalize(p, None, cv_duration, final=False)
+                return
+
+        # final full fit
+        pack = self._make_pack(spec, group)
+        Xd = self._stack(Xt_list, pack)
+        Yd = self._stack(y_list, pack)
+        t0 = time.time()
+        history = pack.fit(Xd, Yd, **_engine_fit_args(fit_args))
+        train_duration = time.time() - t0
+
+        # per-machine adoption + detector finalization
+        offset = len(Xt_list[0]) - pack._n_samples(len(Xt_list[0]))
+        for g_idx, p in enumerate(group):
+            hist = {
+                k: [float(ep[g_idx]) for ep in v] for k, v in history.items()
+            }
+            p.keras_est.adopt_pack_result(
+                spec,
+                pack.state_for_model(g_idx),
+                hist,
+                n_features=Xt_list[g_idx].shape[1],
+                n_features_out=y_list[g_idx].shape[1],
+            )
+            if p.detector is not None:
+                p.detector.scaler.fit(p.y)
+            self._finalize(p, offset, cv_duration, final=True,
+                           train_duration=train_duration / len(group))
+
+        # save + register
+        if self.save_models and self.output_dir:
+            for p in group:
+                out = os.path.join(self.output_dir, p.machine.name)
+                ModelBuilder._save_model(p.model, p.machine, out)
+                if self.model_register_dir:
+                    disk_registry.write_key(
+                        self.model_register_dir,
+                        ModelBuilder(p.machine).cache_key,
+                        out,
+                    )
+        logger.info(
+            "Packed build of %d machines done in %.2fs",
+            len(group), time.time() - t0_all,
+        )
+
+    def _cross_validate_group(
+        self,
+        group: List[MachinePlan],
+        Xt_list: List[np.ndarray],
+        y_list: List[np.ndarray],
+        spec: ModelSpec,
+        fit_args: Dict[str, Any],
+    ):
+        """Packed equivalent of sklearn cross_validate +
+        DiffBasedAnomalyDetector.cross_validate: per fold, fresh packs
+        fit on the train slice, score the test slice per machine,
+        accumulate rolling-min-max thresholds (reference
+        diff.py:176-266, build_model.py:244-289)."""
+        evaluation = group[0].machine.evaluation
+        split_def = evaluation.get(
+            "cv", {"sklearn.model_selection.TimeSeriesSplit": {"n_splits": 3}}
+        )
+        split_obj = serializer.from_definition(split_def)
+        metrics_list = ModelBuilder.metrics_from_list(evaluation.get("metrics"))
+        scoring_scaler_def = evaluation.get("scoring_scaler")
+
+        N = len(Xt_list[0])
+        X_index = group[0].X.index
+        folds = list(split_obj.split(np.zeros((N, 1))))
+
+        # per-machine scoring scalers fitted on full y
+        scoring_scalers = []
+        for p, y_arr in zip(group, y_list):
+            if scoring_scaler_def:
+                sd = scoring_scaler_def
+                if isinstance(sd, str):
+                    sd = {sd: {}}
+                sc = serializer.from_definition(sd)
+                sc.fit(y_arr)
+            else:
+                sc = None
+            scoring_scalers.append(sc)
+
+        per_machine_scores: List[Dict[str, List[float]]] = [
+            {} for _ in group
+        ]
+        window = getattr(group[0].detector, "window", None) if group[0].detector else None
+
+        for fold_i, (train_idx, test_idx) in enumerate(folds):
+            pack = self._make_pack(spec, group)
+            Xd = self._stack([x[train_idx] for x in Xt_list], pack)
+            Yd = self._stack([y[train_idx] for y in y_list], pack)
+            pack.fit(Xd, Yd, **_engine_fit_args(fit_args))
+
+            Xtest = self._stack([x[test_idx] for x in Xt_list], pack)
+            with torch.no_grad():
+                preds = pack.predict(Xtest).float().cpu().numpy()
+
+            for g_idx, p in enumerate(group):
+                y_true_full = y_list[g_idx][test_idx]
+                y_pred = preds[g_idx]
+                y_true = y_true_full[-len(y_pred):]
+                sc = scoring_scalers[g_idx]
+                yt = sc.transform(y_true) if sc is not None else y_true
+                yp = sc.transform(y_pred) if sc is not None else y_pred
+                tags = [t.name for t in p.machine.dataset.target_tag_list]
+                for metric in metrics_list:
+                    mname = metric.__name__.replace("_", "-")
+                    for col, tag in enumerate(tags):
+                        key = f'{mname}-{tag.replace(" ", "-")}'
+                        per_machine_scores[g_idx].setdefault(key, []).append(
+                            float(metric(yt[:, col], yp[:, col]))
+                        )
+                    per_machine_scores[g_idx].setdefault(mname, []).append(
+                        float(metric(yt, yp))
+                    )
+
+                # DiffBased thresholds: fold scaler fitted on y_train
+                if p.detector is not None:
+                    det = p.detector
+                    fold_scaler = sk_clone(det.scaler)
+                    fold_scaler.fit(y_list[g_idx][train_idx])
+                    scaled_mse = pd.Series(
+                        (
+                            (fold_scaler.transform(y_pred)
+                             - fold_scaler.transform(y_true)) ** 2
+                        ).mean(axis=1)
+                    )
+                    mae = pd.DataFrame(np.abs(y_true - y_pred))
+                    agg_thr = scaled_mse.rolling(6).min().max()
+                    tag_thr = mae.rolling(6).min().max()
+                    tag_thr.name = f"fold-{fold_i}"
+                    if not hasattr(det, "aggregate_thresholds_per_fold_"):
+                        det.aggregate_thresholds_per_fold_ = {}
+                        det.feature_thresholds_per_fold_ = pd.DataFrame()
+                        det.smooth_aggregate_thresholds_per_fold_ = {}
+                        det.smooth_feature_thresholds_per_fold_ = pd.DataFrame()
+                    det.aggregate_thresholds_per_fold_[f"fold-{fold_i}"] = agg_thr
+                    det.feature_thresholds_per_fold_ = pd.concat(
+                        [det.feature_thresholds_per_fold_, tag_thr.to_frame().T]
+                    )
+                    det.aggregate_threshold_ = agg_thr
+                    det.feature_thresholds_ = tag_thr
+                    if window is not None:
+                        s_agg = scaled_mse.rolling(window).min().max()
+                        s_tag = mae.rolling(window).min().max()
+                        s_tag.name = f"fold-{fold_i}"
+                        det.smooth_aggregate_thresholds_per_fold_[
+                            f"fold-{fold_i}"
+                        ] = s_agg
+                        det.smooth_feature_thresholds_per_fold_ = pd.concat(
+                            [
+                                det.smooth_feature_thresholds_per_fold_,
+                                s_tag.to_frame().T,
+                            ]
+                        )
+                        det.smooth_aggregate_threshold_ = s_agg
+                        det.smooth_feature_thresholds_ = s_tag
+
+        # assemble fold stats + split metadata per machine
+        for g_idx, p in enumerate(group):
+            scores: Dict[str, Any] = {}
+            for key, vals in per_machine_scores[g_idx].items():
+                arr = np.asarray(vals)
+                entry = {
+                    "fold-mean": arr.mean(),
+                    "fold-std": arr.std(),
+                    "fold-max": arr.max(),
+                    "fold-min": arr.min(),
+                }
+                entry.update(
+                    {f"fold-{i + 1}": float(v) for i, v in enumerate(vals)}
+                )
+                scores[key] = entry
+            p.scores = scores
+            split_metadata: Dict[str, Any] = {}
+            for i, (train_ind, test_ind) in enumerate(folds):
+                split_metadata.update(
+                    {
+                        f"fold-{i + 1}-train-start": X_index[train_ind[0]],
+                        f"fold-{i + 1}-train-end": X_index[train_ind[-1]],
+                        f"fold-{i + 1}-test-start": X_index[test_ind[0]],
+                        f"fold-{i + 1}-test-end": X_index[test_ind[-1]],
+                        f"fold-{i + 1}-n-train": len(train_ind),
+                        f"fold-{i + 1}-n-test": len(test_ind),
+                    }
+                )
+            p.splits = split_metadata
+
+    def _finalize(
+        self,
+        p: MachinePlan,
+        offset: Optional[int],
+        cv_duration: Optional[float],
+        final: bool,
+        train_duration: Optional[float] = None,
+    ):
+        model_meta = (
+            ModelBuilder._extract_metadata_from_model(p.model) if final else {}
+        )
+        p.machine.metadata.build_metadata = BuildMetadata(
+            model=ModelBuildMetadata(
+                model_offset=int(offset) if offset is not None else 0,
+                model_creation_date=(
+                    str(datetime.datetime.now(datetime.timezone.utc).astimezone())
+                    if final
+                    else None
+                ),
+                model_builder_version=gordo_amd.__version__,
+                model_training_duration_sec=train_duration,
+                cross_validation=CrossValidationMetaData(
+                    cv_duration_sec=cv_duration,
+                    scores=p.scores,
+                    splits=p.splits,
+                ),
+                model_meta=model_meta,
+            ),
+            dataset=DatasetBuildMetadata(
+                query_duration_sec=p.query_duration,
+                dataset_meta=p.dataset_meta,
+            ),
+        )
+
+
+def _engine_fit_args(fit_args: Dict[str, Any]) -> Dict[str, Any]:
+    out = {}
+    for k in ("epochs", "batch_size", "shuffle", "verbose"):
+        if k in fit_args:
+            out[k] = fit_args[k]
+    return out

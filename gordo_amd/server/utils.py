@@ -1,0 +1,297 @@
+"""
+Server utilities: dataframe codecs (parquet / nested-JSON), request
+decorators, and the LRU model/metadata caches.
+
+Behavioral spec: gordo/server/utils.py:40-485. The model LRU keeps
+device-resident estimators hot (N_CACHED_MODELS, default 2 like the
+reference — raise it on an MI355X box: 288 GB HBM holds thousands of
+these models).
+"""
+from __future__ import annotations
+
+import functools
+import io
+import logging
+import os
+import pickle
+import re
+import shutil
+import timeit
+import zlib
+from datetime import datetime
+from functools import lru_cache
+from typing import List, Union
+
+import dateutil.parser
+import numpy as np
+import pandas as pd
+import pyarrow as pa
+import pyarrow.parquet as pq
+from flask import Response, g, jsonify, make_response, request
+from sklearn.base import BaseEstimator
+from werkzeug.exceptions import InternalServerError, NotFound, UnprocessableEntity
+
+from .. import serializer
+
+logger = logging.getLogger(__name__)
+
+revision_re = re.compile(r"^\d+$")
+gordo_name_re = re.compile(r"^[a-zA-Z0-9\-]+$")
+
+
+def validate_revision(revision: str) -> bool:
+    return bool(revision_re.match(revision))
+
+
+def validate_gordo_name(gordo_name: str):
+    if gordo_name and not gordo_name_re.match(gordo_name):
+        raise UnprocessableEntity("gordo_name field has wrong format")
+
+
+# ---- dataframe codecs --------------------------------------------------
+def dataframe_into_parquet_bytes(
+    df: pd.DataFrame, compression: str = "snappy"
+) -> bytes:
+    table = pa.Table.from_pandas(df)
+    buf = pa.BufferOutputStream()
+    pq.write_table(table, buf, compression=compression)
+    return buf.getvalue().to_pybytes()
+
+
+def dataframe_from_parquet_bytes(buf: bytes) -> pd.DataFrame:
+    return pq.read_table(io.BytesIO(buf)).to_pandas()
+
+
+def dataframe_to_dict(df: pd.DataFrame) -> dict:
+    """
+    2-level-MultiIndex dataframe → nested JSON-able dict (top-level
+    column name → {sub-column → {index → value}}).
+
+    >>> import pprint
+    >>> columns = pd.MultiIndex.from_tuples(
+    ...     (f"feature{i}", f"sub-feature-{ii}") for i in range(2) for ii in range(2))
+    >>> index = pd.date_range('2019-01-01', '2019-02-01', periods=2)
+    >>> df = pd.DataFrame(np.arange(8).reshape((2, 4)), columns=columns, index=index)
+    >>> pprint.pprint(dataframe_to_dict(df))
+    {'feature0': {'sub-feature-0': {'2019-01-01': 0, '2019-02-01': 4},
+                  'sub-feature-1': {'2019-01-01': 1, '2019-02-01': 5}},
+     'feature1': {'sub-feature-0': {'2019-01-01': 2, '2019-02-01': 6},
+                  'sub-feature-1': {'2019-01-01': 3, '2019-02-01': 7}}}
+    """
+    data = df.copy()
+    if isinstance(data.index, pd.DatetimeIndex):
+        data.index = data.index.astype(str)
+    if isinstance(df.columns, pd.MultiIndex):
+        return {
+            col: (
+                data[col].to_dict()
+                if isinstance(data[col], pd.DataFrame)
+                else pd.DataFrame(data[col]).to_dict()
+            )
+            for col in data.columns.get_level_values(0)
+        }
+    return data.to_dict()
+
+
+def dataframe_from_dict(data: dict) -> pd.DataFrame:
+    """
+    Inverse of :func:`dataframe_to_dict`.
+
+    >>> serialized = {
+    ... 'feature0': {'sub-feature-0': {'2019-01-01': 0, '2019-02-01': 4},
+    ...              'sub-feature-1': {'2019-01-01': 1, '2019-02-01': 5}}}
+    >>> dataframe_from_dict(serialized).shape
+    (2, 2)
+    """
+    if isinstance(data, dict) and any(
+        isinstance(val, dict) for val in data.values()
+    ):
+        try:
+            keys = data.keys()
+            df = pd.concat(
+                (pd.DataFrame.from_dict(data[key]) for key in keys),
+                axis=1,
+                keys=keys,
+            )
+        except (ValueError, AttributeError):
+            df = pd.DataFrame.from_dict(data)
+    else:
+        df = pd.DataFrame.from_dict(data)
+
+    try:
+        df.index = df.index.map(dateutil.parser.isoparse)
+    except (TypeError, ValueError):
+        df.index = df.index.map(int)
+    df.sort_index(inplace=True)
+    return df
+
+
+def parse_iso_datetime(datetime_str: str) -> datetime:
+    parsed = dateutil.parser.isoparse(datetime_str)
+    if parsed.tzinfo is None:
+        raise ValueError(
+            f"Provide timezone to timestamp {datetime_str}. Example: "
+            f"{datetime_str + 'Z'} or {datetime_str + '+00:00'}"
+        )
+    return parsed
+
+
+# ---- request data extraction -------------------------------------------
+def _verify_dataframe(
+    df: pd.DataFrame, expected_columns: List[str]
+) -> Union[Response, pd.DataFrame]:
+    if isinstance(df.columns, pd.MultiIndex):
+        return make_response(
+            (
+                jsonify(
+                    message="Server does not support multi-level dataframes "
+                    f"at this time: {df.columns.tolist()}"
+                ),
+                400,
+            )
+        )
+    if not all(col in df.columns for col in expected_columns):
+        if len(df.columns) != len(expected_columns):
+            return make_response(
+                (
+                    jsonify(
+                        message=f"Unexpected features: was expecting "
+                        f"{expected_columns} length of {len(expected_columns)}, "
+                        f"but got {df.columns} length of {len(df.columns)}"
+                    ),
+                    400,
+                )
+            )
+        df.columns = expected_columns
+        return df
+    return df[expected_columns]
+
+
+def extract_X_y(method):
+    """Pull X (and optional y) out of a POST — JSON dict-of-dicts or
+    multipart parquet files — verify columns against the model's tags,
+    and stash them on ``flask.g``."""
+
+    @functools.wraps(method)
+    def wrapper_method(*args, **kwargs):
+        from .properties import get_tags, get_target_tags
+
+        start_time = timeit.default_timer()
+        if request.method != "POST":
+            raise NotImplementedError(
+                f"Cannot extract X and y from '{request.method}' request."
+            )
+        if request.is_json:
+            if "X" not in (request.json or {}):
+                return make_response(
+                    (jsonify(message='Cannot predict without "X"'), 400)
+                )
+            X = dataframe_from_dict(request.json["X"])
+            y = request.json.get("y")
+            if y is not None:
+                y = dataframe_from_dict(y)
+        else:
+            if "X" not in request.files:
+                return make_response(
+                    (jsonify(message='Cannot predict without "X"'), 400)
+                )
+            X = dataframe_from_parquet_bytes(request.files["X"].read())
+            y = request.files.get("y")
+            if y is not None:
+                y = dataframe_from_parquet_bytes(y.read())
+
+        X = _verify_dataframe(X, [t.name for t in get_tags()])
+        if y is not None:
+            y = _verify_dataframe(y, [t.name for t in get_target_tags()])
+        for data_or_resp in (X, y):
+            if isinstance(data_or_resp, Response):
+                return data_or_resp
+        g.X, g.y = X, y
+        logger.debug(
+            "Time to parse X and y: %ss", timeit.default_timer() - start_time
+        )
+        return method(*args, **kwargs)
+
+    return wrapper_method
+
+
+# ---- caches -------------------------------------------------------------
+@lru_cache(maxsize=int(os.getenv("N_CACHED_MODELS", 2)))
+def load_model(directory: str, name: str) -> BaseEstimator:
+    start_time = timeit.default_timer()
+    model = serializer.load(os.path.join(directory, name))
+    logger.debug("Time to load model: %ss", timeit.default_timer() - start_time)
+    return model
+
+
+def check_metadata_file(directory: str, name: str):
+    if not serializer.metadata_path(os.path.join(directory, name)):
+        raise FileNotFoundError("Unable to load metadata.json file")
+
+
+_n_cached_metadata = int(os.getenv("N_CACHED_METADATA", 250))
+
+
+@lru_cache(maxsize=_n_cached_metadata)
+def _load_compressed_metadata(directory: str, name: str) -> bytes:
+    # stored zlib-compressed-pickled: ~10x smaller resident cache
+    metadata = serializer.load_metadata(os.path.join(directory, name))
+    return zlib.compress(pickle.dumps(metadata))
+
+
+def load_metadata(directory: str, name: str) -> dict:
+    return pickle.loads(zlib.decompress(_load_compressed_metadata(directory, name)))
+
+
+@lru_cache(maxsize=_n_cached_metadata)
+def load_info(directory: str, name: str) -> dict:
+    return serializer.load_info(os.path.join(directory, name))
+
+
+def delete_revision(directory: str, name: str):
+    full_path = os.path.join(directory, name)
+    if not os.path.isfile(os.path.join(full_path, "metadata.json")):
+        raise NotFound("Not found")
+    shutil.rmtree(full_path, ignore_errors=True)
+    if os.path.exists(full_path):
+        raise InternalServerError("Unable to delete this model revision folder")
+    if not os.listdir(directory):
+        shutil.rmtree(directory, ignore_errors=True)
+        if os.path.exists(directory):
+            raise InternalServerError("Unable to delete this revision folder")
+
+
+# ---- view decorators ----------------------------------------------------
+def metadata_required(f):
+    @functools.wraps(f)
+    def wrapper(*args, gordo_project: str, gordo_name: str, **kwargs):
+        validate_gordo_name(gordo_name)
+        g.info = {}
+        try:
+            g.info = load_info(directory=g.collection_dir, name=gordo_name) or {}
+        except FileNotFoundError:
+            pass
+        try:
+            check_metadata_file(g.collection_dir, gordo_name)
+            g.metadata = load_metadata(directory=g.collection_dir, name=gordo_name)
+        except FileNotFoundError:
+            raise NotFound(f"No metadata found for '{gordo_name}'")
+        return f(*args, **kwargs)
+
+    return wrapper
+
+
+def model_required(f):
+    @functools.wraps(f)
+    def wrapper(*args, gordo_project: str, gordo_name: str, **kwargs):
+        validate_gordo_name(gordo_name)
+        try:
+            check_metadata_file(g.collection_dir, gordo_name)
+            g.model = load_model(directory=g.collection_dir, name=gordo_name)
+        except FileNotFoundError:
+            raise NotFound(f"No such model found: '{gordo_name}'")
+        return metadata_required(f)(
+            *args, gordo_project=gordo_project, gordo_name=gordo_name, **kwargs
+        )
+
+    return wrapper
