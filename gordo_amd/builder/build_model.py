@@ -297,28 +297,27 @@ class ModelBuilder:
     ) -> dict:
         """Recurse the estimator graph bottom-up collecting
         get_metadata() dicts (reference build_model.py:516-573)."""
-        metadata = metadata if metadata is not None else {}
+        metadata = dict(metadata) if metadata is not None else {}
+        # a Pipeline's metadata is its final step's
         if isinstance(model, Pipeline):
-            return ModelBuilder._extract_metadata_from_model(
-                model.steps[-1][1], metadata
+            metadata.update(
+                ModelBuilder._extract_metadata_from_model(model.steps[-1][1])
             )
+            return metadata
         if isinstance(model, GordoBase):
             metadata.update(model.get_metadata())
-            return metadata
-        if hasattr(model, "get_params"):
-            for key, val in model.get_params(deep=False).items():
-                if key == "regressor":
-                    continue
-                if isinstance(val, Pipeline):
-                    metadata.update(
-                        ModelBuilder._extract_metadata_from_model(
-                            val.steps[-1][1]
-                        )
-                    )
-                elif isinstance(val, (GordoBase, BaseEstimator)):
-                    metadata.update(
-                        ModelBuilder._extract_metadata_from_model(val)
-                    )
+        # keep recursing into attributes: a GordoBase (e.g. the anomaly
+        # detector) may hold another estimator (its base_estimator) whose
+        # metadata must also surface (reference build_model.py:552-573)
+        for key, val in vars(model).items():
+            if key == "regressor":
+                continue
+            if isinstance(val, Pipeline):
+                metadata.update(
+                    ModelBuilder._extract_metadata_from_model(val.steps[-1][1])
+                )
+            elif isinstance(val, (GordoBase, BaseEstimator)):
+                metadata.update(ModelBuilder._extract_metadata_from_model(val))
         return metadata
 
     # ------------------------------------------------------------------

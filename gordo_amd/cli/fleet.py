@@ -1,0 +1,137 @@
+"""
+``gordo fleet`` — the MI355X-native single-node fan-out.
+
+No reference analog: this replaces the reference's pod-per-model Argo
+workflow with one process per GPU over RCCL/xGMI on a single node
+(SURVEY.md §1 "local fan-out" row). ``gordo fleet build`` builds every
+Machine in a config, packing same-architecture models into grouped
+MFMA GEMM batches per GPU; with --gpus N it relaunches itself under
+``torch.distributed.run`` with one rank per GPU.
+"""
+from __future__ import annotations
+
+import json
+import logging
+import os
+import subprocess
+import sys
+import time
+from typing import Optional
+
+import click
+
+logger = logging.getLogger(__name__)
+
+
+@click.group("fleet")
+def fleet_cli():
+    """Single-node multi-GPU fleet operations (MI355X-native)."""
+
+
+@fleet_cli.command("build")
+@click.option("--machine-config", type=click.Path(exists=True), required=True,
+              envvar="GORDO_FLEET_MACHINE_CONFIG")
+@click.option("--project-name", type=str, required=True,
+              envvar="GORDO_FLEET_PROJECT_NAME")
+@click.option("--output-dir", type=click.Path(), required=True,
+              envvar="GORDO_FLEET_OUTPUT_DIR",
+              help="Model collection dir; models land at <output-dir>/<name>/")
+@click.option("--model-register-dir", type=click.Path(), default=None,
+              envvar="GORDO_FLEET_MODEL_REGISTER_DIR")
+@click.option("--gpus", type=int, default=0,
+              help="Number of GPUs (0 = auto: all visible; runs in-process "
+                   "when 0/1)")
+@click.option("--replace-cache", is_flag=True)
+@click.option("--status-file", type=click.Path(), default=None,
+              help="Write per-machine build status JSON here (rank 0)")
+def fleet_build(
+    machine_config: str,
+    project_name: str,
+    output_dir: str,
+    model_register_dir: Optional[str],
+    gpus: int,
+    replace_cache: bool,
+    status_file: Optional[str],
+):
+    """Build every Machine in the config across the node's GPUs."""
+    import torch
+
+    in_torchrun = "RANK" in os.environ
+    if not in_torchrun:
+        if gpus == 0:
+            gpus = torch.cuda.device_count() if torch.cuda.is_available() else 1
+        if gpus > 1:
+            # relaunch under torch.distributed.run, one rank per GPU
+            cmd = [
+                sys.executable, "-m", "torch.distributed.run",
+                "--nnodes=1", f"--nproc-per-node={gpus}",
+                "--master-addr", "127.0.0.1",
+                "--master-port", os.environ.get("MASTER_PORT", "29517"),
+                "-m", "gordo_amd.cli.fleet_worker",
+            ]
+            env = dict(os.environ)
+            env.update(
+                GORDO_FLEET_MACHINE_CONFIG=machine_config,
+                GORDO_FLEET_PROJECT_NAME=project_name,
+                GORDO_FLEET_OUTPUT_DIR=output_dir,
+                GORDO_FLEET_REPLACE_CACHE="1" if replace_cache else "0",
+            )
+            if model_register_dir:
+                env["GORDO_FLEET_MODEL_REGISTER_DIR"] = model_register_dir
+            if status_file:
+                env["GORDO_FLEET_STATUS_FILE"] = status_file
+            raise SystemExit(subprocess.call(cmd, env=env))
+
+    _run_fleet_build(
+        machine_config, project_name, output_dir, model_register_dir,
+        replace_cache, status_file,
+    )
+
+
+def _run_fleet_build(
+    machine_config: str,
+    project_name: str,
+    output_dir: str,
+    model_register_dir: Optional[str],
+    replace_cache: bool,
+    status_file: Optional[str],
+):
+    from ..parallel import build_fleet, init_distributed
+    from ..workflow import NormalizedConfig
+    from ..workflow.workflow_generator import get_dict_from_yaml
+
+    rank, world = init_distributed()
+    config = get_dict_from_yaml(machine_config)
+    norm = NormalizedConfig(config, project_name=project_name)
+    t0 = time.time()
+    status = build_fleet(
+        norm.machines,
+        output_dir=output_dir,
+        model_register_dir=model_register_dir,
+        replace_cache=replace_cache,
+    )
+    elapsed = time.time() - t0
+    if rank == 0:
+        n_ok = sum(1 for _, err in status if err is None)
+        summary = {
+            "project": project_name,
+            "n_machines": len(norm.machines),
+            "n_ok": n_ok,
+            "n_failed": len(status) - n_ok,
+            "elapsed_sec": elapsed,
+            "machines_per_hour": len(norm.machines) / elapsed * 3600.0,
+            "world_size": world,
+            "status": dict(status),
+        }
+        logger.info(
+            "Fleet build: %d/%d machines OK in %.1fs (%.0f machines/hour, %d ranks)",
+            n_ok, len(norm.machines), elapsed,
+            summary["machines_per_hour"], world,
+        )
+        if status_file:
+            with open(status_file, "w") as f:
+                json.dump(summary, f, indent=2, default=str)
+        if n_ok < len(status):
+            for name, err in status:
+                if err is not None:
+                    logger.error("Machine %s failed: %s", name, err)

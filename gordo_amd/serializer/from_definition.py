@@ -78,8 +78,13 @@ def create_instance(cls, **params):
     if sig is not None:
         for name, value in list(params.items()):
             if isinstance(value, list) and name in sig.parameters:
-                ann = sig.parameters[name].annotation
-                if ann is not inspect.Parameter.empty and is_tuple_type(ann):
+                param = sig.parameters[name]
+                ann = param.annotation
+                # coerce when the annotation OR the default says tuple
+                # (YAML/JSON have no tuples; sklearn validates strictly)
+                if (
+                    ann is not inspect.Parameter.empty and is_tuple_type(ann)
+                ) or isinstance(param.default, tuple):
                     params[name] = tuple(value)
     return cls(**params)
 

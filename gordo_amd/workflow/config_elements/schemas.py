@@ -6,7 +6,7 @@ from __future__ import annotations
 
 from typing import Any, Dict, List, Optional
 
-from pydantic import BaseModel
+from pydantic import BaseModel, ConfigDict
 
 
 class EnvVar(BaseModel):
@@ -46,8 +46,7 @@ class PodRuntime(BaseModel):
     image: Optional[str] = None
     resources: Optional[ResourceRequirements] = None
 
-    class Config:
-        extra = "allow"
+    model_config = ConfigDict(extra="allow")
 
 
 class RemoteLogging(BaseModel):
@@ -65,8 +64,7 @@ class SecurityContext(BaseModel):
     runAsNonRoot: Optional[bool] = None
     allowPrivilegeEscalation: Optional[bool] = None
 
-    class Config:
-        extra = "allow"
+    model_config = ConfigDict(extra="allow")
 
 
 class PodSecurityContext(SecurityContext):

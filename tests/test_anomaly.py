@@ -1,0 +1,160 @@
+import numpy as np
+import pandas as pd
+import pytest
+from sklearn.pipeline import Pipeline
+from sklearn.preprocessing import MinMaxScaler
+
+from gordo_amd.machine.model.anomaly.diff import (
+    DiffBasedAnomalyDetector,
+    DiffBasedKFCVAnomalyDetector,
+)
+from gordo_amd.machine.model.models import KerasAutoEncoder
+
+
+@pytest.fixture(scope="module")
+def Xy():
+    rng = np.random.default_rng(42)
+    t = np.linspace(0, 30, 400)
+    X = pd.DataFrame(
+        {f"tag-{i}": np.sin(t + i) + rng.normal(0, 0.05, len(t)) for i in range(4)}
+    )
+    return X, X.copy()
+
+
+def _detector(**kwargs):
+    return DiffBasedAnomalyDetector(
+        base_estimator=Pipeline(
+            [
+                ("mms", MinMaxScaler()),
+                ("ae", KerasAutoEncoder(kind="feedforward_hourglass", epochs=1)),
+            ]
+        ),
+        **kwargs,
+    )
+
+
+def test_fit_and_anomaly_columns(Xy):
+    X, y = Xy
+    det = _detector(require_thresholds=False)
+    det.fit(X, y)
+    frame = det.anomaly(X, y, frequency=pd.Timedelta(minutes=10))
+    tops = {c[0] for c in frame.columns}
+    assert {
+        "start", "end", "model-input", "model-output",
+        "tag-anomaly-scaled", "total-anomaly-scaled",
+        "tag-anomaly-unscaled", "total-anomaly-unscaled",
+    } <= tops
+    # without cross_validate, no confidence columns
+    assert "anomaly-confidence" not in tops
+    assert "total-anomaly-confidence" not in tops
+
+    # verify scaled tag anomaly against hand-computed pandas
+    pred = det.base_estimator.predict(X.values)
+    expected = np.abs(det.scaler.transform(pred) - det.scaler.transform(y))
+    np.testing.assert_allclose(
+        frame["tag-anomaly-scaled"].values, expected, rtol=1e-5, atol=1e-6
+    )
+    expected_total = (expected ** 2).mean(axis=1)
+    np.testing.assert_allclose(
+        frame["total-anomaly-scaled"].values, expected_total, rtol=1e-5,
+        atol=1e-6,
+    )
+
+
+def test_require_thresholds_raises(Xy):
+    X, y = Xy
+    det = _detector(require_thresholds=True)
+    det.fit(X, y)
+    with pytest.raises(AttributeError):
+        det.anomaly(X, y)
+
+
+def test_cross_validate_sets_thresholds(Xy):
+    X, y = Xy
+    det = _detector(require_thresholds=True)
+    det.cross_validate(X=X, y=y)
+    det.fit(X, y)
+    assert hasattr(det, "aggregate_threshold_")
+    assert hasattr(det, "feature_thresholds_")
+    assert len(det.feature_thresholds_) == X.shape[1]
+    assert len(det.aggregate_thresholds_per_fold_) == 3  # TimeSeriesSplit(3)
+    frame = det.anomaly(X, y, frequency=pd.Timedelta(minutes=10))
+    tops = {c[0] for c in frame.columns}
+    assert "anomaly-confidence" in tops
+    assert "total-anomaly-confidence" in tops
+    np.testing.assert_allclose(
+        frame["total-anomaly-confidence"].values,
+        frame["total-anomaly-scaled"].values / det.aggregate_threshold_,
+        rtol=1e-6,
+    )
+
+
+def test_smoothing_columns(Xy):
+    X, y = Xy
+    det = _detector(require_thresholds=False, window=10)
+    assert det.smoothing_method == "smm"  # default when window given
+    det.fit(X, y)
+    frame = det.anomaly(X, y)
+    tops = {c[0] for c in frame.columns}
+    assert "smooth-tag-anomaly-scaled" in tops
+    assert "smooth-total-anomaly-scaled" in tops
+    # smm == rolling median of the raw series
+    raw = frame["total-anomaly-scaled"]
+    np.testing.assert_allclose(
+        frame["smooth-total-anomaly-scaled"].values,
+        raw.rolling(10).median().values,
+        rtol=1e-6, equal_nan=True,
+    )
+
+
+@pytest.mark.parametrize("method,window", [("sma", 5), ("ewma", 5)])
+def test_smoothing_methods(Xy, method, window):
+    X, y = Xy
+    det = _detector(require_thresholds=False, window=window,
+                    smoothing_method=method)
+    det.fit(X, y)
+    frame = det.anomaly(X, y)
+    raw = frame["total-anomaly-scaled"]
+    if method == "sma":
+        expected = raw.rolling(window).mean()
+    else:
+        expected = raw.ewm(span=window).mean()
+    np.testing.assert_allclose(
+        frame["smooth-total-anomaly-scaled"].values, expected.values,
+        rtol=1e-6, equal_nan=True,
+    )
+
+
+def test_kfcv_detector(Xy):
+    X, y = Xy
+    det = DiffBasedKFCVAnomalyDetector(
+        base_estimator=Pipeline(
+            [
+                ("mms", MinMaxScaler()),
+                ("ae", KerasAutoEncoder(kind="feedforward_hourglass", epochs=1)),
+            ]
+        ),
+        window=20,
+        threshold_percentile=0.99,
+    )
+    det.cross_validate(X=X, y=y)
+    det.fit(X, y)
+    assert np.isfinite(det.aggregate_threshold_)
+    assert len(det.feature_thresholds_) == X.shape[1]
+    frame = det.anomaly(X, y)
+    assert "total-anomaly-confidence" in {c[0] for c in frame.columns}
+
+
+def test_getattr_passthrough():
+    det = _detector()
+    # attribute of the base estimator reachable through the detector
+    assert det.steps[1][0] == "ae"
+
+
+def test_get_params_roundtrip():
+    det = _detector(require_thresholds=False, window=12)
+    params = det.get_params()
+    assert params["window"] == 12
+    assert params["smoothing_method"] == "smm"
+    det2 = DiffBasedAnomalyDetector(**params)
+    assert det2.window == 12

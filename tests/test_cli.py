@@ -1,0 +1,214 @@
+import json
+import os
+
+import pytest
+import yaml
+from click.testing import CliRunner
+
+from gordo_amd.cli import gordo
+from gordo_amd.cli.cli import expand_model, get_all_score_strings
+from gordo_amd.cli.exceptions_reporter import ExceptionsReporter, ReportLevel
+
+
+MACHINE_JSON = {
+    "name": "cli-machine",
+    "project_name": "cli-proj",
+    "dataset": {
+        "type": "RandomDataset",
+        "tag_list": ["a", "b"],
+        "train_start_date": "2019-01-01T00:00:00+00:00",
+        "train_end_date": "2019-01-02T00:00:00+00:00",
+    },
+    "model": {
+        "sklearn.pipeline.Pipeline": {
+            "steps": [
+                "sklearn.preprocessing.MinMaxScaler",
+                {"sklearn.decomposition.PCA": {"n_components": 2}},
+            ]
+        }
+    },
+}
+
+
+def test_build_command(tmp_path):
+    out_dir = tmp_path / "out"
+    runner = CliRunner()
+    result = runner.invoke(
+        gordo,
+        ["build", json.dumps(MACHINE_JSON), str(out_dir), "--print-cv-scores"],
+    )
+    assert result.exit_code == 0, result.output
+    assert (out_dir / "model.pkl").is_file()
+    assert "explained-variance-score" in result.output
+
+
+def test_build_err_name_exit_code(tmp_path):
+    """The reference's fault-injection hook: 'err' in machine name →
+    FileNotFoundError → exit code 30 + exceptions report."""
+    cfg = dict(MACHINE_JSON, name="err-machine")
+    report_file = tmp_path / "report.json"
+    runner = CliRunner()
+    result = runner.invoke(
+        gordo,
+        [
+            "build", json.dumps(cfg), str(tmp_path / "out"),
+            "--exceptions-reporter-file", str(report_file),
+        ],
+    )
+    assert result.exit_code == 30
+    report = json.loads(report_file.read_text())
+    assert report["exit_code"] == 30
+    assert report["type"] == "FileNotFoundError"
+
+
+def test_build_bad_config_exit_code(tmp_path):
+    cfg = dict(MACHINE_JSON, model={"no.such.Model": {}})
+    runner = CliRunner()
+    result = runner.invoke(gordo, ["build", json.dumps(cfg), str(tmp_path)])
+    assert result.exit_code == 2  # ValueError from model validation
+
+
+def test_expand_model():
+    expanded = expand_model(
+        "sklearn.decomposition.PCA:\n  n_components: {{ n }}", {"n": 4}
+    )
+    assert expanded == {"sklearn.decomposition.PCA": {"n_components": 4}}
+    with pytest.raises(ValueError):
+        expand_model("x: {{ missing }}", {})
+
+
+def test_get_all_score_strings():
+    class FakeCV:
+        scores = {"r2-score": {"fold-mean": 0.5, "fold-1": 0.4}}
+
+    class FakeModelMeta:
+        cross_validation = FakeCV()
+
+    class FakeBM:
+        model = FakeModelMeta()
+
+    class FakeMeta:
+        build_metadata = FakeBM()
+
+    class FakeMachine:
+        metadata = FakeMeta()
+
+    scores = get_all_score_strings(FakeMachine())
+    assert "r2-score_fold-mean=0.5" in scores
+
+
+def test_exceptions_reporter_codes():
+    reporter = ExceptionsReporter(
+        ((Exception, 1), (ValueError, 2), (FileNotFoundError, 30))
+    )
+    assert reporter.exception_exit_code(ValueError) == 2
+    assert reporter.exception_exit_code(FileNotFoundError) == 30
+    # subclass resolves to nearest registered ancestor
+    class MyError(ValueError):
+        pass
+
+    assert reporter.exception_exit_code(MyError) == 2
+    assert reporter.exception_exit_code(KeyError) == 1
+    assert reporter.exception_exit_code(None) == 0
+
+
+def test_exceptions_reporter_trim():
+    assert ExceptionsReporter.trim_message("x" * 100, 10) == "x" * 7 + "..."
+    assert ExceptionsReporter.trim_message("short", 10) == "short"
+
+
+def test_workflow_generate(tmp_path):
+    cfg = tmp_path / "cfg.yml"
+    cfg.write_text(
+        """
+machines:
+  - name: wf-m-1
+    dataset: |
+      type: RandomDataset
+      tag_list: [a, b]
+      train_start_date: '2019-01-01T00:00:00+00:00'
+      train_end_date: '2019-01-02T00:00:00+00:00'
+    model: |
+      sklearn.decomposition.PCA:
+        n_components: 2
+"""
+    )
+    runner = CliRunner()
+    result = runner.invoke(
+        gordo,
+        [
+            "workflow", "generate",
+            "--machine-config", str(cfg),
+            "--project-name", "wf-proj",
+        ],
+    )
+    assert result.exit_code == 0, result.output
+    docs = list(yaml.safe_load_all(result.output))
+    assert docs[0]["kind"] == "Workflow"
+    names = {t["name"] for t in docs[0]["spec"]["templates"]}
+    assert {"do-all", "model-builder", "gordo-server"} <= names
+
+
+def test_workflow_generate_gpu_fleet(tmp_path):
+    cfg = tmp_path / "cfg.yml"
+    cfg.write_text(
+        """
+machines:
+  - name: wf-m-1
+    dataset: |
+      type: RandomDataset
+      tag_list: [a, b]
+      train_start_date: '2019-01-01T00:00:00+00:00'
+      train_end_date: '2019-01-02T00:00:00+00:00'
+    model: |
+      sklearn.decomposition.PCA:
+        n_components: 2
+"""
+    )
+    runner = CliRunner()
+    result = runner.invoke(
+        gordo,
+        [
+            "workflow", "generate",
+            "--machine-config", str(cfg),
+            "--project-name", "wf-proj",
+            "--gpu-fleet", "--n-gpus", "8",
+        ],
+    )
+    assert result.exit_code == 0, result.output
+    doc = list(yaml.safe_load_all(result.output))[0]
+    tasks = {
+        t["name"]
+        for t in doc["spec"]["templates"][0]["dag"]["tasks"]
+    }
+    assert "fleet-build" in tasks
+
+
+def test_workflow_split(tmp_path):
+    machines = "\n".join(
+        f"""  - name: wf-m-{i}
+    dataset: |
+      type: RandomDataset
+      tag_list: [a, b]
+      train_start_date: '2019-01-01T00:00:00+00:00'
+      train_end_date: '2019-01-02T00:00:00+00:00'
+    model: |
+      sklearn.decomposition.PCA:
+        n_components: 2"""
+        for i in range(5)
+    )
+    cfg = tmp_path / "cfg.yml"
+    cfg.write_text("machines:\n" + machines)
+    runner = CliRunner()
+    result = runner.invoke(
+        gordo,
+        [
+            "workflow", "generate",
+            "--machine-config", str(cfg),
+            "--project-name", "wf-proj",
+            "--split-workflows", "2",
+        ],
+    )
+    assert result.exit_code == 0, result.output
+    docs = list(yaml.safe_load_all(result.output))
+    assert len(docs) == 3  # 5 machines / 2 per workflow

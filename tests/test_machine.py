@@ -1,0 +1,140 @@
+import pytest
+
+from gordo_amd.core import RandomDataset, SensorTag, normalize_sensor_tag
+from gordo_amd.machine import Machine, load_machine_config, load_model_config
+from gordo_amd.machine.loader import MachineConfigException
+from gordo_amd.machine.validators import ValidUrlString, fix_resource_limits
+
+
+MODEL_DEF = {"sklearn.decomposition.PCA": {"n_components": 2}}
+DATASET_DEF = {
+    "type": "RandomDataset",
+    "train_start_date": "2017-12-25 06:00:00Z",
+    "train_end_date": "2017-12-30 06:00:00Z",
+    "tag_list": ["Tag 1", "Tag 2"],
+}
+
+
+def _machine(**overrides):
+    cfg = dict(name="test-machine", model=MODEL_DEF, dataset=dict(DATASET_DEF))
+    cfg.update(overrides)
+    return Machine.from_config(cfg, project_name="test-proj")
+
+
+def test_from_config_roundtrip():
+    m = _machine()
+    assert m.name == "test-machine"
+    assert m.project_name == "test-proj"
+    assert m.evaluation["cv_mode"] == "full_build"
+    d = m.to_dict()
+    m2 = Machine.from_dict(d)
+    assert m2 == m
+    # yaml/json roundtrips parse
+    assert "test-machine" in m.to_yaml()
+    assert "test-machine" in m.to_json()
+
+
+def test_globals_merge_precedence():
+    config_globals = {
+        "model": {"sklearn.decomposition.PCA": {"n_components": 5}},
+        "runtime": {"server": {"resources": {"requests": {"memory": 1}}}},
+        "evaluation": {"cv_mode": "cross_val_only", "seed": 7},
+        "dataset": {"resolution": "1h"},
+    }
+    m = Machine.from_config(
+        {
+            "name": "m",
+            "dataset": dict(DATASET_DEF),
+            "model": MODEL_DEF,
+            "evaluation": {"cv_mode": "full_build"},
+        },
+        project_name="p",
+        config_globals=config_globals,
+    )
+    # machine model wins over globals model
+    assert m.model == MODEL_DEF
+    # machine evaluation overrides globals, but globals fill gaps
+    assert m.evaluation["cv_mode"] == "full_build"
+    assert m.evaluation["seed"] == 7
+    # reference quirk: globals dataset overlays machine's
+    assert m.dataset.resolution == "1h"
+
+
+def test_invalid_name_rejected():
+    with pytest.raises(ValueError):
+        _machine(name="Bad_Name")
+    with pytest.raises(ValueError):
+        _machine(name="x" * 64)
+
+
+def test_invalid_model_rejected():
+    with pytest.raises(ValueError):
+        _machine(model={"not.a.real.Class": {}})
+
+
+def test_nested_yaml_fields():
+    cfg = load_machine_config(
+        {
+            "name": "m",
+            "model": "sklearn.decomposition.PCA:\n  n_components: 2\n",
+            "dataset": "type: RandomDataset\ntag_list: [a]\n"
+                       "train_start_date: 2019-01-01T00:00:00Z\n"
+                       "train_end_date: 2019-01-02T00:00:00Z\n",
+        }
+    )
+    assert cfg["model"] == {"sklearn.decomposition.PCA": {"n_components": 2}}
+    assert cfg["dataset"]["type"] == "RandomDataset"
+
+
+def test_load_model_config_requires_model():
+    with pytest.raises(MachineConfigException):
+        load_model_config({"name": "m"})
+    with pytest.raises(MachineConfigException):
+        load_machine_config({"model": MODEL_DEF})
+
+
+def test_valid_url_string():
+    assert ValidUrlString.valid_url_string("good-name-01")
+    assert not ValidUrlString.valid_url_string("Bad Name")
+    assert not ValidUrlString.valid_url_string("-leading")
+
+
+def test_fix_resource_limits():
+    res = fix_resource_limits(
+        {"requests": {"memory": 10, "cpu": 5}, "limits": {"memory": 4, "cpu": 9}}
+    )
+    assert res["limits"]["memory"] == 10
+    assert res["limits"]["cpu"] == 9
+
+
+def test_sensor_tag_normalization():
+    assert normalize_sensor_tag("a").name == "a"
+    assert normalize_sensor_tag({"name": "a", "asset": "x"}).asset == "x"
+    assert normalize_sensor_tag(["a", "x"]).asset == "x"
+    assert normalize_sensor_tag(SensorTag("a", "x")) == SensorTag("a", "x")
+
+
+def test_dataset_get_data_shapes():
+    ds = RandomDataset(
+        train_start_date="2019-01-01T00:00:00Z",
+        train_end_date="2019-01-02T00:00:00Z",
+        tag_list=["a", "b", "c"],
+        target_tag_list=["a", "b"],
+    )
+    X, y = ds.get_data()
+    assert list(X.columns) == ["a", "b", "c"]
+    assert list(y.columns) == ["a", "b"]
+    assert len(X) == len(y) > 0
+    meta = ds.get_metadata()
+    assert meta["resolution"] == "10min"
+
+
+def test_dataset_requires_timezone():
+    from gordo_amd.core.exceptions import ConfigException
+
+    with pytest.raises(ConfigException):
+        RandomDataset(
+            train_start_date="2019-01-01T00:00:00",
+            train_end_date="2019-01-02T00:00:00",
+            tag_list=["a"],
+        )

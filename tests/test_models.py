@@ -1,0 +1,163 @@
+import pickle
+
+import numpy as np
+import pytest
+
+from gordo_amd.machine.model import (
+    KerasAutoEncoder,
+    KerasLSTMAutoEncoder,
+    KerasLSTMForecast,
+    create_keras_timeseriesgenerator,
+)
+from gordo_amd.machine.model.factories import (
+    feedforward_hourglass,
+    feedforward_model,
+    lstm_hourglass,
+    lstm_model,
+)
+from gordo_amd.machine.model.factories.utils import hourglass_calc_dims
+from gordo_amd.machine.model.register import register_model_builder
+
+
+@pytest.fixture
+def data():
+    rng = np.random.default_rng(0)
+    return rng.random((240, 6)).astype("float32")
+
+
+def test_hourglass_dims():
+    assert hourglass_calc_dims(0.5, 3, 10) == (8, 7, 5)
+    assert hourglass_calc_dims(0.2, 3, 5) == (4, 2, 1)
+    assert hourglass_calc_dims(1.0, 3, 10) == (10, 10, 10)
+    with pytest.raises(ValueError):
+        hourglass_calc_dims(1.5, 3, 10)
+    with pytest.raises(ValueError):
+        hourglass_calc_dims(0.5, 0, 10)
+
+
+def test_factory_specs():
+    spec = feedforward_hourglass(10)
+    assert [l.units for l in spec.layers] == [8, 7, 5, 5, 7, 8, 10]
+    # l1 on encoder layers except the first
+    assert spec.layers[0].l1_activity == 0.0
+    assert spec.layers[1].l1_activity == pytest.approx(1e-4)
+    assert spec.layers[3].l1_activity == 0.0  # decoder
+    spec = lstm_model(5, lookback_window=7, encoding_dim=(4,),
+                      encoding_func=("tanh",), decoding_dim=(4,),
+                      decoding_func=("tanh",))
+    lstm_layers = [l for l in spec.layers if l.kind == "lstm"]
+    assert [l.return_sequences for l in lstm_layers] == [True, False]
+    assert spec.lookback_window == 7
+
+
+def test_factory_dim_mismatch():
+    with pytest.raises(ValueError):
+        feedforward_model(10, encoding_dim=(4, 2), encoding_func=("tanh",))
+
+
+def test_register_model_builder_validates():
+    with pytest.raises(ValueError):
+        @register_model_builder(type="KerasAutoEncoder")
+        def bad_builder(x):  # no n_features
+            pass
+
+
+def test_autoencoder_fit_predict_score(data):
+    model = KerasAutoEncoder(kind="feedforward_hourglass", epochs=2,
+                             batch_size=32)
+    model.fit(data, data)
+    out = model.predict(data)
+    assert out.shape == data.shape
+    score = model.score(data, data)
+    assert np.isfinite(score)
+    meta = model.get_metadata()
+    assert "history" in meta
+    assert "loss" in meta["history"]
+    assert len(meta["history"]["loss"]) == 2
+    assert "params" in meta["history"]
+
+
+def test_autoencoder_training_reduces_loss(data):
+    model = KerasAutoEncoder(kind="feedforward_hourglass", epochs=15,
+                             batch_size=32)
+    # reconstruct a low-rank signal: loss must drop substantially
+    t = np.linspace(0, 20, len(data))
+    signal = np.stack([np.sin(t + p) for p in np.linspace(0, 1, 6)], axis=1)
+    model.fit(signal.astype("float32"))
+    losses = model.history["loss"]
+    assert losses[-1] < losses[0] * 0.5
+
+
+def test_autoencoder_pickle_roundtrip(data):
+    model = KerasAutoEncoder(kind="feedforward_hourglass", epochs=1)
+    model.fit(data)
+    out1 = model.predict(data)
+    model2 = pickle.loads(pickle.dumps(model))
+    out2 = model2.predict(data)
+    np.testing.assert_allclose(out1, out2, rtol=1e-5, atol=1e-6)
+    assert model2.history == model.history
+
+
+def test_lstm_autoencoder_offset(data):
+    model = KerasLSTMAutoEncoder(kind="lstm_hourglass", lookback_window=12,
+                                 epochs=1)
+    model.fit(data)
+    out = model.predict(data)
+    assert out.shape == (len(data) - 12 + 1, data.shape[1])
+    assert np.isfinite(model.score(data, data))
+
+
+def test_lstm_forecast_offset(data):
+    model = KerasLSTMForecast(kind="lstm_symmetric", lookback_window=8,
+                              epochs=1, dims=(4,), funcs=("tanh",))
+    model.fit(data)
+    out = model.predict(data)
+    assert out.shape == (len(data) - 8, data.shape[1])
+    assert model.get_metadata()["forecast_steps"] == 1
+
+
+def test_lstm_requires_enough_rows():
+    model = KerasLSTMAutoEncoder(kind="lstm_hourglass", lookback_window=100)
+    with pytest.raises(ValueError):
+        model.fit(np.random.rand(50, 3))
+
+
+@pytest.mark.parametrize(
+    "rows,lookback,lookahead,expected_samples",
+    [(100, 20, 0, 81), (100, 20, 1, 80), (100, 20, 3, 78), (10, 2, 0, 9)],
+)
+def test_windower_shapes(rows, lookback, lookahead, expected_samples):
+    X = np.arange(rows * 2, dtype="float64").reshape(rows, 2)
+    y = X.copy()
+    gen = create_keras_timeseriesgenerator(X, y, 10, lookback, lookahead)
+    total = sum(len(gen[i][0]) for i in range(len(gen)))
+    assert total == expected_samples
+    bx, by = gen[0]
+    assert bx.shape[1:] == (lookback, 2)
+    # alignment: target is the row lookback-1+lookahead past window start
+    np.testing.assert_array_equal(bx[0], X[0:lookback])
+    np.testing.assert_array_equal(by[0], y[lookback - 1 + lookahead])
+
+
+def test_windower_negative_lookahead():
+    with pytest.raises(ValueError):
+        create_keras_timeseriesgenerator(np.zeros((10, 1)), None, 1, 2, -1)
+
+
+def test_unknown_kind_raises():
+    with pytest.raises(ValueError):
+        KerasAutoEncoder(kind="not_a_registered_kind")
+
+
+def test_callable_kind_registers():
+    def my_custom_ae(n_features, n_features_out=None, **kwargs):
+        return feedforward_model(
+            n_features, n_features_out, encoding_dim=(4,),
+            encoding_func=("tanh",), decoding_dim=(4,),
+            decoding_func=("tanh",),
+        )
+
+    model = KerasAutoEncoder(kind=my_custom_ae, epochs=1)
+    X = np.random.rand(64, 5).astype("float32")
+    model.fit(X)
+    assert model.predict(X).shape == X.shape

@@ -1,0 +1,245 @@
+"""Packed-engine semantics: a G-model pack must train each model
+exactly as a pack-of-1 does (grouping machines may never change
+results)."""
+import numpy as np
+import pytest
+import torch
+
+from gordo_amd.engine.pack import DensePack, LSTMPack
+from gordo_amd.engine.spec import LayerSpec, ModelSpec
+
+
+def dense_spec(n_features=5, units=(4, 3, 4)):
+    layers = [
+        LayerSpec(kind="dense", units=u, activation="tanh",
+                  l1_activity=(1e-4 if i == 1 else 0.0))
+        for i, u in enumerate(units)
+    ]
+    layers.append(LayerSpec(kind="dense", units=n_features, activation="linear"))
+    return ModelSpec(
+        model_type="feedforward", n_features=n_features,
+        n_features_out=n_features, layers=layers,
+    )
+
+
+def lstm_spec(n_features=4, H=6, lookback=5):
+    return ModelSpec(
+        model_type="lstm", n_features=n_features, n_features_out=n_features,
+        layers=[
+            LayerSpec(kind="lstm", units=H, return_sequences=True),
+            LayerSpec(kind="lstm", units=H, return_sequences=False),
+            LayerSpec(kind="dense", units=n_features, activation="linear"),
+        ],
+        lookback_window=lookback,
+    )
+
+
+def test_dense_pack_of_g_equals_packs_of_1():
+    spec = dense_spec()
+    G, N = 3, 64
+    rng = np.random.default_rng(0)
+    X = rng.random((G, N, spec.n_features)).astype("float32")
+
+    pack = DensePack(spec, G=G, device="cpu", seeds=[11, 22, 33])
+    Xt = torch.from_numpy(X)
+    hist = pack.fit(Xt, Xt.clone(), epochs=3, batch_size=16)
+
+    for g, seed in enumerate([11, 22, 33]):
+        solo = DensePack(spec, G=1, device="cpu", seeds=[seed])
+        Xg = torch.from_numpy(X[g : g + 1])
+        hist1 = solo.fit(Xg, Xg.clone(), epochs=3, batch_size=16)
+        # identical per-epoch losses and final weights
+        for e in range(3):
+            assert hist["loss"][e][g] == pytest.approx(
+                hist1["loss"][e][0], rel=1e-5
+            )
+        sg = pack.state_for_model(g)
+        s1 = solo.state_for_model(0)
+        for k in sg:
+            np.testing.assert_allclose(sg[k], s1[k], rtol=1e-5, atol=1e-7)
+
+
+def test_lstm_pack_of_g_equals_packs_of_1():
+    spec = lstm_spec()
+    G, N = 2, 48
+    rng = np.random.default_rng(1)
+    X = rng.random((G, N, spec.n_features)).astype("float32")
+
+    pack = LSTMPack(spec, G=G, device="cpu", seeds=[5, 6])
+    Xt = torch.from_numpy(X)
+    hist = pack.fit(Xt, Xt.clone(), epochs=2, batch_size=8)
+
+    for g, seed in enumerate([5, 6]):
+        solo = LSTMPack(spec, G=1, device="cpu", seeds=[seed])
+        Xg = torch.from_numpy(X[g : g + 1])
+        hist1 = solo.fit(Xg, Xg.clone(), epochs=2, batch_size=8)
+        for e in range(2):
+            assert hist["loss"][e][g] == pytest.approx(
+                hist1["loss"][e][0], rel=1e-4
+            )
+        sg = pack.state_for_model(g)
+        s1 = solo.state_for_model(0)
+        for k in sg:
+            np.testing.assert_allclose(sg[k], s1[k], rtol=1e-4, atol=1e-6)
+
+
+def test_dense_backward_matches_autograd():
+    """Engine gradients vs torch autograd on an identical model."""
+    spec = dense_spec(n_features=4, units=(3, 2, 3))
+    pack = DensePack(spec, G=1, device="cpu", seeds=[0])
+    X = torch.rand(1, 32, 4)
+    T = torch.rand(1, 32, 4)
+
+    # autograd replica
+    params = {}
+    for name in pack.param_names():
+        params[name] = pack.store.views[name].clone().requires_grad_(True)
+    a = X
+    acts = [a]
+    for i, (_fin, _fout, act, l1) in enumerate(pack.layer_meta):
+        z = torch.baddbmm(params[f"b{i}"].unsqueeze(1), a, params[f"W{i}"])
+        a = torch.tanh(z) if act == "tanh" else z
+        acts.append(a)
+    n = T.shape[1] * T.shape[2]
+    loss = ((a - T) ** 2).sum() / n
+    # add the L1 activity penalties the engine's backward implements
+    for i, (_fin, _fout, act, l1) in enumerate(pack.layer_meta):
+        if l1:
+            loss = loss + l1 * acts[i + 1].abs().sum() / 1.0
+    loss.backward()
+
+    # engine backward (no adam step: snapshot grads before update)
+    acts_e = pack.forward(X)
+    from gordo_amd import ops
+
+    lss, dA = ops.mse_bwd(acts_e[-1], T)
+    for i in range(len(pack.layer_meta) - 1, -1, -1):
+        _fin, _fout, act, l1 = pack.layer_meta[i]
+        dZ = ops.act_l1_bwd(dA, acts_e[i + 1], act, l1)
+        dW, db = ops.grouped_linear_wgrad(acts_e[i], dZ)
+        np.testing.assert_allclose(
+            dW.numpy(), params[f"W{i}"].grad.numpy(), rtol=1e-4, atol=1e-5
+        )
+        np.testing.assert_allclose(
+            db.numpy(), params[f"b{i}"].grad.numpy(), rtol=1e-4, atol=1e-5
+        )
+        if i > 0:
+            dA = ops.grouped_linear_bwd_data(dZ, pack.store.cviews[f"W{i}"])
+
+
+def test_lstm_backward_matches_autograd():
+    spec = lstm_spec(n_features=3, H=4, lookback=6)
+    pack = LSTMPack(spec, G=1, device="cpu", seeds=[7])
+    B, T_len = 8, 6
+    Xw = torch.rand(1, B, T_len, 3)
+    Tgt = torch.rand(1, B, 3)
+
+    params = {
+        name: pack.store.views[name].clone().requires_grad_(True)
+        for name in pack.param_names()
+    }
+
+    def autograd_forward():
+        seq = Xw
+        for li, (fin, H, rs) in enumerate(pack.lstm_meta):
+            h = torch.zeros(1, B, H)
+            c = torch.zeros(1, B, H)
+            hs = []
+            for t in range(T_len):
+                x_t = seq[:, :, t]
+                gates = (
+                    torch.baddbmm(
+                        params[f"bl{li}"].unsqueeze(1), x_t, params[f"Wx{li}"]
+                    )
+                    + torch.bmm(h, params[f"Wh{li}"])
+                )
+                i_g = torch.sigmoid(gates[..., 0 * H : 1 * H])
+                f_g = torch.sigmoid(gates[..., 1 * H : 2 * H])
+                g_g = torch.tanh(gates[..., 2 * H : 3 * H])
+                o_g = torch.sigmoid(gates[..., 3 * H : 4 * H])
+                c = f_g * c + i_g * g_g
+                h = o_g * torch.tanh(c)
+                hs.append(h)
+            seq = torch.stack(hs, dim=2)
+        h_last = seq[:, :, -1]
+        y = torch.baddbmm(params["bd"].unsqueeze(1), h_last, params["Wd"])
+        return y
+
+    y = autograd_forward()
+    n = Tgt.shape[1] * Tgt.shape[2]
+    loss = ((y - Tgt) ** 2).sum() / n
+    loss.backward()
+
+    # engine: run train_batch but capture grads (adam modifies params after)
+    lss = pack.train_batch(Xw, Tgt)
+    assert lss.item() == pytest.approx(loss.item(), rel=1e-4)
+    for name in pack.param_names():
+        np.testing.assert_allclose(
+            pack.store.gviews[name].numpy(),
+            params[name].grad.numpy(),
+            rtol=2e-3, atol=1e-5,
+        )
+
+
+def test_adam_matches_torch():
+    from gordo_amd.ops import reference as ref
+
+    torch.manual_seed(0)
+    p = torch.rand(100)
+    g = torch.randn(100)
+    m = torch.zeros(100)
+    v = torch.zeros(100)
+    p2 = p.clone().requires_grad_(True)
+    opt = torch.optim.Adam([p2], lr=0.01, betas=(0.9, 0.999), eps=1e-7)
+    for step in range(1, 4):
+        ref.adam_step(p, g, m, v, 0.01, 0.9, 0.999, 1e-7, step)
+        p2.grad = g.clone()
+        opt.step()
+    np.testing.assert_allclose(p.numpy(), p2.detach().numpy(), rtol=1e-5,
+                               atol=1e-7)
+
+
+def test_packed_fleet_builder_groups(tmp_path):
+    """PackedFleetBuilder over mixed archs: same results whether built
+    together or separately."""
+    from gordo_amd.parallel import PackedFleetBuilder
+    from gordo_amd.workflow import NormalizedConfig
+
+    def machine(name, tags=4):
+        return {
+            "name": name,
+            "dataset": {
+                "type": "SineWaveDataset",
+                "tag_list": [f"t-{j}" for j in range(tags)],
+                "train_start_date": "2019-01-01T00:00:00+00:00",
+                "train_end_date": "2019-01-03T00:00:00+00:00",
+            },
+            "model": {
+                "gordo_amd.machine.model.models.KerasAutoEncoder": {
+                    "kind": "feedforward_hourglass",
+                    "epochs": 2,
+                }
+            },
+            "evaluation": {"cv_mode": "full_build"},
+        }
+
+    cfg = {"machines": [machine(f"m-{i}") for i in range(3)]}
+    norm = NormalizedConfig(cfg, project_name="p")
+    fb = PackedFleetBuilder(norm.machines, save_models=False)
+    results = dict(fb.build_all())
+    assert all(not isinstance(v, BaseException) for v in results.values())
+
+    # group build == solo build, model for model
+    norm2 = NormalizedConfig(
+        {"machines": [machine("m-1")]}, project_name="p"
+    )
+    fb_solo = PackedFleetBuilder(norm2.machines, save_models=False)
+    solo = dict(fb_solo.build_all())
+    m_group = results["m-1"]
+    m_solo = solo["m-1"]
+    sg = m_group.metadata.build_metadata.model.cross_validation.scores
+    ss = m_solo.metadata.build_metadata.model.cross_validation.scores
+    for key in ss:
+        assert sg[key]["fold-mean"] == pytest.approx(
+            ss[key]["fold-mean"], rel=1e-3, abs=1e-5
+        )

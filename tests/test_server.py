@@ -1,0 +1,188 @@
+import json
+import os
+
+import numpy as np
+import pandas as pd
+import pytest
+
+from gordo_amd.server import utils as server_utils
+
+
+@pytest.fixture
+def X(sensors):
+    rng = np.random.default_rng(1)
+    index = pd.date_range("2019-01-01", periods=50, freq="10min", tz="UTC")
+    return pd.DataFrame(
+        rng.random((50, len(sensors))), columns=sensors, index=index
+    )
+
+
+def _post_json(client, url, X, y=None, **params):
+    payload = {"X": server_utils.dataframe_to_dict(X)}
+    if y is not None:
+        payload["y"] = server_utils.dataframe_to_dict(y)
+    return client.post(url, json=payload, query_string=params)
+
+
+def test_healthcheck(api_client):
+    resp = api_client.get("/healthcheck")
+    assert resp.status_code == 200
+
+
+def test_server_version(api_client):
+    resp = api_client.get("/server-version")
+    assert resp.status_code == 200
+    assert "version" in resp.json
+
+
+def test_metadata_endpoint(api_client, base_route, gordo_name):
+    resp = api_client.get(f"{base_route}/metadata")
+    assert resp.status_code == 200
+    body = resp.json
+    assert body["metadata"]["name"] == gordo_name
+    assert "gordo-server-version" in body
+    assert "revision" in body
+    # /healthcheck alias
+    resp2 = api_client.get(f"{base_route}/healthcheck")
+    assert resp2.status_code == 200
+
+
+def test_models_list(api_client, gordo_project, gordo_name, second_gordo_name):
+    resp = api_client.get(f"/gordo/v0/{gordo_project}/models")
+    assert resp.status_code == 200
+    assert set(resp.json["models"]) >= {gordo_name, second_gordo_name}
+
+
+def test_revisions(api_client, gordo_project, gordo_revision):
+    resp = api_client.get(f"/gordo/v0/{gordo_project}/revisions")
+    assert resp.status_code == 200
+    assert resp.json["latest"] == gordo_revision
+    assert gordo_revision in resp.json["available-revisions"]
+
+
+def test_expected_models(api_client, gordo_project):
+    resp = api_client.get(f"/gordo/v0/{gordo_project}/expected-models")
+    assert resp.status_code == 200
+    assert "expected-models" in resp.json
+
+
+def test_prediction_json(api_client, base_route, X):
+    resp = _post_json(api_client, f"{base_route}/prediction", X)
+    assert resp.status_code == 200
+    data = resp.json["data"]
+    assert "model-input" in data and "model-output" in data
+    assert len(data["model-output"]) == X.shape[1]
+    assert resp.headers.get("revision")
+
+
+def test_prediction_parquet(api_client, base_route, X):
+    import io
+
+    blob = server_utils.dataframe_into_parquet_bytes(X)
+    resp = api_client.post(
+        f"{base_route}/prediction?format=parquet",
+        data={"X": (io.BytesIO(blob), "X")},
+        content_type="multipart/form-data",
+    )
+    assert resp.status_code == 200
+    frame = server_utils.dataframe_from_parquet_bytes(resp.data)
+    assert "model-output" in {c[0] for c in frame.columns}
+    assert len(frame) == len(X)
+
+
+def test_prediction_without_x(api_client, base_route):
+    resp = api_client.post(f"{base_route}/prediction", json={})
+    assert resp.status_code == 400
+
+
+def test_prediction_wrong_columns(api_client, base_route, X):
+    bad = X.iloc[:, :2]
+    resp = _post_json(api_client, f"{base_route}/prediction", bad)
+    assert resp.status_code == 400
+
+
+def test_prediction_unknown_model(api_client, gordo_project, X):
+    resp = _post_json(
+        api_client, f"/gordo/v0/{gordo_project}/no-such-model/prediction", X
+    )
+    assert resp.status_code == 404
+
+
+def test_anomaly_prediction(api_client, base_route, X):
+    resp = _post_json(api_client, f"{base_route}/anomaly/prediction", X, y=X)
+    assert resp.status_code == 200
+    data = resp.json["data"]
+    for key in (
+        "model-input", "model-output",
+        "tag-anomaly-scaled", "total-anomaly-scaled",
+        "tag-anomaly-unscaled", "total-anomaly-unscaled",
+    ):
+        assert key in data, key
+    # smoothed columns dropped by default
+    assert not any(k.startswith("smooth-") for k in data)
+    assert "time-seconds" in resp.json
+
+
+def test_anomaly_prediction_all_columns(
+    api_client, gordo_project, second_gordo_name, X
+):
+    url = f"/gordo/v0/{gordo_project}/{second_gordo_name}/anomaly/prediction"
+    resp = _post_json(api_client, url, X, y=X, all_columns="true")
+    assert resp.status_code == 200
+    assert any(k.startswith("smooth-") for k in resp.json["data"])
+
+
+def test_anomaly_requires_y(api_client, base_route, X):
+    resp = _post_json(api_client, f"{base_route}/anomaly/prediction", X)
+    assert resp.status_code == 400
+
+
+def test_download_model(api_client, base_route):
+    resp = api_client.get(f"{base_route}/download-model")
+    assert resp.status_code == 200
+    from gordo_amd import serializer
+
+    model = serializer.loads(resp.data)
+    assert hasattr(model, "anomaly")
+
+
+def test_revision_query(api_client, base_route, gordo_revision, X):
+    resp = _post_json(
+        api_client, f"{base_route}/prediction", X, revision=gordo_revision
+    )
+    assert resp.status_code == 200
+    assert resp.headers["revision"] == gordo_revision
+
+
+def test_unknown_revision_410(api_client, base_route, X):
+    resp = _post_json(
+        api_client, f"{base_route}/prediction", X, revision="123"
+    )
+    assert resp.status_code == 410
+    resp = _post_json(
+        api_client, f"{base_route}/prediction", X, revision="notanumber"
+    )
+    assert resp.status_code == 410
+
+
+def test_dataframe_dict_roundtrip(X):
+    d = server_utils.dataframe_to_dict(X)
+    df = server_utils.dataframe_from_dict(d)
+    np.testing.assert_allclose(df.values, X.values, rtol=1e-9)
+
+
+def test_parquet_roundtrip(X):
+    blob = server_utils.dataframe_into_parquet_bytes(X)
+    df = server_utils.dataframe_from_parquet_bytes(blob)
+    np.testing.assert_allclose(df.values, X.values)
+
+
+def test_proxy_path_rewrite(flask_app):
+    # Envoy-prefixed request should reach the route
+    resp = flask_app.test_client().get(
+        "/healthcheck",
+        environ_overrides={
+            "HTTP_X_ENVOY_ORIGINAL_PATH": "/gordo/v0/proj/healthcheck"
+        },
+    )
+    assert resp.status_code == 200
