@@ -23,7 +23,7 @@ MACHINE_JSON = {
         "sklearn.pipeline.Pipeline": {
             "steps": [
                 "sklearn.preprocessing.MinMaxScaler",
-                {"sklearn.decomposition.PCA": {"n_components": 2}},
+                "sklearn.linear_model.LinearRegression",
             ]
         }
     },
