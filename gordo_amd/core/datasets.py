@@ -155,6 +155,14 @@ class TimeSeriesDataset(GordoBaseDataset):
         return X, y
 
     def _resample(self, series: pd.Series) -> pd.Series:
+        # fast path: the series is already exactly on the target grid
+        # (synthetic providers generate aligned data) — resampling would
+        # be an expensive no-op per tag.
+        freq = getattr(series.index, "freq", None)
+        if freq is not None and freq == pd.tseries.frequencies.to_offset(
+            self.resolution
+        ):
+            return series
         agg = self.aggregation_methods
         resampled = series.resample(self.resolution)
         if isinstance(agg, str):

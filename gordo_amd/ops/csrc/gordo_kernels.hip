@@ -1,0 +1,638 @@
+// gordo_amd HIP kernels for MI355X (gfx950, CDNA4).
+//
+// The compute core of the framework: grouped (batched-over-models) MFMA
+// GEMMs and the fused pointwise training ops. Every kernel operates on
+// packed [G, ...] tensors where G = models trained in lockstep
+// (SURVEY.md §2.3 kernels K1-K9).
+//
+// GEMM design (cdna_hip_programming.md §5):
+//  * mfma_f32_16x16x32_bf16, fp32 accumulation in AGPRs.
+//  * 64x64 output tile per 256-thread workgroup; 4 waves in a 2x2
+//    layout, each wave owns a 32x32 sub-tile = 2x2 fragments of 16x16.
+//  * K-step 32. A and B tiles staged through LDS with rows padded to
+//    40 bf16 (80 B row stride = 20 dword banks; 16 consecutive rows map
+//    to 16 distinct bank groups, so ds_read_b128 fragment reads are
+//    conflict-free; 80 % 16 == 0 keeps the b128 alignment rule of §6
+//    Guideline 17).
+//  * The B operand is staged TRANSPOSED into LDS (Bt[n][k]) so both
+//    fragment reads are row-contiguous 16-byte ds_read_b128.
+//  * Grid is (g, n-tile, m-tile) flattened with consecutive blocks
+//    sharing the same weight panel, remapped XCD-aware (bijective,
+//    §5.5 T1) so panel reuse hits the same XCD's L2.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include <vector>
+
+#define DEV_INLINE __device__ __forceinline__
+
+using bf16 = __hip_bfloat16;
+
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+
+// activation codes (keep in sync with gordo_amd/ops/reference.py)
+constexpr int ACT_LINEAR = 0, ACT_TANH = 1, ACT_RELU = 2, ACT_SIGMOID = 3;
+
+DEV_INLINE float act_apply(float z, int act) {
+  switch (act) {
+    case ACT_TANH: return tanhf(z);
+    case ACT_RELU: return z > 0.f ? z : 0.f;
+    case ACT_SIGMOID: return 1.f / (1.f + __expf(-z));
+    default: return z;
+  }
+}
+
+DEV_INLINE float act_grad_from_output(float y, int act) {
+  switch (act) {
+    case ACT_TANH: return 1.f - y * y;
+    case ACT_RELU: return y > 0.f ? 1.f : 0.f;
+    case ACT_SIGMOID: return y * (1.f - y);
+    default: return 1.f;
+  }
+}
+
+DEV_INLINE float bf2f(bf16 v) { return __bfloat162float(v); }
+DEV_INLINE bf16 f2bf(float v) { return __float2bfloat16(v); }
+
+// ---------------------------------------------------------------------------
+// Grouped GEMM: C[g, M, N] = op( A[g, M, K] @ B + bias ), B per mode:
+//   BMODE_KN: B[g, K, N] row-major (forward: X @ W)
+//   BMODE_NK: B[g, N, K] row-major (dgrad: dZ @ W^T with W=[K_out? see py])
+// ACC: C += result instead of overwrite (LSTM recurrent gates).
+// ---------------------------------------------------------------------------
+constexpr int BM = 64, BN = 64, BK = 32;
+constexpr int LDT = 40;  // padded LDS row length (bf16 elements)
+
+struct GemmCoord {
+  int g, m0, n0;
+};
+
+DEV_INLINE GemmCoord decode_block(int nblocks, int mt, int nt) {
+  // bijective XCD-aware remap (cdna_hip_programming.md §5 T1):
+  // blocks sharing a weight panel (same g, n-tile) become contiguous per
+  // XCD so the panel stays in one XCD's L2.
+  int bid = blockIdx.x;
+  constexpr int NXCD = 8;
+  if (nblocks >= NXCD) {
+    int q = nblocks / NXCD, r = nblocks % NXCD;
+    int xcd = bid % NXCD, idx = bid / NXCD;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  GemmCoord c;
+  int per_g = mt * nt;
+  c.g = bid / per_g;
+  int rem = bid % per_g;
+  c.n0 = (rem / mt) * BN;
+  c.m0 = (rem % mt) * BM;
+  return c;
+}
+
+template <int BMODE, bool ACC, bool FUSE_BIAS_ACT>
+__global__ __launch_bounds__(256) void grouped_gemm_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ B,
+    const float* __restrict__ bias, bf16* __restrict__ C,
+    int M, int N, int K, int act, int mt, int nt, int nblocks) {
+  __shared__ bf16 sm[2 * BM * LDT];
+  bf16* As = sm;               // [BM][LDT]
+  bf16* Bs = sm + BM * LDT;    // [BN][LDT]  (Bt: [n][k])
+
+  GemmCoord blk = decode_block(nblocks, mt, nt);
+  const int g = blk.g, m0 = blk.m0, n0 = blk.n0;
+  const bf16* Ag = A + (size_t)g * M * K;
+  const bf16* Bg = B + (size_t)g * ((BMODE == 0) ? (size_t)K * N : (size_t)N * K);
+  bf16* Cg = C + (size_t)g * M * N;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;          // 4 waves
+  const int wm = (wid & 1) * 32;     // wave row offset in tile
+  const int wn = (wid >> 1) * 32;    // wave col offset in tile
+
+  f32x4 acc[2][2] = {};
+
+  const int l15 = lane & 15;
+  const int kslot = lane >> 4;  // 0..3 -> k offset kslot*8
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    // ---- stage A[m0..m0+64][k0..k0+32] -> As[row][k] ----
+    // 256 threads x 8 elements; thread t: row=t/4, k8=(t%4)*8
+    {
+      int row = tid >> 2;
+      int kk = (tid & 3) * 8;
+      int gm = m0 + row;
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        int gk = k0 + kk + e;
+        As[row * LDT + kk + e] =
+            (gm < M && gk < K) ? Ag[(size_t)gm * K + gk] : f2bf(0.f);
+      }
+    }
+    // ---- stage B -> Bs[n][k] (transposed for BMODE_KN) ----
+    if (BMODE == 0) {
+      // B[K][N]: thread t reads col n=t/4 over k; scatter into Bs[n][k]
+      int n = tid >> 2;
+      int kk = (tid & 3) * 8;
+      int gn = n0 + n;
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        int gk = k0 + kk + e;
+        Bs[n * LDT + kk + e] =
+            (gn < N && gk < K) ? Bg[(size_t)gk * N + gn] : f2bf(0.f);
+      }
+    } else {
+      // B[N][K]: rows are n — direct copy
+      int n = tid >> 2;
+      int kk = (tid & 3) * 8;
+      int gn = n0 + n;
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        int gk = k0 + kk + e;
+        Bs[n * LDT + kk + e] =
+            (gn < N && gk < K) ? Bg[(size_t)gn * K + gk] : f2bf(0.f);
+      }
+    }
+    __syncthreads();
+
+    // ---- MFMA over the two 16-deep k sub-steps of BK=32 ----
+    #pragma unroll
+    for (int ks = 0; ks < BK; ks += 32) {
+      #pragma unroll
+      for (int fm = 0; fm < 2; ++fm) {
+        // A fragment: rows wm+fm*16+l15, k = ks + kslot*8 .. +8
+        bf16x8 a = *reinterpret_cast<const bf16x8*>(
+            &As[(wm + fm * 16 + l15) * LDT + ks + kslot * 8]);
+        #pragma unroll
+        for (int fn = 0; fn < 2; ++fn) {
+          bf16x8 b = *reinterpret_cast<const bf16x8*>(
+              &Bs[(wn + fn * 16 + l15) * LDT + ks + kslot * 8]);
+          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, b, acc[fm][fn], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: bias + activation, guarded store ----
+  #pragma unroll
+  for (int fm = 0; fm < 2; ++fm) {
+    #pragma unroll
+    for (int fn = 0; fn < 2; ++fn) {
+      int col = n0 + wn + fn * 16 + l15;
+      if (col >= N) continue;
+      float bv = FUSE_BIAS_ACT ? bias[(size_t)g * N + col] : 0.f;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wm + fm * 16 + (lane >> 4) * 4 + r;
+        if (row >= M) continue;
+        float v = acc[fm][fn][r] + bv;
+        if (FUSE_BIAS_ACT) v = act_apply(v, act);
+        size_t off = (size_t)row * N + col;
+        if (ACC) v += bf2f(Cg[off]);
+        Cg[off] = f2bf(v);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Grouped wgrad: dW[g, K, N] (fp32) = A[g, M, K]^T @ dZ[g, M, N].
+// Output tile 64x64 over (K, N); reduction dim is M. Both operands are
+// staged transposed/direct so fragment reads stay row-contiguous:
+//   At[k][m] (transpose of A tile), Zt[n][m] (transpose of dZ tile).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void grouped_wgrad_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ dZ,
+    float* __restrict__ dW, int M, int N, int K, int kt, int nt,
+    int nblocks) {
+  __shared__ bf16 sm[2 * BM * LDT];
+  bf16* As = sm;             // At: [k 64][m 32+pad]
+  bf16* Zs = sm + BM * LDT;  // Zt: [n 64][m 32+pad]
+
+  GemmCoord blk = decode_block(nblocks, kt, nt);
+  const int g = blk.g, k0 = blk.m0, n0 = blk.n0;  // m-slot carries k-tile
+  const bf16* Ag = A + (size_t)g * M * K;
+  const bf16* Zg = dZ + (size_t)g * M * N;
+  float* Wg = dW + (size_t)g * K * N;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wk = (wid & 1) * 32;
+  const int wn = (wid >> 1) * 32;
+  const int l15 = lane & 15;
+  const int mslot = lane >> 4;
+
+  f32x4 acc[2][2] = {};
+
+  for (int m0 = 0; m0 < M; m0 += BK) {
+    // stage A[m0..+32][k0..+64] transposed into As[k][m]
+    {
+      int k = tid >> 2;           // 0..63
+      int mm = (tid & 3) * 8;     // 0..24
+      int gk = k0 + k;
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        int gm = m0 + mm + e;
+        As[k * LDT + mm + e] =
+            (gk < K && gm < M) ? Ag[(size_t)gm * K + gk] : f2bf(0.f);
+      }
+    }
+    // stage dZ[m0..+32][n0..+64] transposed into Zs[n][m]
+    {
+      int n = tid >> 2;
+      int mm = (tid & 3) * 8;
+      int gn = n0 + n;
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        int gm = m0 + mm + e;
+        Zs[n * LDT + mm + e] =
+            (gn < N && gm < M) ? Zg[(size_t)gm * N + gn] : f2bf(0.f);
+      }
+    }
+    __syncthreads();
+
+    #pragma unroll
+    for (int fm = 0; fm < 2; ++fm) {
+      bf16x8 a = *reinterpret_cast<const bf16x8*>(
+          &As[(wk + fm * 16 + l15) * LDT + mslot * 8]);
+      #pragma unroll
+      for (int fn = 0; fn < 2; ++fn) {
+        bf16x8 b = *reinterpret_cast<const bf16x8*>(
+            &Zs[(wn + fn * 16 + l15) * LDT + mslot * 8]);
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a, b, acc[fm][fn], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  #pragma unroll
+  for (int fm = 0; fm < 2; ++fm) {
+    #pragma unroll
+    for (int fn = 0; fn < 2; ++fn) {
+      int col = n0 + wn + fn * 16 + l15;
+      if (col >= N) continue;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = k0 + wk + fm * 16 + (lane >> 4) * 4 + r;
+        if (row >= K) continue;
+        Wg[(size_t)row * N + col] = acc[fm][fn][r];
+      }
+    }
+  }
+}
+
+// column sum: db[g, N] = sum_m dZ[g, m, N]  (fp32 out)
+__global__ void colsum_kernel(const bf16* __restrict__ dZ,
+                              float* __restrict__ db, int M, int N) {
+  int g = blockIdx.y;
+  int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= N) return;
+  const bf16* Zg = dZ + (size_t)g * M * N;
+  float s = 0.f;
+  for (int m = 0; m < M; ++m) s += bf2f(Zg[(size_t)m * N + col]);
+  db[(size_t)g * N + col] = s;
+}
+
+// ---------------------------------------------------------------------------
+// Fused pointwise kernels
+// ---------------------------------------------------------------------------
+
+// dZ = (dA + l1*sign(Y)) * act'(Y)
+__global__ void act_l1_bwd_kernel(const bf16* __restrict__ dA,
+                                  const bf16* __restrict__ Y,
+                                  bf16* __restrict__ dZ, size_t n, int act,
+                                  float l1) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float y = bf2f(Y[i]);
+    float gin = bf2f(dA[i]);
+    if (l1 != 0.f) gin += l1 * (y > 0.f ? 1.f : (y < 0.f ? -1.f : 0.f));
+    dZ[i] = f2bf(gin * act_grad_from_output(y, act));
+  }
+}
+
+// loss[g] = mean((Y-T)^2) over (B*F); dY = 2(Y-T)/(B*F)
+__global__ void mse_bwd_kernel(const bf16* __restrict__ Y,
+                               const bf16* __restrict__ T,
+                               bf16* __restrict__ dY,
+                               float* __restrict__ loss, int per_g) {
+  int g = blockIdx.y;
+  const bf16* Yg = Y + (size_t)g * per_g;
+  const bf16* Tg = T + (size_t)g * per_g;
+  bf16* dYg = dY + (size_t)g * per_g;
+  float inv_n = 1.f / (float)per_g;
+  float local = 0.f;
+  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < (size_t)per_g;
+       i += (size_t)gridDim.x * blockDim.x) {
+    float d = bf2f(Yg[i]) - bf2f(Tg[i]);
+    local += d * d;
+    dYg[i] = f2bf(2.f * d * inv_n);
+  }
+  // wave -> block -> global reduction
+  #pragma unroll
+  for (int off = 32; off; off >>= 1) local += __shfl_down(local, off, 64);
+  __shared__ float warp_part[4];
+  int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+  if (lane == 0) warp_part[wid] = local;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float s = 0.f;
+    for (int w = 0; w < (int)(blockDim.x >> 6); ++w) s += warp_part[w];
+    atomicAdd(&loss[g], s * inv_n);
+  }
+}
+
+// fused Adam over the flat fp32 master buffer + bf16 mirror refresh.
+__global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
+                            float* __restrict__ m, float* __restrict__ v,
+                            bf16* __restrict__ plp, size_t n, float lr,
+                            float b1, float b2, float eps, float bc1,
+                            float bc2) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float gi = g[i];
+    float mi = b1 * m[i] + (1.f - b1) * gi;
+    float vi = b2 * v[i] + (1.f - b2) * gi * gi;
+    m[i] = mi;
+    v[i] = vi;
+    float pi = p[i] - lr / bc1 * mi / (sqrtf(vi / bc2) + eps);
+    p[i] = pi;
+    if (plp) plp[i] = f2bf(pi);
+  }
+}
+
+// LSTM pointwise fwd: gates[g,b,4H] pre-act (+c_prev f32) ->
+// h bf16, c f32, gact bf16 (keras gate order i,f,g,o).
+__global__ void lstm_pw_fwd_kernel(const bf16* __restrict__ gates,
+                                   const float* __restrict__ c_prev,
+                                   bf16* __restrict__ h, float* __restrict__ c,
+                                   bf16* __restrict__ gact, size_t rows,
+                                   int H) {
+  size_t idx = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t n = rows * H;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; idx < n; idx += stride) {
+    size_t row = idx / H;
+    int hh = idx % H;
+    const bf16* grow = gates + row * 4 * H;
+    float i_g = 1.f / (1.f + __expf(-bf2f(grow[hh])));
+    float f_g = 1.f / (1.f + __expf(-bf2f(grow[H + hh])));
+    float g_g = tanhf(bf2f(grow[2 * H + hh]));
+    float o_g = 1.f / (1.f + __expf(-bf2f(grow[3 * H + hh])));
+    float cc = f_g * c_prev[idx] + i_g * g_g;
+    c[idx] = cc;
+    h[idx] = f2bf(o_g * tanhf(cc));
+    bf16* ga = gact + row * 4 * H;
+    ga[hh] = f2bf(i_g);
+    ga[H + hh] = f2bf(f_g);
+    ga[2 * H + hh] = f2bf(g_g);
+    ga[3 * H + hh] = f2bf(o_g);
+  }
+}
+
+// LSTM pointwise bwd -> dgates (pre-act) bf16, dc_prev f32
+__global__ void lstm_pw_bwd_kernel(
+    const bf16* __restrict__ dh, const float* __restrict__ dc_next,
+    const bf16* __restrict__ gact, const float* __restrict__ c,
+    const float* __restrict__ c_prev, bf16* __restrict__ dgates,
+    float* __restrict__ dc_prev, size_t rows, int H) {
+  size_t idx = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t n = rows * H;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; idx < n; idx += stride) {
+    size_t row = idx / H;
+    int hh = idx % H;
+    const bf16* ga = gact + row * 4 * H;
+    float i_g = bf2f(ga[hh]);
+    float f_g = bf2f(ga[H + hh]);
+    float g_g = bf2f(ga[2 * H + hh]);
+    float o_g = bf2f(ga[3 * H + hh]);
+    float tc = tanhf(c[idx]);
+    float dhv = bf2f(dh[idx]);
+    float dc = dc_next[idx] + dhv * o_g * (1.f - tc * tc);
+    float di = dc * g_g;
+    float df = dc * c_prev[idx];
+    float dg = dc * i_g;
+    float do_ = dhv * tc;
+    dc_prev[idx] = dc * f_g;
+    bf16* dgrow = dgates + row * 4 * H;
+    dgrow[hh] = f2bf(di * i_g * (1.f - i_g));
+    dgrow[H + hh] = f2bf(df * f_g * (1.f - f_g));
+    dgrow[2 * H + hh] = f2bf(dg * (1.f - g_g * g_g));
+    dgrow[3 * H + hh] = f2bf(do_ * o_g * (1.f - o_g));
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Host-side launchers (torch extension API)
+// ---------------------------------------------------------------------------
+namespace {
+
+#define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
+
+inline int ceil_div(int64_t a, int64_t b) { return (int)((a + b - 1) / b); }
+
+torch::Tensor to_bf16c(const torch::Tensor& t) {
+  return t.to(torch::kBFloat16).contiguous();
+}
+
+hipStream_t cur_stream() {
+  return at::cuda::getCurrentCUDAStream().stream();
+}
+
+torch::Tensor grouped_linear_fwd(torch::Tensor X, torch::Tensor W,
+                                 torch::Tensor b, int64_t act) {
+  CHECK_GPU(X);
+  auto Xc = to_bf16c(X);
+  auto Wc = to_bf16c(W);
+  auto bc = b.to(torch::kFloat32).contiguous();
+  int G = Xc.size(0), M = Xc.size(1), K = Xc.size(2), N = Wc.size(2);
+  TORCH_CHECK(Wc.size(1) == K, "W K mismatch");
+  auto Y = torch::empty({G, M, N}, Xc.options());
+  int mt = ceil_div(M, BM), nt = ceil_div(N, BN);
+  int nblocks = G * mt * nt;
+  hipLaunchKernelGGL((grouped_gemm_kernel<0, false, true>), dim3(nblocks),
+                     dim3(256), 0, cur_stream(),
+                     (const bf16*)Xc.data_ptr(), (const bf16*)Wc.data_ptr(),
+                     bc.data_ptr<float>(), (bf16*)Y.data_ptr(), M, N, K,
+                     (int)act, mt, nt, nblocks);
+  return Y;
+}
+
+torch::Tensor grouped_linear_bwd_data(torch::Tensor dZ, torch::Tensor W) {
+  CHECK_GPU(dZ);
+  auto Zc = to_bf16c(dZ);
+  auto Wc = to_bf16c(W);
+  int G = Zc.size(0), M = Zc.size(1), N = Zc.size(2);
+  int Kout = Wc.size(1);
+  TORCH_CHECK(Wc.size(2) == N, "W N mismatch");
+  auto dX = torch::empty({G, M, Kout}, Zc.options());
+  // dX = dZ @ W^T: output cols = Kout, inner dim = N; W[Kout][N] = BMODE_NK
+  int mt = ceil_div(M, BM), nt = ceil_div(Kout, BN);
+  int nblocks = G * mt * nt;
+  hipLaunchKernelGGL((grouped_gemm_kernel<1, false, false>), dim3(nblocks),
+                     dim3(256), 0, cur_stream(),
+                     (const bf16*)Zc.data_ptr(), (const bf16*)Wc.data_ptr(),
+                     nullptr, (bf16*)dX.data_ptr(), M, Kout, N, ACT_LINEAR,
+                     mt, nt, nblocks);
+  return dX;
+}
+
+torch::Tensor grouped_gemm_acc(torch::Tensor A, torch::Tensor B,
+                               torch::Tensor C) {
+  CHECK_GPU(A);
+  auto Ac = to_bf16c(A);
+  auto Bc = to_bf16c(B);
+  TORCH_CHECK(C.is_contiguous() && C.scalar_type() == torch::kBFloat16,
+              "C must be contiguous bf16");
+  int G = Ac.size(0), M = Ac.size(1), K = Ac.size(2), N = Bc.size(2);
+  int mt = ceil_div(M, BM), nt = ceil_div(N, BN);
+  int nblocks = G * mt * nt;
+  hipLaunchKernelGGL((grouped_gemm_kernel<0, true, false>), dim3(nblocks),
+                     dim3(256), 0, cur_stream(),
+                     (const bf16*)Ac.data_ptr(), (const bf16*)Bc.data_ptr(),
+                     nullptr, (bf16*)C.data_ptr(), M, N, K, ACT_LINEAR, mt,
+                     nt, nblocks);
+  return C;
+}
+
+std::vector<torch::Tensor> grouped_linear_wgrad(torch::Tensor X,
+                                                torch::Tensor dZ) {
+  CHECK_GPU(X);
+  auto Xc = to_bf16c(X);
+  auto Zc = to_bf16c(dZ);
+  int G = Xc.size(0), M = Xc.size(1), K = Xc.size(2), N = Zc.size(2);
+  auto dW = torch::empty({G, K, N},
+                         Xc.options().dtype(torch::kFloat32));
+  auto db = torch::empty({G, N}, Xc.options().dtype(torch::kFloat32));
+  int kt = ceil_div(K, BM), nt = ceil_div(N, BN);
+  int nblocks = G * kt * nt;
+  hipLaunchKernelGGL(grouped_wgrad_kernel, dim3(nblocks), dim3(256), 0,
+                     cur_stream(), (const bf16*)Xc.data_ptr(),
+                     (const bf16*)Zc.data_ptr(), dW.data_ptr<float>(), M, N,
+                     K, kt, nt, nblocks);
+  hipLaunchKernelGGL(colsum_kernel, dim3(ceil_div(N, 256), G), dim3(256), 0,
+                     cur_stream(), (const bf16*)Zc.data_ptr(),
+                     db.data_ptr<float>(), M, N);
+  return {dW, db};
+}
+
+torch::Tensor act_l1_bwd(torch::Tensor dA, torch::Tensor Y, int64_t act,
+                         double l1) {
+  CHECK_GPU(dA);
+  auto dAc = to_bf16c(dA);
+  auto Yc = to_bf16c(Y);
+  auto dZ = torch::empty_like(dAc);
+  size_t n = dAc.numel();
+  int blocks = (int)std::min<size_t>((n + 255) / 256, 2048);
+  hipLaunchKernelGGL(act_l1_bwd_kernel, dim3(blocks), dim3(256), 0,
+                     cur_stream(), (const bf16*)dAc.data_ptr(),
+                     (const bf16*)Yc.data_ptr(), (bf16*)dZ.data_ptr(), n,
+                     (int)act, (float)l1);
+  return dZ;
+}
+
+std::vector<torch::Tensor> mse_bwd(torch::Tensor Y, torch::Tensor T) {
+  CHECK_GPU(Y);
+  auto Yc = to_bf16c(Y);
+  auto Tc = to_bf16c(T);
+  int G = Yc.size(0);
+  int64_t per_g = Yc.numel() / G;
+  auto dY = torch::empty_like(Yc);
+  auto loss = torch::zeros({G}, Yc.options().dtype(torch::kFloat32));
+  int blocks = (int)std::min<int64_t>((per_g + 255) / 256, 512);
+  hipLaunchKernelGGL(mse_bwd_kernel, dim3(blocks, G), dim3(256), 0,
+                     cur_stream(), (const bf16*)Yc.data_ptr(),
+                     (const bf16*)Tc.data_ptr(), (bf16*)dY.data_ptr(),
+                     loss.data_ptr<float>(), (int)per_g);
+  return {loss, dY};
+}
+
+void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+               torch::Tensor v, double lr, double b1, double b2, double eps,
+               int64_t step, c10::optional<torch::Tensor> plp) {
+  CHECK_GPU(p);
+  size_t n = p.numel();
+  float bc1 = 1.f - powf((float)b1, (float)step);
+  float bc2 = 1.f - powf((float)b2, (float)step);
+  bf16* plp_ptr = nullptr;
+  if (plp.has_value()) plp_ptr = (bf16*)plp->data_ptr();
+  int blocks = (int)std::min<size_t>((n + 255) / 256, 4096);
+  hipLaunchKernelGGL(adam_kernel, dim3(blocks), dim3(256), 0, cur_stream(),
+                     p.data_ptr<float>(), g.data_ptr<float>(),
+                     m.data_ptr<float>(), v.data_ptr<float>(), plp_ptr, n,
+                     (float)lr, (float)b1, (float)b2, (float)eps, bc1, bc2);
+}
+
+std::vector<torch::Tensor> lstm_pointwise_fwd(torch::Tensor gates,
+                                              torch::Tensor c_prev) {
+  CHECK_GPU(gates);
+  auto gc = to_bf16c(gates);
+  auto cp = c_prev.to(torch::kFloat32).contiguous();
+  int H4 = gc.size(-1);
+  int H = H4 / 4;
+  size_t rows = gc.numel() / H4;
+  auto sizes = gc.sizes().vec();
+  sizes.back() = H;
+  auto h = torch::empty(sizes, gc.options());
+  auto c = torch::empty(sizes, gc.options().dtype(torch::kFloat32));
+  auto gact = torch::empty_like(gc);
+  size_t n = rows * H;
+  int blocks = (int)std::min<size_t>((n + 255) / 256, 2048);
+  hipLaunchKernelGGL(lstm_pw_fwd_kernel, dim3(blocks), dim3(256), 0,
+                     cur_stream(), (const bf16*)gc.data_ptr(),
+                     cp.data_ptr<float>(), (bf16*)h.data_ptr(),
+                     c.data_ptr<float>(), (bf16*)gact.data_ptr(), rows, H);
+  return {h, c, gact};
+}
+
+std::vector<torch::Tensor> lstm_pointwise_bwd(torch::Tensor dh,
+                                              torch::Tensor dc_next,
+                                              torch::Tensor gact,
+                                              torch::Tensor c,
+                                              torch::Tensor c_prev) {
+  CHECK_GPU(dh);
+  auto dhc = to_bf16c(dh);
+  auto dcn = dc_next.to(torch::kFloat32).contiguous();
+  auto gc = to_bf16c(gact);
+  auto cc = c.to(torch::kFloat32).contiguous();
+  auto cpc = c_prev.to(torch::kFloat32).contiguous();
+  int H = dhc.size(-1);
+  size_t rows = dhc.numel() / H;
+  auto dgates = torch::empty_like(gc);
+  auto dc_prev = torch::empty_like(cc);
+  size_t n = rows * H;
+  int blocks = (int)std::min<size_t>((n + 255) / 256, 2048);
+  hipLaunchKernelGGL(lstm_pw_bwd_kernel, dim3(blocks), dim3(256), 0,
+                     cur_stream(), (const bf16*)dhc.data_ptr(),
+                     dcn.data_ptr<float>(), (const bf16*)gc.data_ptr(),
+                     cc.data_ptr<float>(), cpc.data_ptr<float>(),
+                     (bf16*)dgates.data_ptr(), dc_prev.data_ptr<float>(),
+                     rows, H);
+  return {dgates, dc_prev};
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("grouped_linear_fwd", &grouped_linear_fwd,
+          "Y = act(X@W + b) per group (MFMA)");
+  mod.def("grouped_linear_bwd_data", &grouped_linear_bwd_data,
+          "dX = dZ@W^T per group (MFMA)");
+  mod.def("grouped_linear_wgrad", &grouped_linear_wgrad,
+          "dW = X^T@dZ, db = colsum(dZ) per group (MFMA)");
+  mod.def("grouped_gemm_acc", &grouped_gemm_acc, "C += A@B per group (MFMA)");
+  mod.def("act_l1_bwd", &act_l1_bwd, "fused activation+L1 backward");
+  mod.def("mse_bwd", &mse_bwd, "fused per-model MSE loss + grad");
+  mod.def("adam_step", &adam_step, "fused Adam + bf16 mirror refresh");
+  mod.def("lstm_pointwise_fwd", &lstm_pointwise_fwd, "LSTM cell fwd");
+  mod.def("lstm_pointwise_bwd", &lstm_pointwise_bwd, "LSTM cell bwd");
+}

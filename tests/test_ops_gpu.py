@@ -1,0 +1,174 @@
+"""HIP kernel numerics vs the CPU fp32 reference (gordo_amd.ops.reference).
+
+Every op is compared against the plain PyTorch fp32 oracle at bf16
+tolerances. Shapes cover the framework's hot cases: odd feature counts
+(50 tags, hourglass dims 38/28/19), edge tiles, and LSTM gate widths.
+"""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from gordo_amd import ops
+from gordo_amd.ops import reference as ref
+
+
+def require_hip():
+    assert ops.hip_available(), (
+        "HIP extension not built — GPU tests must not fall back to eager"
+    )
+
+
+def _rand(*shape, seed=0):
+    g = torch.Generator().manual_seed(seed)
+    return torch.rand(*shape, generator=g) * 2 - 1
+
+
+def to_dev_bf16(t):
+    return t.to("cuda", torch.bfloat16)
+
+
+# bf16 has ~3 decimal digits; accumulation in fp32.
+RTOL, ATOL = 2e-2, 2e-2
+
+
+@pytest.mark.parametrize(
+    "G,M,K,N",
+    [
+        (1, 64, 64, 64),
+        (3, 100, 50, 38),     # 50-tag hourglass first layer
+        (5, 33, 19, 28),      # odd everything, edge tiles
+        (2, 256, 40, 160),    # LSTM gates shape (4H=160)
+        (1, 4176, 50, 152),   # big-M LSTM x-side GEMM
+    ],
+)
+@pytest.mark.parametrize("act", ["linear", "tanh", "sigmoid", "relu"])
+def test_grouped_linear_fwd(G, M, K, N, act):
+    require_hip()
+    X, W = _rand(G, M, K, seed=1), _rand(G, K, N, seed=2)
+    b = _rand(G, N, seed=3)
+    want = ref.grouped_linear_fwd(X, W, b, ref.act_code(act))
+    got = ops.grouped_linear_fwd(
+        to_dev_bf16(X), to_dev_bf16(W), b.cuda(), act
+    ).float().cpu()
+    torch.testing.assert_close(got, want, rtol=RTOL, atol=ATOL)
+
+
+def test_gemm_not_transposed():
+    """Asymmetric operand check (guide §3): catches silent C^T."""
+    require_hip()
+    G, M, K, N = 1, 32, 32, 48
+    X = torch.zeros(G, M, K)
+    for i in range(M):
+        X[0, i, i % K] = 1.0  # permuted identity-ish
+    W = torch.arange(K * N, dtype=torch.float32).view(G, K, N) / (K * N)
+    b = torch.zeros(G, N)
+    want = ref.grouped_linear_fwd(X, W, b, 0)
+    got = ops.grouped_linear_fwd(
+        to_dev_bf16(X), to_dev_bf16(W), b.cuda(), 0
+    ).float().cpu()
+    torch.testing.assert_close(got, want, rtol=RTOL, atol=ATOL)
+
+
+@pytest.mark.parametrize(
+    "G,M,N,K", [(2, 100, 38, 50), (3, 65, 17, 33), (1, 1000, 152, 40)]
+)
+def test_grouped_linear_bwd_data(G, M, N, K):
+    require_hip()
+    dZ, W = _rand(G, M, N, seed=4), _rand(G, K, N, seed=5)
+    want = ref.grouped_linear_bwd_data(dZ, W)
+    got = ops.grouped_linear_bwd_data(
+        to_dev_bf16(dZ), to_dev_bf16(W)
+    ).float().cpu()
+    torch.testing.assert_close(got, want, rtol=RTOL, atol=ATOL)
+
+
+@pytest.mark.parametrize(
+    "G,M,K,N", [(2, 200, 50, 38), (1, 4176, 40, 160), (4, 37, 19, 21)]
+)
+def test_grouped_linear_wgrad(G, M, K, N):
+    require_hip()
+    X, dZ = _rand(G, M, K, seed=6), _rand(G, M, N, seed=7) * 0.1
+    wantW, wantb = ref.grouped_linear_wgrad(X, dZ)
+    gotW, gotb = ops.grouped_linear_wgrad(to_dev_bf16(X), to_dev_bf16(dZ))
+    # fp32 accumulation over large M: scale tolerance with sqrt(M)
+    tol = dict(rtol=3e-2, atol=2e-2 * max(1.0, (M / 256) ** 0.5))
+    torch.testing.assert_close(gotW.cpu(), wantW, **tol)
+    torch.testing.assert_close(gotb.cpu(), wantb, **tol)
+
+
+def test_grouped_gemm_acc():
+    require_hip()
+    G, M, K, N = 3, 64, 24, 96
+    A, B = _rand(G, M, K, seed=8), _rand(G, K, N, seed=9)
+    C0 = _rand(G, M, N, seed=10)
+    want = C0.clone()
+    ref.grouped_gemm_acc(A, B, want)
+    got = to_dev_bf16(C0).contiguous()
+    ops.grouped_gemm_acc(to_dev_bf16(A), to_dev_bf16(B), got)
+    torch.testing.assert_close(got.float().cpu(), want, rtol=3e-2, atol=3e-2)
+
+
+@pytest.mark.parametrize("act,l1", [("tanh", 0.0), ("tanh", 1e-4),
+                                    ("linear", 1e-4), ("sigmoid", 0.0)])
+def test_act_l1_bwd(act, l1):
+    require_hip()
+    dA, Y = _rand(2, 100, 30, seed=11), _rand(2, 100, 30, seed=12)
+    want = ref.act_l1_bwd(dA, Y, ref.act_code(act), l1)
+    got = ops.act_l1_bwd(to_dev_bf16(dA), to_dev_bf16(Y), act, l1)
+    torch.testing.assert_close(got.float().cpu(), want, rtol=2e-2, atol=2e-3)
+
+
+def test_mse_bwd():
+    require_hip()
+    Y, T = _rand(3, 500, 50, seed=13), _rand(3, 500, 50, seed=14)
+    want_loss, want_dY = ref.mse_bwd(Y, T)
+    got_loss, got_dY = ops.mse_bwd(to_dev_bf16(Y), to_dev_bf16(T))
+    torch.testing.assert_close(got_loss.cpu(), want_loss, rtol=1e-2, atol=1e-3)
+    torch.testing.assert_close(
+        got_dY.float().cpu(), want_dY, rtol=2e-2, atol=1e-4
+    )
+
+
+def test_adam_step():
+    require_hip()
+    n = 10007
+    p = _rand(n, seed=15)
+    g = _rand(n, seed=16)
+    m = torch.zeros(n)
+    v = torch.zeros(n)
+    pd, gd, md, vd = (t.cuda() for t in (p, g, m, v))
+    plp = pd.to(torch.bfloat16)
+    for step in (1, 2, 3):
+        ref.adam_step(p, g, m, v, 0.01, 0.9, 0.999, 1e-7, step)
+        ops.adam_step(pd, gd, md, vd, 0.01, 0.9, 0.999, 1e-7, step, plp)
+    torch.testing.assert_close(pd.cpu(), p, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(md.cpu(), m, rtol=1e-5, atol=1e-7)
+    torch.testing.assert_close(plp.float().cpu(), p, rtol=1e-2, atol=1e-2)
+
+
+def test_lstm_pointwise_roundtrip():
+    require_hip()
+    G, B, H = 2, 64, 38
+    gates = _rand(G, B, 4 * H, seed=17) * 2
+    c_prev = _rand(G, B, H, seed=18)
+    want_h, want_c, want_gact = ref.lstm_pointwise_fwd(gates, c_prev)
+    got_h, got_c, got_gact = ops.lstm_pointwise_fwd(
+        to_dev_bf16(gates), c_prev.cuda()
+    )
+    torch.testing.assert_close(got_h.float().cpu(), want_h, rtol=2e-2, atol=1e-2)
+    torch.testing.assert_close(got_c.cpu(), want_c, rtol=2e-2, atol=1e-2)
+
+    dh = _rand(G, B, H, seed=19)
+    dc_next = _rand(G, B, H, seed=20) * 0.1
+    want_dg, want_dcp = ref.lstm_pointwise_bwd(
+        dh, dc_next, want_gact, want_c, c_prev
+    )
+    got_dg, got_dcp = ops.lstm_pointwise_bwd(
+        to_dev_bf16(dh), dc_next.cuda(), got_gact, got_c, c_prev.cuda()
+    )
+    torch.testing.assert_close(
+        got_dg.float().cpu(), want_dg, rtol=3e-2, atol=2e-2
+    )
+    torch.testing.assert_close(got_dcp.cpu(), want_dcp, rtol=3e-2, atol=2e-2)
