@@ -330,9 +330,20 @@ class LSTMPack(BasePack):
             self.store.views["Wd"][g].copy_(_glorot_uniform((fin, fout), gen))
 
     # ---- sequence forward/backward ------------------------------------
+    def _use_fused(self) -> bool:
+        """The fused on-device sequence-scan kernels (lstm_seq.hip) run
+        the whole T-step recurrence in one launch per layer with Wh
+        resident in LDS; used on GPU whenever every layer fits the
+        kernel's H<=64 geometry."""
+        if self.device.type != "cuda":
+            return False
+        return all(ops.lstm_seq_available(H) for _fin, H, _rs in self.lstm_meta)
+
     def _forward_seq(self, Xw: torch.Tensor, keep: bool):
         """Xw: [G, B, T, F]. Returns (y, cache)."""
         G, B, T, _ = Xw.shape
+        if self._use_fused():
+            return self._forward_seq_fused(Xw, keep)
         seq = Xw
         cache = []
         for li, (fin, H, rs) in enumerate(self.lstm_meta):
@@ -381,11 +392,91 @@ class LSTMPack(BasePack):
             cache.append(dict(h_last=h_last, y=y))
         return y, cache
 
+    def _forward_seq_fused(self, Xw: torch.Tensor, keep: bool):
+        """GPU path: one big x-side GEMM + one fused sequence-scan
+        kernel per layer."""
+        G, B, T, _ = Xw.shape
+        seq = Xw
+        cache = []
+        for li, (fin, H, rs) in enumerate(self.lstm_meta):
+            Wx = self.store.cviews[f"Wx{li}"]
+            Wh = self.store.cviews[f"Wh{li}"]
+            b = self.store.views[f"bl{li}"]
+            xW = ops.grouped_linear_fwd(
+                seq.reshape(G, B * T, fin), Wx, b, "linear"
+            ).view(G, B, T, 4 * H)
+            hs, cs, gacts = ops.lstm_seq_fwd(xW, Wh)
+            cache.append(
+                dict(
+                    seq_in=seq if keep else None,
+                    hs=hs,
+                    cs=cs if keep else None,
+                    gacts=gacts if keep else None,
+                )
+            )
+            seq = hs
+        h_last = seq[:, :, -1].contiguous()
+        fin, fout, act = self.dense_meta
+        y = ops.grouped_linear_fwd(
+            h_last, self.store.cviews["Wd"], self.store.views["bd"], act
+        )
+        if keep:
+            cache.append(dict(h_last=h_last, y=y))
+        return y, cache
+
     def predict_windows(self, Xw: torch.Tensor) -> torch.Tensor:
         y, _ = self._forward_seq(self._to_compute(Xw), keep=False)
         return y
 
+    def _train_batch_fused(self, Xw, Tb) -> torch.Tensor:
+        G, B, T, _ = Xw.shape
+        y, cache = self._forward_seq_fused(Xw, keep=True)
+        loss, dY = ops.mse_bwd(y, Tb.to(y.dtype))
+
+        fin, fout, act = self.dense_meta
+        head = cache[-1]
+        dZ = ops.act_l1_bwd(dY, head["y"], act, 0.0)
+        dWd, dbd = ops.grouped_linear_wgrad(head["h_last"], dZ)
+        self.store.gviews["Wd"].copy_(dWd)
+        self.store.gviews["bd"].copy_(dbd)
+        dh_last = ops.grouped_linear_bwd_data(dZ, self.store.cviews["Wd"])
+
+        dSeq = None
+        for li in range(len(self.lstm_meta) - 1, -1, -1):
+            fin, H, rs = self.lstm_meta[li]
+            lc = cache[li]
+            Wh = self.store.cviews[f"Wh{li}"]
+            Wx = self.store.cviews[f"Wx{li}"]
+            last_only = dSeq is None
+            dG_flat = ops.lstm_seq_bwd(
+                dh_last if last_only else dSeq,
+                lc["gacts"], lc["cs"], Wh, last_only,
+            ).view(G, B * T, 4 * H)
+            hs = lc["hs"]
+            h_prev_all = torch.cat(
+                [torch.zeros_like(hs[:, :, :1]), hs[:, :, :-1]], dim=2
+            )
+            dWx, dbl = ops.grouped_linear_wgrad(
+                lc["seq_in"].reshape(G, B * T, fin), dG_flat
+            )
+            dWh, _ = ops.grouped_linear_wgrad(
+                h_prev_all.reshape(G, B * T, H), dG_flat
+            )
+            self.store.gviews[f"Wx{li}"].copy_(dWx)
+            self.store.gviews[f"Wh{li}"].copy_(dWh)
+            self.store.gviews[f"bl{li}"].copy_(dbl)
+            if li > 0:
+                dSeq = ops.grouped_linear_bwd_data(dG_flat, Wx).view(
+                    G, B, T, fin
+                )
+
+        a = self.spec.adam_params
+        self.store.adam_step(a["lr"], a["beta1"], a["beta2"], a["eps"])
+        return loss
+
     def train_batch(self, Xw, Tb) -> torch.Tensor:
+        if self._use_fused():
+            return self._train_batch_fused(Xw, Tb)
         G, B, T, _ = Xw.shape
         y, cache = self._forward_seq(Xw, keep=True)
         loss, dY = ops.mse_bwd(y, Tb.to(y.dtype))

@@ -106,6 +106,20 @@ def adam_step(p, g, m, v, lr, beta1, beta2, eps, step, p_lp=None):
     return ref.adam_step(p, g, m, v, lr, beta1, beta2, eps, step, p_lp)
 
 
+def lstm_seq_fwd(xW, Wh):
+    """Fused on-device LSTM forward scan (GPU only, H<=64)."""
+    return _require_hip().lstm_seq_fwd(xW, Wh)
+
+
+def lstm_seq_bwd(dSeq, gacts, cs, Wh, last_only):
+    """Fused on-device LSTM backward (BPTT) scan (GPU only, H<=64)."""
+    return _require_hip().lstm_seq_bwd(dSeq, gacts, cs, Wh, last_only)
+
+
+def lstm_seq_available(H: int) -> bool:
+    return hip_available() and H <= 64
+
+
 def lstm_pointwise_fwd(gates, c_prev):
     if _on_gpu(gates):
         return _require_hip().lstm_pointwise_fwd(gates, c_prev)

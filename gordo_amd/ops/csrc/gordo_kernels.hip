@@ -622,7 +622,19 @@ std::vector<torch::Tensor> lstm_pointwise_bwd(torch::Tensor dh,
 
 }  // namespace
 
+// fused LSTM sequence-scan entry points (lstm_seq.hip)
+namespace gordo_lstm {
+std::vector<torch::Tensor> lstm_seq_fwd(torch::Tensor xW, torch::Tensor Wh);
+torch::Tensor lstm_seq_bwd(torch::Tensor dSeq, torch::Tensor gacts,
+                           torch::Tensor cs, torch::Tensor Wh,
+                           bool last_only);
+}  // namespace gordo_lstm
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("lstm_seq_fwd", &gordo_lstm::lstm_seq_fwd,
+          "fused LSTM forward sequence scan (Wh resident in LDS)");
+  mod.def("lstm_seq_bwd", &gordo_lstm::lstm_seq_bwd,
+          "fused LSTM backward (BPTT) sequence scan");
   mod.def("grouped_linear_fwd", &grouped_linear_fwd,
           "Y = act(X@W + b) per group (MFMA)");
   mod.def("grouped_linear_bwd_data", &grouped_linear_bwd_data,

@@ -1,0 +1,342 @@
+// Fused LSTM sequence-scan kernels for MI355X (gfx950).
+//
+// The reference's per-timestep Keras LSTM maps
+// to a sequential h/c recurrence that launch-per-timestep execution
+// makes launch-bound (~576 launches per batch at lookback 144). These
+// kernels run the ENTIRE sequence loop on-device:
+//
+//   lstm_seq_fwd : per (model g, 64-window row tile) workgroup, the
+//     recurrent weight matrix Wh is staged TRANSPOSED into LDS once,
+//     h/c live in LDS across timesteps, and each step is
+//     {MFMA h@Wh -> gates (+= precomputed xW_t from HBM) -> fused
+//     sigmoid/tanh gate math -> h/c update}, writing the h sequence,
+//     c sequence and activated gates needed by BPTT.
+//   lstm_seq_bwd : the reverse-time scan: fused gate backward +
+//     MFMA dgates@Wh^T carry, emitting pre-activation gate grads for
+//     the batched wgrad GEMMs.
+//
+// Supported geometry: H <= 64 (4H <= 256); the Python engine falls
+// back to the per-timestep kernels beyond that. One workgroup = 4
+// waves; the x-side GEMM (x@Wx + b over all B*T rows) stays a single
+// big grouped MFMA GEMM in gordo_kernels.hip.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include <vector>
+
+#define DEV_INLINE __device__ __forceinline__
+
+using bf16 = __hip_bfloat16;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+
+DEV_INLINE float lbf2f(bf16 v) { return __bfloat162float(v); }
+DEV_INLINE bf16 lf2bf(float v) { return __float2bfloat16(v); }
+DEV_INLINE float sigmoidf_(float x) { return 1.f / (1.f + __expf(-x)); }
+
+constexpr int ROWS = 64;   // window rows per workgroup
+constexpr int LDK = 72;    // padded K-row length for h / WhT tiles (bf16)
+
+// ---------------------------------------------------------------------------
+// Forward scan.
+//   xW    [G, B, T, 4H]  bf16 (x@Wx + b, precomputed)
+//   Wh    [G, H, 4H]     bf16
+//   hs    [G, B, T, H]   bf16 out
+//   cs    [G, B, T, H]   f32  out
+//   gacts [G, B, T, 4H]  bf16 out (activated i,f,g,o)
+// grid.x = G * ceil(B/64); dynamic LDS.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void lstm_seq_fwd_kernel(
+    const bf16* __restrict__ xW, const bf16* __restrict__ Wh,
+    bf16* __restrict__ hs, float* __restrict__ cs,
+    bf16* __restrict__ gacts, int B, int T, int H, int ldg) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int H4 = 4 * H;
+  // carves (all 16B aligned: LDK and ldg are multiples of 8 bf16)
+  bf16* WhT = reinterpret_cast<bf16*>(smem);              // [H4][LDK]
+  bf16* hS = WhT + (size_t)H4 * LDK;                      // [ROWS][LDK]
+  bf16* gS = hS + (size_t)ROWS * LDK;                     // [ROWS][ldg]
+  float* cS = reinterpret_cast<float*>(gS + (size_t)ROWS * ldg);  // [ROWS][H]
+
+  const int g = blockIdx.x / ((B + ROWS - 1) / ROWS);
+  const int r0 = (blockIdx.x % ((B + ROWS - 1) / ROWS)) * ROWS;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int l15 = lane & 15;
+  const int kslot = lane >> 4;
+
+  const bf16* Whg = Wh + (size_t)g * H * H4;
+  const bf16* xWg = xW + ((size_t)g * B + r0) * T * H4;
+  bf16* hsg = hs + ((size_t)g * B + r0) * T * H;
+  float* csg = cs + ((size_t)g * B + r0) * T * H;
+  bf16* gag = gacts + ((size_t)g * B + r0) * T * H4;
+  const int rows_here = min(ROWS, B - r0);
+
+  // ---- one-time: stage WhT[n][k=h] (transposed, zero-padded) ----
+  for (int i = tid; i < H4 * LDK; i += 256) {
+    int n = i / LDK, k = i % LDK;
+    WhT[i] = (k < H) ? Whg[(size_t)k * H4 + n] : lf2bf(0.f);
+  }
+  // zero h, c
+  for (int i = tid; i < ROWS * LDK; i += 256) hS[i] = lf2bf(0.f);
+  for (int i = tid; i < ROWS * H; i += 256) cS[i] = 0.f;
+  __syncthreads();
+
+  const int wcol0 = wid * 64;       // this wave's 64-column slice of 4H
+  const bool wave_active = wcol0 < H4;
+
+  for (int t = 0; t < T; ++t) {
+    // ---- gates = h @ Wh (MFMA) + xW_t ----
+    if (wave_active) {
+      f32x4 acc[4][4] = {};
+      for (int kk = 0; kk < H; kk += 32) {
+        #pragma unroll
+        for (int fm = 0; fm < 4; ++fm) {
+          bf16x8 a = *reinterpret_cast<const bf16x8*>(
+              &hS[(fm * 16 + l15) * LDK + kk + kslot * 8]);
+          #pragma unroll
+          for (int fn = 0; fn < 4; ++fn) {
+            int col = wcol0 + fn * 16 + l15;
+            bf16x8 b = *reinterpret_cast<const bf16x8*>(
+                &WhT[(size_t)min(col, H4 - 1) * LDK + kk + kslot * 8]);
+            acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a, b, acc[fm][fn], 0, 0, 0);
+          }
+        }
+      }
+      #pragma unroll
+      for (int fm = 0; fm < 4; ++fm) {
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+          int col = wcol0 + fn * 16 + l15;
+          if (col >= H4) continue;
+          #pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            int row = fm * 16 + (lane >> 4) * 4 + r;
+            float xv = (row < rows_here)
+                           ? lbf2f(xWg[((size_t)row * T + t) * H4 + col])
+                           : 0.f;
+            gS[row * ldg + col] = lf2bf(acc[fm][fn][r] + xv);
+          }
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- fused gate math + h/c update + outputs ----
+    for (int e = tid; e < ROWS * H; e += 256) {
+      int row = e / H, hh = e % H;
+      const bf16* grow = &gS[row * ldg];
+      float i_g = sigmoidf_(lbf2f(grow[hh]));
+      float f_g = sigmoidf_(lbf2f(grow[H + hh]));
+      float g_g = tanhf(lbf2f(grow[2 * H + hh]));
+      float o_g = sigmoidf_(lbf2f(grow[3 * H + hh]));
+      float cc = f_g * cS[row * H + hh] + i_g * g_g;
+      float hv = o_g * tanhf(cc);
+      cS[row * H + hh] = cc;
+      hS[row * LDK + hh] = lf2bf(hv);
+      if (row < rows_here) {
+        size_t base = ((size_t)row * T + t) * H + hh;
+        hsg[base] = lf2bf(hv);
+        csg[base] = cc;
+        size_t gbase = ((size_t)row * T + t) * H4;
+        gag[gbase + hh] = lf2bf(i_g);
+        gag[gbase + H + hh] = lf2bf(f_g);
+        gag[gbase + 2 * H + hh] = lf2bf(g_g);
+        gag[gbase + 3 * H + hh] = lf2bf(o_g);
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Backward scan (BPTT).
+//   dSeq    [G, B, T, H] bf16 — upstream grads on the h sequence, or
+//           when last_only: [G, B, H], grads only on h_{T-1}.
+//   gacts   [G, B, T, 4H] bf16, cs [G, B, T, H] f32 (from forward)
+//   Wh      [G, H, 4H] bf16
+//   dG      [G, B, T, 4H] bf16 out — pre-activation gate grads.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void lstm_seq_bwd_kernel(
+    const bf16* __restrict__ dSeq, const bf16* __restrict__ gacts,
+    const float* __restrict__ cs, const bf16* __restrict__ Wh,
+    bf16* __restrict__ dG, int B, int T, int H, int ldg, int last_only) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int H4 = 4 * H;
+  bf16* WhN = reinterpret_cast<bf16*>(smem);               // [H][ldg] native
+  bf16* dgS = WhN + (size_t)H * ldg;                       // [ROWS][ldg]
+  bf16* dhS = dgS + (size_t)ROWS * ldg;                    // [ROWS][LDK]
+  float* dcS = reinterpret_cast<float*>(dhS + (size_t)ROWS * LDK);  // [ROWS][H]
+
+  const int g = blockIdx.x / ((B + ROWS - 1) / ROWS);
+  const int r0 = (blockIdx.x % ((B + ROWS - 1) / ROWS)) * ROWS;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int l15 = lane & 15;
+  const int kslot = lane >> 4;
+
+  const bf16* Whg = Wh + (size_t)g * H * H4;
+  const bf16* gag = gacts + ((size_t)g * B + r0) * T * H4;
+  const float* csg = cs + ((size_t)g * B + r0) * T * H;
+  const bf16* dSg = last_only ? dSeq + ((size_t)g * B + r0) * H
+                              : dSeq + ((size_t)g * B + r0) * T * H;
+  bf16* dGg = dG + ((size_t)g * B + r0) * T * H4;
+  const int rows_here = min(ROWS, B - r0);
+
+  // stage Wh native [H][4H] zero-padded to [H][ldg]
+  for (int i = tid; i < H * ldg; i += 256) {
+    int h = i / ldg, n = i % ldg;
+    WhN[i] = (n < H4) ? Whg[(size_t)h * H4 + n] : lf2bf(0.f);
+  }
+  for (int i = tid; i < ROWS * LDK; i += 256) dhS[i] = lf2bf(0.f);
+  for (int i = tid; i < ROWS * H; i += 256) dcS[i] = 0.f;
+  // zero dgS once: the gate-backward phase writes only cols [0, 4H),
+  // but the MFMA K loop reads through the padded columns.
+  for (int i = tid; i < ROWS * ldg; i += 256) dgS[i] = lf2bf(0.f);
+  __syncthreads();
+
+  for (int t = T - 1; t >= 0; --t) {
+    // ---- fused gate backward ----
+    for (int e = tid; e < ROWS * H; e += 256) {
+      int row = e / H, hh = e % H;
+      float dh = lbf2f(dhS[row * LDK + hh]);
+      if (row < rows_here) {
+        if (last_only) {
+          if (t == T - 1) dh += lbf2f(dSg[(size_t)row * H + hh]);
+        } else {
+          dh += lbf2f(dSg[((size_t)row * T + t) * H + hh]);
+        }
+      }
+      size_t gbase = ((size_t)row * T + t) * H4;
+      size_t cbase = ((size_t)row * T + t) * H + hh;
+      float i_g = 0.5f, f_g = 0.5f, g_g = 0.f, o_g = 0.5f, cc = 0.f,
+            cp = 0.f;
+      if (row < rows_here) {
+        i_g = lbf2f(gag[gbase + hh]);
+        f_g = lbf2f(gag[gbase + H + hh]);
+        g_g = lbf2f(gag[gbase + 2 * H + hh]);
+        o_g = lbf2f(gag[gbase + 3 * H + hh]);
+        cc = csg[cbase];
+        cp = (t > 0) ? csg[cbase - H] : 0.f;
+      }
+      float tc = tanhf(cc);
+      float dc = dcS[row * H + hh] + dh * o_g * (1.f - tc * tc);
+      float di = dc * g_g;
+      float df = dc * cp;
+      float dg = dc * i_g;
+      float do_ = dh * tc;
+      dcS[row * H + hh] = dc * f_g;
+      float vi = di * i_g * (1.f - i_g);
+      float vf = df * f_g * (1.f - f_g);
+      float vg = dg * (1.f - g_g * g_g);
+      float vo = do_ * o_g * (1.f - o_g);
+      dgS[row * ldg + hh] = lf2bf(vi);
+      dgS[row * ldg + H + hh] = lf2bf(vf);
+      dgS[row * ldg + 2 * H + hh] = lf2bf(vg);
+      dgS[row * ldg + 3 * H + hh] = lf2bf(vo);
+      if (row < rows_here) {
+        dGg[gbase + hh] = lf2bf(vi);
+        dGg[gbase + H + hh] = lf2bf(vf);
+        dGg[gbase + 2 * H + hh] = lf2bf(vg);
+        dGg[gbase + 3 * H + hh] = lf2bf(vo);
+      }
+    }
+    // zero the LDS pad columns once is unnecessary: ldg cols >= H4 are
+    // never written but also never read as MFMA K (K loop runs over H4).
+    __syncthreads();
+
+    // ---- dh_carry = dgates @ Wh^T (MFMA): out [ROWS][H] ----
+    // wave w owns rows w*16..w*16+15 (fm = w), cols 0..63 (fn 0..3).
+    {
+      f32x4 acc[4] = {};
+      for (int kk = 0; kk < H4; kk += 32) {
+        bf16x8 a = *reinterpret_cast<const bf16x8*>(
+            &dgS[(wid * 16 + l15) * ldg + kk + kslot * 8]);
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+          int col = fn * 16 + l15;  // h index
+          bf16x8 b = *reinterpret_cast<const bf16x8*>(
+              &WhN[(size_t)min(col, H - 1) * ldg + kk + kslot * 8]);
+          acc[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, b, acc[fn], 0, 0, 0);
+        }
+      }
+      __syncthreads();  // dgS reads done before dhS overwrite? different
+                        // arrays — barrier orders dhS consumers above.
+      #pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int col = fn * 16 + l15;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = wid * 16 + (lane >> 4) * 4 + r;
+          if (col < H) dhS[row * LDK + col] = lf2bf(acc[fn][r]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// ---------------------------------------------------------------------------
+namespace gordo_lstm {
+
+// pad 4H up to the MFMA K-step (32) plus 8: the K loop's last
+// fragment read may touch columns [H4, ru32(H4)); they must exist in
+// the row and be zero.
+inline int pad_ldg(int h4) { return ((h4 + 31) & ~31) + 8; }
+
+std::vector<torch::Tensor> lstm_seq_fwd(torch::Tensor xW, torch::Tensor Wh) {
+  TORCH_CHECK(xW.is_cuda() && xW.dim() == 4, "xW must be [G,B,T,4H] on GPU");
+  auto xc = xW.to(torch::kBFloat16).contiguous();
+  auto Whc = Wh.to(torch::kBFloat16).contiguous();
+  int G = xc.size(0), B = xc.size(1), T = xc.size(2), H4 = xc.size(3);
+  int H = H4 / 4;
+  TORCH_CHECK(H <= 64, "lstm_seq_fwd supports H <= 64");
+  int ldg = pad_ldg(H4);
+  auto hs = torch::empty({G, B, T, H}, xc.options());
+  auto cs = torch::empty({G, B, T, H}, xc.options().dtype(torch::kFloat32));
+  auto gacts = torch::empty({G, B, T, H4}, xc.options());
+  size_t lds = (size_t)H4 * LDK * 2 + (size_t)ROWS * LDK * 2 +
+               (size_t)ROWS * ldg * 2 + (size_t)ROWS * H * 4;
+  TORCH_CHECK(lds <= 160 * 1024, "LDS budget exceeded");
+  int blocks = G * ((B + ROWS - 1) / ROWS);
+  hipLaunchKernelGGL(lstm_seq_fwd_kernel, dim3(blocks), dim3(256), lds,
+                     at::cuda::getCurrentCUDAStream().stream(),
+                     (const bf16*)xc.data_ptr(), (const bf16*)Whc.data_ptr(),
+                     (bf16*)hs.data_ptr(), cs.data_ptr<float>(),
+                     (bf16*)gacts.data_ptr(), B, T, H, ldg);
+  return {hs, cs, gacts};
+}
+
+torch::Tensor lstm_seq_bwd(torch::Tensor dSeq, torch::Tensor gacts,
+                           torch::Tensor cs, torch::Tensor Wh,
+                           bool last_only) {
+  TORCH_CHECK(gacts.is_cuda() && gacts.dim() == 4, "gacts must be [G,B,T,4H]");
+  auto dc = dSeq.to(torch::kBFloat16).contiguous();
+  auto gc = gacts.to(torch::kBFloat16).contiguous();
+  auto cc = cs.to(torch::kFloat32).contiguous();
+  auto Whc = Wh.to(torch::kBFloat16).contiguous();
+  int G = gc.size(0), B = gc.size(1), T = gc.size(2), H4 = gc.size(3);
+  int H = H4 / 4;
+  TORCH_CHECK(H <= 64, "lstm_seq_bwd supports H <= 64");
+  int ldg = pad_ldg(H4);
+  auto dG = torch::empty_like(gc);
+  size_t lds = (size_t)H * ldg * 2 + (size_t)ROWS * ldg * 2 +
+               (size_t)ROWS * LDK * 2 + (size_t)ROWS * H * 4;
+  TORCH_CHECK(lds <= 160 * 1024, "LDS budget exceeded");
+  int blocks = G * ((B + ROWS - 1) / ROWS);
+  hipLaunchKernelGGL(lstm_seq_bwd_kernel, dim3(blocks), dim3(256), lds,
+                     at::cuda::getCurrentCUDAStream().stream(),
+                     (const bf16*)dc.data_ptr(), (const bf16*)gc.data_ptr(),
+                     cc.data_ptr<float>(), (const bf16*)Whc.data_ptr(),
+                     (bf16*)dG.data_ptr(), B, T, H, ldg, last_only ? 1 : 0);
+  return dG;
+}
+
+}  // namespace gordo_lstm

@@ -30,7 +30,7 @@ def to_dev_bf16(t):
 
 
 # bf16 has ~3 decimal digits; accumulation in fp32.
-RTOL, ATOL = 2e-2, 2e-2
+RTOL, ATOL = 2.5e-2, 2.5e-2
 
 
 @pytest.mark.parametrize(
@@ -172,3 +172,64 @@ def test_lstm_pointwise_roundtrip():
         got_dg.float().cpu(), want_dg, rtol=3e-2, atol=2e-2
     )
     torch.testing.assert_close(got_dcp.cpu(), want_dcp, rtol=3e-2, atol=2e-2)
+
+
+@pytest.mark.parametrize("G,B,T,H,last_only", [
+    (2, 64, 12, 38, False),
+    (1, 100, 20, 42, True),   # edge row tile, odd H
+    (3, 32, 144, 25, True),   # full lookback
+])
+def test_lstm_seq_fused_vs_reference(G, B, T, H, last_only):
+    """Fused sequence-scan kernels vs the per-timestep fp32 oracle."""
+    require_hip()
+    H4 = 4 * H
+    xW = _rand(G, B, T, H4, seed=30)
+    Wh = _rand(G, H, H4, seed=31) * 0.3
+
+    # reference forward (fp32, per-timestep)
+    hs_ref = torch.empty(G, B, T, H)
+    cs_ref = torch.empty(G, B, T, H)
+    ga_ref = torch.empty(G, B, T, H4)
+    h = torch.zeros(G, B, H)
+    c = torch.zeros(G, B, H)
+    for t in range(T):
+        gates = xW[:, :, t] + torch.bmm(h, Wh)
+        h, c, ga = ref.lstm_pointwise_fwd(gates, c)
+        hs_ref[:, :, t] = h
+        cs_ref[:, :, t] = c
+        ga_ref[:, :, t] = ga
+
+    got_hs, got_cs, got_ga = ops.lstm_seq_fwd(to_dev_bf16(xW), to_dev_bf16(Wh))
+    torch.testing.assert_close(
+        got_hs.float().cpu(), hs_ref, rtol=5e-2, atol=3e-2
+    )
+    torch.testing.assert_close(got_cs.cpu(), cs_ref, rtol=5e-2, atol=3e-2)
+
+    # reference backward
+    if last_only:
+        dSeq = _rand(G, B, H, seed=32)
+    else:
+        dSeq = _rand(G, B, T, H, seed=32)
+    dG_ref = torch.empty(G, B, T, H4)
+    dh = torch.zeros(G, B, H)
+    dc = torch.zeros(G, B, H)
+    for t in range(T - 1, -1, -1):
+        dh_t = dh.clone()
+        if last_only:
+            if t == T - 1:
+                dh_t += dSeq
+        else:
+            dh_t += dSeq[:, :, t]
+        c_prev = cs_ref[:, :, t - 1] if t > 0 else torch.zeros(G, B, H)
+        dgates, dc = ref.lstm_pointwise_bwd(
+            dh_t, dc, ga_ref[:, :, t], cs_ref[:, :, t], c_prev
+        )
+        dG_ref[:, :, t] = dgates
+        dh = torch.bmm(dgates, Wh.transpose(1, 2))
+
+    got_dG = ops.lstm_seq_bwd(
+        to_dev_bf16(dSeq), got_ga, got_cs, to_dev_bf16(Wh), last_only
+    )
+    torch.testing.assert_close(
+        got_dG.float().cpu(), dG_ref, rtol=8e-2, atol=4e-2
+    )
