@@ -100,7 +100,16 @@ def main():
     ap.add_argument("--steps", type=int, default=2)
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--machines-per-gpu", type=int, default=MACHINES_PER_GPU)
+    ap.add_argument("--verbose", action="store_true",
+                    help="log per-phase build timings to stderr")
     args = ap.parse_args()
+
+    import logging
+
+    logging.basicConfig(
+        level=logging.INFO if args.verbose else logging.WARNING,
+        stream=sys.stderr,
+    )
 
     from gordo_amd.machine import Machine
     from gordo_amd.parallel import init_distributed, shard_machines
