@@ -208,10 +208,16 @@ __global__ __launch_bounds__(256) void grouped_gemm_kernel(
 __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ dZ,
     float* __restrict__ dW, int M, int N, int K, int kt, int nt,
-    int nblocks) {
+    int nblocks, int mchunk) {
   __shared__ bf16 sm[2 * BM * LDT];
   bf16* As = sm;             // At: [k 64][m 32+pad]
   bf16* Zs = sm + BM * LDT;  // Zt: [n 64][m 32+pad]
+
+  // split-M: blockIdx.y selects an M-chunk; partials atomicAdd into the
+  // fp32 dW (the reduction dim M is B*T ~ 37k for LSTM wgrads — without
+  // the split only G*kt*nt workgroups exist and the chip idles).
+  const int m_begin = blockIdx.y * mchunk;
+  const int m_end = min(M, m_begin + mchunk);
 
   GemmCoord blk = decode_block(nblocks, kt, nt);
   const int g = blk.g, k0 = blk.m0, n0 = blk.n0;  // m-slot carries k-tile
@@ -229,7 +235,7 @@ __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
 
   f32x4 acc[2][2] = {};
 
-  for (int m0 = 0; m0 < M; m0 += BK) {
+  for (int m0 = m_begin; m0 < m_end; m0 += BK) {
     // stage A[m0..+32][k0..+64] transposed into As[k][m]
     {
       int k = tid >> 2;           // 0..63
@@ -271,6 +277,7 @@ __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
     __syncthreads();
   }
 
+  const bool single_chunk = gridDim.y == 1;
   #pragma unroll
   for (int fm = 0; fm < 2; ++fm) {
     #pragma unroll
@@ -281,22 +288,31 @@ __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
       for (int r = 0; r < 4; ++r) {
         int row = k0 + wk + fm * 16 + (lane >> 4) * 4 + r;
         if (row >= K) continue;
-        Wg[(size_t)row * N + col] = acc[fm][fn][r];
+        if (single_chunk)
+          Wg[(size_t)row * N + col] = acc[fm][fn][r];
+        else
+          atomicAdd(&Wg[(size_t)row * N + col], acc[fm][fn][r]);
       }
     }
   }
 }
 
-// column sum: db[g, N] = sum_m dZ[g, m, N]  (fp32 out)
+// column sum: db[g, N] = sum_m dZ[g, m, N]  (fp32 out, split-M partials)
 __global__ void colsum_kernel(const bf16* __restrict__ dZ,
-                              float* __restrict__ db, int M, int N) {
-  int g = blockIdx.y;
+                              float* __restrict__ db, int M, int N,
+                              int mchunk) {
+  int g = blockIdx.z;
   int col = blockIdx.x * blockDim.x + threadIdx.x;
   if (col >= N) return;
+  int m_begin = blockIdx.y * mchunk;
+  int m_end = min(M, m_begin + mchunk);
   const bf16* Zg = dZ + (size_t)g * M * N;
   float s = 0.f;
-  for (int m = 0; m < M; ++m) s += bf2f(Zg[(size_t)m * N + col]);
-  db[(size_t)g * N + col] = s;
+  for (int m = m_begin; m < m_end; ++m) s += bf2f(Zg[(size_t)m * N + col]);
+  if (gridDim.y == 1)
+    db[(size_t)g * N + col] = s;
+  else
+    atomicAdd(&db[(size_t)g * N + col], s);
 }
 
 // ---------------------------------------------------------------------------
@@ -510,18 +526,28 @@ std::vector<torch::Tensor> grouped_linear_wgrad(torch::Tensor X,
   auto Xc = to_bf16c(X);
   auto Zc = to_bf16c(dZ);
   int G = Xc.size(0), M = Xc.size(1), K = Xc.size(2), N = Zc.size(2);
-  auto dW = torch::empty({G, K, N},
-                         Xc.options().dtype(torch::kFloat32));
-  auto db = torch::empty({G, N}, Xc.options().dtype(torch::kFloat32));
   int kt = ceil_div(K, BM), nt = ceil_div(N, BN);
   int nblocks = G * kt * nt;
-  hipLaunchKernelGGL(grouped_wgrad_kernel, dim3(nblocks), dim3(256), 0,
-                     cur_stream(), (const bf16*)Xc.data_ptr(),
-                     (const bf16*)Zc.data_ptr(), dW.data_ptr<float>(), M, N,
-                     K, kt, nt, nblocks);
-  hipLaunchKernelGGL(colsum_kernel, dim3(ceil_div(N, 256), G), dim3(256), 0,
-                     cur_stream(), (const bf16*)Zc.data_ptr(),
-                     db.data_ptr<float>(), M, N);
+  // choose the M split so the grid comfortably fills 256 CUs
+  int target_chunks = std::max(1, 2048 / std::max(nblocks, 1));
+  int mchunk = std::max(BK, (int)(((M + target_chunks - 1) / target_chunks
+                                   + BK - 1) / BK * BK));
+  int nchunks = ceil_div(M, mchunk);
+  auto opts = Xc.options().dtype(torch::kFloat32);
+  auto dW = nchunks > 1 ? torch::zeros({G, K, N}, opts)
+                        : torch::empty({G, K, N}, opts);
+  auto db = nchunks > 1 ? torch::zeros({G, N}, opts)
+                        : torch::empty({G, N}, opts);
+  hipLaunchKernelGGL(grouped_wgrad_kernel, dim3(nblocks, nchunks),
+                     dim3(256), 0, cur_stream(),
+                     (const bf16*)Xc.data_ptr(), (const bf16*)Zc.data_ptr(),
+                     dW.data_ptr<float>(), M, N, K, kt, nt, nblocks, mchunk);
+  int db_chunks = std::max(1, std::min(64, M / 1024));
+  int db_mchunk = ceil_div(M, db_chunks);
+  hipLaunchKernelGGL(colsum_kernel,
+                     dim3(ceil_div(N, 256), ceil_div(M, db_mchunk), G),
+                     dim3(256), 0, cur_stream(), (const bf16*)Zc.data_ptr(),
+                     db.data_ptr<float>(), M, N, db_mchunk);
   return {dW, db};
 }
 
