@@ -37,8 +37,11 @@ DEV_INLINE float lbf2f(bf16 v) { return __bfloat162float(v); }
 DEV_INLINE bf16 lf2bf(float v) { return __float2bfloat16(v); }
 DEV_INLINE float sigmoidf_(float x) { return 1.f / (1.f + __expf(-x)); }
 
-constexpr int ROWS = 64;   // window rows per workgroup
 constexpr int LDK = 72;    // padded K-row length for h / WhT tiles (bf16)
+// ROWS (window rows per workgroup) is a template parameter: 64 rows for
+// big launches, 32 when the grid would underfill the 256 CUs (doubles
+// the number of independent workgroups on the bench's G*ceil(B/64)
+// ~ 250-workgroup shape).
 
 // ---------------------------------------------------------------------------
 // Forward scan.
@@ -49,6 +52,7 @@ constexpr int LDK = 72;    // padded K-row length for h / WhT tiles (bf16)
 //   gacts [G, B, T, 4H]  bf16 out (activated i,f,g,o)
 // grid.x = G * ceil(B/64); dynamic LDS.
 // ---------------------------------------------------------------------------
+template <int ROWS>
 __global__ __launch_bounds__(256) void lstm_seq_fwd_kernel(
     const bf16* __restrict__ xW, const bf16* __restrict__ Wh,
     bf16* __restrict__ hs, float* __restrict__ cs,
@@ -92,10 +96,11 @@ __global__ __launch_bounds__(256) void lstm_seq_fwd_kernel(
   for (int t = 0; t < T; ++t) {
     // ---- gates = h @ Wh (MFMA) + xW_t ----
     if (wave_active) {
-      f32x4 acc[4][4] = {};
+      constexpr int FM = ROWS / 16;
+      f32x4 acc[FM][4] = {};
       for (int kk = 0; kk < H; kk += 32) {
         #pragma unroll
-        for (int fm = 0; fm < 4; ++fm) {
+        for (int fm = 0; fm < FM; ++fm) {
           bf16x8 a = *reinterpret_cast<const bf16x8*>(
               &hS[(fm * 16 + l15) * LDK + kk + kslot * 8]);
           #pragma unroll
@@ -109,7 +114,7 @@ __global__ __launch_bounds__(256) void lstm_seq_fwd_kernel(
         }
       }
       #pragma unroll
-      for (int fm = 0; fm < 4; ++fm) {
+      for (int fm = 0; fm < FM; ++fm) {
         #pragma unroll
         for (int fn = 0; fn < 4; ++fn) {
           int col = wcol0 + fn * 16 + l15;
@@ -162,6 +167,7 @@ __global__ __launch_bounds__(256) void lstm_seq_fwd_kernel(
 //   Wh      [G, H, 4H] bf16
 //   dG      [G, B, T, 4H] bf16 out — pre-activation gate grads.
 // ---------------------------------------------------------------------------
+template <int ROWS>
 __global__ __launch_bounds__(256) void lstm_seq_bwd_kernel(
     const bf16* __restrict__ dSeq, const bf16* __restrict__ gacts,
     const float* __restrict__ cs, const bf16* __restrict__ Wh,
@@ -253,7 +259,7 @@ __global__ __launch_bounds__(256) void lstm_seq_bwd_kernel(
 
     // ---- dh_carry = dgates @ Wh^T (MFMA): out [ROWS][H] ----
     // wave w owns rows w*16..w*16+15 (fm = w), cols 0..63 (fn 0..3).
-    {
+    if (wid < ROWS / 16) {
       f32x4 acc[4] = {};
       for (int kk = 0; kk < H4; kk += 32) {
         bf16x8 a = *reinterpret_cast<const bf16x8*>(
@@ -267,8 +273,6 @@ __global__ __launch_bounds__(256) void lstm_seq_bwd_kernel(
               a, b, acc[fn], 0, 0, 0);
         }
       }
-      __syncthreads();  // dgS reads done before dhS overwrite? different
-                        // arrays — barrier orders dhS consumers above.
       #pragma unroll
       for (int fn = 0; fn < 4; ++fn) {
         int col = fn * 16 + l15;
@@ -291,6 +295,11 @@ namespace gordo_lstm {
 // the row and be zero.
 inline int pad_ldg(int h4) { return ((h4 + 31) & ~31) + 8; }
 
+inline int pick_rows(int G, int B) {
+  // 32-row tiles when 64-row tiles would underfill the 256 CUs
+  return (G * ((B + 63) / 64) >= 512) ? 64 : 32;
+}
+
 std::vector<torch::Tensor> lstm_seq_fwd(torch::Tensor xW, torch::Tensor Wh) {
   TORCH_CHECK(xW.is_cuda() && xW.dim() == 4, "xW must be [G,B,T,4H] on GPU");
   auto xc = xW.to(torch::kBFloat16).contiguous();
@@ -302,15 +311,24 @@ std::vector<torch::Tensor> lstm_seq_fwd(torch::Tensor xW, torch::Tensor Wh) {
   auto hs = torch::empty({G, B, T, H}, xc.options());
   auto cs = torch::empty({G, B, T, H}, xc.options().dtype(torch::kFloat32));
   auto gacts = torch::empty({G, B, T, H4}, xc.options());
-  size_t lds = (size_t)H4 * LDK * 2 + (size_t)ROWS * LDK * 2 +
-               (size_t)ROWS * ldg * 2 + (size_t)ROWS * H * 4;
+  int rows = pick_rows(G, B);
+  size_t lds = (size_t)H4 * LDK * 2 + (size_t)rows * LDK * 2 +
+               (size_t)rows * ldg * 2 + (size_t)rows * H * 4;
   TORCH_CHECK(lds <= 160 * 1024, "LDS budget exceeded");
-  int blocks = G * ((B + ROWS - 1) / ROWS);
-  hipLaunchKernelGGL(lstm_seq_fwd_kernel, dim3(blocks), dim3(256), lds,
-                     at::cuda::getCurrentCUDAStream().stream(),
-                     (const bf16*)xc.data_ptr(), (const bf16*)Whc.data_ptr(),
-                     (bf16*)hs.data_ptr(), cs.data_ptr<float>(),
-                     (bf16*)gacts.data_ptr(), B, T, H, ldg);
+  int blocks = G * ((B + rows - 1) / rows);
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  if (rows == 64)
+    hipLaunchKernelGGL(lstm_seq_fwd_kernel<64>, dim3(blocks), dim3(256), lds,
+                       stream, (const bf16*)xc.data_ptr(),
+                       (const bf16*)Whc.data_ptr(), (bf16*)hs.data_ptr(),
+                       cs.data_ptr<float>(), (bf16*)gacts.data_ptr(), B, T, H,
+                       ldg);
+  else
+    hipLaunchKernelGGL(lstm_seq_fwd_kernel<32>, dim3(blocks), dim3(256), lds,
+                       stream, (const bf16*)xc.data_ptr(),
+                       (const bf16*)Whc.data_ptr(), (bf16*)hs.data_ptr(),
+                       cs.data_ptr<float>(), (bf16*)gacts.data_ptr(), B, T, H,
+                       ldg);
   return {hs, cs, gacts};
 }
 
@@ -327,15 +345,24 @@ torch::Tensor lstm_seq_bwd(torch::Tensor dSeq, torch::Tensor gacts,
   TORCH_CHECK(H <= 64, "lstm_seq_bwd supports H <= 64");
   int ldg = pad_ldg(H4);
   auto dG = torch::empty_like(gc);
-  size_t lds = (size_t)H * ldg * 2 + (size_t)ROWS * ldg * 2 +
-               (size_t)ROWS * LDK * 2 + (size_t)ROWS * H * 4;
+  int rows = pick_rows(G, B);
+  size_t lds = (size_t)H * ldg * 2 + (size_t)rows * ldg * 2 +
+               (size_t)rows * LDK * 2 + (size_t)rows * H * 4;
   TORCH_CHECK(lds <= 160 * 1024, "LDS budget exceeded");
-  int blocks = G * ((B + ROWS - 1) / ROWS);
-  hipLaunchKernelGGL(lstm_seq_bwd_kernel, dim3(blocks), dim3(256), lds,
-                     at::cuda::getCurrentCUDAStream().stream(),
-                     (const bf16*)dc.data_ptr(), (const bf16*)gc.data_ptr(),
-                     cc.data_ptr<float>(), (const bf16*)Whc.data_ptr(),
-                     (bf16*)dG.data_ptr(), B, T, H, ldg, last_only ? 1 : 0);
+  int blocks = G * ((B + rows - 1) / rows);
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  if (rows == 64)
+    hipLaunchKernelGGL(lstm_seq_bwd_kernel<64>, dim3(blocks), dim3(256), lds,
+                       stream, (const bf16*)dc.data_ptr(),
+                       (const bf16*)gc.data_ptr(), cc.data_ptr<float>(),
+                       (const bf16*)Whc.data_ptr(), (bf16*)dG.data_ptr(), B,
+                       T, H, ldg, last_only ? 1 : 0);
+  else
+    hipLaunchKernelGGL(lstm_seq_bwd_kernel<32>, dim3(blocks), dim3(256), lds,
+                       stream, (const bf16*)dc.data_ptr(),
+                       (const bf16*)gc.data_ptr(), cc.data_ptr<float>(),
+                       (const bf16*)Whc.data_ptr(), (bf16*)dG.data_ptr(), B,
+                       T, H, ldg, last_only ? 1 : 0);
   return dG;
 }
 

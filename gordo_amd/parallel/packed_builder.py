@@ -276,9 +276,16 @@ class PackedFleetBuilder:
         cv_mode = str(evaluation.get("cv_mode", "full_build")).lower()
         cv_duration = None
         t0_all = time.time()
+        # ONE pack serves every fold fit and the final fit: weight init
+        # (per-model glorot + orthogonal QR on CPU) costs ~1 s per pack,
+        # so fold fits reset to an init snapshot instead of re-creating.
+        pack = self._make_pack(spec, group)
+        init_snapshot = pack.store.p32.clone()
         if cv_mode in ("cross_val_only", "full_build"):
             t0 = time.time()
-            self._cross_validate_group(group, Xt_list, y_list, spec, fit_args)
+            self._cross_validate_group(
+                group, Xt_list, y_list, spec, fit_args, pack, init_snapshot
+            )
             cv_duration = time.time() - t0
             if cv_mode == "cross_val_only":
                 for p in group:
@@ -286,7 +293,7 @@ class PackedFleetBuilder:
                 return
 
         # final full fit
-        pack = self._make_pack(spec, group)
+        self._reset_pack(pack, init_snapshot)
         Xd = self._stack(Xt_list, pack)
         Yd = self._stack(y_list, pack)
         t0 = time.time()
@@ -327,6 +334,12 @@ class PackedFleetBuilder:
             len(group), time.time() - t0_all,
         )
 
+    @staticmethod
+    def _reset_pack(pack, init_snapshot: torch.Tensor):
+        pack.store.p32.copy_(init_snapshot)
+        pack.store.reset_adam()
+        pack.store.sync_lp()
+
     def _cross_validate_group(
         self,
         group: List[MachinePlan],
@@ -334,6 +347,8 @@ class PackedFleetBuilder:
         y_list: List[np.ndarray],
         spec: ModelSpec,
         fit_args: Dict[str, Any],
+        pack,
+        init_snapshot: torch.Tensor,
     ):
         """Packed equivalent of sklearn cross_validate +
         DiffBasedAnomalyDetector.cross_validate: per fold, fresh packs
@@ -372,7 +387,7 @@ class PackedFleetBuilder:
 
         for fold_i, (train_idx, test_idx) in enumerate(folds):
             t_f0 = time.time()
-            pack = self._make_pack(spec, group)
+            self._reset_pack(pack, init_snapshot)
             Xd = self._stack([x[train_idx] for x in Xt_list], pack)
             Yd = self._stack([y[train_idx] for y in y_list], pack)
             pack.fit(Xd, Yd, **_engine_fit_args(fit_args))
