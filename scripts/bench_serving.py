@@ -1,0 +1,140 @@
+#!/usr/bin/env python3
+"""
+Serving benchmark — ml_server predictions/sec (BASELINE.md metric #2,
+replicating reference benchmarks/test_ml_server.py's payload shape:
+100-sample x n_tags JSON POSTs to /prediction and /anomaly/prediction).
+
+Trains a 50-tag feedforward DiffBased model (BASELINE config #2), dumps
+it into a model-collection dir, serves it through the Flask app
+in-process and measures sustained request throughput with concurrent
+worker threads. On a GPU box the model's forward runs through the HIP
+grouped-GEMM path.
+"""
+import argparse
+import json
+import os
+import statistics
+import sys
+import tempfile
+import threading
+import time
+
+import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+N_TAGS = 50
+ROWS = 100
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--rounds", type=int, default=200)
+    ap.add_argument("--threads", type=int, default=8)
+    ap.add_argument("--endpoint", default="anomaly",
+                    choices=["anomaly", "prediction", "both"])
+    args = ap.parse_args()
+
+    import pandas as pd
+    import torch
+
+    from gordo_amd import serializer
+    from gordo_amd.builder import local_build
+    from gordo_amd.server.utils import dataframe_to_dict
+
+    sensors = [f"bench-tag-{i}" for i in range(N_TAGS)]
+    tag_block = "\n".join(f"        - {s}" for s in sensors)
+    config = f"""
+machines:
+  - dataset: |
+      tags:
+{tag_block}
+      train_start_date: '2019-01-01T00:00:00+00:00'
+      train_end_date: '2019-01-08T00:00:00+00:00'
+      type: SineWaveDataset
+    model: |
+      gordo.machine.model.anomaly.diff.DiffBasedAnomalyDetector:
+        require_thresholds: true
+        base_estimator:
+          sklearn.pipeline.Pipeline:
+            steps:
+            - sklearn.preprocessing.MinMaxScaler
+            - gordo.machine.model.models.KerasAutoEncoder:
+                kind: feedforward_hourglass
+                epochs: 3
+    name: serve-bench
+"""
+    with tempfile.TemporaryDirectory() as td:
+        collection = os.path.join(td, "1577836800000")
+        for model, machine in local_build(config):
+            d = os.path.join(collection, machine.name)
+            serializer.dump(model, d, metadata=json.loads(machine.to_json()),
+                            info={})
+        os.environ["MODEL_COLLECTION_DIR"] = collection
+        from gordo_amd.server.server import build_app
+
+        app = build_app()
+        app.testing = True
+
+        X = pd.DataFrame(np.random.random((ROWS, N_TAGS)), columns=sensors)
+        payload = {"X": dataframe_to_dict(X), "y": dataframe_to_dict(X)}
+        results = {}
+        for endpoint in (
+            ["anomaly", "prediction"] if args.endpoint == "both"
+            else [args.endpoint]
+        ):
+            url = (
+                "/gordo/v0/bench/serve-bench/anomaly/prediction"
+                if endpoint == "anomaly"
+                else "/gordo/v0/bench/serve-bench/prediction"
+            )
+            # warmup + correctness
+            client = app.test_client()
+            resp = client.post(url, json=payload)
+            assert resp.status_code == 200, resp.data[:300]
+
+            latencies = []
+            lock = threading.Lock()
+            counter = {"n": 0}
+
+            def worker():
+                c = app.test_client()
+                while True:
+                    with lock:
+                        if counter["n"] >= args.rounds:
+                            return
+                        counter["n"] += 1
+                    t0 = time.perf_counter()
+                    r = c.post(url, json=payload)
+                    dt = time.perf_counter() - t0
+                    assert r.status_code == 200
+                    with lock:
+                        latencies.append(dt)
+
+            t0 = time.time()
+            threads = [
+                threading.Thread(target=worker) for _ in range(args.threads)
+            ]
+            for t in threads:
+                t.start()
+            for t in threads:
+                t.join()
+            elapsed = time.time() - t0
+            rps = len(latencies) / elapsed
+            results[endpoint] = {
+                "requests_per_sec": rps,
+                "predictions_per_sec": rps * ROWS,
+                "mean_latency_ms": statistics.mean(latencies) * 1000,
+                "p50_latency_ms": statistics.median(latencies) * 1000,
+                "rounds": len(latencies),
+                "threads": args.threads,
+                "payload_rows": ROWS,
+                "n_tags": N_TAGS,
+                "device": "cuda" if torch.cuda.is_available() else "cpu",
+            }
+        print(json.dumps({"metric": "ml_server predictions/sec",
+                          "results": results}))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,106 @@
+"""Client tests against the in-process Flask server (the reference's
+"mock ML-server mesh" pattern: every HTTP call routes into the test
+client — tests/conftest.py:333-422 there; a transport shim here)."""
+import threading
+
+import dateutil.parser
+import pandas as pd
+import pytest
+
+from gordo_amd.client import Client, ForwardPredictionsToDisk
+
+
+class FlaskSession:
+    """requests.Session-compatible shim over a Flask test client."""
+
+    def __init__(self, flask_app):
+        self.client = flask_app.test_client()
+        self.lock = threading.Lock()
+        self.fail_next = 0  # fault injection: 500s before succeeding
+
+    def request(self, method, url, params=None, json=None, **kwargs):
+        path = url.split("://", 1)[-1].split("/", 1)[1]
+        with self.lock:
+            if self.fail_next > 0:
+                self.fail_next -= 1
+                return FakeResponse(500)
+            resp = self.client.open(
+                "/" + path, method=method, json=json, query_string=params
+            )
+        return resp
+
+
+class FakeResponse:
+    def __init__(self, status_code):
+        self.status_code = status_code
+
+
+@pytest.fixture
+def client(flask_app, gordo_project):
+    return Client(
+        project=gordo_project,
+        host="server",
+        port=80,
+        scheme="http",
+        session=FlaskSession(flask_app),
+        parallelism=2,
+        n_retries=3,
+    )
+
+
+def test_get_revisions(client, gordo_revision):
+    revs = client.get_revisions()
+    assert revs["latest"] == gordo_revision
+
+
+def test_machine_names(client, gordo_name, second_gordo_name):
+    names = client.get_machine_names()
+    assert {gordo_name, second_gordo_name} <= set(names)
+
+
+def test_get_metadata(client, gordo_name):
+    meta = client.get_metadata()
+    assert gordo_name in meta
+    assert meta[gordo_name]["name"] == gordo_name
+
+
+def test_download_model(client, gordo_name):
+    models = client.download_model(targets=[gordo_name])
+    assert hasattr(models[gordo_name], "anomaly")
+
+
+def test_predict(client, gordo_name):
+    start = dateutil.parser.isoparse("2019-01-01T00:00:00+00:00")
+    end = dateutil.parser.isoparse("2019-01-02T00:00:00+00:00")
+    results = client.predict(start, end, targets=[gordo_name])
+    assert len(results) == 1
+    name, frame, errors = results[0]
+    assert name == gordo_name
+    assert errors == []
+    assert len(frame) > 0
+    assert "total-anomaly-scaled" in {c[0] for c in frame.columns}
+
+
+def test_predict_batched(client, gordo_name):
+    client.batch_size = 50  # force multiple POST batches
+    start = dateutil.parser.isoparse("2019-01-01T00:00:00+00:00")
+    end = dateutil.parser.isoparse("2019-01-02T00:00:00+00:00")
+    name, frame, errors = client.predict(start, end, targets=[gordo_name])[0]
+    assert errors == []
+    assert len(frame) == 144  # one day at 10min resolution
+
+
+def test_retry_on_5xx(client, gordo_name):
+    client.session.fail_next = 2  # two 500s, then success
+    names = client.get_machine_names()
+    assert gordo_name in names
+
+
+def test_forward_to_disk(tmp_path, client, gordo_name):
+    start = dateutil.parser.isoparse("2019-01-01T00:00:00+00:00")
+    end = dateutil.parser.isoparse("2019-01-02T00:00:00+00:00")
+    name, frame, errors = client.predict(start, end, targets=[gordo_name])[0]
+    fwd = ForwardPredictionsToDisk(str(tmp_path))
+    fwd.forward_predictions(frame, name)
+    stored = pd.read_parquet(tmp_path / f"{name}.parquet")
+    assert len(stored) == len(frame)
