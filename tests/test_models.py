@@ -78,7 +78,8 @@ def test_autoencoder_fit_predict_score(data):
 
 
 def test_autoencoder_training_reduces_loss(data):
-    model = KerasAutoEncoder(kind="feedforward_hourglass", epochs=15,
+    np.random.seed(0)  # pins the pack weight-init seed
+    model = KerasAutoEncoder(kind="feedforward_hourglass", epochs=25,
                              batch_size=32)
     # reconstruct a low-rank signal: loss must drop substantially
     t = np.linspace(0, 20, len(data))
