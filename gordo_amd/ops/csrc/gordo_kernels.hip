@@ -37,9 +37,11 @@ typedef __attribute__((ext_vector_type(8))) short bf16x8;
 // activation codes (keep in sync with gordo_amd/ops/reference.py)
 constexpr int ACT_LINEAR = 0, ACT_TANH = 1, ACT_RELU = 2, ACT_SIGMOID = 3;
 
+DEV_INLINE float fast_tanhf(float x);
+
 DEV_INLINE float act_apply(float z, int act) {
   switch (act) {
-    case ACT_TANH: return tanhf(z);
+    case ACT_TANH: return fast_tanhf(z);
     case ACT_RELU: return z > 0.f ? z : 0.f;
     case ACT_SIGMOID: return 1.f / (1.f + __expf(-z));
     default: return z;
@@ -53,6 +55,13 @@ DEV_INLINE float act_grad_from_output(float y, int act) {
     case ACT_SIGMOID: return y * (1.f - y);
     default: return 1.f;
   }
+}
+
+DEV_INLINE float fast_tanhf(float x) {
+  // tanh via one __expf: the ocml tanh is an order of magnitude slower
+  // and the LSTM gate math is transcendental-bound.
+  float e = __expf(2.f * x);
+  return (e - 1.f) / (e + 1.f);
 }
 
 DEV_INLINE float bf2f(bf16 v) { return __bfloat162float(v); }
@@ -207,8 +216,8 @@ __global__ __launch_bounds__(256) void grouped_gemm_kernel(
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ dZ,
-    float* __restrict__ dW, int M, int N, int K, int kt, int nt,
-    int nblocks, int mchunk) {
+    float* __restrict__ dW, float* __restrict__ db, int M, int N, int K,
+    int kt, int nt, int nblocks, int mchunk) {
   __shared__ bf16 sm[2 * BM * LDT];
   bf16* As = sm;             // At: [k 64][m 32+pad]
   bf16* Zs = sm + BM * LDT;  // Zt: [n 64][m 32+pad]
@@ -234,6 +243,7 @@ __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
   const int mslot = lane >> 4;
 
   f32x4 acc[2][2] = {};
+  float db_part = 0.f;
 
   for (int m0 = m_begin; m0 < m_end; m0 += BK) {
     // stage A[m0..+32][k0..+64] transposed into As[k][m]
@@ -248,7 +258,9 @@ __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
             (gk < K && gm < M) ? Ag[(size_t)gm * K + gk] : f2bf(0.f);
       }
     }
-    // stage dZ[m0..+32][n0..+64] transposed into Zs[n][m]
+    // stage dZ[m0..+32][n0..+64] transposed into Zs[n][m]; k-tile-0
+    // blocks fold the bias-grad column sum into the same pass (what a
+    // separate colsum kernel did with a latency-bound column walk).
     {
       int n = tid >> 2;
       int mm = (tid & 3) * 8;
@@ -256,8 +268,9 @@ __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
       #pragma unroll
       for (int e = 0; e < 8; ++e) {
         int gm = m0 + mm + e;
-        Zs[n * LDT + mm + e] =
-            (gn < N && gm < M) ? Zg[(size_t)gm * N + gn] : f2bf(0.f);
+        bf16 v = (gn < N && gm < M) ? Zg[(size_t)gm * N + gn] : f2bf(0.f);
+        Zs[n * LDT + mm + e] = v;
+        if (k0 == 0) db_part += bf2f(v);
       }
     }
     __syncthreads();
@@ -277,6 +290,10 @@ __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
     __syncthreads();
   }
 
+  if (k0 == 0) {
+    int gn = n0 + (tid >> 2);
+    if (gn < N) atomicAdd(&db[(size_t)g * N + gn], db_part);
+  }
   const bool single_chunk = gridDim.y == 1;
   #pragma unroll
   for (int fm = 0; fm < 2; ++fm) {
@@ -297,23 +314,6 @@ __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
   }
 }
 
-// column sum: db[g, N] = sum_m dZ[g, m, N]  (fp32 out, split-M partials)
-__global__ void colsum_kernel(const bf16* __restrict__ dZ,
-                              float* __restrict__ db, int M, int N,
-                              int mchunk) {
-  int g = blockIdx.z;
-  int col = blockIdx.x * blockDim.x + threadIdx.x;
-  if (col >= N) return;
-  int m_begin = blockIdx.y * mchunk;
-  int m_end = min(M, m_begin + mchunk);
-  const bf16* Zg = dZ + (size_t)g * M * N;
-  float s = 0.f;
-  for (int m = m_begin; m < m_end; ++m) s += bf2f(Zg[(size_t)m * N + col]);
-  if (gridDim.y == 1)
-    db[(size_t)g * N + col] = s;
-  else
-    atomicAdd(&db[(size_t)g * N + col], s);
-}
 
 // ---------------------------------------------------------------------------
 // Fused pointwise kernels
@@ -401,11 +401,11 @@ __global__ void lstm_pw_fwd_kernel(const bf16* __restrict__ gates,
     const bf16* grow = gates + row * 4 * H;
     float i_g = 1.f / (1.f + __expf(-bf2f(grow[hh])));
     float f_g = 1.f / (1.f + __expf(-bf2f(grow[H + hh])));
-    float g_g = tanhf(bf2f(grow[2 * H + hh]));
+    float g_g = fast_tanhf(bf2f(grow[2 * H + hh]));
     float o_g = 1.f / (1.f + __expf(-bf2f(grow[3 * H + hh])));
     float cc = f_g * c_prev[idx] + i_g * g_g;
     c[idx] = cc;
-    h[idx] = f2bf(o_g * tanhf(cc));
+    h[idx] = f2bf(o_g * fast_tanhf(cc));
     bf16* ga = gact + row * 4 * H;
     ga[hh] = f2bf(i_g);
     ga[H + hh] = f2bf(f_g);
@@ -431,7 +431,7 @@ __global__ void lstm_pw_bwd_kernel(
     float f_g = bf2f(ga[H + hh]);
     float g_g = bf2f(ga[2 * H + hh]);
     float o_g = bf2f(ga[3 * H + hh]);
-    float tc = tanhf(c[idx]);
+    float tc = fast_tanhf(c[idx]);
     float dhv = bf2f(dh[idx]);
     float dc = dc_next[idx] + dhv * o_g * (1.f - tc * tc);
     float di = dc * g_g;
@@ -536,18 +536,12 @@ std::vector<torch::Tensor> grouped_linear_wgrad(torch::Tensor X,
   auto opts = Xc.options().dtype(torch::kFloat32);
   auto dW = nchunks > 1 ? torch::zeros({G, K, N}, opts)
                         : torch::empty({G, K, N}, opts);
-  auto db = nchunks > 1 ? torch::zeros({G, N}, opts)
-                        : torch::empty({G, N}, opts);
+  auto db = torch::zeros({G, N}, opts);  // always atomically accumulated
   hipLaunchKernelGGL(grouped_wgrad_kernel, dim3(nblocks, nchunks),
                      dim3(256), 0, cur_stream(),
                      (const bf16*)Xc.data_ptr(), (const bf16*)Zc.data_ptr(),
-                     dW.data_ptr<float>(), M, N, K, kt, nt, nblocks, mchunk);
-  int db_chunks = std::max(1, std::min(64, M / 1024));
-  int db_mchunk = ceil_div(M, db_chunks);
-  hipLaunchKernelGGL(colsum_kernel,
-                     dim3(ceil_div(N, 256), ceil_div(M, db_mchunk), G),
-                     dim3(256), 0, cur_stream(), (const bf16*)Zc.data_ptr(),
-                     db.data_ptr<float>(), M, N, db_mchunk);
+                     dW.data_ptr<float>(), db.data_ptr<float>(), M, N, K,
+                     kt, nt, nblocks, mchunk);
   return {dW, db};
 }
 

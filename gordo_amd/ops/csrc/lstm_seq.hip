@@ -36,6 +36,10 @@ typedef __attribute__((ext_vector_type(8))) short bf16x8;
 DEV_INLINE float lbf2f(bf16 v) { return __bfloat162float(v); }
 DEV_INLINE bf16 lf2bf(float v) { return __float2bfloat16(v); }
 DEV_INLINE float sigmoidf_(float x) { return 1.f / (1.f + __expf(-x)); }
+DEV_INLINE float fast_tanhf_(float x) {
+  float e = __expf(2.f * x);
+  return (e - 1.f) / (e + 1.f);
+}
 
 constexpr int LDK = 72;    // padded K-row length for h / WhT tiles (bf16)
 // ROWS (window rows per workgroup) is a template parameter: 64 rows for
@@ -138,10 +142,10 @@ __global__ __launch_bounds__(256) void lstm_seq_fwd_kernel(
       const bf16* grow = &gS[row * ldg];
       float i_g = sigmoidf_(lbf2f(grow[hh]));
       float f_g = sigmoidf_(lbf2f(grow[H + hh]));
-      float g_g = tanhf(lbf2f(grow[2 * H + hh]));
+      float g_g = fast_tanhf_(lbf2f(grow[2 * H + hh]));
       float o_g = sigmoidf_(lbf2f(grow[3 * H + hh]));
       float cc = f_g * cS[row * H + hh] + i_g * g_g;
-      float hv = o_g * tanhf(cc);
+      float hv = o_g * fast_tanhf_(cc);
       cS[row * H + hh] = cc;
       hS[row * LDK + hh] = lf2bf(hv);
       if (row < rows_here) {
@@ -231,7 +235,7 @@ __global__ __launch_bounds__(256) void lstm_seq_bwd_kernel(
         cc = csg[cbase];
         cp = (t > 0) ? csg[cbase - H] : 0.f;
       }
-      float tc = tanhf(cc);
+      float tc = fast_tanhf_(cc);
       float dc = dcS[row * H + hh] + dh * o_g * (1.f - tc * tc);
       float di = dc * g_g;
       float df = dc * cp;
