@@ -61,7 +61,7 @@ DEV_INLINE float fast_tanhf(float x) {
   // tanh via one __expf: the ocml tanh is an order of magnitude slower
   // and the LSTM gate math is transcendental-bound.
   float e = __expf(2.f * x);
-  return (e - 1.f) / (e + 1.f);
+  return 1.f - 2.f / (e + 1.f);  // saturates to 1 when e overflows to inf
 }
 
 DEV_INLINE float bf2f(bf16 v) { return __bfloat162float(v); }

@@ -38,7 +38,7 @@ DEV_INLINE bf16 lf2bf(float v) { return __float2bfloat16(v); }
 DEV_INLINE float sigmoidf_(float x) { return 1.f / (1.f + __expf(-x)); }
 DEV_INLINE float fast_tanhf_(float x) {
   float e = __expf(2.f * x);
-  return (e - 1.f) / (e + 1.f);
+  return 1.f - 2.f / (e + 1.f);  // saturates to 1 when e overflows to inf
 }
 
 constexpr int LDK = 72;    // padded K-row length for h / WhT tiles (bf16)
