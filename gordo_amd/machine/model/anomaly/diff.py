@@ -227,6 +227,59 @@ class DiffBasedAnomalyDetector(AnomalyDetectorBase):
             return metric.ewm(span=self.window).mean()
         raise ValueError(f"Unknown smoothing_method {self.smoothing_method!r}")
 
+    # rows below which the H2D round-trip outweighs the fused kernel
+    _DEVICE_SCORE_MIN_ROWS = 512
+
+    def _fused_device_scores(self, data: pd.DataFrame, y: pd.DataFrame):
+        """Serving hot path: one fused HIP kernel (ops.anomaly_score,
+        kernel K9) computes every residual column family when a GPU is
+        present and the scaler is a fitted MinMaxScaler-style affine
+        transform. Returns (tag_scaled, total_scaled, tag_unscaled,
+        total_unscaled) as numpy, or None to use the pandas path."""
+        try:
+            import torch
+
+            from .... import ops
+
+            if (
+                len(data) < self._DEVICE_SCORE_MIN_ROWS
+                or not torch.cuda.is_available()
+                or not ops.hip_available()
+                or not hasattr(self.scaler, "scale_")
+                or not hasattr(self.scaler, "min_")
+            ):
+                return None
+            out = torch.as_tensor(
+                np.ascontiguousarray(
+                    data["model-output"].to_numpy(dtype=np.float32)
+                ),
+                device="cuda",
+            )
+            yt = torch.as_tensor(
+                np.ascontiguousarray(
+                    y.to_numpy(dtype=np.float32)[-len(data):, :]
+                ),
+                device="cuda",
+            )
+            scale = torch.as_tensor(
+                np.asarray(self.scaler.scale_, dtype=np.float32), device="cuda"
+            )
+            minv = torch.as_tensor(
+                np.asarray(self.scaler.min_, dtype=np.float32), device="cuda"
+            )
+            ts, tots, tu, totu, _, _ = ops.anomaly_score(
+                out, yt, scale, minv, None, 1.0
+            )
+            return (
+                ts.cpu().numpy(),
+                tots.cpu().numpy(),
+                tu.cpu().numpy(),
+                totu.cpu().numpy(),
+            )
+        except Exception:  # any wobble → exact pandas path
+            logger.debug("device anomaly scoring unavailable", exc_info=True)
+            return None
+
     def anomaly(
         self,
         X: pd.DataFrame,
@@ -249,36 +302,38 @@ class DiffBasedAnomalyDetector(AnomalyDetectorBase):
             frequency=frequency,
         )
 
-        model_out_scaled = pd.DataFrame(
-            self.scaler.transform(data["model-output"]),
-            columns=data["model-output"].columns,
-            index=data.index,
-        )
+        scored = self._fused_device_scores(data, y)
+        if scored is not None:
+            ts, tots, tu, totu = scored
+        else:
+            model_out_scaled = self.scaler.transform(data["model-output"])
+            scaled_y = self.scaler.transform(y)
+            ts = np.abs(model_out_scaled - scaled_y[-len(data):, :])
+            tots = np.square(ts).mean(axis=1)
+            tu = np.abs(
+                data["model-output"].to_numpy() - y.to_numpy()[-len(data):, :]
+            )
+            totu = np.square(tu).mean(axis=1)
 
-        # scaled per-tag anomaly, with y offset to match model output
-        scaled_y = self.scaler.transform(y)
-        tag_anomaly_scaled = np.abs(model_out_scaled - scaled_y[-len(data):, :])
-        tag_anomaly_scaled.columns = pd.MultiIndex.from_product(
-            (("tag-anomaly-scaled",), tag_anomaly_scaled.columns)
+        tag_anomaly_scaled = pd.DataFrame(
+            ts,
+            index=data.index,
+            columns=pd.MultiIndex.from_product(
+                (("tag-anomaly-scaled",), y.columns.tolist())
+            ),
         )
         data = data.join(tag_anomaly_scaled)
-        data["total-anomaly-scaled"] = np.square(
-            data["tag-anomaly-scaled"]
-        ).mean(axis=1)
+        data["total-anomaly-scaled"] = tots
 
         unscaled_abs_diff = pd.DataFrame(
-            data=np.abs(
-                data["model-output"].to_numpy() - y.to_numpy()[-len(data):, :]
-            ),
+            data=tu,
             index=data.index,
             columns=pd.MultiIndex.from_product(
                 (("tag-anomaly-unscaled",), y.columns.tolist())
             ),
         )
         data = data.join(unscaled_abs_diff)
-        data["total-anomaly-unscaled"] = np.square(
-            data["tag-anomaly-unscaled"]
-        ).mean(axis=1)
+        data["total-anomaly-unscaled"] = totu
 
         if self.window is not None and self.smoothing_method is not None:
             smooth_tag_anomaly_scaled = self._smoothing(tag_anomaly_scaled)

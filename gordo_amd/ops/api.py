@@ -120,6 +120,13 @@ def lstm_seq_available(H: int) -> bool:
     return hip_available() and H <= 64
 
 
+def anomaly_score(out, y, scale, minv, feat_thr, agg_thr):
+    """Fused serving-path anomaly scoring (GPU only; callers fall back
+    to numpy/pandas on CPU — machine/model/anomaly/diff.py)."""
+    return _require_hip().anomaly_score(out, y, scale, minv, feat_thr,
+                                        float(agg_thr))
+
+
 def lstm_pointwise_fwd(gates, c_prev):
     if _on_gpu(gates):
         return _require_hip().lstm_pointwise_fwd(gates, c_prev)

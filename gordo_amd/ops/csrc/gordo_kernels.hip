@@ -642,6 +642,14 @@ std::vector<torch::Tensor> lstm_pointwise_bwd(torch::Tensor dh,
 
 }  // namespace
 
+// fused anomaly scoring entry point (anomaly.hip)
+namespace gordo_anomaly {
+std::vector<torch::Tensor> anomaly_score(
+    torch::Tensor out, torch::Tensor y, torch::Tensor scale,
+    torch::Tensor minv, c10::optional<torch::Tensor> feat_thr,
+    double agg_thr);
+}  // namespace gordo_anomaly
+
 // fused LSTM sequence-scan entry points (lstm_seq.hip)
 namespace gordo_lstm {
 std::vector<torch::Tensor> lstm_seq_fwd(torch::Tensor xW, torch::Tensor Wh);
@@ -651,6 +659,8 @@ torch::Tensor lstm_seq_bwd(torch::Tensor dSeq, torch::Tensor gacts,
 }  // namespace gordo_lstm
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("anomaly_score", &gordo_anomaly::anomaly_score,
+          "fused DiffBased anomaly scoring (serving hot path)");
   mod.def("lstm_seq_fwd", &gordo_lstm::lstm_seq_fwd,
           "fused LSTM forward sequence scan (Wh resident in LDS)");
   mod.def("lstm_seq_bwd", &gordo_lstm::lstm_seq_bwd,

@@ -235,3 +235,78 @@ def test_lstm_seq_fused_vs_reference(G, B, T, H, last_only):
     torch.testing.assert_close(
         got_dG.float().cpu(), dG_ref, rtol=8e-2, atol=4e-2
     )
+
+
+def test_anomaly_score_kernel_vs_pandas():
+    """K9 fused serving kernel vs the exact pandas/numpy formulas."""
+    require_hip()
+    import numpy as np
+    from sklearn.preprocessing import MinMaxScaler
+
+    rng = np.random.default_rng(5)
+    N, F = 1000, 50
+    out = rng.random((N, F)).astype("float32")
+    y = rng.random((N, F)).astype("float32")
+    scaler = MinMaxScaler().fit(rng.random((200, F)) * 3)
+    thr = rng.random(F).astype("float32") + 0.1
+    agg = 0.37
+
+    ts_ref = np.abs(scaler.transform(out) - scaler.transform(y))
+    tots_ref = (ts_ref ** 2).mean(axis=1)
+    tu_ref = np.abs(out - y)
+    totu_ref = (tu_ref ** 2).mean(axis=1)
+
+    ts, tots, tu, totu, conf, tconf = ops.anomaly_score(
+        torch.as_tensor(out, device="cuda"),
+        torch.as_tensor(y, device="cuda"),
+        torch.as_tensor(scaler.scale_.astype("float32"), device="cuda"),
+        torch.as_tensor(scaler.min_.astype("float32"), device="cuda"),
+        torch.as_tensor(thr, device="cuda"),
+        agg,
+    )
+    np.testing.assert_allclose(ts.cpu().numpy(), ts_ref, rtol=1e-5, atol=1e-6)
+    np.testing.assert_allclose(tots.cpu().numpy(), tots_ref, rtol=1e-5,
+                               atol=1e-6)
+    np.testing.assert_allclose(tu.cpu().numpy(), tu_ref, rtol=1e-6, atol=1e-7)
+    np.testing.assert_allclose(totu.cpu().numpy(), totu_ref, rtol=1e-5,
+                               atol=1e-7)
+    np.testing.assert_allclose(conf.cpu().numpy(), tu_ref / thr, rtol=1e-5,
+                               atol=1e-6)
+    np.testing.assert_allclose(tconf.cpu().numpy(), tots_ref / agg,
+                               rtol=1e-5, atol=1e-6)
+
+
+def test_anomaly_frame_device_path_matches_pandas():
+    """DiffBased.anomaly at >=512 rows takes the fused kernel path; the
+    frame must equal the pandas path bitwise-close."""
+    require_hip()
+    import numpy as np
+    import pandas as pd
+    from sklearn.pipeline import Pipeline
+    from sklearn.preprocessing import MinMaxScaler
+
+    from gordo_amd.machine.model.anomaly.diff import DiffBasedAnomalyDetector
+    from gordo_amd.machine.model.models import KerasAutoEncoder
+
+    rng = np.random.default_rng(6)
+    X = pd.DataFrame(rng.random((800, 10)),
+                     columns=[f"t{i}" for i in range(10)])
+    det = DiffBasedAnomalyDetector(
+        base_estimator=Pipeline([
+            ("mms", MinMaxScaler()),
+            ("ae", KerasAutoEncoder(kind="feedforward_hourglass", epochs=1)),
+        ]),
+        require_thresholds=False,
+    )
+    det.fit(X, X)
+    frame_gpu = det.anomaly(X, X)
+    # force the pandas path and compare
+    det._DEVICE_SCORE_MIN_ROWS = 10 ** 9
+    frame_cpu = det.anomaly(X, X)
+    for col in ("tag-anomaly-scaled", "total-anomaly-scaled",
+                "tag-anomaly-unscaled", "total-anomaly-unscaled"):
+        np.testing.assert_allclose(
+            np.asarray(frame_gpu[col], dtype=float),
+            np.asarray(frame_cpu[col], dtype=float),
+            rtol=1e-4, atol=1e-5,
+        )
