@@ -1,0 +1,186 @@
+import pytest
+
+from gordo_amd.util import disk_registry
+from gordo_amd.util.text import replace_all_non_ascii_chars
+from gordo_amd.util.utils import capture_args
+from gordo_amd.util.version import (
+    GordoPR,
+    GordoRelease,
+    GordoSHA,
+    GordoSpecial,
+    parse_version,
+)
+import gordo_amd
+
+
+def test_disk_registry_roundtrip(tmp_path):
+    reg = tmp_path / "reg"
+    assert disk_registry.get_value(reg, "missing") is None
+    disk_registry.write_key(reg, "abc" * 40, "/some/path")
+    assert disk_registry.get_value(reg, "abc" * 40) == "/some/path"
+    # overwrite
+    disk_registry.write_key(reg, "abc" * 40, "/other/path")
+    assert disk_registry.get_value(reg, "abc" * 40) == "/other/path"
+    assert disk_registry.delete_value(reg, "abc" * 40) is True
+    assert disk_registry.delete_value(reg, "abc" * 40) is False
+
+
+def test_disk_registry_sanitizes_keys(tmp_path):
+    disk_registry.write_key(tmp_path, "../../evil key", "v")
+    assert disk_registry.get_value(tmp_path, "../../evil key") == "v"
+    # nothing escaped the registry dir
+    assert not (tmp_path.parent / "evil key").exists()
+
+
+def test_capture_args():
+    class A:
+        @capture_args
+        def __init__(self, x, y=2, **kwargs):
+            pass
+
+    a = A(1, y=5, extra="e")
+    assert a._params == {"x": 1, "y": 5, "extra": "e"}
+
+
+def test_replace_non_ascii():
+    assert replace_all_non_ascii_chars("héllo wörld", "_") == "h_llo w_rld"
+    assert replace_all_non_ascii_chars("ascii") == "ascii"
+
+
+@pytest.mark.parametrize(
+    "tag,expected",
+    [
+        ("1.2.3", GordoRelease(1, 2, 3)),
+        ("1.2", GordoRelease(1, 2)),
+        ("3", GordoRelease(3)),
+        ("latest", GordoSpecial("latest")),
+        ("stable", GordoSpecial("stable")),
+        ("pr-42", GordoPR(42)),
+        ("deadbeefcafe", GordoSHA("deadbeefcafe")),
+    ],
+)
+def test_docker_version_parse(tag, expected):
+    assert parse_version(tag) == expected
+
+
+def test_docker_version_roundtrip():
+    for tag in ("1.2.3", "latest", "pr-7"):
+        assert parse_version(tag).get_version() == tag
+
+
+def test_package_version_parse():
+    assert gordo_amd.parse_version("1.2.3") == (1, 2, False)
+    assert gordo_amd.parse_version("1.2.3.dev1")[2] is True
+    with pytest.raises(ValueError):
+        gordo_amd.parse_version("nope")
+
+
+def test_image_pull_policy():
+    from gordo_amd.workflow.workflow_generator import default_image_pull_policy
+
+    assert default_image_pull_policy("1.2.3") == "IfNotPresent"
+    assert default_image_pull_policy("1.2") == "Always"
+    assert default_image_pull_policy("latest") == "Always"
+
+
+def test_inf_imputer():
+    import numpy as np
+
+    from gordo_amd.machine.model.transformers import InfImputer
+
+    X = np.array([[1.0, np.inf], [-np.inf, 2.0], [3.0, 4.0]])
+    imp = InfImputer(strategy="minmax", delta=1.0)
+    out = imp.fit_transform(X)
+    assert np.isfinite(out).all()
+    assert out[0, 1] == pytest.approx(5.0)  # max(2,4)+1
+    assert out[1, 0] == pytest.approx(0.0)  # min(1,3)-1
+
+    imp2 = InfImputer(inf_fill_value=99.0, neg_inf_fill_value=-99.0)
+    out2 = imp2.fit_transform(X)
+    assert out2[0, 1] == 99.0
+    assert out2[1, 0] == -99.0
+
+
+def test_multiply_by():
+    from gordo_amd.machine.model.transformer_funcs.general import multiply_by
+
+    assert multiply_by(3, 4) == 12
+
+
+def test_reporters_base_roundtrip():
+    from gordo_amd.reporters import PostgresReporter
+
+    rep = PostgresReporter(host="myhost", port=5432)
+    params = rep.get_params()
+    assert params["host"] == "myhost"
+    d = rep.to_dict()
+    assert "PostgresReporter" in next(iter(d))
+
+
+def test_metadata_dataclasses_roundtrip():
+    from gordo_amd.machine.metadata import BuildMetadata, Metadata
+
+    m = Metadata(user_defined={"a": 1})
+    d = m.to_dict()
+    m2 = Metadata.from_dict(d)
+    assert m2.user_defined == {"a": 1}
+    assert isinstance(m2.build_metadata, BuildMetadata)
+
+
+def test_find_path_in_dict():
+    from gordo_amd.server.properties import find_path_in_dict
+
+    assert find_path_in_dict(["a", "b"], {"a": {"b": 3}}) == 3
+    with pytest.raises(KeyError):
+        find_path_in_dict(["a", "missing"], {"a": {}})
+
+
+def test_prometheus_metrics_app():
+    from flask import Flask
+    from prometheus_client.registry import CollectorRegistry
+
+    from gordo_amd.server.prometheus import GordoServerPrometheusMetrics
+
+    app = Flask("prom-test")
+
+    @app.route("/gordo/v0/<gordo_project>/<gordo_name>/x")
+    def route(gordo_project, gordo_name):
+        return "ok"
+
+    registry = CollectorRegistry()
+    metrics = GordoServerPrometheusMetrics(
+        args_labels=[("gordo_project", "project"), ("gordo_name", "model")],
+        info={"version": "1.0.0"},
+        registry=registry,
+    )
+    metrics.prepare_app(app)
+    client = app.test_client()
+    assert client.get("/gordo/v0/p1/m1/x").status_code == 200
+    assert client.get("/gordo/v0/p1/m1/x").status_code == 200
+    count = registry.get_sample_value(
+        "gordo_server_requests_total",
+        {
+            "method": "GET",
+            "path": "/gordo/v0/<gordo_project>/<gordo_name>/x",
+            "status_code": "200",
+            "project": "p1",
+            "model": "m1",
+            "version": "1.0.0",
+        },
+    )
+    assert count == 2.0
+
+
+def test_prometheus_sidecar_app():
+    from gordo_amd.server.prometheus.server import build_app
+
+    client = build_app().test_client()
+    assert client.get("/healthcheck").status_code == 200
+    resp = client.get("/metrics")
+    assert resp.status_code == 200
+
+
+def test_mlflow_batching():
+    from gordo_amd.reporters.mlflow import batch_log_items
+
+    assert batch_log_items(list(range(5)), 2) == [[0, 1], [2, 3], [4]]
