@@ -106,7 +106,7 @@ class PackedFleetBuilder:
         output_dir: Optional[str] = None,
         model_register_dir: Optional[str] = None,
         device: Optional[str] = None,
-        data_workers: int = 8,
+        data_workers: int = 16,
         save_models: bool = True,
         replace_cache: bool = False,
     ):
