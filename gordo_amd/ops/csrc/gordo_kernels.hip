@@ -67,6 +67,15 @@ DEV_INLINE float fast_tanhf(float x) {
 DEV_INLINE float bf2f(bf16 v) { return __bfloat162float(v); }
 DEV_INLINE bf16 f2bf(float v) { return __float2bfloat16(v); }
 
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
+
+// gather 8 bf16 into registers, store as ONE 16-byte ds_write_b128:
+// element-wise 2-byte staging writes were the dominant LDS bank
+// conflict source (profiles/lstm_pmc_r01.txt).
+DEV_INLINE void lds_store8(bf16* dst, const bf16 (&v)[8]) {
+  *reinterpret_cast<u32x4*>(dst) = *reinterpret_cast<const u32x4*>(v);
+}
+
 // ---------------------------------------------------------------------------
 // Grouped GEMM: C[g, M, N] = op( A[g, M, K] @ B + bias ), B per mode:
 //   BMODE_KN: B[g, K, N] row-major (forward: X @ W)
@@ -133,12 +142,13 @@ __global__ __launch_bounds__(256) void grouped_gemm_kernel(
       int row = tid >> 2;
       int kk = (tid & 3) * 8;
       int gm = m0 + row;
+      bf16 v[8];
       #pragma unroll
       for (int e = 0; e < 8; ++e) {
         int gk = k0 + kk + e;
-        As[row * LDT + kk + e] =
-            (gm < M && gk < K) ? Ag[(size_t)gm * K + gk] : f2bf(0.f);
+        v[e] = (gm < M && gk < K) ? Ag[(size_t)gm * K + gk] : f2bf(0.f);
       }
+      lds_store8(&As[row * LDT + kk], v);
     }
     // ---- stage B -> Bs[n][k] (transposed for BMODE_KN) ----
     if (BMODE == 0) {
@@ -146,23 +156,25 @@ __global__ __launch_bounds__(256) void grouped_gemm_kernel(
       int n = tid >> 2;
       int kk = (tid & 3) * 8;
       int gn = n0 + n;
+      bf16 v[8];
       #pragma unroll
       for (int e = 0; e < 8; ++e) {
         int gk = k0 + kk + e;
-        Bs[n * LDT + kk + e] =
-            (gn < N && gk < K) ? Bg[(size_t)gk * N + gn] : f2bf(0.f);
+        v[e] = (gn < N && gk < K) ? Bg[(size_t)gk * N + gn] : f2bf(0.f);
       }
+      lds_store8(&Bs[n * LDT + kk], v);
     } else {
       // B[N][K]: rows are n — direct copy
       int n = tid >> 2;
       int kk = (tid & 3) * 8;
       int gn = n0 + n;
+      bf16 v[8];
       #pragma unroll
       for (int e = 0; e < 8; ++e) {
         int gk = k0 + kk + e;
-        Bs[n * LDT + kk + e] =
-            (gn < N && gk < K) ? Bg[(size_t)gn * K + gk] : f2bf(0.f);
+        v[e] = (gn < N && gk < K) ? Bg[(size_t)gn * K + gk] : f2bf(0.f);
       }
+      lds_store8(&Bs[n * LDT + kk], v);
     }
     __syncthreads();
 
@@ -251,12 +263,13 @@ __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
       int k = tid >> 2;           // 0..63
       int mm = (tid & 3) * 8;     // 0..24
       int gk = k0 + k;
+      bf16 v[8];
       #pragma unroll
       for (int e = 0; e < 8; ++e) {
         int gm = m0 + mm + e;
-        As[k * LDT + mm + e] =
-            (gk < K && gm < M) ? Ag[(size_t)gm * K + gk] : f2bf(0.f);
+        v[e] = (gk < K && gm < M) ? Ag[(size_t)gm * K + gk] : f2bf(0.f);
       }
+      lds_store8(&As[k * LDT + mm], v);
     }
     // stage dZ[m0..+32][n0..+64] transposed into Zs[n][m]; k-tile-0
     // blocks fold the bias-grad column sum into the same pass (what a
@@ -265,13 +278,14 @@ __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
       int n = tid >> 2;
       int mm = (tid & 3) * 8;
       int gn = n0 + n;
+      bf16 v[8];
       #pragma unroll
       for (int e = 0; e < 8; ++e) {
         int gm = m0 + mm + e;
-        bf16 v = (gn < N && gm < M) ? Zg[(size_t)gm * N + gn] : f2bf(0.f);
-        Zs[n * LDT + mm + e] = v;
-        if (k0 == 0) db_part += bf2f(v);
+        v[e] = (gn < N && gm < M) ? Zg[(size_t)gm * N + gn] : f2bf(0.f);
+        if (k0 == 0) db_part += bf2f(v[e]);
       }
+      lds_store8(&Zs[n * LDT + mm], v);
     }
     __syncthreads();
 
