@@ -20,6 +20,7 @@ from __future__ import annotations
 
 import logging
 import math
+import os
 from typing import Dict, List, Optional, Sequence, Tuple
 
 import numpy as np
@@ -83,6 +84,12 @@ class _ParamStore:
             if self.compute_dtype != torch.float32
             else None
         )
+        # device-resident Adam step counter (hipGraph-capturable)
+        self.step_buf = (
+            torch.zeros(1, dtype=torch.int32, device=dev)
+            if dev.type == "cuda"
+            else None
+        )
         for name, shape in self._shapes:
             off, n = self._offsets[name]
             self.views[name] = self.p32[off : off + n].view(*shape)
@@ -105,12 +112,15 @@ class _ParamStore:
         ops.adam_step(
             self.p32, self.g32, self.m, self.v,
             lr, beta1, beta2, eps, self.step_count, self.plp,
+            step_buf=self.step_buf,
         )
 
     def reset_adam(self):
         self.m.zero_()
         self.v.zero_()
         self.step_count = 0
+        if self.step_buf is not None:
+            self.step_buf.zero_()
 
 
 class BasePack:
@@ -178,6 +188,69 @@ class BasePack:
         return out
 
     # ---- the epoch loop ------------------------------------------------
+    # hipGraph capture of the steady-state train_batch: the whole
+    # fwd+bwd+Adam kernel sequence replays as ONE graph launch per
+    # full-size batch (shapes static; the shuffled gather runs eagerly
+    # into static buffers; the ragged last batch stays eager). Disable
+    # with GORDO_HIPGRAPH=0.
+    _graph_enabled = os.environ.get("GORDO_HIPGRAPH", "1") != "0"
+
+    def _graph_train_step(self, Xb: torch.Tensor, Tb: torch.Tensor):
+        """Replay (capturing on first use) the train_batch graph for
+        this batch shape. Returns the live loss tensor, or None when
+        capture is unavailable for this configuration."""
+        if not (self._graph_enabled and self.device.type == "cuda"):
+            return None
+        key = (tuple(Xb.shape), tuple(Tb.shape))
+        cached = getattr(self, "_graph_cache", None)
+        if cached is None:
+            cached = self._graph_cache = {}
+        entry = cached.get(key)
+        if entry is None:
+            if len(cached) >= 4:  # bound distinct shapes per pack
+                return None
+            try:
+                sx = torch.empty_like(Xb)
+                st = torch.empty_like(Tb)
+                sx.copy_(Xb)
+                st.copy_(Tb)
+                # warmup + capture run real optimizer steps — snapshot
+                # the training state and restore it afterwards so
+                # capture does not perturb the training trajectory.
+                store = self.store
+                snap = (
+                    store.p32.clone(), store.m.clone(), store.v.clone(),
+                    store.step_buf.clone(), store.step_count,
+                )
+                s = torch.cuda.Stream()
+                s.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(s):
+                    self.train_batch(sx, st)
+                torch.cuda.current_stream().wait_stream(s)
+                graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(graph):
+                    loss = self.train_batch(sx, st)
+                store.p32.copy_(snap[0])
+                store.m.copy_(snap[1])
+                store.v.copy_(snap[2])
+                store.step_buf.copy_(snap[3])
+                store.step_count = snap[4]
+                store.sync_lp()
+                entry = cached[key] = (graph, sx, st, loss)
+            except Exception:
+                logger.warning(
+                    "hipGraph capture failed; falling back to eager",
+                    exc_info=True,
+                )
+                self._graph_enabled = False
+                return None
+        graph, sx, st, loss = entry
+        sx.copy_(Xb)
+        st.copy_(Tb)
+        graph.replay()
+        self.store.step_count += 1  # device step_buf is the truth
+        return loss
+
     def fit(
         self,
         X: torch.Tensor,
@@ -217,7 +290,9 @@ class BasePack:
                 if idx.shape[1] == 0:
                     continue
                 Xb, Tb = self._gather_batch(X, Y, idx)
-                loss = self.train_batch(Xb, Tb)
+                loss = self._graph_train_step(Xb, Tb)
+                if loss is None:
+                    loss = self.train_batch(Xb, Tb)
                 epoch_loss += loss * idx.shape[1]
                 samples_seen += idx.shape[1]
             epoch_loss = (epoch_loss / max(samples_seen, 1)).cpu().tolist()

@@ -97,11 +97,14 @@ def mse_bwd(Y, T) -> Tuple[torch.Tensor, torch.Tensor]:
     return ref.mse_bwd(Y, T)
 
 
-def adam_step(p, g, m, v, lr, beta1, beta2, eps, step, p_lp=None):
+def adam_step(p, g, m, v, lr, beta1, beta2, eps, step, p_lp=None,
+              step_buf=None):
     if _on_gpu(p):
+        if step_buf is None:
+            raise ValueError("GPU adam_step needs a device step_buf")
         return _require_hip().adam_step(
             p, g, m, v, float(lr), float(beta1), float(beta2), float(eps),
-            int(step), p_lp,
+            int(step), p_lp, step_buf,
         )
     return ref.adam_step(p, g, m, v, lr, beta1, beta2, eps, step, p_lp)
 

@@ -380,11 +380,21 @@ __global__ void mse_bwd_kernel(const bf16* __restrict__ Y,
 }
 
 // fused Adam over the flat fp32 master buffer + bf16 mirror refresh.
+// The step counter lives ON DEVICE (step_buf) so the kernel sequence is
+// hipGraph-capturable: bias correction is recomputed from memory at
+// every replay instead of being frozen into the captured kernel args.
+__global__ void adam_bump_kernel(int* step_buf) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *step_buf += 1;
+}
+
 __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
                             float* __restrict__ m, float* __restrict__ v,
                             bf16* __restrict__ plp, size_t n, float lr,
-                            float b1, float b2, float eps, float bc1,
-                            float bc2) {
+                            float b1, float b2, float eps,
+                            const int* __restrict__ step_buf) {
+  float fstep = (float)*step_buf;
+  float bc1 = 1.f - __powf(b1, fstep);
+  float bc2 = 1.f - __powf(b2, fstep);
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   size_t stride = (size_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) {
@@ -592,18 +602,22 @@ std::vector<torch::Tensor> mse_bwd(torch::Tensor Y, torch::Tensor T) {
 
 void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, double lr, double b1, double b2, double eps,
-               int64_t step, c10::optional<torch::Tensor> plp) {
+               int64_t step, c10::optional<torch::Tensor> plp,
+               torch::Tensor step_buf) {
   CHECK_GPU(p);
+  TORCH_CHECK(step_buf.is_cuda() && step_buf.scalar_type() == torch::kInt32,
+              "step_buf must be an int32 GPU tensor");
   size_t n = p.numel();
-  float bc1 = 1.f - powf((float)b1, (float)step);
-  float bc2 = 1.f - powf((float)b2, (float)step);
   bf16* plp_ptr = nullptr;
   if (plp.has_value()) plp_ptr = (bf16*)plp->data_ptr();
   int blocks = (int)std::min<size_t>((n + 255) / 256, 4096);
+  hipLaunchKernelGGL(adam_bump_kernel, dim3(1), dim3(64), 0, cur_stream(),
+                     step_buf.data_ptr<int>());
   hipLaunchKernelGGL(adam_kernel, dim3(blocks), dim3(256), 0, cur_stream(),
                      p.data_ptr<float>(), g.data_ptr<float>(),
                      m.data_ptr<float>(), v.data_ptr<float>(), plp_ptr, n,
-                     (float)lr, (float)b1, (float)b2, (float)eps, bc1, bc2);
+                     (float)lr, (float)b1, (float)b2, (float)eps,
+                     step_buf.data_ptr<int>());
 }
 
 std::vector<torch::Tensor> lstm_pointwise_fwd(torch::Tensor gates,
