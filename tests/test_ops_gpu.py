@@ -142,9 +142,11 @@ def test_adam_step():
     v = torch.zeros(n)
     pd, gd, md, vd = (t.cuda() for t in (p, g, m, v))
     plp = pd.to(torch.bfloat16)
+    step_buf = torch.zeros(1, dtype=torch.int32, device="cuda")
     for step in (1, 2, 3):
         ref.adam_step(p, g, m, v, 0.01, 0.9, 0.999, 1e-7, step)
-        ops.adam_step(pd, gd, md, vd, 0.01, 0.9, 0.999, 1e-7, step, plp)
+        ops.adam_step(pd, gd, md, vd, 0.01, 0.9, 0.999, 1e-7, step, plp,
+                      step_buf=step_buf)
     torch.testing.assert_close(pd.cpu(), p, rtol=1e-5, atol=1e-6)
     torch.testing.assert_close(md.cpu(), m, rtol=1e-5, atol=1e-7)
     torch.testing.assert_close(plp.float().cpu(), p, rtol=1e-2, atol=1e-2)
