@@ -302,13 +302,14 @@ class PackedFleetBuilder:
 
         # per-machine adoption + detector finalization
         offset = len(Xt_list[0]) - pack._n_samples(len(Xt_list[0]))
+        all_states = pack.states_for_all_models()
         for g_idx, p in enumerate(group):
             hist = {
                 k: [float(ep[g_idx]) for ep in v] for k, v in history.items()
             }
             p.keras_est.adopt_pack_result(
                 spec,
-                pack.state_for_model(g_idx),
+                all_states[g_idx],
                 hist,
                 n_features=Xt_list[g_idx].shape[1],
                 n_features_out=y_list[g_idx].shape[1],
@@ -318,9 +319,9 @@ class PackedFleetBuilder:
             self._finalize(p, offset, cv_duration, final=True,
                            train_duration=train_duration / len(group))
 
-        # save + register
+        # save + register (parallel: pickle+json dumps are I/O bound)
         if self.save_models and self.output_dir:
-            for p in group:
+            def save_one(p):
                 out = os.path.join(self.output_dir, p.machine.name)
                 ModelBuilder._save_model(p.model, p.machine, out)
                 if self.model_register_dir:
@@ -329,6 +330,9 @@ class PackedFleetBuilder:
                         ModelBuilder(p.machine).cache_key,
                         out,
                     )
+
+            with concurrent.futures.ThreadPoolExecutor(8) as ex:
+                list(ex.map(save_one, group))
         logger.info(
             "Packed build of %d machines done in %.2fs",
             len(group), time.time() - t0_all,
