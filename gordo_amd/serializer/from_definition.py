@@ -23,7 +23,7 @@ import inspect
 import logging
 from typing import Any, Dict, List, Union
 
-from ..core.import_utils import import_location, resolve_alias
+from ..core.import_utils import import_location
 from .utils import is_tuple_type
 
 logger = logging.getLogger(__name__)

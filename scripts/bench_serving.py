@@ -24,7 +24,8 @@ import numpy as np
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 N_TAGS = 50
-ROWS = 100
+ROWS = 100  # overridable: --rows (10k-row requests exercise the fused
+            # K9 device scoring path, BASELINE config #5)
 
 
 def main():
@@ -33,6 +34,7 @@ def main():
     ap.add_argument("--threads", type=int, default=8)
     ap.add_argument("--endpoint", default="anomaly",
                     choices=["anomaly", "prediction", "both"])
+    ap.add_argument("--rows", type=int, default=ROWS)
     args = ap.parse_args()
 
     import pandas as pd
@@ -76,7 +78,8 @@ machines:
         app = build_app()
         app.testing = True
 
-        X = pd.DataFrame(np.random.random((ROWS, N_TAGS)), columns=sensors)
+        X = pd.DataFrame(np.random.random((args.rows, N_TAGS)),
+                         columns=sensors)
         payload = {"X": dataframe_to_dict(X), "y": dataframe_to_dict(X)}
         results = {}
         for endpoint in (
@@ -123,12 +126,12 @@ machines:
             rps = len(latencies) / elapsed
             results[endpoint] = {
                 "requests_per_sec": rps,
-                "predictions_per_sec": rps * ROWS,
+                "predictions_per_sec": rps * args.rows,
                 "mean_latency_ms": statistics.mean(latencies) * 1000,
                 "p50_latency_ms": statistics.median(latencies) * 1000,
                 "rounds": len(latencies),
                 "threads": args.threads,
-                "payload_rows": ROWS,
+                "payload_rows": args.rows,
                 "n_tags": N_TAGS,
                 "device": "cuda" if torch.cuda.is_available() else "cpu",
             }

@@ -1,0 +1,16 @@
+#!/usr/bin/env bash
+# CI shard runner (the analog of the reference's scripts/tests.sh lanes).
+# Usage: scripts/tests.sh [lane]
+#   cpu       — the full CPU suite (default)
+#   gpu       — GPU-marked tests (needs an MI355X)
+#   bench     — benchmark harnesses
+#   doctest   — doctest lane only
+set -e
+LANE="${1:-cpu}"
+case "$LANE" in
+  cpu)     exec python -m pytest tests/ -q -m "not gpu" ;;
+  gpu)     exec python -m pytest tests/ -q -m gpu ;;
+  bench)   exec python -m pytest benchmarks/ -q -s ;;
+  doctest) exec python -m pytest tests/test_doctests.py -q ;;
+  *) echo "unknown lane: $LANE" >&2; exit 2 ;;
+esac
