@@ -553,10 +553,17 @@ class LSTMPack(BasePack):
             self.store.gviews[f"Wx{li}"].copy_(dWx)
             self.store.gviews[f"Wh{li}"].copy_(dWh)
             self.store.gviews[f"bl{li}"].copy_(dbl)
+            # free this layer's BPTT cache before descending: the
+            # allocator reuses it for the next layer's dSeq/transients
+            # (whole-batch caches are the memory ceiling at G~100+).
+            cache[li] = None
+            del lc, hs, h_prev_all
+            prev_dSeq = dSeq
             if li > 0:
                 dSeq = ops.grouped_linear_bwd_data(dG_flat, Wx).view(
                     G, B, T, fin
                 )
+            del dG_flat, prev_dSeq
 
         a = self.spec.adam_params
         self.store.adam_step(a["lr"], a["beta1"], a["beta2"], a["eps"])

@@ -200,6 +200,13 @@ class PackedFleetBuilder:
                 p.machine.name, int(p.machine.evaluation.get("seed", 0))
             )
 
+    # pack-size caps: BPTT caches are ~B*T*(11H)*layers bytes per LSTM
+    # model per batch (plus hipGraph retention); 128 LSTM models per
+    # pack keeps a 288 GB GPU comfortable at lookback 144 while still
+    # amortizing kernel launches. Dense models are ~100x lighter.
+    MAX_PACK_LSTM = 128
+    MAX_PACK_DENSE = 1024
+
     def _group(self, plans: List[MachinePlan]) -> List[List[MachinePlan]]:
         groups: Dict[Any, List[MachinePlan]] = {}
         for p in plans:
@@ -221,7 +228,16 @@ class PackedFleetBuilder:
                 p.detector is not None,
             )
             groups.setdefault(key, []).append(p)
-        return list(groups.values())
+        out: List[List[MachinePlan]] = []
+        for key, members in groups.items():
+            cap = (
+                self.MAX_PACK_LSTM
+                if members[0].spec.model_type == "lstm"
+                else self.MAX_PACK_DENSE
+            )
+            for s0 in range(0, len(members), cap):
+                out.append(members[s0 : s0 + cap])
+        return out
 
     # ---- the packed group build -----------------------------------------
     def _pack_cls(self, spec: ModelSpec):
