@@ -35,6 +35,10 @@ def main():
     ap.add_argument("--endpoint", default="anomaly",
                     choices=["anomaly", "prediction", "both"])
     ap.add_argument("--rows", type=int, default=ROWS)
+    ap.add_argument("--direct", action="store_true",
+                    help="measure model.anomaly() directly (no HTTP/JSON): "
+                         "the batched inference engine path, BASELINE "
+                         "config #5 shape")
     args = ap.parse_args()
 
     import pandas as pd
@@ -73,6 +77,31 @@ machines:
             serializer.dump(model, d, metadata=json.loads(machine.to_json()),
                             info={})
         os.environ["MODEL_COLLECTION_DIR"] = collection
+
+        if args.direct:
+            import pandas as pd
+
+            model = serializer.load(os.path.join(collection, "serve-bench"))
+            X = pd.DataFrame(np.random.random((args.rows, N_TAGS)),
+                             columns=sensors)
+            model.anomaly(X, X)  # warmup (captures/caches)
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
+            t0 = time.time()
+            for _ in range(args.rounds):
+                model.anomaly(X, X)
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
+            dt = (time.time() - t0) / args.rounds
+            print(json.dumps({
+                "metric": "batched anomaly() rows/sec (no HTTP)",
+                "rows_per_call": args.rows,
+                "ms_per_call": dt * 1000,
+                "rows_per_sec": args.rows / dt,
+                "device": "cuda" if torch.cuda.is_available() else "cpu",
+            }))
+            return
+
         from gordo_amd.server.server import build_app
 
         app = build_app()
