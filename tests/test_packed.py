@@ -243,3 +243,36 @@ def test_packed_fleet_builder_groups(tmp_path):
         assert sg[key]["fold-mean"] == pytest.approx(
             ss[key]["fold-mean"], rel=1e-3, abs=1e-5
         )
+
+
+def test_group_splitting_caps_pack_size():
+    """Oversized architecture groups split into capped packs (memory
+    ceiling at lookback-144 BPTT); results must be unaffected since
+    pack membership never changes per-model math."""
+    from gordo_amd.parallel import PackedFleetBuilder
+    from gordo_amd.workflow import NormalizedConfig
+
+    machines = [
+        {
+            "name": f"cap-{i}",
+            "dataset": {
+                "type": "SineWaveDataset",
+                "tag_list": ["a", "b", "c"],
+                "train_start_date": "2019-01-01T00:00:00+00:00",
+                "train_end_date": "2019-01-02T00:00:00+00:00",
+            },
+            "model": {
+                "gordo_amd.machine.model.models.KerasAutoEncoder": {
+                    "kind": "feedforward_hourglass",
+                    "epochs": 1,
+                }
+            },
+        }
+        for i in range(7)
+    ]
+    norm = NormalizedConfig({"machines": machines}, project_name="p")
+    fb = PackedFleetBuilder(norm.machines, save_models=False)
+    fb.MAX_PACK_DENSE = 3
+    results = dict(fb.build_all())
+    assert len(results) == 7
+    assert all(not isinstance(v, BaseException) for v in results.values())
