@@ -216,7 +216,23 @@ def extract_X_y(method):
 
 
 # ---- caches -------------------------------------------------------------
-@lru_cache(maxsize=int(os.getenv("N_CACHED_MODELS", 2)))
+def _default_model_cache_size() -> int:
+    # reference default is 2 (k8s CPU-pod memory envelope,
+    # gordo/server/utils.py:334); an MI355X box serves thousands of
+    # these models from 288 GB HBM, so the GPU default is 512.
+    if os.getenv("N_CACHED_MODELS"):
+        return int(os.environ["N_CACHED_MODELS"])
+    try:
+        import torch
+
+        if torch.cuda.is_available():
+            return 512
+    except ImportError:
+        pass
+    return 2
+
+
+@lru_cache(maxsize=_default_model_cache_size())
 def load_model(directory: str, name: str) -> BaseEstimator:
     start_time = timeit.default_timer()
     model = serializer.load(os.path.join(directory, name))
