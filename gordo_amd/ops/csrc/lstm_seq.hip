@@ -101,6 +101,25 @@ __global__ __launch_bounds__(256) void lstm_seq_fwd_kernel(
     // ---- gates = h @ Wh (MFMA) + xW_t ----
     if (wave_active) {
       constexpr int FM = ROWS / 16;
+      // prefetch this step's x-side gate values BEFORE the MFMA loop so
+      // the HBM latency hides under the matrix work (the epilogue then
+      // reads registers, not memory)
+      bf16 xv[FM][4][4];
+      #pragma unroll
+      for (int fm = 0; fm < FM; ++fm) {
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+          int col = wcol0 + fn * 16 + l15;
+          #pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            int row = fm * 16 + (lane >> 4) * 4 + r;
+            xv[fm][fn][r] =
+                (col < H4 && row < rows_here)
+                    ? xWg[((size_t)row * T + t) * H4 + col]
+                    : lf2bf(0.f);
+          }
+        }
+      }
       f32x4 acc[FM][4] = {};
       for (int kk = 0; kk < H; kk += 32) {
         #pragma unroll
@@ -126,10 +145,8 @@ __global__ __launch_bounds__(256) void lstm_seq_fwd_kernel(
           #pragma unroll
           for (int r = 0; r < 4; ++r) {
             int row = fm * 16 + (lane >> 4) * 4 + r;
-            float xv = (row < rows_here)
-                           ? lbf2f(xWg[((size_t)row * T + t) * H4 + col])
-                           : 0.f;
-            gS[row * ldg + col] = lf2bf(acc[fm][fn][r] + xv);
+            gS[row * ldg + col] =
+                lf2bf(acc[fm][fn][r] + lbf2f(xv[fm][fn][r]));
           }
         }
       }
