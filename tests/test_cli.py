@@ -212,3 +212,41 @@ def test_workflow_split(tmp_path):
     assert result.exit_code == 0, result.output
     docs = list(yaml.safe_load_all(result.output))
     assert len(docs) == 3  # 5 machines / 2 per workflow
+
+
+def test_fleet_build_cli(tmp_path):
+    """`gordo fleet build` end to end in-process (single rank)."""
+    cfg = tmp_path / "cfg.yml"
+    cfg.write_text(
+        """
+machines:
+  - name: fleet-cli-m
+    dataset: |
+      type: SineWaveDataset
+      tag_list: [a, b, c]
+      train_start_date: '2019-01-01T00:00:00+00:00'
+      train_end_date: '2019-01-02T00:00:00+00:00'
+    model: |
+      gordo.machine.model.models.KerasAutoEncoder:
+        kind: feedforward_hourglass
+        epochs: 1
+"""
+    )
+    out_dir = tmp_path / "models"
+    status = tmp_path / "status.json"
+    runner = CliRunner()
+    result = runner.invoke(
+        gordo,
+        [
+            "fleet", "build",
+            "--machine-config", str(cfg),
+            "--project-name", "fleet-cli",
+            "--output-dir", str(out_dir),
+            "--gpus", "1",
+            "--status-file", str(status),
+        ],
+    )
+    assert result.exit_code == 0, result.output
+    summary = json.loads(status.read_text())
+    assert summary["n_ok"] == 1
+    assert (out_dir / "fleet-cli-m" / "model.pkl").is_file()
