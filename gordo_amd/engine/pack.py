@@ -124,7 +124,8 @@ class _ParamStore:
 
 
 class BasePack:
-    def __init__(self, spec: ModelSpec, G: int, device=None, seeds=None):
+    def __init__(self, spec: ModelSpec, G: int, device=None, seeds=None,
+                 init_p32: Optional[torch.Tensor] = None):
         self.spec = spec
         self.G = G
         self.device = torch.device(
@@ -136,7 +137,12 @@ class BasePack:
         assert len(self.seeds) == G
         self._declare_params()
         self.store.allocate()
-        self._init_weights()
+        if init_p32 is not None:
+            # clone-init: copy a sibling pack's flat parameter snapshot
+            # (per-model glorot + orthogonal QR costs ~1 s per pack)
+            self.store.p32.copy_(init_p32)
+        else:
+            self._init_weights()
         self.store.sync_lp()
 
     # -- interface -------------------------------------------------------

@@ -17,6 +17,7 @@ per-machine ModelBuilder on CPU.
 from __future__ import annotations
 
 import concurrent.futures
+import contextlib
 import datetime
 import hashlib
 import json
@@ -50,6 +51,8 @@ from ..machine.model.models import KerasBaseEstimator, KerasLSTMBaseEstimator
 from ..util import disk_registry
 
 logger = logging.getLogger(__name__)
+
+_nullcontext = contextlib.nullcontext
 
 
 # ---------------------------------------------------------------------------
@@ -243,12 +246,14 @@ class PackedFleetBuilder:
     def _pack_cls(self, spec: ModelSpec):
         return LSTMPack if spec.model_type == "lstm" else DensePack
 
-    def _make_pack(self, spec: ModelSpec, group: List[MachinePlan]):
+    def _make_pack(self, spec: ModelSpec, group: List[MachinePlan],
+                   init_p32=None):
         return self._pack_cls(spec)(
             spec,
             G=len(group),
             device=self.device,
             seeds=[p.seed for p in group],
+            init_p32=init_p32,
         )
 
     def _stack(self, arrays: List[np.ndarray], pack) -> torch.Tensor:
@@ -354,6 +359,74 @@ class PackedFleetBuilder:
             len(group), time.time() - t0_all,
         )
 
+    # fold fits are independent — on GPU they run concurrently on
+    # separate streams with per-fold packs (the chip is underfilled by
+    # one pack's sequence-scan kernels); bounded so concurrent BPTT
+    # caches stay well inside HBM.
+    MAX_CONCURRENT_FOLD_MODELS = 256
+
+    def _fit_folds(
+        self, folds, group, Xt_list, y_list, spec, fit_args, pack,
+        init_snapshot,
+    ):
+        import threading
+
+        def run_fold(fold_pack, train_idx, test_idx, out, fold_i, stream):
+            try:
+                ctx = (
+                    torch.cuda.stream(stream)
+                    if stream is not None
+                    else _nullcontext()
+                )
+                with ctx:
+                    t0 = time.time()
+                    Xd = self._stack([x[train_idx] for x in Xt_list], fold_pack)
+                    Yd = self._stack([y[train_idx] for y in y_list], fold_pack)
+                    fold_pack.fit(Xd, Yd, **_engine_fit_args(fit_args))
+                    t_fit = time.time() - t0
+                    t0 = time.time()
+                    Xtest = self._stack(
+                        [x[test_idx] for x in Xt_list], fold_pack
+                    )
+                    with torch.no_grad():
+                        preds = fold_pack.predict(Xtest).float().cpu().numpy()
+                    out[fold_i] = (preds, t_fit, time.time() - t0)
+            except Exception as e:  # surface via the caller
+                out[fold_i] = e
+
+        concurrent = (
+            self.device != "cpu"
+            and torch.cuda.is_available()
+            and len(group) * len(folds) <= self.MAX_CONCURRENT_FOLD_MODELS
+        )
+        out: Dict[int, Any] = {}
+        if concurrent:
+            threads = []
+            for fold_i, (train_idx, test_idx) in enumerate(folds):
+                fold_pack = self._make_pack(spec, group,
+                                            init_p32=init_snapshot)
+                stream = torch.cuda.Stream()
+                threads.append(
+                    threading.Thread(
+                        target=run_fold,
+                        args=(fold_pack, train_idx, test_idx, out, fold_i,
+                              stream),
+                    )
+                )
+            for t in threads:
+                t.start()
+            for t in threads:
+                t.join()
+            torch.cuda.synchronize()
+        else:
+            for fold_i, (train_idx, test_idx) in enumerate(folds):
+                self._reset_pack(pack, init_snapshot)
+                run_fold(pack, train_idx, test_idx, out, fold_i, None)
+        for fold_i, res in out.items():
+            if isinstance(res, Exception):
+                raise res
+        return out
+
     @staticmethod
     def _reset_pack(pack, init_snapshot: torch.Tensor):
         pack.store.p32.copy_(init_snapshot)
@@ -405,19 +478,13 @@ class PackedFleetBuilder:
         ]
         window = getattr(group[0].detector, "window", None) if group[0].detector else None
 
+        fold_preds = self._fit_folds(
+            folds, group, Xt_list, y_list, spec, fit_args, pack, init_snapshot
+        )
+
         for fold_i, (train_idx, test_idx) in enumerate(folds):
             t_f0 = time.time()
-            self._reset_pack(pack, init_snapshot)
-            Xd = self._stack([x[train_idx] for x in Xt_list], pack)
-            Yd = self._stack([y[train_idx] for y in y_list], pack)
-            pack.fit(Xd, Yd, **_engine_fit_args(fit_args))
-            t_fit = time.time() - t_f0
-
-            t_p0 = time.time()
-            Xtest = self._stack([x[test_idx] for x in Xt_list], pack)
-            with torch.no_grad():
-                preds = pack.predict(Xtest).float().cpu().numpy()
-            t_pred = time.time() - t_p0
+            preds, t_fit, t_pred = fold_preds[fold_i]
             t_s0 = time.time()
 
             for g_idx, p in enumerate(group):
