@@ -213,6 +213,9 @@ class BasePack:
     # into static buffers; the ragged last batch stays eager). Disable
     # with GORDO_HIPGRAPH=0.
     _graph_enabled = os.environ.get("GORDO_HIPGRAPH", "1") != "0"
+    # hip stream-capture mode is process-global: captures from two
+    # threads at once crash the process. One capture at a time.
+    _graph_capture_lock = __import__("threading").Lock()
 
     def _graph_train_step(self, Xb: torch.Tensor, Tb: torch.Tensor):
         """Replay (capturing on first use) the train_batch graph for
@@ -241,14 +244,15 @@ class BasePack:
                     store.p32.clone(), store.m.clone(), store.v.clone(),
                     store.step_buf.clone(), store.step_count,
                 )
-                s = torch.cuda.Stream()
-                s.wait_stream(torch.cuda.current_stream())
-                with torch.cuda.stream(s):
-                    self.train_batch(sx, st)
-                torch.cuda.current_stream().wait_stream(s)
-                graph = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(graph):
-                    loss = self.train_batch(sx, st)
+                with BasePack._graph_capture_lock:
+                    s = torch.cuda.Stream()
+                    s.wait_stream(torch.cuda.current_stream())
+                    with torch.cuda.stream(s):
+                        self.train_batch(sx, st)
+                    torch.cuda.current_stream().wait_stream(s)
+                    graph = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(graph):
+                        loss = self.train_batch(sx, st)
                 store.p32.copy_(snap[0])
                 store.m.copy_(snap[1])
                 store.v.copy_(snap[2])

@@ -405,6 +405,10 @@ class PackedFleetBuilder:
             for fold_i, (train_idx, test_idx) in enumerate(folds):
                 fold_pack = self._make_pack(spec, group,
                                             init_p32=init_snapshot)
+                # no graph capture in concurrent fold packs: capture is
+                # process-global and the folds are throughput-bound on
+                # real kernels anyway
+                fold_pack._graph_enabled = False
                 stream = torch.cuda.Stream()
                 threads.append(
                     threading.Thread(
