@@ -251,7 +251,11 @@ class BasePack:
                         self.train_batch(sx, st)
                     torch.cuda.current_stream().wait_stream(s)
                     graph = torch.cuda.CUDAGraph()
-                    with torch.cuda.graph(graph):
+                    # thread_local: other builder threads keep launching
+                    # on their own streams while this one captures
+                    with torch.cuda.graph(
+                        graph, capture_error_mode="thread_local"
+                    ):
                         loss = self.train_batch(sx, st)
                 store.p32.copy_(snap[0])
                 store.m.copy_(snap[1])
