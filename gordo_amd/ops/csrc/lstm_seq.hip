@@ -308,6 +308,230 @@ __global__ __launch_bounds__(256) void lstm_seq_bwd_kernel(
   }
 }
 
+
+// ===========================================================================
+// v2 scan kernels — require H % 16 == 0 (the engine pads hidden sizes
+// to 16; pad units are mathematically inert). With 16-aligned gate
+// blocks, the four gate values of hidden unit hh land in the SAME lane
+// of the MFMA D-fragments, so the entire gate pointwise phase runs in
+// registers: no gate LDS round trip and no barriers in the T loop
+// (the only LDS use is the Wh stage plus a wave-private 16-row h/dg
+// bounce to re-shape D-layout -> A-layout between steps). Waves are
+// fully independent -> 4 blocks/CU of free-running waves.
+// ===========================================================================
+
+template <int HF>  // HF = H/16; H = 16*HF <= 64
+__global__ __launch_bounds__(256) void lstm_seq_fwd_v2_kernel(
+    const bf16* __restrict__ xW, const bf16* __restrict__ Wh,
+    bf16* __restrict__ hs, float* __restrict__ cs,
+    bf16* __restrict__ gacts, int B, int T) {
+  constexpr int H = 16 * HF;
+  constexpr int H4 = 4 * H;
+  constexpr int FN = 4 * HF;
+  constexpr int KK = (H + 31) / 32;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* WhT = reinterpret_cast<bf16*>(smem);            // [H4][LDK] shared
+  bf16* hP_all = WhT + (size_t)H4 * LDK;                // [4][16][LDK]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int l15 = lane & 15;
+  const int kslot = lane >> 4;
+  bf16* hP = hP_all + (size_t)wid * 16 * LDK;
+
+  const int nb = (B + 63) / 64;
+  const int g = blockIdx.x / nb;
+  const int r0 = (blockIdx.x % nb) * 64 + wid * 16;
+  const bf16* Whg = Wh + (size_t)g * H * H4;
+  const bf16* xWg = xW + ((size_t)g * B + r0) * T * H4;
+  bf16* hsg = hs + ((size_t)g * B + r0) * T * H;
+  float* csg = cs + ((size_t)g * B + r0) * T * H;
+  bf16* gag = gacts + ((size_t)g * B + r0) * T * H4;
+  const int rows_here = max(0, min(16, B - r0));
+
+  for (int i = tid; i < H4 * LDK; i += 256) {
+    int n = i / LDK, k = i % LDK;
+    WhT[i] = (k < H) ? Whg[(size_t)k * H4 + n] : lf2bf(0.f);
+  }
+  for (int i = lane; i < 16 * LDK; i += 64) hP[i] = lf2bf(0.f);
+  __syncthreads();  // WhT ready — the only barrier
+
+  float cReg[HF][4] = {};
+  const int myrow = kslot * 4;  // this lane's 4 rows: myrow..myrow+3
+
+  for (int t = 0; t < T; ++t) {
+    // prefetch the x-side gate tile for this step (in-lane positions)
+    bf16 xv[FN][4];
+    #pragma unroll
+    for (int fn = 0; fn < FN; ++fn) {
+      int col = fn * 16 + l15;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = myrow + r;
+        xv[fn][r] = (row < rows_here)
+                        ? xWg[((size_t)row * T + t) * H4 + col]
+                        : lf2bf(0.f);
+      }
+    }
+    // h A-fragments from the wave-private bounce buffer
+    bf16x8 afrag[KK];
+    #pragma unroll
+    for (int kk = 0; kk < KK; ++kk)
+      afrag[kk] = *reinterpret_cast<const bf16x8*>(
+          &hP[l15 * LDK + kk * 32 + kslot * 8]);
+    // gates = h @ Wh
+    f32x4 acc[FN];
+    #pragma unroll
+    for (int fn = 0; fn < FN; ++fn) {
+      f32x4 a = {};
+      #pragma unroll
+      for (int kk = 0; kk < KK; ++kk) {
+        bf16x8 b = *reinterpret_cast<const bf16x8*>(
+            &WhT[(size_t)(fn * 16 + l15) * LDK + kk * 32 + kslot * 8]);
+        a = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[kk], b, a, 0, 0,
+                                                    0);
+      }
+      acc[fn] = a;
+    }
+    // fused gate math — entirely in registers (same lane holds all 4
+    // gate values of each (row, hh))
+    #pragma unroll
+    for (int hf = 0; hf < HF; ++hf) {
+      int hh = hf * 16 + l15;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = myrow + r;
+        float i_g = sigmoidf_(acc[hf][r] + lbf2f(xv[hf][r]));
+        float f_g = sigmoidf_(acc[HF + hf][r] + lbf2f(xv[HF + hf][r]));
+        float g_g =
+            fast_tanhf_(acc[2 * HF + hf][r] + lbf2f(xv[2 * HF + hf][r]));
+        float o_g = sigmoidf_(acc[3 * HF + hf][r] + lbf2f(xv[3 * HF + hf][r]));
+        float cc = f_g * cReg[hf][r] + i_g * g_g;
+        float hv = o_g * fast_tanhf_(cc);
+        cReg[hf][r] = cc;
+        hP[row * LDK + hh] = lf2bf(hv);
+        if (row < rows_here) {
+          size_t base = ((size_t)row * T + t) * H + hh;
+          hsg[base] = lf2bf(hv);
+          csg[base] = cc;
+          size_t gb = ((size_t)row * T + t) * H4;
+          gag[gb + hh] = lf2bf(i_g);
+          gag[gb + H + hh] = lf2bf(f_g);
+          gag[gb + 2 * H + hh] = lf2bf(g_g);
+          gag[gb + 3 * H + hh] = lf2bf(o_g);
+        }
+      }
+    }
+  }
+}
+
+template <int HF>
+__global__ __launch_bounds__(256) void lstm_seq_bwd_v2_kernel(
+    const bf16* __restrict__ dSeq, const bf16* __restrict__ gacts,
+    const float* __restrict__ cs, const bf16* __restrict__ Wh,
+    bf16* __restrict__ dG, int B, int T, int ldg, int last_only) {
+  constexpr int H = 16 * HF;
+  constexpr int H4 = 4 * H;
+  constexpr int KK4 = (H4 + 31) / 32;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* WhN = reinterpret_cast<bf16*>(smem);             // [H][ldg] shared
+  bf16* dgP_all = WhN + (size_t)H * ldg;                 // [4][16][ldg]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int l15 = lane & 15;
+  const int kslot = lane >> 4;
+  bf16* dgP = dgP_all + (size_t)wid * 16 * ldg;
+
+  const int nb = (B + 63) / 64;
+  const int g = blockIdx.x / nb;
+  const int r0 = (blockIdx.x % nb) * 64 + wid * 16;
+  const bf16* Whg = Wh + (size_t)g * H * H4;
+  const bf16* gag = gacts + ((size_t)g * B + r0) * T * H4;
+  const float* csg = cs + ((size_t)g * B + r0) * T * H;
+  const bf16* dSg = last_only ? dSeq + ((size_t)g * B + r0) * H
+                              : dSeq + ((size_t)g * B + r0) * T * H;
+  bf16* dGg = dG + ((size_t)g * B + r0) * T * H4;
+  const int rows_here = max(0, min(16, B - r0));
+
+  for (int i = tid; i < H * ldg; i += 256) {
+    int h = i / ldg, n = i % ldg;
+    WhN[i] = (n < H4) ? Whg[(size_t)h * H4 + n] : lf2bf(0.f);
+  }
+  for (int i = lane; i < 16 * ldg; i += 64) dgP[i] = lf2bf(0.f);
+  __syncthreads();  // WhN ready
+
+  float dhReg[HF][4] = {};
+  float dcReg[HF][4] = {};
+  const int myrow = kslot * 4;
+
+  for (int t = T - 1; t >= 0; --t) {
+    // fused gate backward — in registers, in-lane
+    #pragma unroll
+    for (int hf = 0; hf < HF; ++hf) {
+      int hh = hf * 16 + l15;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = myrow + r;
+        float dh = dhReg[hf][r];
+        float i_g = 0.5f, f_g = 0.5f, g_g = 0.f, o_g = 0.5f, cc = 0.f,
+              cp = 0.f;
+        if (row < rows_here) {
+          if (last_only) {
+            if (t == T - 1) dh += lbf2f(dSg[(size_t)row * H + hh]);
+          } else {
+            dh += lbf2f(dSg[((size_t)row * T + t) * H + hh]);
+          }
+          size_t gb = ((size_t)row * T + t) * H4;
+          i_g = lbf2f(gag[gb + hh]);
+          f_g = lbf2f(gag[gb + H + hh]);
+          g_g = lbf2f(gag[gb + 2 * H + hh]);
+          o_g = lbf2f(gag[gb + 3 * H + hh]);
+          size_t cb = ((size_t)row * T + t) * H + hh;
+          cc = csg[cb];
+          cp = (t > 0) ? csg[cb - H] : 0.f;
+        }
+        float tc = fast_tanhf_(cc);
+        float dc = dcReg[hf][r] + dh * o_g * (1.f - tc * tc);
+        float vi = dc * g_g * i_g * (1.f - i_g);
+        float vf = dc * cp * f_g * (1.f - f_g);
+        float vg = dc * i_g * (1.f - g_g * g_g);
+        float vo = dh * tc * o_g * (1.f - o_g);
+        dcReg[hf][r] = dc * f_g;
+        dgP[row * ldg + hh] = lf2bf(vi);
+        dgP[row * ldg + H + hh] = lf2bf(vf);
+        dgP[row * ldg + 2 * H + hh] = lf2bf(vg);
+        dgP[row * ldg + 3 * H + hh] = lf2bf(vo);
+        if (row < rows_here) {
+          size_t gb = ((size_t)row * T + t) * H4;
+          dGg[gb + hh] = lf2bf(vi);
+          dGg[gb + H + hh] = lf2bf(vf);
+          dGg[gb + 2 * H + hh] = lf2bf(vg);
+          dGg[gb + 3 * H + hh] = lf2bf(vo);
+        }
+      }
+    }
+    // dh_carry = dgates @ Wh^T (wave-local; D-layout output stays in
+    // registers for the next step's pointwise)
+    #pragma unroll
+    for (int hf = 0; hf < HF; ++hf) {
+      f32x4 a = {};
+      #pragma unroll
+      for (int kk = 0; kk < KK4; ++kk) {
+        bf16x8 af = *reinterpret_cast<const bf16x8*>(
+            &dgP[l15 * ldg + kk * 32 + kslot * 8]);
+        bf16x8 b = *reinterpret_cast<const bf16x8*>(
+            &WhN[(size_t)(hf * 16 + l15) * ldg + kk * 32 + kslot * 8]);
+        a = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b, a, 0, 0, 0);
+      }
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) dhReg[hf][r] = a[r];
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 namespace gordo_lstm {
 
@@ -338,6 +562,31 @@ std::vector<torch::Tensor> lstm_seq_fwd(torch::Tensor xW, torch::Tensor Wh) {
   TORCH_CHECK(lds <= 160 * 1024, "LDS budget exceeded");
   int blocks = G * ((B + rows - 1) / rows);
   auto stream = at::cuda::getCurrentCUDAStream().stream();
+  if (H % 16 == 0) {
+    // v2: register-resident gate math, no barriers in the T loop
+    int blocks2 = G * ((B + 63) / 64);
+    size_t lds2 = (size_t)H4 * LDK * 2 + (size_t)4 * 16 * LDK * 2;
+    auto xp = (const bf16*)xc.data_ptr();
+    auto wp = (const bf16*)Whc.data_ptr();
+    auto hp = (bf16*)hs.data_ptr();
+    auto cp = cs.data_ptr<float>();
+    auto gp = (bf16*)gacts.data_ptr();
+    switch (H / 16) {
+      case 1: hipLaunchKernelGGL(lstm_seq_fwd_v2_kernel<1>, dim3(blocks2),
+                                 dim3(256), lds2, stream, xp, wp, hp, cp, gp,
+                                 B, T); break;
+      case 2: hipLaunchKernelGGL(lstm_seq_fwd_v2_kernel<2>, dim3(blocks2),
+                                 dim3(256), lds2, stream, xp, wp, hp, cp, gp,
+                                 B, T); break;
+      case 3: hipLaunchKernelGGL(lstm_seq_fwd_v2_kernel<3>, dim3(blocks2),
+                                 dim3(256), lds2, stream, xp, wp, hp, cp, gp,
+                                 B, T); break;
+      default: hipLaunchKernelGGL(lstm_seq_fwd_v2_kernel<4>, dim3(blocks2),
+                                 dim3(256), lds2, stream, xp, wp, hp, cp, gp,
+                                 B, T); break;
+    }
+    return {hs, cs, gacts};
+  }
   if (rows == 64)
     hipLaunchKernelGGL(lstm_seq_fwd_kernel<64>, dim3(blocks), dim3(256), lds,
                        stream, (const bf16*)xc.data_ptr(),
@@ -372,6 +621,31 @@ torch::Tensor lstm_seq_bwd(torch::Tensor dSeq, torch::Tensor gacts,
   TORCH_CHECK(lds <= 160 * 1024, "LDS budget exceeded");
   int blocks = G * ((B + rows - 1) / rows);
   auto stream = at::cuda::getCurrentCUDAStream().stream();
+  if (H % 16 == 0) {
+    int blocks2 = G * ((B + 63) / 64);
+    size_t lds2 = (size_t)H * ldg * 2 + (size_t)4 * 16 * ldg * 2;
+    auto dp = (const bf16*)dc.data_ptr();
+    auto gp = (const bf16*)gc.data_ptr();
+    auto cp = cc.data_ptr<float>();
+    auto wp = (const bf16*)Whc.data_ptr();
+    auto op = (bf16*)dG.data_ptr();
+    int lo = last_only ? 1 : 0;
+    switch (H / 16) {
+      case 1: hipLaunchKernelGGL(lstm_seq_bwd_v2_kernel<1>, dim3(blocks2),
+                                 dim3(256), lds2, stream, dp, gp, cp, wp, op,
+                                 B, T, ldg, lo); break;
+      case 2: hipLaunchKernelGGL(lstm_seq_bwd_v2_kernel<2>, dim3(blocks2),
+                                 dim3(256), lds2, stream, dp, gp, cp, wp, op,
+                                 B, T, ldg, lo); break;
+      case 3: hipLaunchKernelGGL(lstm_seq_bwd_v2_kernel<3>, dim3(blocks2),
+                                 dim3(256), lds2, stream, dp, gp, cp, wp, op,
+                                 B, T, ldg, lo); break;
+      default: hipLaunchKernelGGL(lstm_seq_bwd_v2_kernel<4>, dim3(blocks2),
+                                 dim3(256), lds2, stream, dp, gp, cp, wp, op,
+                                 B, T, ldg, lo); break;
+    }
+    return dG;
+  }
   if (rows == 64)
     hipLaunchKernelGGL(lstm_seq_bwd_kernel<64>, dim3(blocks), dim3(256), lds,
                        stream, (const bf16*)dc.data_ptr(),

@@ -312,3 +312,57 @@ def test_anomaly_frame_device_path_matches_pandas():
             np.asarray(frame_cpu[col], dtype=float),
             rtol=1e-4, atol=1e-5,
         )
+
+
+@pytest.mark.parametrize("G,B,T,H,last_only", [
+    (2, 64, 12, 16, False),
+    (1, 100, 20, 48, True),    # Hp=48 (the padded hourglass-42 case)
+    (2, 30, 144, 32, True),    # full lookback, ragged rows
+    (1, 64, 10, 64, False),    # HF=4
+])
+def test_lstm_seq_v2_vs_reference(G, B, T, H, last_only):
+    """The v2 (register-resident, H%16==0) scan kernels vs the
+    per-timestep fp32 oracle — same harness as the v1 test."""
+    require_hip()
+    H4 = 4 * H
+    xW = _rand(G, B, T, H4, seed=40)
+    Wh = _rand(G, H, H4, seed=41) * 0.3
+
+    hs_ref = torch.empty(G, B, T, H)
+    cs_ref = torch.empty(G, B, T, H)
+    ga_ref = torch.empty(G, B, T, H4)
+    h = torch.zeros(G, B, H)
+    c = torch.zeros(G, B, H)
+    for t in range(T):
+        gates = xW[:, :, t] + torch.bmm(h, Wh)
+        h, c, ga = ref.lstm_pointwise_fwd(gates, c)
+        hs_ref[:, :, t] = h
+        cs_ref[:, :, t] = c
+        ga_ref[:, :, t] = ga
+
+    got_hs, got_cs, got_ga = ops.lstm_seq_fwd(to_dev_bf16(xW), to_dev_bf16(Wh))
+    torch.testing.assert_close(got_hs.float().cpu(), hs_ref, rtol=5e-2,
+                               atol=3e-2)
+    torch.testing.assert_close(got_cs.cpu(), cs_ref, rtol=5e-2, atol=3e-2)
+
+    dSeq = _rand(G, B, H, seed=42) if last_only else _rand(G, B, T, H, seed=42)
+    dG_ref = torch.empty(G, B, T, H4)
+    dh = torch.zeros(G, B, H)
+    dc = torch.zeros(G, B, H)
+    for t in range(T - 1, -1, -1):
+        dh_t = dh.clone()
+        if last_only:
+            if t == T - 1:
+                dh_t += dSeq
+        else:
+            dh_t += dSeq[:, :, t]
+        c_prev = cs_ref[:, :, t - 1] if t > 0 else torch.zeros(G, B, H)
+        dgates, dc = ref.lstm_pointwise_bwd(dh_t, dc, ga_ref[:, :, t],
+                                            cs_ref[:, :, t], c_prev)
+        dG_ref[:, :, t] = dgates
+        dh = torch.bmm(dgates, Wh.transpose(1, 2))
+
+    got_dG = ops.lstm_seq_bwd(to_dev_bf16(dSeq), got_ga, got_cs,
+                              to_dev_bf16(Wh), last_only)
+    torch.testing.assert_close(got_dG.float().cpu(), dG_ref, rtol=8e-2,
+                               atol=4e-2)
