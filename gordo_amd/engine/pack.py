@@ -399,32 +399,20 @@ class LSTMPack(BasePack):
     analog of create_keras_timeseriesgenerator, reference
     models.py:713-793)."""
 
-    # Hidden sizes are padded up to a multiple of 16: the pad units are
-    # mathematically inert (zero weights/bias → h_pad = c_pad = 0
-    # through training, zero gradients), and H % 16 == 0 lets the fused
-    # v2 sequence-scan kernel keep the whole gate pointwise phase
-    # lane-local (the 4 gate values of hidden unit hh land in the same
-    # MFMA output lane when the gate blocks are 16-aligned).
-    @staticmethod
-    def _pad16(h: int) -> int:
-        return (h + 15) & ~15
-
     def _declare_params(self):
         spec = self.spec
-        self.lstm_meta = []  # (fin, H_padded, return_sequences)
-        self.lstm_true_h = []
+        self.lstm_meta = []  # (fin, H, return_sequences)
         fin = spec.n_features
         lstm_layers = [l for l in spec.layers if l.kind == "lstm"]
         dense_layers = [l for l in spec.layers if l.kind == "dense"]
         assert len(dense_layers) == 1, "LSTM spec needs exactly one output dense layer"
         for i, layer in enumerate(lstm_layers):
-            Hp = self._pad16(layer.units)
-            self.store.declare(f"Wx{i}", (self.G, fin, 4 * Hp))
-            self.store.declare(f"Wh{i}", (self.G, Hp, 4 * Hp))
-            self.store.declare(f"bl{i}", (self.G, 4 * Hp))
-            self.lstm_meta.append((fin, Hp, layer.return_sequences))
-            self.lstm_true_h.append(layer.units)
-            fin = Hp
+            H = layer.units
+            self.store.declare(f"Wx{i}", (self.G, fin, 4 * H))
+            self.store.declare(f"Wh{i}", (self.G, H, 4 * H))
+            self.store.declare(f"bl{i}", (self.G, 4 * H))
+            self.lstm_meta.append((fin, H, layer.return_sequences))
+            fin = H
         out_layer = dense_layers[0]
         self.dense_meta = (fin, out_layer.units, out_layer.activation)
         self.store.declare("Wd", (self.G, fin, out_layer.units))
@@ -433,31 +421,15 @@ class LSTMPack(BasePack):
     def _init_weights(self):
         for g in range(self.G):
             gen = torch.Generator().manual_seed(int(self.seeds[g]) & 0x7FFFFFFF)
-            for i, (fin, Hp, _rs) in enumerate(self.lstm_meta):
-                H = self.lstm_true_h[i]
-                fin_true = (
-                    self.spec.n_features if i == 0 else self.lstm_true_h[i - 1]
-                )
-                # glorot limits from the TRUE fan dims; pad stays zero
-                wx = torch.zeros(fin, 4 * Hp)
-                wh = torch.zeros(Hp, 4 * Hp)
-                b = torch.zeros(4 * Hp)
-                # Keras glorot limit over the WHOLE (fin, 4H) kernel
-                limit = math.sqrt(6.0 / (fin_true + 4 * H))
-                for q in range(4):
-                    wx[:fin_true, q * Hp : q * Hp + H] = (
-                        torch.rand((fin_true, H), generator=gen) * 2 - 1
-                    ) * limit
-                    wh[:H, q * Hp : q * Hp + H] = _orthogonal((H, H), gen)
-                b[Hp : Hp + H] = 1.0  # unit forget-gate bias (Keras default)
-                self.store.views[f"Wx{i}"][g].copy_(wx)
+            for i, (fin, H, _rs) in enumerate(self.lstm_meta):
+                self.store.views[f"Wx{i}"][g].copy_(_glorot_uniform((fin, 4 * H), gen))
+                wh = torch.cat([_orthogonal((H, H), gen) for _ in range(4)], dim=1)
                 self.store.views[f"Wh{i}"][g].copy_(wh)
+                b = torch.zeros(4 * H)
+                b[H : 2 * H] = 1.0  # unit forget-gate bias (Keras default)
                 self.store.views[f"bl{i}"][g].copy_(b)
             fin, fout, _act = self.dense_meta
-            h_true = self.lstm_true_h[-1]
-            wd = torch.zeros(fin, fout)
-            wd[:h_true] = _glorot_uniform((h_true, fout), gen)
-            self.store.views["Wd"][g].copy_(wd)
+            self.store.views["Wd"][g].copy_(_glorot_uniform((fin, fout), gen))
 
     # ---- sequence forward/backward ------------------------------------
     def _use_fused(self) -> bool:
