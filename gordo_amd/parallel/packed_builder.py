@@ -54,6 +54,10 @@ logger = logging.getLogger(__name__)
 
 _nullcontext = contextlib.nullcontext
 
+# deterministic weight-init snapshots keyed by (arch, seeds) — see
+# PackedFleetBuilder._build_group
+_INIT_CACHE: Dict[Any, torch.Tensor] = {}
+
 
 # ---------------------------------------------------------------------------
 @dataclass
@@ -304,9 +308,17 @@ class PackedFleetBuilder:
         t0_all = time.time()
         # ONE pack serves every fold fit and the final fit: weight init
         # (per-model glorot + orthogonal QR on CPU) costs ~1 s per pack,
-        # so fold fits reset to an init snapshot instead of re-creating.
-        pack = self._make_pack(spec, group)
+        # so fold fits reset to an init snapshot instead of re-creating;
+        # the init itself is deterministic in (arch, per-machine seeds)
+        # and memoized across builds.
+        init_key = (spec.arch_key(), tuple(p.seed for p in group))
+        cached_init = _INIT_CACHE.get(init_key)
+        pack = self._make_pack(spec, group, init_p32=cached_init)
         init_snapshot = pack.store.p32.clone()
+        if cached_init is None:
+            if len(_INIT_CACHE) >= 8:
+                _INIT_CACHE.pop(next(iter(_INIT_CACHE)))
+            _INIT_CACHE[init_key] = init_snapshot.cpu()
         if cv_mode in ("cross_val_only", "full_build"):
             t0 = time.time()
             self._cross_validate_group(
