@@ -306,6 +306,22 @@ class PackedFleetBuilder:
         cv_mode = str(evaluation.get("cv_mode", "full_build")).lower()
         cv_duration = None
         t0_all = time.time()
+        fit_state = {}
+
+        def final_fit(stream):
+            ctx = (
+                torch.cuda.stream(stream) if stream is not None
+                else _nullcontext()
+            )
+            with ctx:
+                self._reset_pack(pack, init_snapshot)
+                Xd = self._stack(Xt_list, pack)
+                Yd = self._stack(y_list, pack)
+                t0 = time.time()
+                fit_state["history"] = pack.fit(
+                    Xd, Yd, **_engine_fit_args(fit_args)
+                )
+                fit_state["duration"] = time.time() - t0
         # ONE pack serves every fold fit and the final fit: weight init
         # (per-model glorot + orthogonal QR on CPU) costs ~1 s per pack,
         # so fold fits reset to an init snapshot instead of re-creating;
@@ -320,23 +336,40 @@ class PackedFleetBuilder:
                 _INIT_CACHE.pop(next(iter(_INIT_CACHE)))
             _INIT_CACHE[init_key] = init_snapshot.cpu()
         if cv_mode in ("cross_val_only", "full_build"):
+            # the final fit is independent of the fold fits (fresh init
+            # either way): on GPU it overlaps the concurrent fold packs
+            # on its own stream.
+            final_thread = None
+            if (
+                cv_mode == "full_build"
+                and self.device != "cpu"
+                and torch.cuda.is_available()
+                and len(group) * 4 <= self.MAX_CONCURRENT_FOLD_MODELS
+            ):
+                import threading
+
+                final_thread = threading.Thread(
+                    target=final_fit, args=(torch.cuda.Stream(),)
+                )
+                final_thread.start()
             t0 = time.time()
             self._cross_validate_group(
                 group, Xt_list, y_list, spec, fit_args, pack, init_snapshot
             )
             cv_duration = time.time() - t0
+            if final_thread is not None:
+                final_thread.join()
+                torch.cuda.synchronize()
             if cv_mode == "cross_val_only":
                 for p in group:
                     self._finalize(p, None, cv_duration, final=False)
                 return
 
-        # final full fit
-        self._reset_pack(pack, init_snapshot)
-        Xd = self._stack(Xt_list, pack)
-        Yd = self._stack(y_list, pack)
-        t0 = time.time()
-        history = pack.fit(Xd, Yd, **_engine_fit_args(fit_args))
-        train_duration = time.time() - t0
+        # final full fit (already done concurrently when possible)
+        if "history" not in fit_state:
+            final_fit(None)
+        history = fit_state["history"]
+        train_duration = fit_state["duration"]
 
         # per-machine adoption + detector finalization
         offset = len(Xt_list[0]) - pack._n_samples(len(Xt_list[0]))
