@@ -413,7 +413,10 @@ class PackedFleetBuilder:
     # separate streams with per-fold packs (the chip is underfilled by
     # one pack's sequence-scan kernels); bounded so concurrent BPTT
     # caches stay well inside HBM.
-    MAX_CONCURRENT_FOLD_MODELS = 256
+    # 4 concurrent fits x 128-model packs ~ 80 GB of transient BPTT
+    # caches at lookback 144 — comfortable in 288 GB (fold packs run
+    # without graph capture, which halves retention)
+    MAX_CONCURRENT_FOLD_MODELS = 640
 
     def _fit_folds(
         self, folds, group, Xt_list, y_list, spec, fit_args, pack,
