@@ -46,7 +46,10 @@ from ..machine.metadata import (
     DatasetBuildMetadata,
     ModelBuildMetadata,
 )
-from ..machine.model.anomaly.diff import DiffBasedAnomalyDetector
+from ..machine.model.anomaly.diff import (
+    DiffBasedAnomalyDetector,
+    DiffBasedKFCVAnomalyDetector,
+)
 from ..machine.model.models import KerasBaseEstimator, KerasLSTMBaseEstimator
 from ..util import disk_registry
 
@@ -85,6 +88,13 @@ def _classify(model) -> Tuple[Optional[DiffBasedAnomalyDetector], Optional[list]
     [DiffBased*Detector(base_estimator=)] [Pipeline(pre..., )] KerasEst."""
     detector = None
     inner = model
+    if isinstance(model, DiffBasedKFCVAnomalyDetector):
+        # KFCV thresholds are quantiles of the full reassembled
+        # validation series (reference diff.py:566-635) — different
+        # math than the packed rolling-min-max path; these machines
+        # take the exact per-machine ModelBuilder path (their inner
+        # estimator still trains on the GPU as a pack of 1).
+        return None, None, None
     if isinstance(model, DiffBasedAnomalyDetector):
         detector = model
         inner = model.base_estimator

@@ -276,3 +276,55 @@ def test_group_splitting_caps_pack_size():
     results = dict(fb.build_all())
     assert len(results) == 7
     assert all(not isinstance(v, BaseException) for v in results.values())
+
+
+def test_kfcv_machines_take_exact_path(tmp_path):
+    """KFCV detectors must go through ModelBuilder (quantile threshold
+    semantics), not the packed rolling-threshold path."""
+    from gordo_amd.parallel import PackedFleetBuilder
+    from gordo_amd.workflow import NormalizedConfig
+
+    cfg = {
+        "machines": [
+            {
+                "name": "kfcv-m",
+                "dataset": {
+                    "type": "SineWaveDataset",
+                    "tag_list": ["a", "b", "c"],
+                    "train_start_date": "2019-01-01T00:00:00+00:00",
+                    "train_end_date": "2019-01-03T00:00:00+00:00",
+                },
+                "model": {
+                    "gordo_amd.machine.model.anomaly.diff.DiffBasedKFCVAnomalyDetector": {
+                        "window": 12,
+                        "base_estimator": {
+                            "gordo_amd.machine.model.models.KerasAutoEncoder": {
+                                "kind": "feedforward_hourglass",
+                                "epochs": 1,
+                            }
+                        },
+                    }
+                },
+                "evaluation": {
+                    "cv_mode": "full_build",
+                    "cv": {
+                        "sklearn.model_selection.KFold": {
+                            "n_splits": 3, "shuffle": True,
+                            "random_state": 0,
+                        }
+                    },
+                },
+            }
+        ]
+    }
+    norm = NormalizedConfig(cfg, project_name="p")
+    fb = PackedFleetBuilder(norm.machines, output_dir=str(tmp_path))
+    results = dict(fb.build_all())
+    machine = results["kfcv-m"]
+    assert not isinstance(machine, BaseException), repr(machine)
+    from gordo_amd import serializer
+
+    model = serializer.load(str(tmp_path / "kfcv-m"))
+    # quantile thresholds set (scalar aggregate, per-tag series)
+    assert hasattr(model, "aggregate_threshold_")
+    assert len(model.feature_thresholds_) == 3
