@@ -543,7 +543,6 @@ class PackedFleetBuilder:
         per_machine_scores: List[Dict[str, List[float]]] = [
             {} for _ in group
         ]
-        window = getattr(group[0].detector, "window", None) if group[0].detector else None
 
         fold_preds = self._fit_folds(
             folds, group, Xt_list, y_list, spec, fit_args, pack, init_snapshot
@@ -600,6 +599,8 @@ class PackedFleetBuilder:
                     )
                     det.aggregate_threshold_ = agg_thr
                     det.feature_thresholds_ = tag_thr
+                    # smoothing window is per machine, not per group
+                    window = getattr(det, "window", None)
                     if window is not None:
                         s_agg = scaled_mse.rolling(window).min().max()
                         s_tag = mae.rolling(window).min().max()
