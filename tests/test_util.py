@@ -184,3 +184,58 @@ def test_mlflow_batching():
     from gordo_amd.reporters.mlflow import batch_log_items
 
     assert batch_log_items(list(range(5)), 2) == [[0, 1], [2, 3], [4]]
+
+
+def test_gpu_only_ops_fail_loudly_on_cpu():
+    """GPU-only fused ops must raise, never silently fall back."""
+    import torch
+
+    from gordo_amd import ops
+
+    if torch.cuda.is_available():
+        pytest.skip("CPU-only check")
+    a = torch.rand(1, 4, 8)
+    with pytest.raises((RuntimeError, Exception)):
+        ops.lstm_seq_fwd(torch.rand(1, 2, 3, 16), torch.rand(1, 4, 16))
+    with pytest.raises(ValueError):
+        # GPU adam requires the device step counter
+        class FakeCuda:
+            is_cuda = True
+        ops.adam_step  # signature check only: the api guards step_buf
+
+
+def test_serializer_callbacks_and_params():
+    from gordo_amd.serializer import build_callbacks, load_params_from_definition
+
+    cbs = build_callbacks(["sklearn.preprocessing.MinMaxScaler"])
+    from sklearn.preprocessing import MinMaxScaler
+
+    assert isinstance(cbs[0], MinMaxScaler)
+    params = load_params_from_definition(
+        {"scaler": {"sklearn.preprocessing.MinMaxScaler": {}},
+         "n": 3}
+    )
+    assert isinstance(params["scaler"], MinMaxScaler)
+    assert params["n"] == 3
+
+
+def test_influx_provider_gated():
+    from gordo_amd.core import InfluxDataProvider
+
+    p = InfluxDataProvider()
+    with pytest.raises(RuntimeError):
+        list(p.load_series(None, None, []))
+
+
+def test_engine_spec_roundtrip():
+    from gordo_amd.engine.spec import LayerSpec, ModelSpec
+
+    spec = ModelSpec(
+        model_type="lstm", n_features=5, n_features_out=5,
+        layers=[LayerSpec(kind="lstm", units=8)],
+        lookback_window=12,
+    )
+    d = spec.to_dict()
+    spec2 = ModelSpec.from_dict(d)
+    assert spec2.arch_key() == spec.arch_key()
+    assert spec2.adam_params["lr"] == 0.001
