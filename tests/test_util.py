@@ -194,14 +194,8 @@ def test_gpu_only_ops_fail_loudly_on_cpu():
 
     if torch.cuda.is_available():
         pytest.skip("CPU-only check")
-    a = torch.rand(1, 4, 8)
-    with pytest.raises((RuntimeError, Exception)):
+    with pytest.raises(Exception):
         ops.lstm_seq_fwd(torch.rand(1, 2, 3, 16), torch.rand(1, 4, 16))
-    with pytest.raises(ValueError):
-        # GPU adam requires the device step counter
-        class FakeCuda:
-            is_cuda = True
-        ops.adam_step  # signature check only: the api guards step_buf
 
 
 def test_serializer_callbacks_and_params():
