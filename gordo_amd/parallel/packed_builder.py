@@ -88,13 +88,6 @@ def _classify(model) -> Tuple[Optional[DiffBasedAnomalyDetector], Optional[list]
     [DiffBased*Detector(base_estimator=)] [Pipeline(pre..., )] KerasEst."""
     detector = None
     inner = model
-    if isinstance(model, DiffBasedKFCVAnomalyDetector):
-        # KFCV thresholds are quantiles of the full reassembled
-        # validation series (reference diff.py:566-635) — different
-        # math than the packed rolling-min-max path; these machines
-        # take the exact per-machine ModelBuilder path (their inner
-        # estimator still trains on the GPU as a pack of 1).
-        return None, None, None
     if isinstance(model, DiffBasedAnomalyDetector):
         detector = model
         inner = model.base_estimator
@@ -548,6 +541,20 @@ class PackedFleetBuilder:
             folds, group, Xt_list, y_list, spec, fit_args, pack, init_snapshot
         )
 
+        # KFCV detectors: thresholds are quantiles of the smoothed
+        # validation metric over the FULL reassembled prediction series
+        # (reference diff.py:566-635) — accumulate per-machine y_pred
+        # across folds here, finalize after the fold loop.
+        kfcv_pred = {
+            g_idx: np.zeros_like(y_list[g_idx])
+            for g_idx, p in enumerate(group)
+            if isinstance(p.detector, DiffBasedKFCVAnomalyDetector)
+        }
+        kfcv_mse = {
+            g_idx: np.full(len(y_list[g_idx]), np.nan, dtype=np.float64)
+            for g_idx in kfcv_pred
+        }
+
         for fold_i, (train_idx, test_idx) in enumerate(folds):
             t_f0 = time.time()
             preds, t_fit, t_pred = fold_preds[fold_i]
@@ -574,7 +581,22 @@ class PackedFleetBuilder:
                     )
 
                 # DiffBased thresholds: fold scaler fitted on y_train
-                if p.detector is not None:
+                if isinstance(p.detector, DiffBasedKFCVAnomalyDetector):
+                    det = p.detector
+                    fold_scaler = sk_clone(det.scaler)
+                    fold_scaler.fit(y_list[g_idx][train_idx])
+                    pred_full = preds[g_idx]
+                    # KFold test slices are contiguous index sets; the
+                    # model output covers the whole slice (offset 0 for
+                    # dense; windows handled below for LSTM)
+                    off = len(test_idx) - len(pred_full)
+                    rows = test_idx[off:]
+                    kfcv_pred[g_idx][rows] = pred_full
+                    kfcv_mse[g_idx][rows] = (
+                        (fold_scaler.transform(pred_full)
+                         - fold_scaler.transform(y_list[g_idx][rows])) ** 2
+                    ).mean(axis=1)
+                elif p.detector is not None:
                     det = p.detector
                     fold_scaler = sk_clone(det.scaler)
                     fold_scaler.fit(y_list[g_idx][train_idx])
@@ -620,6 +642,15 @@ class PackedFleetBuilder:
                 "  fold %d: fit %.2fs, predict %.2fs, score+thresholds %.2fs",
                 fold_i, t_fit, t_pred, time.time() - t_s0,
             )
+
+        # finalize KFCV thresholds over the reassembled series
+        for g_idx in kfcv_pred:
+            det = group[g_idx].detector
+            y_full = y_list[g_idx]
+            mse_series = pd.Series(kfcv_mse[g_idx])
+            det.aggregate_threshold_ = det._calculate_threshold(mse_series)
+            abs_err = pd.DataFrame(np.abs(y_full - kfcv_pred[g_idx]))
+            det.feature_thresholds_ = det._calculate_threshold(abs_err)
 
         # assemble fold stats + split metadata per machine
         for g_idx, p in enumerate(group):

@@ -278,9 +278,10 @@ def test_group_splitting_caps_pack_size():
     assert all(not isinstance(v, BaseException) for v in results.values())
 
 
-def test_kfcv_machines_take_exact_path(tmp_path):
-    """KFCV detectors must go through ModelBuilder (quantile threshold
-    semantics), not the packed rolling-threshold path."""
+def test_kfcv_machines_quantile_thresholds(tmp_path):
+    """KFCV detectors get quantile thresholds (over the reassembled
+    validation series) from the packed path — not the DiffBased
+    rolling-min-max math."""
     from gordo_amd.parallel import PackedFleetBuilder
     from gordo_amd.workflow import NormalizedConfig
 
