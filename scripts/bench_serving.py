@@ -35,6 +35,9 @@ def main():
     ap.add_argument("--endpoint", default="anomaly",
                     choices=["anomaly", "prediction", "both"])
     ap.add_argument("--rows", type=int, default=ROWS)
+    ap.add_argument("--n-models", type=int, default=1,
+                    help="serve N distinct models round-robin (the "
+                         "many-model LRU/HBM-residency story)")
     ap.add_argument("--direct", action="store_true",
                     help="measure model.anomaly() directly (no HTTP/JSON): "
                          "the batched inference engine path, BASELINE "
@@ -50,11 +53,10 @@ def main():
 
     sensors = [f"bench-tag-{i}" for i in range(N_TAGS)]
     tag_block = "\n".join(f"        - {s}" for s in sensors)
-    config = f"""
-machines:
+    machine_block = """
   - dataset: |
       tags:
-{tag_block}
+{tags}
       train_start_date: '2019-01-01T00:00:00+00:00'
       train_end_date: '2019-01-08T00:00:00+00:00'
       type: SineWaveDataset
@@ -68,8 +70,12 @@ machines:
             - gordo.machine.model.models.KerasAutoEncoder:
                 kind: feedforward_hourglass
                 epochs: 3
-    name: serve-bench
+    name: serve-bench{i}
 """
+    config = "machines:" + "".join(
+        machine_block.format(tags=tag_block, i=("" if i == 0 else f"-{i}"))
+        for i in range(args.n_models)
+    )
     with tempfile.TemporaryDirectory() as td:
         collection = os.path.join(td, "1577836800000")
         for model, machine in local_build(config):
@@ -115,11 +121,16 @@ machines:
             ["anomaly", "prediction"] if args.endpoint == "both"
             else [args.endpoint]
         ):
-            url = (
-                "/gordo/v0/bench/serve-bench/anomaly/prediction"
-                if endpoint == "anomaly"
-                else "/gordo/v0/bench/serve-bench/prediction"
-            )
+            model_names = ["serve-bench"] + [
+                f"serve-bench-{i}" for i in range(1, args.n_models)
+            ]
+            urls = [
+                f"/gordo/v0/bench/{name}/"
+                + ("anomaly/prediction" if endpoint == "anomaly"
+                   else "prediction")
+                for name in model_names
+            ]
+            url = urls[0]
             # warmup + correctness
             client = app.test_client()
             resp = client.post(url, json=payload)
@@ -136,8 +147,9 @@ machines:
                         if counter["n"] >= args.rounds:
                             return
                         counter["n"] += 1
+                        my_url = urls[counter["n"] % len(urls)]
                     t0 = time.perf_counter()
-                    r = c.post(url, json=payload)
+                    r = c.post(my_url, json=payload)
                     dt = time.perf_counter() - t0
                     assert r.status_code == 200
                     with lock:
@@ -162,6 +174,7 @@ machines:
                 "threads": args.threads,
                 "payload_rows": args.rows,
                 "n_tags": N_TAGS,
+                "n_models": args.n_models,
                 "device": "cuda" if torch.cuda.is_available() else "cpu",
             }
         print(json.dumps({"metric": "ml_server predictions/sec",
