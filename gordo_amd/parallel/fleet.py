@@ -38,9 +38,18 @@ def init_distributed(backend: Optional[str] = None) -> Tuple[int, int]:
     if "RANK" not in os.environ or "WORLD_SIZE" not in os.environ:
         return 0, 1
     if backend is None:
+        backend = os.environ.get("GORDO_DIST_BACKEND")
+    if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
     if backend == "nccl":
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", 0)))
+        local_rank = int(os.environ.get("LOCAL_RANK", 0))
+        if local_rank >= torch.cuda.device_count():
+            # oversubscribed (more ranks than GPUs, e.g. a CPU-parallel
+            # test on a 1-GPU box): RCCL cannot place the rank — fall
+            # back to gloo instead of crashing
+            backend = "gloo"
+        else:
+            torch.cuda.set_device(local_rank)
     dist.init_process_group(backend=backend)
     return dist.get_rank(), dist.get_world_size()
 
