@@ -39,6 +39,8 @@ def test_fleet_build_two_ranks(tmp_path):
     env = dict(os.environ)
     env.update(
         GORDO_DIST_BACKEND="gloo",  # 2 CPU ranks even on a 1-GPU box
+        CUDA_VISIBLE_DEVICES="",    # children must not touch the GPU a
+                                    # parent test process may hold
         GORDO_FLEET_MACHINE_CONFIG=str(cfg_path),
         GORDO_FLEET_PROJECT_NAME="fleet-proj",
         GORDO_FLEET_OUTPUT_DIR=str(out_dir),
