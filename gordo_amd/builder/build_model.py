@@ -214,7 +214,13 @@ class ModelBuilder:
         import torch
 
         logger.info("Setting random seed: %r", seed)
-        torch.manual_seed(seed)
+        # Seed the CPU generator ONLY: torch.manual_seed also resets
+        # every CUDA generator's state, which orphans live captured
+        # hipGraphs (their destructors then throw c10::Error
+        # "The graph should be registered to the state" inside GC and
+        # terminate the process — HIPGeneratorImpl.cpp:158). The engine
+        # draws all randomness from explicit CPU generators.
+        torch.default_generator.manual_seed(seed)
         np.random.seed(seed)
         random.seed(seed)
 

@@ -184,9 +184,9 @@ def test_lstm_backward_matches_autograd():
 def test_adam_matches_torch():
     from gordo_amd.ops import reference as ref
 
-    torch.manual_seed(0)
-    p = torch.rand(100)
-    g = torch.randn(100)
+    gen = torch.Generator().manual_seed(0)
+    p = torch.rand(100, generator=gen)
+    g = torch.randn(100, generator=gen)
     m = torch.zeros(100)
     v = torch.zeros(100)
     p2 = p.clone().requires_grad_(True)
