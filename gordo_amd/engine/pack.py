@@ -232,6 +232,14 @@ class BasePack:
             if len(cached) >= 4:  # bound distinct shapes per pack
                 return None
             try:
+                # flush pending GC NOW: a garbage pack's CUDAGraph
+                # destructor running inside the upcoming capture (or
+                # inside another thread's capture) hits HIP's
+                # generator-state unregister check and terminates the
+                # process
+                import gc
+
+                gc.collect()
                 sx = torch.empty_like(Xb)
                 st = torch.empty_like(Tb)
                 sx.copy_(Xb)
@@ -277,6 +285,14 @@ class BasePack:
         graph.replay()
         self.store.step_count += 1  # device step_buf is the truth
         return loss
+
+    def release_graphs(self):
+        """Drop captured graphs deterministically (outside any capture)
+        instead of at arbitrary GC time."""
+        cached = getattr(self, "_graph_cache", None)
+        if cached:
+            with BasePack._graph_capture_lock:
+                cached.clear()
 
     def fit(
         self,

@@ -407,6 +407,7 @@ class PackedFleetBuilder:
 
             with concurrent.futures.ThreadPoolExecutor(8) as ex:
                 list(ex.map(save_one, group))
+        pack.release_graphs()
         logger.info(
             "Packed build of %d machines done in %.2fs",
             len(group), time.time() - t0_all,
