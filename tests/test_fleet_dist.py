@@ -61,9 +61,10 @@ def test_fleet_build_two_ranks(tmp_path):
     )
     assert proc.returncode == 0, proc.stderr[-3000:]
     summary = json.loads(status_file.read_text())
-    assert summary["n_machines"] == 4
-    assert summary["n_ok"] == 4
-    assert summary["world_size"] == 2
+    debug = (summary, proc.stdout[-1500:], proc.stderr[-1500:])
+    assert summary["n_machines"] == 4, debug
+    assert summary["n_ok"] == 4, debug
+    assert summary["world_size"] == 2, debug
     # every machine saved in the shared collection layout
     for i in range(4):
         d = out_dir / f"fleet-m-{i}"
