@@ -74,7 +74,7 @@ with tempfile.TemporaryDirectory() as tmp:
     results = PackedFleetBuilder(norm.machines, output_dir=collection).build_all()
     for name, res in results:
         ok = not isinstance(res, BaseException)
-        print(f"built {name}: {'OK' if ok else res!r}")
+        print(f"built {name}: " + ("OK" if ok else repr(res)))
 
     os.environ["MODEL_COLLECTION_DIR"] = collection
     from gordo_amd.server.server import build_app
@@ -98,6 +98,6 @@ with tempfile.TemporaryDirectory() as tmp:
     start = dateutil.parser.isoparse("2019-01-01T00:00:00+00:00")
     end = dateutil.parser.isoparse("2019-01-02T00:00:00+00:00")
     for name, frame, errors in client.predict(start, end):
+        conf = frame["total-anomaly-confidence"].to_numpy().max()
         print(f"{name}: {len(frame)} scored rows, errors={errors}, "
-              f"max total-anomaly-confidence="
-              f"{float(frame['total-anomaly-confidence'].max().iloc[0]):.3f}")
+              f"max total-anomaly-confidence={float(conf):.3f}")

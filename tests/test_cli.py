@@ -250,3 +250,86 @@ machines:
     summary = json.loads(status.read_text())
     assert summary["n_ok"] == 1
     assert (out_dir / "fleet-cli-m" / "model.pkl").is_file()
+
+
+def test_workflow_generate_options(tmp_path):
+    """Option combinations render into the manifest (the reference's
+    golden-config lane, test_workflow_generator.py:145-813 shape)."""
+    cfg = tmp_path / "cfg.yml"
+    cfg.write_text(
+        """
+machines:
+  - name: wf-opt-m
+    dataset: |
+      type: RandomDataset
+      tag_list: [a, b]
+      train_start_date: '2019-01-01T00:00:00+00:00'
+      train_end_date: '2019-01-02T00:00:00+00:00'
+    model: |
+      sklearn.decomposition.PCA:
+        n_components: 2
+"""
+    )
+    runner = CliRunner()
+    result = runner.invoke(
+        gordo,
+        [
+            "workflow", "generate",
+            "--machine-config", str(cfg),
+            "--project-name", "wf-proj",
+            "--project-revision", "1577000000000",
+            "--builder-retries", "7",
+            "--keda-enabled",
+            "--docker-registry", "my.registry",
+            "--docker-repository", "my-repo",
+            "--resource-labels", '{"team": "x"}',
+            "--custom-model-builder-envs",
+            '[{"name": "EXTRA", "value": "1"}]',
+        ],
+    )
+    assert result.exit_code == 0, result.output
+    out = result.output
+    assert "limit: 7" in out                       # builder retries
+    assert "keda.sh/v1alpha1" in out               # KEDA scaler emitted
+    assert "my.registry/my-repo/gordo-base" in out # image coordinates
+    assert "team: x" in out                        # resource labels
+    assert "EXTRA" in out                          # custom builder env
+    assert "1577000000000" in out                  # revision in paths
+    docs = list(yaml.safe_load_all(out))
+    assert docs[0]["kind"] == "Workflow"
+
+
+def test_workflow_crd_unwrap(tmp_path):
+    """Configs wrapped in the Gordo CRD form (spec.config) unwrap."""
+    cfg = tmp_path / "crd.yml"
+    cfg.write_text(
+        """
+apiVersion: equinor.com/v1
+kind: Gordo
+metadata:
+  name: crd-proj
+spec:
+  config:
+    machines:
+      - name: crd-m
+        dataset: |
+          type: RandomDataset
+          tag_list: [a, b]
+          train_start_date: '2019-01-01T00:00:00+00:00'
+          train_end_date: '2019-01-02T00:00:00+00:00'
+        model: |
+          sklearn.decomposition.PCA:
+            n_components: 2
+"""
+    )
+    runner = CliRunner()
+    result = runner.invoke(
+        gordo,
+        [
+            "workflow", "generate",
+            "--machine-config", str(cfg),
+            "--project-name", "crd-proj",
+        ],
+    )
+    assert result.exit_code == 0, result.output
+    assert "build-crd-m" in result.output
