@@ -292,36 +292,57 @@ class KerasAutoEncoder(KerasBaseEstimator, TransformerMixin):
 
 
 class KerasRawModelRegressor(KerasAutoEncoder):
-    """Build a model from a raw ``spec:`` definition — a Sequential of
-    Dense layers with a ``compile:`` section (reference
-    models.py:401-460; only the Dense subset is meaningful here, which
-    is what the reference's own example uses)."""
+    """Build a model from a raw definition — ``kind`` is a dict with a
+    ``spec:`` (a Sequential of Dense layers) and an optional
+    ``compile:`` section, exactly the reference's raw-config shape
+    (models.py:401-460). Only the Dense subset is meaningful here,
+    which is what the reference's own example uses."""
+
+    _expected_keys = ("spec", "compile")
 
     def __init__(self, kind: Union[dict, str, Callable] = "raw", **kwargs):
         if isinstance(kind, dict):
-            kwargs.setdefault("spec", kind)
-            kind = "raw"
-        self._raw_kind = kind
+            # accept both {"spec": {...}, "compile": {...}} (reference
+            # shape) and a bare {Sequential: {...}} spec
+            if "spec" in kind:
+                kwargs.setdefault("spec", kind["spec"])
+                if "compile" in kind:
+                    kwargs.setdefault("compile", kind["compile"])
+            else:
+                kwargs.setdefault("spec", kind)
+            self._raw_kind = kind
+        else:
+            self._raw_kind = kind
         super().__init__(kind="feedforward_model", **kwargs)
         self.kind = "raw"
 
     def load_kind(self, kind):
         return "feedforward_model"
 
+    def __repr__(self):
+        return f"{self.__class__.__name__}(kind: {self._raw_kind!r})"
+
     def get_params(self, deep=True):
         params = {"kind": self._raw_kind}
-        params.update(self.kwargs)
+        params.update(
+            {k: v for k, v in self.kwargs.items() if k not in ("spec", "compile")}
+        )
         return params
 
     def build_pack_spec(self, n_features, n_features_out=None):
         spec_def = self.kwargs.get("spec")
         if not spec_def:
             raise ValueError("KerasRawModelRegressor requires a 'spec'")
-        return _parse_raw_spec(spec_def, n_features, n_features_out or n_features)
+        return _parse_raw_spec(
+            spec_def,
+            self.kwargs.get("compile") or {},
+            n_features,
+            n_features_out or n_features,
+        )
 
 
 def _parse_raw_spec(
-    spec_def: dict, n_features: int, n_features_out: int
+    spec_def: dict, compile_def: dict, n_features: int, n_features_out: int
 ) -> ModelSpec:
     """Parse a keras-like raw Sequential spec into a dense ModelSpec."""
     if not isinstance(spec_def, dict) or len(spec_def) != 1:
@@ -351,9 +372,6 @@ def _parse_raw_spec(
             raise ValueError(f"Unparsable layer definition {layer_def!r}")
     if not layers:
         raise ValueError("raw spec contains no layers")
-    compile_def = {}
-    if "compile" in (spec_def.get(seq_key) or {}):
-        compile_def = spec_def[seq_key]["compile"] or {}
     return ModelSpec(
         model_type="feedforward",
         n_features=n_features,

@@ -162,3 +162,39 @@ def test_callable_kind_registers():
     X = np.random.rand(64, 5).astype("float32")
     model.fit(X)
     assert model.predict(X).shape == X.shape
+
+
+def test_raw_model_regressor_reference_shape():
+    """The reference's exact raw-config shape (models.py:401-460
+    docstring): kind = {compile: ..., spec: {Sequential: {layers}}}."""
+    import yaml
+
+    from gordo_amd.machine.model import KerasRawModelRegressor
+
+    config = yaml.safe_load(
+        """
+compile:
+  loss: mse
+  optimizer: adam
+spec:
+  tensorflow.keras.models.Sequential:
+    layers:
+      - tensorflow.keras.layers.Dense:
+          units: 4
+          input_shape: [4]
+      - tensorflow.keras.layers.Dense:
+          units: 1
+"""
+    )
+    model = KerasRawModelRegressor(kind=config, epochs=1)
+    X = np.random.rand(30, 4).astype("float32")
+    y = np.random.rand(30, 1).astype("float32")
+    model.fit(X, y)
+    out = model.predict(X)
+    assert out.shape == (30, 1)
+    # sklearn-clone compatible params
+    params = model.get_params()
+    assert params["kind"] == config
+    clone = KerasRawModelRegressor(**params)
+    clone.fit(X, y)
+    assert clone.predict(X).shape == (30, 1)
