@@ -1,5 +1,6 @@
 import json
 import os
+import sys
 
 import numpy as np
 import pandas as pd
@@ -186,3 +187,88 @@ def test_proxy_path_rewrite(flask_app):
         },
     )
     assert resp.status_code == 200
+
+
+def test_server_timing_header(api_client):
+    resp = api_client.get("/healthcheck")
+    assert "Server-Timing" in resp.headers
+    assert "request_walltime_s" in resp.headers["Server-Timing"]
+
+
+def test_invalid_gordo_name_422(api_client, gordo_project, X):
+    resp = _post_json(
+        api_client, f"/gordo/v0/{gordo_project}/bad_name!/prediction", X
+    )
+    assert resp.status_code == 422
+
+
+def test_metadata_404_unknown_model(api_client, gordo_project):
+    resp = api_client.get(f"/gordo/v0/{gordo_project}/no-such/metadata")
+    assert resp.status_code == 404
+
+
+def test_delete_revision_endpoint(
+    flask_app, gordo_project, gordo_name, model_collection_directory
+):
+    import shutil
+
+    # create an OLD revision alongside the current one
+    parent = os.path.dirname(model_collection_directory)
+    old_rev = os.path.join(parent, "1000000000000")
+    src = os.path.join(model_collection_directory, gordo_name)
+    shutil.copytree(src, os.path.join(old_rev, gordo_name),
+                    dirs_exist_ok=True)
+    client = flask_app.test_client()
+
+    # the old revision serves
+    resp = client.get(
+        f"/gordo/v0/{gordo_project}/{gordo_name}/metadata",
+        query_string={"revision": "1000000000000"},
+    )
+    assert resp.status_code == 200
+
+    # deleting the CURRENT revision is refused
+    cur = os.path.basename(model_collection_directory)
+    resp = client.delete(
+        f"/gordo/v0/{gordo_project}/{gordo_name}/revision/{cur}"
+    )
+    assert resp.status_code == 409
+
+    # non-numeric revision refused
+    resp = client.delete(
+        f"/gordo/v0/{gordo_project}/{gordo_name}/revision/abc"
+    )
+    assert resp.status_code == 422
+
+    # deleting the old revision works and removes it from disk
+    resp = client.delete(
+        f"/gordo/v0/{gordo_project}/{gordo_name}/revision/1000000000000"
+    )
+    assert resp.status_code == 200
+    assert not os.path.exists(old_rev)
+
+
+def test_with_prometheus_enabled(model_collection_directory,
+                                 trained_model_directories):
+    from prometheus_client.registry import CollectorRegistry
+
+    from gordo_amd.server.server import build_app
+
+    os.environ["MODEL_COLLECTION_DIR"] = model_collection_directory
+    app = build_app(
+        config={"ENABLE_PROMETHEUS": True, "PROJECT": "prom-proj"},
+        prometheus_registry=CollectorRegistry(),
+    )
+    app.testing = True
+    resp = app.test_client().get("/healthcheck")
+    assert resp.status_code == 200
+
+
+def test_run_cmd():
+    import subprocess
+
+    from gordo_amd.server.server import run_cmd
+
+    run_cmd([sys.executable, "-c", "print('ok')"])
+    with pytest.raises(subprocess.CalledProcessError):
+        run_cmd([sys.executable, "-c", "raise SystemExit(3)"])
