@@ -115,8 +115,16 @@ def _resolve_value(value: Any, instantiate_strings: bool = False) -> Any:
         obj = _maybe_import(value)
         if isinstance(obj, str):
             return obj
-        if instantiate_strings and inspect.isclass(obj):
-            return _construct(obj, None)
+        if inspect.isclass(obj):
+            # Estimator classes given as bare strings are instantiated with
+            # defaults (reference gordo/serializer/from_definition.py:290-303:
+            # `{"base_estimator": "sklearn...RandomForestRegressor"}` →
+            # RandomForestRegressor()); other classes stay classes unless the
+            # caller asked for instantiation.
+            from sklearn.base import BaseEstimator
+
+            if instantiate_strings or issubclass(obj, BaseEstimator):
+                return _construct(obj, None)
         return obj
     return value
 

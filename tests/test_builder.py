@@ -194,3 +194,191 @@ def test_local_build(config_str):
     for model, machine in results:
         machine_check(machine)
         assert machine.project_name == "local-build"
+
+
+class _ScalingRegressor:
+    """Predicts X * multiplier — used to probe per-feature error scaling."""
+
+    def __init__(self, multiplier):
+        self.multiplier = multiplier
+
+    def fit(self, X, y):
+        return self
+
+    def predict(self, X):
+        return np.asarray(X) * self.multiplier
+
+
+@pytest.mark.parametrize(
+    "scaler", [None, "sklearn.preprocessing.MinMaxScaler"]
+)
+def test_build_metrics_dict_scaler(scaler):
+    """With a scoring scaler, a 20% error on a large-scale feature and on
+    a small-scale feature weigh the same (reference
+    tests/gordo/builder/test_builder.py::test_get_metrics_dict_scaler)."""
+    import pandas as pd
+    import sklearn.metrics
+
+    y = pd.DataFrame(
+        np.array([[1, 1], [2, 2], [3, 3], [4, 4], [5, 5]]) * [1, 100],
+        columns=["Tag 1", "Tag 2"],
+    )
+    metrics_dict = ModelBuilder.build_metrics_dict(
+        [sklearn.metrics.mean_squared_error], y, scaler=scaler
+    )
+    metric_func = metrics_dict["mean-squared-error"]
+    mse_feat1_wrong = metric_func(_ScalingRegressor(np.array([0.8, 1.0])), y, y)
+    mse_feat2_wrong = metric_func(_ScalingRegressor(np.array([1.0, 0.8])), y, y)
+    if scaler:
+        assert np.isclose(mse_feat1_wrong, mse_feat2_wrong)
+    else:
+        assert not np.isclose(mse_feat1_wrong, mse_feat2_wrong)
+
+
+def test_determine_offset_helper():
+    from sklearn.linear_model import LinearRegression
+    from sklearn.multioutput import MultiOutputRegressor
+
+    from gordo_amd.machine.model import models
+
+    X = np.random.RandomState(0).random((100, 10))
+    y = np.random.RandomState(1).random((100, 10))
+    cases = [
+        (
+            models.KerasLSTMAutoEncoder(
+                kind="lstm_hourglass", lookback_window=10, epochs=1
+            ),
+            9,
+        ),
+        (
+            models.KerasLSTMForecast(
+                kind="lstm_symmetric", lookback_window=13, epochs=1
+            ),
+            13,
+        ),
+        (models.KerasAutoEncoder(kind="feedforward_hourglass", epochs=1), 0),
+        (MultiOutputRegressor(LinearRegression()), 0),
+    ]
+    for model, expected in cases:
+        model.fit(X, y)
+        assert ModelBuilder._determine_offset(model, X) == expected, model
+
+
+def test_output_dir_nested(tmp_path):
+    """Builder creates missing intermediate directories."""
+    out = tmp_path / "some" / "sub" / "directories"
+    machine = make_machine(model=SKLEARN_MODEL)
+    ModelBuilder(machine).build(output_dir=str(out))
+    assert (out / "model.pkl").is_file()
+    assert (out / "metadata.json").is_file()
+
+
+def test_output_scores_metadata_aggregate():
+    """Aggregate fold-mean == mean of the per-tag fold-means (reference
+    test_output_scores_metadata), through a TransformedTargetRegressor."""
+    model = {
+        "gordo.machine.model.anomaly.diff.DiffBasedAnomalyDetector": {
+            "require_thresholds": False,
+            "scaler": "sklearn.preprocessing.MinMaxScaler",
+            "base_estimator": {
+                "sklearn.compose.TransformedTargetRegressor": {
+                    "transformer": "sklearn.preprocessing.MinMaxScaler",
+                    "regressor": {
+                        "sklearn.pipeline.Pipeline": {
+                            "steps": [
+                                "sklearn.preprocessing.MinMaxScaler",
+                                {
+                                    "gordo.machine.model.models.KerasAutoEncoder": {
+                                        "kind": "feedforward_hourglass",
+                                        "compression_factor": 0.5,
+                                        "encoding_layers": 1,
+                                        "func": "tanh",
+                                        "out_func": "linear",
+                                        "epochs": 1,
+                                    }
+                                },
+                            ]
+                        }
+                    },
+                }
+            },
+        }
+    }
+    _, machine_out = ModelBuilder(make_machine(model=model)).build()
+    scores = machine_out.metadata.build_metadata.model.cross_validation.scores
+    for s in (
+        "explained-variance-score",
+        "r2-score",
+        "mean-squared-error",
+        "mean-absolute-error",
+    ):
+        per_tag = (scores[f"{s}-Tag-1"]["fold-mean"]
+                   + scores[f"{s}-Tag-2"]["fold-mean"]) / 2
+        assert per_tag == pytest.approx(scores[s]["fold-mean"])
+
+
+@pytest.mark.parametrize(
+    "metrics_",
+    (
+        ["sklearn.metrics.r2_score"],
+        ["r2_score"],  # bare names resolve in sklearn.metrics
+        None,
+        ["sklearn.metrics.r2_score",
+         "sklearn.metrics.explained_variance_score"],
+    ),
+)
+def test_model_builder_metrics_list(metrics_):
+    evaluation = {"cv_mode": "full_build"}
+    if metrics_:
+        evaluation["metrics"] = metrics_
+    machine = make_machine(
+        model={
+            "sklearn.multioutput.MultiOutputRegressor": {
+                "estimator": "sklearn.linear_model.LinearRegression"
+            }
+        },
+        evaluation=evaluation,
+    )
+    _, machine_out = ModelBuilder(machine).build()
+    expected = metrics_ or [
+        "sklearn.metrics.explained_variance_score",
+        "sklearn.metrics.r2_score",
+        "sklearn.metrics.mean_squared_error",
+        "sklearn.metrics.mean_absolute_error",
+    ]
+    scores = machine_out.metadata.build_metadata.model.cross_validation.scores
+    assert all(
+        m.split(".")[-1].replace("_", "-") in scores for m in expected
+    )
+
+
+@pytest.mark.parametrize(
+    "cv",
+    (
+        {"sklearn.model_selection.TimeSeriesSplit": {
+            "n_splits": 5, "max_train_size": 10}},
+        {"sklearn.model_selection.ShuffleSplit": {
+            "n_splits": 5, "random_state": 0}},
+        None,
+    ),
+)
+def test_n_splits_from_config(cv):
+    """Arbitrary CV splitters configured via evaluation.cv are honored
+    (reference test_n_splits_from_config) — fold count in metadata."""
+    evaluation = {"cv_mode": "cross_val_only"}
+    if cv:
+        evaluation["cv"] = cv
+    machine = make_machine(
+        model={
+            "sklearn.multioutput.MultiOutputRegressor": {
+                "estimator": "sklearn.linear_model.LinearRegression"
+            }
+        },
+        evaluation=evaluation,
+    )
+    _, machine_out = ModelBuilder(machine).build()
+    scores = machine_out.metadata.build_metadata.model.cross_validation.scores
+    n_expected = 5 if cv else 3
+    folds = [k for k in scores["r2-score"] if k.startswith("fold-")]
+    # fold-mean/std/max/min + fold-1..n
+    assert len(folds) == n_expected + 4
