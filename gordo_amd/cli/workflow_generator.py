@@ -168,6 +168,10 @@ def workflow_generate(
         "n_gpus": n_gpus,
         "service_account": runtime.get("service_account", "gordo-workflow"),
         "models_pvc": runtime.get("models_pvc", "gordo-models"),
+        # the server serves every model in the project, independent of how
+        # builds were chunked into workflows (reference
+        # test_workflow_generator.py::test_expected_models_in_workflow)
+        "expected_models_json": json.dumps([m.name for m in norm.machines]),
     }
 
     template = load_workflow_template(workflow_template)

@@ -333,3 +333,89 @@ spec:
     )
     assert result.exit_code == 0, result.output
     assert "build-crd-m" in result.output
+
+
+def _wf_config(tmp_path, n=1, globals_yaml=""):
+    machines = "\n".join(
+        f"""  - name: wf-many-{i}
+    dataset: |
+      type: RandomDataset
+      tag_list: [a, b]
+      train_start_date: '2019-01-01T00:00:00+00:00'
+      train_end_date: '2019-01-02T00:00:00+00:00'
+    model: |
+      sklearn.decomposition.PCA:
+        n_components: 2"""
+        for i in range(n)
+    )
+    cfg = tmp_path / "cfg.yml"
+    cfg.write_text(f"machines:\n{machines}\n{globals_yaml}")
+    return cfg
+
+
+def test_workflow_runtime_resources_override(tmp_path):
+    """globals.runtime.builder/server resources land in the manifest, with
+    limits bumped to >= requests (reference
+    test_workflow_generator.py::test_runtime_overrides_builder)."""
+    cfg = _wf_config(
+        tmp_path,
+        globals_yaml="""
+globals:
+  runtime: |
+    builder:
+      resources:
+        requests: {memory: 9999, cpu: 7777}
+        limits: {memory: 1, cpu: 8888}
+    server:
+      resources:
+        requests: {memory: 6666}
+        limits: {memory: 6667}
+""",
+    )
+    result = CliRunner().invoke(
+        gordo,
+        ["workflow", "generate", "--machine-config", str(cfg),
+         "--project-name", "wf-res"],
+    )
+    assert result.exit_code == 0, result.output
+    out = result.output
+    assert "9999" in out and "7777" in out
+    assert "6666" in out and "6667" in out
+    # requests.memory 9999 > limits.memory 1 → limit fixed up to 9999
+    docs = list(yaml.safe_load_all(out))
+    dumped = yaml.safe_dump(docs[0])
+    assert "memory: 1\n" not in dumped
+
+
+def test_workflow_model_names_embedded_and_replicas(tmp_path):
+    cfg = _wf_config(tmp_path, n=3)
+    result = CliRunner().invoke(
+        gordo,
+        ["workflow", "generate", "--machine-config", str(cfg),
+         "--project-name", "wf-embed", "--n-servers", "3"],
+    )
+    assert result.exit_code == 0, result.output
+    doc = list(yaml.safe_load_all(result.output))[0]
+    tasks = {t["name"] for t in doc["spec"]["templates"][0]["dag"]["tasks"]}
+    for i in range(3):
+        assert f"build-wf-many-{i}" in tasks
+    assert "replicas: 3" in result.output
+
+
+def test_workflow_expected_models_env(tmp_path):
+    """Server deployment carries EXPECTED_MODELS with EVERY machine name,
+    even when builds are split across workflows."""
+    cfg = _wf_config(tmp_path, n=5)
+    result = CliRunner().invoke(
+        gordo,
+        ["workflow", "generate", "--machine-config", str(cfg),
+         "--project-name", "wf-exp", "--split-workflows", "2"],
+    )
+    assert result.exit_code == 0, result.output
+    import re
+
+    m = re.search(r"EXPECTED_MODELS\s*\n\s*value: '(\[[^']*\])'",
+                  result.output)
+    assert m, result.output[:2000]
+    names = json.loads(m.group(1))
+    assert names == [f"wf-many-{i}" for i in range(5)]
