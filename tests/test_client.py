@@ -104,3 +104,49 @@ def test_forward_to_disk(tmp_path, client, gordo_name):
     fwd.forward_predictions(frame, name)
     stored = pd.read_parquet(tmp_path / f"{name}.parquet")
     assert len(stored) == len(frame)
+
+
+def test_status_code_exception_mapping(client, gordo_name):
+    """4xx statuses map to typed exceptions (reference
+    tests/gordo/client/test_client.py::test__handle_response_errors)."""
+    from gordo_amd.client.client import (
+        BadGordoRequest,
+        HttpUnprocessableEntity,
+        NotFound,
+        ResourceGone,
+    )
+
+    with pytest.raises(HttpUnprocessableEntity):
+        client._request("GET", client._url("bad_name!/metadata"))
+    with pytest.raises(NotFound):
+        client._request("GET", client._url("no-such-model/metadata"))
+    with pytest.raises(ResourceGone):
+        client._request(
+            "GET", client._url("models"), params={"revision": "123"}
+        )
+    with pytest.raises(BadGordoRequest):
+        # POST prediction without an X payload → 400
+        client._request(
+            "POST", client._url(f"{gordo_name}/prediction"), json={}
+        )
+
+
+def test_unknown_revision_raises(flask_app, gordo_project):
+    from gordo_amd.client.client import ResourceGone
+
+    bad = Client(
+        project=gordo_project,
+        host="server",
+        port=80,
+        scheme="http",
+        session=FlaskSession(flask_app),
+        revision="123",
+    )
+    with pytest.raises(ResourceGone):
+        bad.get_machine_names()
+
+
+def test_metadata_specific_targets(client, gordo_name, second_gordo_name):
+    md = client.get_metadata(targets=[gordo_name])
+    assert set(md) == {gordo_name}
+    assert md[gordo_name]["name"] == gordo_name
