@@ -138,3 +138,62 @@ def test_dataset_requires_timezone():
             train_end_date="2019-01-02T00:00:00",
             tag_list=["a"],
         )
+
+
+def test_load_globals_config_nested_yaml():
+    """Every machine-level field given as a YAML string parses (reference
+    tests/gordo/machine/test_loader.py::test_load_globals_config)."""
+    import yaml as _yaml
+
+    from gordo_amd.machine.loader import (
+        load_globals_config,
+        load_machine_config,
+    )
+
+    config = _yaml.safe_load(
+        """
+dataset: |
+  tags:
+  - tag1
+  - tag2
+  train_end_date: '2022-01-15T00:00:00+00:00'
+  train_start_date: '2021-12-25T00:00:00+00:00'
+model: |
+  gordo.machine.model.anomaly.diff.DiffBasedAnomalyDetector:
+    base_estimator:
+      sklearn.pipeline.Pipeline:
+        steps:
+          - sklearn.preprocessing.MinMaxScaler
+          - gordo.machine.model.models.KerasAutoEncoder:
+              kind: feedforward_hourglass
+runtime: |
+  builder:
+    resources:
+      requests:
+        memory: 1000
+metadata: |
+  key1: value1
+evaluation: |
+  cv_mode: no_cv
+"""
+    )
+    out = load_globals_config(dict(config))
+    assert out["dataset"]["tags"] == ["tag1", "tag2"]
+    assert out["runtime"]["builder"]["resources"]["requests"]["memory"] == 1000
+    assert out["metadata"] == {"key1": "value1"}
+    assert out["evaluation"] == {"cv_mode": "no_cv"}
+    model_key = next(iter(out["model"]))
+    assert model_key.endswith("DiffBasedAnomalyDetector")
+
+    # machine config: same parsing + name survives
+    mc = load_machine_config({"name": "m1", **config})
+    assert mc["name"] == "m1"
+    assert mc["evaluation"] == {"cv_mode": "no_cv"}
+
+
+def test_load_globals_config_bad_yaml_raises():
+    from gordo_amd.machine.loader import load_globals_config
+
+    with pytest.raises(Exception) as exc:
+        load_globals_config({"model": ": not :\n - valid yaml ["})
+    assert "model" in str(exc.value)

@@ -233,3 +233,21 @@ def test_engine_spec_roundtrip():
     spec2 = ModelSpec.from_dict(d)
     assert spec2.arch_key() == spec.arch_key()
     assert spec2.adam_params["lr"] == 0.001
+
+
+def test_disk_registry_overwrite_and_delete(tmp_path):
+    from gordo_amd.util import disk_registry
+
+    # get from a registry dir that doesn't exist → None, no creation
+    missing = tmp_path / "never-created"
+    assert disk_registry.get_value(missing, "k") is None
+    assert not missing.exists()
+
+    reg = tmp_path / "reg"
+    disk_registry.write_key(reg, "k", "v1")
+    disk_registry.write_key(reg, "k", "v2")  # silent overwrite
+    assert disk_registry.get_value(reg, "k") == "v2"
+
+    assert disk_registry.delete_value(reg, "k") is True
+    assert disk_registry.get_value(reg, "k") is None
+    assert disk_registry.delete_value(reg, "k") is False  # double delete
