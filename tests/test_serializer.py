@@ -124,3 +124,67 @@ def test_dumps_loads_bytes():
 def test_from_definition_rejects_bad_path():
     with pytest.raises((ImportError, ValueError)):
         serializer.from_definition({"no.such.module.Klass": {}})
+
+
+@pytest.mark.parametrize(
+    "definition",
+    [
+        """
+sklearn.multioutput.MultiOutputRegressor:
+  estimator: sklearn.ensemble.RandomForestRegressor
+""",
+        """
+sklearn.multioutput.MultiOutputRegressor:
+  estimator:
+    sklearn.ensemble.RandomForestRegressor:
+      n_estimators: 20
+""",
+        """
+sklearn.multioutput.MultiOutputRegressor:
+  estimator:
+    sklearn.pipeline.Pipeline:
+      steps:
+        - sklearn.ensemble.RandomForestRegressor:
+            n_estimators: 20
+""",
+        """
+sklearn.multioutput.MultiOutputRegressor:
+  estimator:
+    sklearn.pipeline.Pipeline:
+      steps:
+        - sklearn.cluster.FeatureAgglomeration:
+            n_clusters: 2
+            pooling_func: numpy.mean
+        - sklearn.linear_model.LinearRegression
+""",
+    ],
+)
+def test_models_as_parameters(definition):
+    """Estimator-valued params in every spelling: bare class string,
+    class-with-kwargs dict, nested Pipeline, callable param
+    (numpy.mean) — reference test_serializer_from_definition.py:27-70."""
+    import numpy as np
+    from sklearn.multioutput import MultiOutputRegressor
+
+    import yaml
+
+    model = serializer.from_definition(yaml.safe_load(definition))
+    assert isinstance(model, MultiOutputRegressor)
+    X, y = np.random.random((10, 10)), np.random.random((10, 2))
+    model.fit(X, y)
+    model.predict(X)
+
+
+def test_into_definition_captures_kwargs():
+    """into_definition records non-default init kwargs so the definition
+    round-trips (reference test_captures_kwarg_to_init)."""
+    from sklearn.decomposition import PCA
+    from sklearn.pipeline import Pipeline
+
+    pipe = Pipeline([("pca", PCA(n_components=3, whiten=True))])
+    definition = serializer.into_definition(pipe)
+    step = definition["sklearn.pipeline.Pipeline"]["steps"][0]
+    key = next(k for k in step if k.endswith(".PCA"))
+    params = step[key]
+    assert params["n_components"] == 3
+    assert params["whiten"] is True
