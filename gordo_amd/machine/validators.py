@@ -132,11 +132,18 @@ def fix_resource_limits(resources: dict) -> dict:
     for key in ("cpu", "memory"):
         req, lim = requests.get(key), limits.get(key)
         if req is not None and lim is not None:
+            # values must be plain ints (the unit scale is fixed by the
+            # config schema); reference raises rather than guessing at
+            # k8s quantity strings (validators.py:173 there)
             try:
-                if int(req) > int(lim):
-                    limits[key] = req
-            except (TypeError, ValueError):
-                pass
+                req_i, lim_i = int(req), int(lim)
+            except (TypeError, ValueError) as e:
+                raise ValueError(
+                    f"resource {key!r} values must be ints; got "
+                    f"requests={req!r} limits={lim!r}"
+                ) from e
+            if req_i > lim_i:
+                limits[key] = req_i
     if limits:
         resources["limits"] = limits
     return resources

@@ -272,3 +272,64 @@ def test_run_cmd():
     run_cmd([sys.executable, "-c", "print('ok')"])
     with pytest.raises(subprocess.CalledProcessError):
         run_cmd([sys.executable, "-c", "raise SystemExit(3)"])
+
+
+@pytest.mark.parametrize(
+    "expect_multi_lvl, data",
+    [
+        (False, {"col1": [0, 1, 2, 3], "col2": [0, 1, 2, 3]}),
+        (True, {("ft1", "col1"): [0, 1], ("ft1", "col2"): [0, 1]}),
+        (True, {"ft1": {"col1": [0, 1, 2]}, "ft2": {"col1": [0, 1, 2]}}),
+        (False, [[0, 1, 2], [0, 1, 2]]),
+    ],
+)
+def test_dataframe_from_dict_shapes(expect_multi_lvl, data):
+    """Raw payload shapes map deterministically to flat vs MultiIndex
+    columns (reference tests/gordo/server/test_utils.py:78-88)."""
+    df = server_utils.dataframe_from_dict(data)
+    assert isinstance(df.columns, pd.MultiIndex) == expect_multi_lvl
+
+
+def test_multilevel_roundtrip():
+    df = pd.DataFrame(
+        np.random.default_rng(0).random((10, 4)),
+        columns=pd.MultiIndex.from_product((("f1", "f2"), ("c1", "c2"))),
+        index=pd.date_range("2016-01-01", "2016-02-01", periods=10),
+    )
+    c = server_utils.dataframe_from_dict(server_utils.dataframe_to_dict(df))
+    assert isinstance(c.columns, pd.MultiIndex)
+    assert df.columns.tolist() == c.columns.tolist()
+    assert df.index.tolist() == c.index.tolist()
+    np.testing.assert_allclose(df.values, c.values)
+    # the input frame's own DatetimeIndex is not mutated
+    assert isinstance(df.index, pd.DatetimeIndex)
+
+
+@pytest.mark.parametrize(
+    "index",
+    (
+        list(range(10)),
+        [str(i) for i in (3, 1, 4, 0, 9, 2, 8, 5, 7, 6)],
+        pd.date_range("2020-01-01", "2020-01-02", periods=10),
+        pd.date_range("2020-01-01", "2020-01-02", periods=10).astype(
+            str
+        ).tolist()[::-1],
+    ),
+)
+def test_dataframe_from_dict_ordering(index):
+    """from_dict parses string indexes as datetimes or ints and sorts
+    ascending (reference test_utils.py:109-131)."""
+    import dateutil.parser
+
+    df = pd.DataFrame(np.random.default_rng(1).random((10, 5)))
+    df.index = index
+    original = df.copy()
+    if isinstance(original.index[0], str):
+        try:
+            original.index = original.index.map(dateutil.parser.isoparse)
+        except ValueError:
+            original.index = original.index.map(int)
+    original = original.sort_index()
+    out = server_utils.dataframe_from_dict(server_utils.dataframe_to_dict(df))
+    assert list(out.index) == list(original.index)
+    np.testing.assert_allclose(out.values, original.values)

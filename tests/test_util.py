@@ -251,3 +251,23 @@ def test_disk_registry_overwrite_and_delete(tmp_path):
     assert disk_registry.delete_value(reg, "k") is True
     assert disk_registry.get_value(reg, "k") is None
     assert disk_registry.delete_value(reg, "k") is False  # double delete
+
+
+def test_fix_resource_limits_edges():
+    from gordo_amd.machine.validators import fix_resource_limits
+
+    # works without requests / without limits
+    assert fix_resource_limits({"limits": {"cpu": 4}})["limits"]["cpu"] == 4
+    assert (
+        fix_resource_limits({"requests": {"cpu": 5}})["requests"]["cpu"] == 5
+    )
+    # untouched when limits already >= requests
+    res = fix_resource_limits(
+        {"requests": {"cpu": 5}, "limits": {"cpu": 6}}
+    )
+    assert res["limits"]["cpu"] == 6
+    # non-int values are an error, not a silent skip
+    with pytest.raises(ValueError):
+        fix_resource_limits(
+            {"requests": {"memory": "1M"}, "limits": {"memory": 3}}
+        )
