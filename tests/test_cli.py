@@ -419,3 +419,39 @@ def test_workflow_expected_models_env(tmp_path):
     assert m, result.output[:2000]
     names = json.loads(m.group(1))
     assert names == [f"wf-many-{i}" for i in range(5)]
+
+
+def test_normalized_config_docker_images():
+    """Default unified image + per-section custom overrides survive the
+    globals merge (reference test_normalized_config.py)."""
+    from gordo_amd.workflow import NormalizedConfig
+
+    norm = NormalizedConfig(
+        {"machines": [], "globals": {"runtime": {}}}, "test"
+    )
+    rt = norm.globals["runtime"]
+    for section in (
+        "deployer", "server", "prometheus_metrics_server", "builder",
+        "client",
+    ):
+        assert rt[section]["image"] == "gordo-base"
+
+    norm2 = NormalizedConfig(
+        {
+            "machines": [],
+            "globals": {
+                "runtime": {
+                    "deployer": {"image": "my-deployer"},
+                    "server": {"image": "my-server"},
+                    "builder": {"image": "my-builder"},
+                }
+            },
+        },
+        "test",
+    )
+    rt2 = norm2.globals["runtime"]
+    assert rt2["deployer"]["image"] == "my-deployer"
+    assert rt2["server"]["image"] == "my-server"
+    assert rt2["builder"]["image"] == "my-builder"
+    # untouched sections keep the default
+    assert rt2["client"]["image"] == "gordo-base"
