@@ -198,3 +198,87 @@ spec:
     clone = KerasRawModelRegressor(**params)
     clone.fit(X, y)
     assert clone.predict(X).shape == (30, 1)
+
+
+def test_windower_exact_content_lb3_loah0():
+    """Exact window/target values (reference test_model.py:239-260)."""
+    from gordo_amd.machine.model.models import create_keras_timeseriesgenerator
+
+    X = np.array([[0, 1], [2, 3], [4, 5], [6, 7], [8, 9]])
+    gen = create_keras_timeseriesgenerator(
+        X, X.copy(), batch_size=2, lookback_window=3, lookahead=0
+    )
+    assert gen[0][0].tolist() == [
+        [[0, 1], [2, 3], [4, 5]], [[2, 3], [4, 5], [6, 7]]
+    ]
+    assert gen[0][1].tolist() == [[4, 5], [6, 7]]
+    assert gen[1][0].tolist() == [[[4, 5], [6, 7], [8, 9]]]
+    assert gen[1][1].tolist() == [[8, 9]]
+
+
+def test_windower_exact_content_lb2_loah1():
+    from gordo_amd.machine.model.models import create_keras_timeseriesgenerator
+
+    X = np.array([[0, 1], [2, 3], [4, 5], [6, 7], [8, 9]])
+    gen = create_keras_timeseriesgenerator(
+        X, X.copy(), batch_size=2, lookback_window=2, lookahead=1
+    )
+    assert gen[0][0].tolist() == [[[0, 1], [2, 3]], [[2, 3], [4, 5]]]
+    assert gen[0][1].tolist() == [[4, 5], [6, 7]]
+    assert gen[1][0].tolist() == [[[4, 5], [6, 7]]]
+    assert gen[1][1].tolist() == [[8, 9]]
+
+
+def test_windower_exact_content_lb3_loah2():
+    from gordo_amd.machine.model.models import create_keras_timeseriesgenerator
+
+    X = np.array([[0, 1], [2, 3], [4, 5], [6, 7], [8, 9]])
+    gen = create_keras_timeseriesgenerator(
+        X, X.copy(), batch_size=2, lookback_window=3, lookahead=2
+    )
+    assert gen[0][0].tolist() == [[[0, 1], [2, 3], [4, 5]]]
+    assert gen[0][1].tolist() == [[8, 9]]
+
+
+def test_autoencoder_accepts_1d_array():
+    """1-D inputs are treated as a single feature column (reference
+    test_keras_ae_reshapes_array / test_keras_forecast_reshapes_array)."""
+    from gordo_amd.machine.model.models import (
+        KerasAutoEncoder,
+        KerasLSTMForecast,
+    )
+
+    x = np.random.RandomState(0).random(100)
+    model = KerasAutoEncoder(kind="feedforward_hourglass", epochs=1)
+    model.fit(x)
+    out = model.predict(x)
+    assert out.shape[0] == 100
+
+    f = KerasLSTMForecast(kind="lstm_symmetric", lookback_window=5, epochs=1)
+    f.fit(x)
+    assert f.predict(x).shape[0] == 100 - 5
+
+
+def test_hourglass_validation_raises():
+    from gordo_amd.machine.model.factories.utils import hourglass_calc_dims
+
+    with pytest.raises(ValueError):
+        hourglass_calc_dims(compression_factor=2, encoding_layers=2, n_features=10)
+    with pytest.raises(ValueError):
+        hourglass_calc_dims(compression_factor=-1, encoding_layers=2, n_features=10)
+    with pytest.raises(ValueError):
+        hourglass_calc_dims(compression_factor=0.5, encoding_layers=0, n_features=10)
+
+
+def test_hourglass_compression_factor_dims():
+    """compression_factor sweep (reference
+    test_feedforward_hourglass_compression_factors)."""
+    from gordo_amd.machine.model.factories.utils import hourglass_calc_dims
+
+    assert tuple(hourglass_calc_dims(0.2, 3, 10)) == (7, 5, 2)
+    assert tuple(hourglass_calc_dims(0.5, 2, 100)) == (75, 50)
+    assert tuple(hourglass_calc_dims(1.0, 3, 10)) == (10, 10, 10)
+    # the reference's large-n case: smallest layer floors at 1 node
+    assert tuple(hourglass_calc_dims(0.0, 3, 100000)) == (66667, 33334, 1)
+    # floor of 1 dimension even at compression 0
+    assert hourglass_calc_dims(0.0, 2, 10)[-1] >= 1
