@@ -109,6 +109,17 @@ class DiffBasedAnomalyDetector(AnomalyDetectorBase):
             and self.smooth_aggregate_threshold_ is not None
         ):
             metadata["smooth-aggregate-threshold"] = self.smooth_aggregate_threshold_
+        if (
+            getattr(self, "smooth_feature_thresholds_per_fold_", None) is not None
+            and len(self.smooth_feature_thresholds_per_fold_) > 0
+        ):
+            metadata["smooth-feature-thresholds-per-fold"] = (
+                self.smooth_feature_thresholds_per_fold_.to_dict()
+            )
+        if getattr(self, "smooth_aggregate_thresholds_per_fold_", None):
+            metadata["smooth-aggregate-thresholds-per-fold"] = (
+                self.smooth_aggregate_thresholds_per_fold_
+            )
         if isinstance(self.base_estimator, GordoBase):
             metadata.update(self.base_estimator.get_metadata())
         else:

@@ -158,3 +158,66 @@ def test_get_params_roundtrip():
     assert params["smoothing_method"] == "smm"
     det2 = DiffBasedAnomalyDetector(**params)
     assert det2.window == 12
+
+
+@pytest.mark.parametrize("mode", ["tscv", "tscv_win", "kfcv"])
+def test_diff_detector_metadata_lifecycle(mode):
+    """get_metadata keys before and after cross_validate (reference
+    test_anomaly_detectors.py::test_diff_detector_get_metadata)."""
+    import pandas as pd
+    from sklearn.linear_model import LinearRegression
+    from sklearn.multioutput import MultiOutputRegressor
+
+    from gordo_amd.machine.model.anomaly.diff import (
+        DiffBasedAnomalyDetector,
+        DiffBasedKFCVAnomalyDetector,
+    )
+
+    X = pd.DataFrame(np.random.RandomState(0).random((200, 5)))
+    y = pd.DataFrame(np.random.RandomState(1).random((200, 2)))
+    base = MultiOutputRegressor(LinearRegression())
+    if mode == "tscv":
+        model = DiffBasedAnomalyDetector(base_estimator=base)
+    elif mode == "tscv_win":
+        model = DiffBasedAnomalyDetector(base_estimator=base, window=144)
+    else:
+        model = DiffBasedKFCVAnomalyDetector(base_estimator=base)
+
+    md = model.get_metadata()
+    assert "base_estimator" in md and "scaler" in md and "shuffle" in md
+    for key in (
+        "feature-thresholds", "aggregate-threshold",
+        "feature-thresholds-per-fold", "aggregate-thresholds-per-fold",
+    ):
+        assert key not in md
+
+    # thresholds appear only after the detector's own cross_validate
+    assert not hasattr(model, "feature_thresholds_")
+    model.fit(X, y)
+    assert not hasattr(model, "feature_thresholds_")
+    model.cross_validate(X=X, y=y)
+    md = model.get_metadata()
+
+    assert isinstance(md["feature-thresholds"], list)
+    assert len(md["feature-thresholds"]) == 2
+    assert "aggregate-threshold" in md
+    assert isinstance(model.feature_thresholds_, pd.Series)
+    assert model.feature_thresholds_.notna().all()
+
+    if mode != "kfcv":
+        assert "feature-thresholds-per-fold" in md
+        assert "aggregate-thresholds-per-fold" in md
+        assert isinstance(model.feature_thresholds_per_fold_, pd.DataFrame)
+    if mode != "tscv":
+        assert "window" in md
+        assert "smoothing-method" in md
+    if mode == "tscv_win":
+        for key in (
+            "smooth-feature-thresholds",
+            "smooth-aggregate-threshold",
+            "smooth-feature-thresholds-per-fold",
+            "smooth-aggregate-thresholds-per-fold",
+        ):
+            assert key in md, key
+    if mode == "kfcv":
+        assert "threshold-percentile" in md
