@@ -266,8 +266,23 @@ def run_server_cli(
     )
 
 
+@click.command("run-metrics-server")
+@click.option("--host", type=HostIP(), default="0.0.0.0",
+              envvar="GORDO_METRICS_SERVER_HOST")
+@click.option("--port", type=click.IntRange(1, 65535), default=5000,
+              envvar="GORDO_METRICS_SERVER_PORT")
+def run_metrics_server_cli(host, port):
+    """Run the standalone prometheus /metrics sidecar app (the server
+    pod's second container — reference template :1147-1180 /
+    gordo/server/prometheus/server.py)."""
+    from ..server.prometheus.server import build_app
+
+    build_app().run(host=host, port=port)
+
+
 gordo.add_command(build)
 gordo.add_command(run_server_cli)
+gordo.add_command(run_metrics_server_cli)
 
 from .workflow_generator import workflow_cli  # noqa: E402
 

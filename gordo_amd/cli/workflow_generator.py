@@ -172,6 +172,13 @@ def workflow_generate(
         # builds were chunked into workflows (reference
         # test_workflow_generator.py::test_expected_models_in_workflow)
         "expected_models_json": json.dumps([m.name for m in norm.machines]),
+        # pod/container security contexts from runtime globals (validated
+        # by the pydantic schemas via NormalizedConfig.prepare_runtime)
+        "pod_security_context": runtime.get("pod_security_context"),
+        "security_context": runtime.get("security_context"),
+        "prometheus_sidecar": bool(
+            runtime.get("prometheus_metrics_server", {}).get("enable", True)
+        ),
     }
 
     template = load_workflow_template(workflow_template)

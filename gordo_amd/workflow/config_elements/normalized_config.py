@@ -14,7 +14,13 @@ from typing import Any, Dict, List, Optional, Type
 
 from pydantic import TypeAdapter
 
-from .schemas import BuilderPodRuntime, PodRuntime, Volume
+from .schemas import (
+    BuilderPodRuntime,
+    PodRuntime,
+    PodSecurityContext,
+    SecurityContext,
+    Volume,
+)
 from ..workflow_generator.helpers import patch_dict
 from ...machine import Machine, load_globals_config, load_machine_config
 from ...machine.validators import fix_runtime
@@ -146,6 +152,16 @@ class NormalizedConfig:
                 runtime[name] = pod_runtime.model_dump(exclude_none=True)
 
         prepare_pod_runtime("builder", BuilderPodRuntime)
+        if "pod_security_context" in runtime:
+            psc = TypeAdapter(PodSecurityContext).validate_python(
+                runtime["pod_security_context"]
+            )
+            runtime["pod_security_context"] = psc.model_dump(exclude_none=True)
+        if "security_context" in runtime:
+            sc = TypeAdapter(SecurityContext).validate_python(
+                runtime["security_context"]
+            )
+            runtime["security_context"] = sc.model_dump(exclude_none=True)
         if "volumes" in runtime:
             volumes = TypeAdapter(List[Volume]).validate_python(runtime["volumes"])
             runtime["volumes"] = [

@@ -455,3 +455,65 @@ def test_normalized_config_docker_images():
     assert rt2["builder"]["image"] == "my-builder"
     # untouched sections keep the default
     assert rt2["client"]["image"] == "gordo-base"
+
+
+def test_workflow_security_contexts_and_sidecar(tmp_path):
+    """runtime.pod_security_context / security_context validate through
+    the pydantic schemas and render into the server manifest; the
+    prometheus metrics sidecar container is emitted (reference
+    test_pod_security_context / test_security_context /
+    prometheus sidecar template:1147-1180)."""
+    cfg = _wf_config(
+        tmp_path,
+        globals_yaml="""
+globals:
+  runtime: |
+    pod_security_context:
+      runAsUser: 1000
+      fsGroup: 2000
+    security_context:
+      runAsNonRoot: true
+""",
+    )
+    result = CliRunner().invoke(
+        gordo,
+        ["workflow", "generate", "--machine-config", str(cfg),
+         "--project-name", "wf-sec"],
+    )
+    assert result.exit_code == 0, result.output
+    out = result.output
+    assert "runAsUser: 1000" in out
+    assert "fsGroup: 2000" in out
+    assert "runAsNonRoot: true" in out
+    assert "run-metrics-server" in out  # sidecar container
+    assert "containerPort: 5000" in out
+    list(yaml.safe_load_all(out))  # manifest stays valid YAML
+
+
+def test_workflow_bad_security_context_rejected(tmp_path):
+    cfg = _wf_config(
+        tmp_path,
+        globals_yaml="""
+globals:
+  runtime: |
+    pod_security_context:
+      runAsUser: not-an-int
+""",
+    )
+    result = CliRunner().invoke(
+        gordo,
+        ["workflow", "generate", "--machine-config", str(cfg),
+         "--project-name", "wf-sec-bad"],
+    )
+    assert result.exit_code != 0
+
+
+def test_run_metrics_server_app():
+    from gordo_amd.server.prometheus.server import build_app
+
+    app = build_app()
+    app.testing = True
+    c = app.test_client()
+    assert c.get("/healthcheck").status_code == 200
+    resp = c.get("/metrics")
+    assert resp.status_code == 200
