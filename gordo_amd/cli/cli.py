@@ -292,5 +292,9 @@ from .fleet import fleet_cli  # noqa: E402
 
 gordo.add_command(fleet_cli)
 
+from .client import client_cli  # noqa: E402
+
+gordo.add_command(client_cli)
+
 if __name__ == "__main__":
     gordo()

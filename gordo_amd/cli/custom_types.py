@@ -69,3 +69,18 @@ def key_value_par(val: str) -> Tuple[str, Any]:
     except yaml.YAMLError:
         parsed = value
     return key, parsed
+
+
+class IsoFormatDateTime(click.ParamType):
+    """ISO-8601 datetime click param (tz-aware), e.g.
+    2020-01-01T00:00:00+00:00."""
+
+    name = "isodatetime"
+
+    def convert(self, value, param, ctx):
+        import dateutil.parser
+
+        try:
+            return dateutil.parser.isoparse(value)
+        except ValueError:
+            self.fail(f"{value!r} is not a valid ISO datetime", param, ctx)

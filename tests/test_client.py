@@ -150,3 +150,98 @@ def test_metadata_specific_targets(client, gordo_name, second_gordo_name):
     md = client.get_metadata(targets=[gordo_name])
     assert set(md) == {gordo_name}
     assert md[gordo_name]["name"] == gordo_name
+
+
+@pytest.fixture
+def cli_env(flask_app, monkeypatch, gordo_project):
+    """Route `gordo client` CLI commands through the in-process server."""
+    from gordo_amd.cli import client as cli_client
+
+    def make(ctx_obj):
+        return Client(
+            project=ctx_obj["project"],
+            host="server",
+            port=80,
+            scheme="http",
+            parallelism=ctx_obj.get("parallelism", 2),
+            metadata=ctx_obj.get("metadata"),
+            session=FlaskSession(flask_app),
+        )
+
+    monkeypatch.setattr(cli_client, "make_client", make)
+    return gordo_project
+
+
+@pytest.mark.parametrize(
+    "args",
+    [
+        ["--help"],
+        ["--project", "p", "predict", "--help"],
+        ["--project", "p", "metadata", "--help"],
+        ["--project", "p", "download-model", "--help"],
+    ],
+)
+def test_client_cli_exists(args):
+    from click.testing import CliRunner
+
+    from gordo_amd.cli.cli import gordo
+
+    out = CliRunner().invoke(gordo, ["client"] + args)
+    assert out.exit_code == 0, out.output
+
+
+def test_client_cli_metadata(cli_env, gordo_name, tmp_path):
+    import json as _json
+
+    from click.testing import CliRunner
+
+    from gordo_amd.cli.cli import gordo
+
+    out = CliRunner().invoke(
+        gordo,
+        ["client", "--project", cli_env, "metadata", "--target", gordo_name],
+    )
+    assert out.exit_code == 0, out.output
+    assert gordo_name in out.output
+
+    out_file = tmp_path / "metadata.json"
+    out = CliRunner().invoke(
+        gordo,
+        ["client", "--project", cli_env, "metadata",
+         "--output-file", str(out_file), "--target", gordo_name],
+    )
+    assert out.exit_code == 0, out.output
+    assert gordo_name in _json.loads(out_file.read_text())
+
+
+def test_client_cli_download_model(cli_env, gordo_name, tmp_path):
+    from click.testing import CliRunner
+    from sklearn.base import BaseEstimator
+
+    from gordo_amd import serializer
+    from gordo_amd.cli.cli import gordo
+
+    out = CliRunner().invoke(
+        gordo,
+        ["client", "--project", cli_env, "download-model", str(tmp_path),
+         "--target", gordo_name],
+    )
+    assert out.exit_code == 0, out.output
+    model = serializer.load(str(tmp_path / gordo_name))
+    assert isinstance(model, BaseEstimator)
+
+
+def test_client_cli_predict(cli_env, gordo_name, tmp_path):
+    from click.testing import CliRunner
+
+    from gordo_amd.cli.cli import gordo
+
+    out = CliRunner().invoke(
+        gordo,
+        ["client", "--project", cli_env, "predict",
+         "2019-01-01T00:00:00Z", "2019-01-01T06:00:00Z",
+         "--target", gordo_name, "--output-dir", str(tmp_path)],
+    )
+    assert out.exit_code == 0, out.output
+    saved = list(tmp_path.glob("*.csv.gz"))
+    assert len(saved) == 1 and gordo_name in saved[0].name
