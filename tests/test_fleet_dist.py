@@ -51,7 +51,9 @@ def test_fleet_build_two_ranks(tmp_path):
         [
             sys.executable, "-m", "torch.distributed.run",
             "--nnodes=1", "--nproc-per-node=2",
-            "--master-addr", "127.0.0.1", "--master-port", "29531",
+            # standalone picks a free rendezvous port: a fixed port can
+            # collide with the driver's own torchrun on a shared GPU box
+            "--standalone", "--local-addr", "127.0.0.1",
             "-m", "gordo_amd.cli.fleet_worker",
         ],
         env=env,
