@@ -197,3 +197,35 @@ def test_load_globals_config_bad_yaml_raises():
     with pytest.raises(Exception) as exc:
         load_globals_config({"model": ": not :\n - valid yaml ["})
     assert "model" in str(exc.value)
+
+
+def test_machine_report_runs_reporters(tmp_path):
+    """machine.report() instantiates & runs every runtime reporter
+    (reference tests/gordo/machine/test_machine.py::test_builder_with_reporter,
+    without the live postgres)."""
+    from gordo_amd.machine import Machine
+    from gordo_amd.reporters.base import BaseReporter
+
+    reported = []
+
+    class _CollectingReporter(BaseReporter):
+        def report(self, machine):
+            reported.append(machine.name)
+
+    m = Machine.from_config(
+        {
+            "name": "report-m",
+            "model": {"sklearn.decomposition.PCA": {"n_components": 2}},
+            "dataset": {
+                "type": "RandomDataset",
+                "tag_list": ["a", "b", "c"],
+                "train_start_date": "2019-01-01T00:00:00Z",
+                "train_end_date": "2019-01-02T00:00:00Z",
+            },
+            "runtime": {"reporters": []},
+        },
+        project_name="p",
+    )
+    m.runtime["reporters"] = [_CollectingReporter()]
+    m.report()
+    assert reported == ["report-m"]
