@@ -63,9 +63,10 @@ def _create_anomaly_response(start_time: Optional[float] = None):
             mimetype="application/octet-stream",
         )
     context: Dict[Any, Any] = {}
-    context["data"] = utils.dataframe_to_dict(anomaly_df)
     context["time-seconds"] = f"{timeit.default_timer() - start_time:.4f}"
-    return make_response(jsonify(context), context.pop("status-code", 200))
+    return utils.frame_json_response(
+        context, anomaly_df, context.pop("status-code", 200)
+    )
 
 
 @anomaly_blueprint.route(

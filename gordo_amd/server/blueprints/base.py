@@ -79,8 +79,9 @@ def post_prediction():
             io.BytesIO(server_utils.dataframe_into_parquet_bytes(data)),
             mimetype="application/octet-stream",
         )
-    context["data"] = server_utils.dataframe_to_dict(data)
-    return make_response((jsonify(context), context.pop("status-code", 200)))
+    return server_utils.frame_json_response(
+        context, data, context.pop("status-code", 200)
+    )
 
 
 @base_blueprint.route(

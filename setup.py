@@ -41,7 +41,27 @@ def _hip_extension():
     return [ext], {"build_ext": BuildExtension}
 
 
+def _fastjson_extension():
+    """Plain C++ (no HIP/torch) response-frame JSON encoder."""
+    try:
+        import pybind11
+        from setuptools import Extension
+    except ImportError:
+        return []
+    return [
+        Extension(
+            name="gordo_amd.server._gordo_fastjson",
+            sources=[os.path.join("gordo_amd", "server", "csrc",
+                                  "fastjson.cpp")],
+            include_dirs=[pybind11.get_include()],
+            extra_compile_args=["-O3", "-std=c++17"],
+            language="c++",
+        )
+    ]
+
+
 ext_modules, cmdclass = _hip_extension()
+ext_modules += _fastjson_extension()
 
 setup(
     name="gordo-amd",

@@ -333,3 +333,57 @@ def test_dataframe_from_dict_ordering(index):
     out = server_utils.dataframe_from_dict(server_utils.dataframe_to_dict(df))
     assert list(out.index) == list(original.index)
     np.testing.assert_allclose(out.values, original.values)
+
+
+def test_fastjson_frame_encoder_equivalence():
+    """The C++ response encoder is byte-identical to
+    json.dumps(dataframe_to_dict(df)) on response-shaped frames,
+    including NaN/inf, integral floats and escaped keys."""
+    fj = pytest.importorskip("gordo_amd.server._gordo_fastjson")
+    import json
+
+    rng = np.random.default_rng(3)
+    df = pd.DataFrame(
+        rng.random((100, 100)),
+        columns=pd.MultiIndex.from_product(
+            (("model-input", "model-output"), [f"t{i}" for i in range(50)])
+        ),
+        index=pd.date_range("2020-01-01", periods=100, freq="10min"),
+    )
+    df.iloc[0, 0] = float("nan")
+    df.iloc[1, 1] = float("inf")
+    df.iloc[2, 2] = -float("inf")
+    df.iloc[3, 3] = 1.0
+    fast = fj.encode_frame(
+        df.index.astype(str).tolist(),
+        [c[0] for c in df.columns],
+        [c[1] for c in df.columns],
+        df.values,
+    )
+    ref = json.dumps(server_utils.dataframe_to_dict(df)).encode()
+    assert fast == ref
+
+    df2 = pd.DataFrame(
+        [[1.5]],
+        columns=pd.MultiIndex.from_tuples([('a"b', "c\\d\te")]),
+        index=["i\n1"],
+    )
+    fast2 = fj.encode_frame(
+        ["i\n1"], ['a"b'], ["c\\d\te"], df2.values
+    )
+    assert fast2 == json.dumps(server_utils.dataframe_to_dict(df2)).encode()
+
+    # non-ascii keys are refused (callers fall back to the python codec)
+    with pytest.raises(ValueError):
+        fj.encode_frame(["ø"], ["a"], ["b"], np.zeros((1, 1)))
+
+
+def test_prediction_json_fast_path_matches(api_client, base_route, X):
+    """The endpoint response through frame_json_response parses to the
+    same structure as the python codec path."""
+    resp = _post_json(api_client, f"{base_route}/prediction", X)
+    assert resp.status_code == 200
+    assert resp.mimetype == "application/json"
+    data = resp.json["data"]
+    frame = server_utils.dataframe_from_dict(data)
+    assert len(frame) == len(X)
