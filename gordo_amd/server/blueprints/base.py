@@ -154,7 +154,14 @@ def get_revision_list(gordo_project: str):
         )
         available_revisions = [g.current_revision]
     return jsonify(
-        {"latest": g.current_revision, "available-revisions": available_revisions}
+        {
+            # latest = what the server was deployed with; revision = the
+            # one this request asked for (reference
+            # test_gordo_server.py::test_list_revisions keys)
+            "latest": g.current_revision,
+            "revision": g.revision,
+            "available-revisions": available_revisions,
+        }
     )
 
 

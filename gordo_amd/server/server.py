@@ -95,6 +95,9 @@ def build_app(
         prometheus_metrics = GordoServerPrometheusMetrics(
             args_labels=[("gordo_project", "project"), ("gordo_name", "model")],
             info={"version": gordo_amd.__version__},
+            # probes shouldn't pollute the request metrics (reference
+            # tests/gordo/server/test_prometheus.py::test_ignore)
+            ignore_paths=["/healthcheck"],
             registry=prometheus_registry,
         )
         prometheus_metrics.prepare_app(app)

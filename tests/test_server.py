@@ -387,3 +387,72 @@ def test_prediction_json_fast_path_matches(api_client, base_route, X):
     data = resp.json["data"]
     frame = server_utils.dataframe_from_dict(data)
     assert len(frame) == len(X)
+
+
+def test_revisions_echo_requested_revision(
+    api_client, base_route, gordo_project, gordo_revision,
+    model_collection_directory,
+):
+    """/revisions carries latest + the revision this request resolved
+    (reference test_gordo_server.py::test_list_revisions keys)."""
+    resp = api_client.get(f"/gordo/v0/{gordo_project}/revisions")
+    assert set(resp.json.keys()) == {
+        "latest", "available-revisions", "revision"
+    }
+    assert resp.json["revision"] == gordo_revision
+
+    # a second on-disk revision can be requested explicitly
+    import shutil
+
+    parent = os.path.dirname(model_collection_directory)
+    other = os.path.join(parent, "1000000000001")
+    shutil.copytree(model_collection_directory, other, dirs_exist_ok=True)
+    try:
+        resp = api_client.get(
+            f"/gordo/v0/{gordo_project}/revisions",
+            query_string={"revision": "1000000000001"},
+        )
+        assert resp.json["revision"] == "1000000000001"
+        assert resp.json["latest"] == gordo_revision
+    finally:
+        shutil.rmtree(other)
+
+
+def test_models_list_missing_collection_dir():
+    from gordo_amd.server.server import build_app
+
+    os.environ["MODEL_COLLECTION_DIR"] = os.path.join(
+        "does", "not", "exist", "1"
+    )
+    app = build_app(config={"ENABLE_PROMETHEUS": False})
+    app.testing = True
+    resp = app.test_client().get("/gordo/v0/any-proj/models")
+    assert resp.status_code == 200
+    assert resp.json["models"] == []
+
+
+def test_prometheus_ignores_healthcheck(model_collection_directory,
+                                        trained_model_directories):
+    from prometheus_client.registry import CollectorRegistry
+
+    from gordo_amd.server.server import build_app
+
+    os.environ["MODEL_COLLECTION_DIR"] = model_collection_directory
+    registry = CollectorRegistry()
+    app = build_app(
+        config={"ENABLE_PROMETHEUS": True, "PROJECT": "prom-proj2"},
+        prometheus_registry=registry,
+    )
+    app.testing = True
+    client = app.test_client()
+    client.get("/healthcheck")
+    client.get("/server-version")
+    # /healthcheck not sampled; /server-version is
+    names = {
+        (s.labels.get("path"),)
+        for metric in registry.collect()
+        for s in metric.samples
+        if metric.name == "gordo_server_requests"
+    }
+    assert ("/healthcheck",) not in names
+    assert ("/server-version",) in names
