@@ -10,7 +10,7 @@ per-machine client pods.
 """
 import json
 import os
-from typing import List, Optional
+from typing import List
 
 import click
 

@@ -10,12 +10,12 @@ waves, deterministic per (tag, seed).
 from __future__ import annotations
 
 import hashlib
-from typing import Iterable, List, Optional, Dict, Any
+from typing import Iterable, Optional, Dict, Any
 
 import numpy as np
 import pandas as pd
 
-from .sensor_tag import SensorTag, normalize_sensor_tag
+from .sensor_tag import SensorTag
 from .import_utils import import_location
 
 

@@ -20,7 +20,7 @@ import pandas as pd
 from .base import GordoBaseDataset
 from .data_providers import DataProvider, RandomDataProvider, load_data_provider
 from .exceptions import ConfigException, InsufficientDataError
-from .sensor_tag import SensorTag, normalize_sensor_tag
+from .sensor_tag import normalize_sensor_tag
 
 logger = logging.getLogger(__name__)
 

@@ -21,13 +21,13 @@ from __future__ import annotations
 import logging
 import math
 import os
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import Dict, List, Optional, Tuple
 
 import numpy as np
 import torch
 
 from .. import ops
-from .spec import LayerSpec, ModelSpec
+from .spec import ModelSpec
 
 logger = logging.getLogger(__name__)
 

@@ -16,7 +16,6 @@ from typing import Any, Dict, List, Optional
 import yaml
 
 from .encoders import MachineJSONEncoder, MachineSafeDumper
-from .loader import load_machine_config
 from .metadata import Metadata
 from .validators import (
     ValidDataset,
@@ -26,7 +25,7 @@ from .validators import (
     ValidUrlString,
 )
 from ..core.base import GordoBaseDataset
-from ..core.sensor_tag import SensorTag, normalize_sensor_tag
+from ..core.sensor_tag import SensorTag
 
 
 class Machine:

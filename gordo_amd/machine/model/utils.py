@@ -6,8 +6,8 @@ from __future__ import annotations
 
 import functools
 import logging
-from datetime import datetime, timedelta
-from typing import Iterable, List, Optional, Union
+from datetime import timedelta
+from typing import List, Optional, Union
 
 import numpy as np
 import pandas as pd

@@ -7,7 +7,6 @@ from __future__ import annotations
 import datetime
 import logging
 import re
-from typing import Any
 
 import pandas as pd
 

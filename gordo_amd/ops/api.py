@@ -9,7 +9,6 @@ the fp32 reference implementations run (the test oracle / CPU lane).
 """
 from __future__ import annotations
 
-import os
 from typing import Optional, Tuple
 
 import torch

@@ -50,7 +50,7 @@ from ..machine.model.anomaly.diff import (
     DiffBasedAnomalyDetector,
     DiffBasedKFCVAnomalyDetector,
 )
-from ..machine.model.models import KerasBaseEstimator, KerasLSTMBaseEstimator
+from ..machine.model.models import KerasBaseEstimator
 from ..util import disk_registry
 
 logger = logging.getLogger(__name__)

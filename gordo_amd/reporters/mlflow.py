@@ -12,7 +12,6 @@ from __future__ import annotations
 
 import logging
 from contextlib import contextmanager
-from datetime import datetime
 from typing import Any, Dict, List, Tuple
 
 from .base import BaseReporter

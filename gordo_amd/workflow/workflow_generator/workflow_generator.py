@@ -9,13 +9,13 @@ from __future__ import annotations
 
 import io
 import os
-from typing import Any, Dict, Optional, Union
+from typing import Any, Dict, Union
 
 import dateutil.parser
 import jinja2
 import yaml
 
-from ...util.version import GordoRelease, GordoSHA, GordoSpecial, parse_version
+from ...util.version import GordoRelease, GordoSHA, parse_version
 
 
 def _timestamp_constructor(loader, node):
