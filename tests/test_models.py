@@ -385,3 +385,18 @@ def test_imputer_from_definition(config_str):
     if hasattr(obj, "steps"):
         obj = obj.steps[-1][1]
     assert isinstance(obj, InfImputer)
+
+
+def test_metric_wrapper_scaler_equalizes():
+    from sklearn.metrics import mean_squared_error
+    from sklearn.preprocessing import MinMaxScaler
+
+    from gordo_amd.machine.model.utils import metric_wrapper
+
+    y = np.array([[1, 1], [2, 2], [3, 3], [4, 4], [5, 5]]) * [1, 100]
+    noscaler = metric_wrapper(mean_squared_error)
+    assert not np.isclose(noscaler(y, y * [0.8, 1]), noscaler(y, y * [1, 0.8]))
+    scaled = metric_wrapper(
+        mean_squared_error, scaler=MinMaxScaler().fit(y)
+    )
+    assert np.isclose(scaled(y, y * [0.8, 1]), scaled(y, y * [1, 0.8]))
