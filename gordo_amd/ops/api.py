@@ -109,8 +109,22 @@ def adam_step(p, g, m, v, lr, beta1, beta2, eps, step, p_lp=None,
 
 
 def lstm_seq_fwd(xW, Wh):
-    """Fused on-device LSTM forward scan (GPU only, H<=64)."""
-    return _require_hip().lstm_seq_fwd(xW, Wh)
+    """Fused on-device LSTM forward scan (GPU only, H<=64).
+
+    ``GORDO_LSTM_V3=1`` selects the software-pipelined v3 kernel
+    (double-buffered x-gate prefetch) — opt-in until GPU-validated
+    (ROADMAP round-2 lever #1)."""
+    import os as _os
+
+    ext = _require_hip()
+    if _os.environ.get("GORDO_LSTM_V3") == "1":
+        return ext.lstm_seq_fwd_v3(xW, Wh)
+    return ext.lstm_seq_fwd(xW, Wh)
+
+
+def lstm_seq_fwd_v3(xW, Wh):
+    """The pipelined forward scan, directly (for A/B tests)."""
+    return _require_hip().lstm_seq_fwd_v3(xW, Wh)
 
 
 def lstm_seq_bwd(dSeq, gacts, cs, Wh, last_only):

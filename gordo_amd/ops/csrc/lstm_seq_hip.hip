@@ -34,6 +34,12 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(8))) short bf16x8;
 
 DEV_INLINE float lbf2f(bf16 v) { return __bfloat162float(v); }
+DEV_INLINE unsigned short bf16_bits(bf16 v) {
+  return __builtin_bit_cast(unsigned short, v);
+}
+DEV_INLINE bf16 bits_bf16(unsigned short b) {
+  return __builtin_bit_cast(bf16, b);
+}
 DEV_INLINE bf16 lf2bf(float v) { return __float2bfloat16(v); }
 DEV_INLINE float sigmoidf_(float x) { return 1.f / (1.f + __expf(-x)); }
 DEV_INLINE float fast_tanhf_(float x) {
@@ -308,6 +314,171 @@ __global__ __launch_bounds__(256) void lstm_seq_bwd_kernel(
   }
 }
 
+
+// ===========================================================================
+// v3 forward scan — v1 structure plus a software-pipelined double
+// buffer for the x-side gate tile: while the pointwise phase of step t
+// runs (LDS + global stores, no dependence on xW[t+1]), the loads of
+// xW[t+1] are already in flight into registers, so their HBM latency
+// hides under the gate math and the barrier instead of stalling the
+// MFMA epilogue of step t+1. The tile is kept packed two-bf16-per-u32
+// to halve the register footprint (2 buffers x FM x 4 cols x 2 pairs).
+// DORMANT until GPU-validated: dispatched only via GORDO_LSTM_V3=1
+// (ROADMAP round-2 lever #1).
+// ===========================================================================
+
+template <int ROWS>
+__global__ __launch_bounds__(256) void lstm_seq_fwd_v3_kernel(
+    const bf16* __restrict__ xW, const bf16* __restrict__ Wh,
+    bf16* __restrict__ hs, float* __restrict__ cs,
+    bf16* __restrict__ gacts, int B, int T, int H, int ldg) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int H4 = 4 * H;
+  bf16* WhT = reinterpret_cast<bf16*>(smem);              // [H4][LDK]
+  bf16* hS = WhT + (size_t)H4 * LDK;                      // [ROWS][LDK]
+  bf16* gS = hS + (size_t)ROWS * LDK;                     // [ROWS][ldg]
+  float* cS = reinterpret_cast<float*>(gS + (size_t)ROWS * ldg);  // [ROWS][H]
+
+  const int g = blockIdx.x / ((B + ROWS - 1) / ROWS);
+  const int r0 = (blockIdx.x % ((B + ROWS - 1) / ROWS)) * ROWS;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int l15 = lane & 15;
+  const int kslot = lane >> 4;
+
+  const bf16* Whg = Wh + (size_t)g * H * H4;
+  const bf16* xWg = xW + ((size_t)g * B + r0) * T * H4;
+  bf16* hsg = hs + ((size_t)g * B + r0) * T * H;
+  float* csg = cs + ((size_t)g * B + r0) * T * H;
+  bf16* gag = gacts + ((size_t)g * B + r0) * T * H4;
+  const int rows_here = min(ROWS, B - r0);
+
+  for (int i = tid; i < H4 * LDK; i += 256) {
+    int n = i / LDK, k = i % LDK;
+    WhT[i] = (k < H) ? Whg[(size_t)k * H4 + n] : lf2bf(0.f);
+  }
+  for (int i = tid; i < ROWS * LDK; i += 256) hS[i] = lf2bf(0.f);
+  for (int i = tid; i < ROWS * H; i += 256) cS[i] = 0.f;
+  __syncthreads();
+
+  const int wcol0 = wid * 64;
+  const bool wave_active = wcol0 < H4;
+  constexpr int FM = ROWS / 16;
+
+  // packed x-gate tile: [fm][fn][pair] = (row 2p, row 2p+1) as 2x bf16
+  unsigned int xv[FM][4][2];
+
+  // prologue: prefetch t = 0
+  if (wave_active) {
+    #pragma unroll
+    for (int fm = 0; fm < FM; ++fm) {
+      #pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int col = wcol0 + fn * 16 + l15;
+        #pragma unroll
+        for (int p = 0; p < 2; ++p) {
+          unsigned int pk = 0;
+          #pragma unroll
+          for (int q = 0; q < 2; ++q) {
+            int row = fm * 16 + kslot * 4 + p * 2 + q;
+            bf16 v = (col < H4 && row < rows_here)
+                         ? xWg[((size_t)row * T + 0) * H4 + col]
+                         : lf2bf(0.f);
+            pk |= (unsigned int)bf16_bits(v) << (16 * q);
+          }
+          xv[fm][fn][p] = pk;
+        }
+      }
+    }
+  }
+
+  for (int t = 0; t < T; ++t) {
+    if (wave_active) {
+      f32x4 acc[FM][4] = {};
+      for (int kk = 0; kk < H; kk += 32) {
+        #pragma unroll
+        for (int fm = 0; fm < FM; ++fm) {
+          bf16x8 a = *reinterpret_cast<const bf16x8*>(
+              &hS[(fm * 16 + l15) * LDK + kk + kslot * 8]);
+          #pragma unroll
+          for (int fn = 0; fn < 4; ++fn) {
+            int col = wcol0 + fn * 16 + l15;
+            bf16x8 b = *reinterpret_cast<const bf16x8*>(
+                &WhT[(size_t)min(col, H4 - 1) * LDK + kk + kslot * 8]);
+            acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a, b, acc[fm][fn], 0, 0, 0);
+          }
+        }
+      }
+      #pragma unroll
+      for (int fm = 0; fm < FM; ++fm) {
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+          int col = wcol0 + fn * 16 + l15;
+          if (col >= H4) continue;
+          #pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            int row = fm * 16 + kslot * 4 + r;
+            bf16 xb = bits_bf16(
+                (unsigned short)(xv[fm][fn][r >> 1] >> (16 * (r & 1))));
+            gS[row * ldg + col] = lf2bf(acc[fm][fn][r] + lbf2f(xb));
+          }
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- prefetch t+1 BEFORE the pointwise phase: these loads have no
+    // dependence on this step's gate math, so they overlap it ----
+    if (wave_active && t + 1 < T) {
+      #pragma unroll
+      for (int fm = 0; fm < FM; ++fm) {
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+          int col = wcol0 + fn * 16 + l15;
+          #pragma unroll
+          for (int p = 0; p < 2; ++p) {
+            unsigned int pk = 0;
+            #pragma unroll
+            for (int q = 0; q < 2; ++q) {
+              int row = fm * 16 + kslot * 4 + p * 2 + q;
+              bf16 v = (col < H4 && row < rows_here)
+                           ? xWg[((size_t)row * T + (t + 1)) * H4 + col]
+                           : lf2bf(0.f);
+              pk |= (unsigned int)bf16_bits(v) << (16 * q);
+            }
+            xv[fm][fn][p] = pk;
+          }
+        }
+      }
+    }
+
+    for (int e = tid; e < ROWS * H; e += 256) {
+      int row = e / H, hh = e % H;
+      const bf16* grow = &gS[row * ldg];
+      float i_g = sigmoidf_(lbf2f(grow[hh]));
+      float f_g = sigmoidf_(lbf2f(grow[H + hh]));
+      float g_g = fast_tanhf_(lbf2f(grow[2 * H + hh]));
+      float o_g = sigmoidf_(lbf2f(grow[3 * H + hh]));
+      float cc = f_g * cS[row * H + hh] + i_g * g_g;
+      float hv = o_g * fast_tanhf_(cc);
+      cS[row * H + hh] = cc;
+      hS[row * LDK + hh] = lf2bf(hv);
+      if (row < rows_here) {
+        size_t base = ((size_t)row * T + t) * H + hh;
+        hsg[base] = lf2bf(hv);
+        csg[base] = cc;
+        size_t gbase = ((size_t)row * T + t) * H4;
+        gag[gbase + hh] = lf2bf(i_g);
+        gag[gbase + H + hh] = lf2bf(f_g);
+        gag[gbase + 2 * H + hh] = lf2bf(g_g);
+        gag[gbase + 3 * H + hh] = lf2bf(o_g);
+      }
+    }
+    __syncthreads();
+  }
+}
 
 // ===========================================================================
 // v2 scan kernels — require H % 16 == 0 (the engine pads hidden sizes
@@ -596,6 +767,42 @@ std::vector<torch::Tensor> lstm_seq_fwd(torch::Tensor xW, torch::Tensor Wh) {
   else
     hipLaunchKernelGGL(lstm_seq_fwd_kernel<32>, dim3(blocks), dim3(256), lds,
                        stream, (const bf16*)xc.data_ptr(),
+                       (const bf16*)Whc.data_ptr(), (bf16*)hs.data_ptr(),
+                       cs.data_ptr<float>(), (bf16*)gacts.data_ptr(), B, T, H,
+                       ldg);
+  return {hs, cs, gacts};
+}
+
+std::vector<torch::Tensor> lstm_seq_fwd_v3(torch::Tensor xW,
+                                            torch::Tensor Wh) {
+  // identical contract to lstm_seq_fwd, pipelined kernel (always the
+  // barriered layout, never the v2 path: v3 is measured against v1 on
+  // the same shapes)
+  TORCH_CHECK(xW.is_cuda() && xW.dim() == 4, "xW must be [G,B,T,4H] on GPU");
+  auto xc = xW.to(torch::kBFloat16).contiguous();
+  auto Whc = Wh.to(torch::kBFloat16).contiguous();
+  int G = xc.size(0), B = xc.size(1), T = xc.size(2), H4 = xc.size(3);
+  int H = H4 / 4;
+  TORCH_CHECK(H <= 64, "lstm_seq_fwd_v3 supports H <= 64");
+  int ldg = pad_ldg(H4);
+  auto hs = torch::empty({G, B, T, H}, xc.options());
+  auto cs = torch::empty({G, B, T, H}, xc.options().dtype(torch::kFloat32));
+  auto gacts = torch::empty({G, B, T, H4}, xc.options());
+  int rows = pick_rows(G, B);
+  size_t lds = (size_t)H4 * LDK * 2 + (size_t)rows * LDK * 2 +
+               (size_t)rows * ldg * 2 + (size_t)rows * H * 4;
+  TORCH_CHECK(lds <= 160 * 1024, "LDS budget exceeded");
+  int blocks = G * ((B + rows - 1) / rows);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  if (rows == 64)
+    hipLaunchKernelGGL(lstm_seq_fwd_v3_kernel<64>, dim3(blocks), dim3(256),
+                       lds, stream, (const bf16*)xc.data_ptr(),
+                       (const bf16*)Whc.data_ptr(), (bf16*)hs.data_ptr(),
+                       cs.data_ptr<float>(), (bf16*)gacts.data_ptr(), B, T, H,
+                       ldg);
+  else
+    hipLaunchKernelGGL(lstm_seq_fwd_v3_kernel<32>, dim3(blocks), dim3(256),
+                       lds, stream, (const bf16*)xc.data_ptr(),
                        (const bf16*)Whc.data_ptr(), (bf16*)hs.data_ptr(),
                        cs.data_ptr<float>(), (bf16*)gacts.data_ptr(), B, T, H,
                        ldg);

@@ -366,3 +366,27 @@ def test_lstm_seq_v2_vs_reference(G, B, T, H, last_only):
                               to_dev_bf16(Wh), last_only)
     torch.testing.assert_close(got_dG.float().cpu(), dG_ref, rtol=8e-2,
                                atol=4e-2)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("G,B,T,H", [
+    (2, 64, 12, 38),
+    (1, 100, 20, 42),    # edge row tile, odd H
+    (3, 32, 144, 25),    # full lookback
+    (4, 130, 50, 60),    # near the H cap, ragged rows (64 would hit
+                          # the v2 dispatch in lstm_seq_fwd: H%16==0)
+])
+def test_lstm_seq_v3_matches_v1(G, B, T, H):
+    """The pipelined v3 forward scan is numerically identical to the
+    shipping v1 kernel — same layout, same math, only the x-gate tile
+    is double-buffered. Exact equality expected (bit-identical inputs,
+    same op order)."""
+    require_hip()
+    H4 = 4 * H
+    xW = to_dev_bf16(_rand(G, B, T, H4, seed=40))
+    Wh = to_dev_bf16(_rand(G, H, H4, seed=41) * 0.3)
+    v1_hs, v1_cs, v1_ga = ops.lstm_seq_fwd(xW, Wh)
+    v3_hs, v3_cs, v3_ga = ops.lstm_seq_fwd_v3(xW, Wh)
+    assert torch.equal(v1_hs, v3_hs)
+    assert torch.equal(v1_cs, v3_cs)
+    assert torch.equal(v1_ga, v3_ga)
