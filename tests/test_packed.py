@@ -329,3 +329,26 @@ def test_kfcv_machines_quantile_thresholds(tmp_path):
     # quantile thresholds set (scalar aggregate, per-tag series)
     assert hasattr(model, "aggregate_threshold_")
     assert len(model.feature_thresholds_) == 3
+
+
+@pytest.mark.parametrize("w", [6, 7, 144])
+def test_trail_min_max_matches_pandas(w):
+    """The O(n) scipy threshold helper is exactly pandas
+    rolling(w).min().max() for 1-D and 2-D inputs, short series
+    included."""
+    import pandas as pd
+
+    from gordo_amd.parallel.packed_builder import _trail_min_max
+
+    rng = np.random.default_rng(0)
+    a = rng.random(300)
+    A = rng.random((300, 7))
+    assert np.isclose(
+        _trail_min_max(a, w), pd.Series(a).rolling(w).min().max()
+    )
+    np.testing.assert_allclose(
+        _trail_min_max(A, w), pd.DataFrame(A).rolling(w).min().max().values
+    )
+    # shorter than the window: NaN like pandas
+    assert np.isnan(_trail_min_max(a[: w - 1], w))
+    assert np.all(np.isnan(_trail_min_max(A[: w - 1], w)))
