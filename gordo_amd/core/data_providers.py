@@ -60,6 +60,21 @@ class DataProvider:
     ) -> Iterable[pd.Series]:
         raise NotImplementedError()
 
+    def load_frame(
+        self,
+        train_start_date: pd.Timestamp,
+        train_end_date: pd.Timestamp,
+        tag_list: Iterable[SensorTag],
+        resolution: str = "10T",
+    ) -> Optional[pd.DataFrame]:
+        """Whole-frame fast path: all tags in ONE aligned DataFrame
+        (None -> caller falls back to per-series load_series + join).
+        Synthetic providers emit naturally aligned data, so the
+        per-series resample/concat of the general path is pure
+        overhead (~30%% of a build step's data phase at bench shape).
+        Values must be identical to load_series."""
+        return None
+
     def __repr__(self):
         return f"{type(self).__name__}({self._params!r})"
 
@@ -102,21 +117,32 @@ class SineWaveDataProvider(DataProvider):
         super().__init__(noise=noise, **kwargs)
         self.noise = noise
 
+    def _tag_values(self, tag, t):
+        rng = np.random.default_rng(_tag_seed(tag, 1))
+        freq = rng.uniform(0.002, 0.05)
+        phase = rng.uniform(0, 2 * np.pi)
+        amp = rng.uniform(0.5, 2.0)
+        offset = rng.uniform(-1.0, 1.0)
+        return (
+            amp * np.sin(2 * np.pi * freq * t + phase)
+            + offset
+            + rng.standard_normal(len(t)) * self.noise
+        )
+
     def load_series(self, train_start_date, train_end_date, tag_list, resolution="10T"):
         index = _time_index(train_start_date, train_end_date, resolution)
         t = np.arange(len(index), dtype=np.float64)
         for tag in tag_list:
-            rng = np.random.default_rng(_tag_seed(tag, 1))
-            freq = rng.uniform(0.002, 0.05)
-            phase = rng.uniform(0, 2 * np.pi)
-            amp = rng.uniform(0.5, 2.0)
-            offset = rng.uniform(-1.0, 1.0)
-            values = (
-                amp * np.sin(2 * np.pi * freq * t + phase)
-                + offset
-                + rng.standard_normal(len(index)) * self.noise
-            )
-            yield pd.Series(values, index=index, name=tag.name)
+            yield pd.Series(self._tag_values(tag, t), index=index, name=tag.name)
+
+    def load_frame(self, train_start_date, train_end_date, tag_list, resolution="10T"):
+        index = _time_index(train_start_date, train_end_date, resolution)
+        t = np.arange(len(index), dtype=np.float64)
+        tags = list(tag_list)
+        out = np.empty((len(index), len(tags)), dtype=np.float64)
+        for j, tag in enumerate(tags):
+            out[:, j] = self._tag_values(tag, t)
+        return pd.DataFrame(out, index=index, columns=[tg.name for tg in tags])
 
 
 class InfluxDataProvider(DataProvider):

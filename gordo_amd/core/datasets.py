@@ -15,6 +15,7 @@ import logging
 from typing import Any, Dict, List, Optional, Tuple, Union
 
 import dateutil.parser
+import numpy as np
 import pandas as pd
 
 from .base import GordoBaseDataset
@@ -116,20 +117,31 @@ class TimeSeriesDataset(GordoBaseDataset):
         for t in self.target_tag_list:
             all_tags.setdefault(t.name, t)
 
-        series_list = []
-        for series in self.data_provider.load_series(
+        frame = self.data_provider.load_frame(
             self.train_start_date,
             self.train_end_date,
             list(all_tags.values()),
             resolution=self.resolution,
-        ):
-            series_list.append(self._resample(series))
+        )
+        if frame is not None:
+            # aligned single-frame fast path (values identical to the
+            # per-series join; provider contract in DataProvider.load_frame)
+            frame = frame[~np.isnan(frame.values).any(axis=1)]
+        else:
+            series_list = []
+            for series in self.data_provider.load_series(
+                self.train_start_date,
+                self.train_end_date,
+                list(all_tags.values()),
+                resolution=self.resolution,
+            ):
+                series_list.append(self._resample(series))
 
-        if not series_list:
-            raise InsufficientDataError("Data provider returned no series")
+            if not series_list:
+                raise InsufficientDataError("Data provider returned no series")
 
-        frame = pd.concat(series_list, axis=1, join="inner")
-        frame = frame.dropna(how="any")
+            frame = pd.concat(series_list, axis=1, join="inner")
+            frame = frame.dropna(how="any")
 
         if self.row_filter:
             frame = self._apply_row_filter(frame)

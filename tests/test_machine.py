@@ -229,3 +229,27 @@ def test_machine_report_runs_reporters(tmp_path):
     m.runtime["reporters"] = [_CollectingReporter()]
     m.report()
     assert reported == ["report-m"]
+
+
+def test_sinewave_load_frame_bit_identical():
+    """The provider's whole-frame fast path produces bit-identical data
+    to the per-series join path."""
+    import gordo_amd.core.data_providers as dpm
+    from gordo_amd.core import SineWaveDataset
+
+    ds = SineWaveDataset(
+        tag_list=[f"s-{j}" for j in range(5)],
+        train_start_date="2019-01-01T00:00:00Z",
+        train_end_date="2019-01-02T00:00:00Z",
+    )
+    X1, y1 = ds.get_data()
+    orig = dpm.SineWaveDataProvider.load_frame
+    dpm.SineWaveDataProvider.load_frame = dpm.DataProvider.load_frame
+    try:
+        X2, y2 = ds.get_data()
+    finally:
+        dpm.SineWaveDataProvider.load_frame = orig
+    assert (X1.values == X2.values).all()
+    assert list(X1.columns) == list(X2.columns)
+    assert (X1.index == X2.index).all()
+    assert (y1.values == y2.values).all()
