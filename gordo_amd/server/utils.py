@@ -333,10 +333,25 @@ def frame_json_response(context: dict, frame: pd.DataFrame, status: int = 200):
 
     from flask import jsonify, make_response
 
+    # each top-level family must be one contiguous run: the C++ encoder
+    # emits one object per run, and a repeated key would silently drop
+    # data in the client's parser
+    grouped = True
+    if isinstance(frame.columns, pd.MultiIndex):
+        seen = set()
+        prev = object()
+        for top, _ in frame.columns:
+            if top != prev:
+                if top in seen:
+                    grouped = False
+                    break
+                seen.add(top)
+                prev = top
     if (
         _gordo_fastjson is not None
         and isinstance(frame.columns, pd.MultiIndex)
         and frame.columns.nlevels == 2
+        and grouped
         and frame.values.dtype.kind == "f"
     ):
         index = (

@@ -456,3 +456,24 @@ def test_prometheus_ignores_healthcheck(model_collection_directory,
     }
     assert ("/healthcheck",) not in names
     assert ("/server-version",) in names
+
+
+def test_frame_json_response_nonadjacent_families():
+    """Interleaved top-level column families fall back to the python
+    codec rather than emitting duplicate JSON keys."""
+    from gordo_amd.server.utils import frame_json_response
+    from flask import Flask
+
+    df = pd.DataFrame(
+        [[1.0, 2.0, 3.0]],
+        columns=pd.MultiIndex.from_tuples(
+            [("a", "x"), ("b", "x"), ("a", "y")]
+        ),
+        index=["i"],
+    )
+    app = Flask("t")
+    with app.test_request_context():
+        resp = frame_json_response({}, df)
+    data = json.loads(resp.get_data())["data"]
+    assert data["a"] == {"x": {"i": 1.0}, "y": {"i": 3.0}}
+    assert data["b"] == {"x": {"i": 2.0}}
