@@ -382,3 +382,21 @@ def test_n_splits_from_config(cv):
     folds = [k for k in scores["r2-score"] if k.startswith("fold-")]
     # fold-mean/std/max/min + fold-1..n
     assert len(folds) == n_expected + 4
+
+
+def test_cv_mode_build_only(tmp_path):
+    """build_only: model saved, no CV run, scores empty (reference
+    tests/gordo/cli/test_cli.py::test_build_cv_mode_build_only)."""
+    import json
+    import os
+
+    machine = make_machine(
+        model=SKLEARN_MODEL, evaluation={"cv_mode": "build_only"}
+    )
+    ModelBuilder(machine).build(output_dir=str(tmp_path))
+    md = json.load(open(tmp_path / "metadata.json"))
+    cv = md["metadata"]["build_metadata"]["model"]["cross_validation"]
+    assert cv["cv_duration_sec"] is None
+    assert cv["scores"] == {}
+    assert sorted(json.load(open(tmp_path / "info.json"))) == ["checksum"]
+    assert (tmp_path / "model.pkl").is_file()
