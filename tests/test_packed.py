@@ -352,3 +352,73 @@ def test_trail_min_max_matches_pandas(w):
     # shorter than the window: NaN like pandas
     assert np.isnan(_trail_min_max(a[: w - 1], w))
     assert np.all(np.isnan(_trail_min_max(A[: w - 1], w)))
+
+
+def test_packed_diffbased_thresholds_match_modelbuilder(tmp_path):
+    """A DiffBased machine built in a pack carries thresholds that
+    agree with ModelBuilder's sklearn CV path within training noise.
+
+    They are NOT bit-equal by design: sklearn's cross_validate clones
+    the estimator per fold (each clone draws a fresh init from the
+    advancing RNG) while the pack reuses one memoized seeded init for
+    all folds (the +56% build-throughput lever). Thresholds are
+    rolling-extrema, so that per-fold init difference shows up at the
+    ~10-15%% level; structural errors (wrong fold slices, wrong scaler,
+    wrong rolling window) show up at >2x and are what this guards."""
+    from gordo_amd.builder import ModelBuilder
+    from gordo_amd.parallel import PackedFleetBuilder
+    from gordo_amd.workflow import NormalizedConfig
+
+    cfg = {
+        "name": "thr-m",
+        "dataset": {
+            "type": "SineWaveDataset",
+            "tag_list": [f"t-{j}" for j in range(4)],
+            "train_start_date": "2019-01-01T00:00:00+00:00",
+            "train_end_date": "2019-01-04T00:00:00+00:00",
+        },
+        "model": {
+            "gordo_amd.machine.model.anomaly.diff.DiffBasedAnomalyDetector": {
+                "window": 12,
+                "base_estimator": {
+                    "sklearn.pipeline.Pipeline": {
+                        "steps": [
+                            "sklearn.preprocessing.MinMaxScaler",
+                            {
+                                "gordo_amd.machine.model.models."
+                                "KerasAutoEncoder": {
+                                    "kind": "feedforward_hourglass",
+                                    "epochs": 2,
+                                }
+                            },
+                        ]
+                    }
+                },
+            }
+        },
+        "evaluation": {"cv_mode": "full_build"},
+    }
+    norm = NormalizedConfig({"machines": [cfg]}, project_name="p")
+    packed = dict(
+        PackedFleetBuilder(norm.machines, save_models=False).build_all()
+    )["thr-m"]
+
+    norm2 = NormalizedConfig({"machines": [cfg]}, project_name="p")
+    _, solo = ModelBuilder(norm2.machines[0]).build()
+
+    pm = packed.metadata.build_metadata.model.model_meta
+    sm = solo.metadata.build_metadata.model.model_meta
+    for key in (
+        "aggregate-threshold",
+        "feature-thresholds",
+        "smooth-aggregate-threshold",
+        "smooth-feature-thresholds",
+    ):
+        assert key in pm and key in sm, key
+        np.testing.assert_allclose(
+            np.asarray(pm[key], dtype=float),
+            np.asarray(sm[key], dtype=float),
+            rtol=0.6,  # factor-2 structural guard, see docstring
+            atol=1e-4,
+            err_msg=key,
+        )
