@@ -21,7 +21,7 @@ from __future__ import annotations
 import logging
 import math
 import os
-from typing import Dict, List, Optional, Tuple
+from typing import Any, Dict, List, Optional, Tuple
 
 import numpy as np
 import torch
@@ -306,6 +306,7 @@ class BasePack:
         batch_size: int = 256,
         shuffle: bool = True,
         verbose: int = 0,
+        early_stopping: Optional[Dict[str, Any]] = None,
         **_,
     ) -> Dict[str, list]:
         """
@@ -317,6 +318,17 @@ class BasePack:
         G, N = X.shape[0], X.shape[1]
         adam = self.spec.adam_params
         history: Dict[str, list] = {"loss": [], "accuracy": []}
+        # EarlyStopping (keras semantics, restore_best_weights=False):
+        # training ends when EVERY model in the pack has gone `patience`
+        # epochs without improving its train loss by `min_delta` (models
+        # train in lockstep, so the group stops when all have stalled;
+        # for G=1 this is exactly the keras behavior on `loss`).
+        es_patience = es_min_delta = None
+        if early_stopping is not None:
+            es_patience = int(early_stopping.get("patience", 0))
+            es_min_delta = float(early_stopping.get("min_delta", 0.0))
+            es_best = np.full(G, np.inf)
+            es_wait = np.zeros(G, dtype=np.int64)
         n_batches = max(1, math.ceil(self._n_samples(N) / batch_size))
         gens = [torch.Generator().manual_seed(int(s) & 0x7FFFFFFF) for s in self.seeds]
         for epoch in range(epochs):
@@ -350,6 +362,17 @@ class BasePack:
                     "epoch %d/%d mean-loss=%.6g", epoch + 1, epochs,
                     float(np.mean(epoch_loss)),
                 )
+            if es_patience is not None:
+                losses = np.asarray(epoch_loss)
+                improved = losses < (es_best - es_min_delta)
+                es_best = np.minimum(es_best, losses)
+                es_wait = np.where(improved, 0, es_wait + 1)
+                if epoch > 0 and (es_wait >= es_patience).all():
+                    if verbose:
+                        logger.info(
+                            "early stopping at epoch %d/%d", epoch + 1, epochs
+                        )
+                    break
         return history
 
     def _n_samples(self, N: int) -> int:

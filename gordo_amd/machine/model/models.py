@@ -173,7 +173,11 @@ class KerasBaseEstimator(BaseEstimator, GordoBase):
         Yd = torch.from_numpy(y).unsqueeze(0).to(pack.device, pack.compute_dtype)
         fit_args = dict(self.fit_args())
         fit_args.update(self.extract_supported_fit_args(kwargs))
-        history = pack.fit(Xd, Yd, **_clean_fit_args(fit_args))
+        early_stopping = _parse_early_stopping(fit_args.get("callbacks"))
+        history = pack.fit(
+            Xd, Yd, early_stopping=early_stopping,
+            **_clean_fit_args(fit_args),
+        )
         self.adopt_pack_result(
             spec, pack.state_for_model(0), _history_for_model(history, 0)
         )
@@ -260,6 +264,33 @@ class KerasBaseEstimator(BaseEstimator, GordoBase):
     def __call__(self):
         # factory-style call used by some sklearn clone paths
         return self
+
+
+def _parse_early_stopping(callbacks) -> Optional[Dict[str, Any]]:
+    """Extract EarlyStopping settings from a keras-style callbacks list
+    (reference configs: [{"tensorflow.keras.callbacks.EarlyStopping":
+    {"monitor": ..., "patience": ...}}]). Only EarlyStopping has an
+    engine analog; other callbacks are ignored. The engine has no
+    validation split, so val_* monitors fall back to train loss."""
+    for cb in callbacks or []:
+        if isinstance(cb, str) and cb.rsplit(".", 1)[-1] == "EarlyStopping":
+            return {"patience": 0, "min_delta": 0.0}
+        if isinstance(cb, dict) and len(cb) == 1:
+            key = next(iter(cb))
+            if key.rsplit(".", 1)[-1] == "EarlyStopping":
+                params = cb[key] or {}
+                monitor = params.get("monitor", "val_loss")
+                if str(monitor).startswith("val"):
+                    logger.warning(
+                        "EarlyStopping monitor %r: the engine trains "
+                        "without a validation split; monitoring train "
+                        "loss instead", monitor,
+                    )
+                return {
+                    "patience": int(params.get("patience", 0)),
+                    "min_delta": float(params.get("min_delta", 0.0)),
+                }
+    return None
 
 
 def _clean_fit_args(args: Dict[str, Any]) -> Dict[str, Any]:
@@ -446,7 +477,11 @@ class KerasLSTMBaseEstimator(KerasBaseEstimator, TransformerMixin, metaclass=abc
         Yd = torch.from_numpy(y).unsqueeze(0).to(pack.device, pack.compute_dtype)
         fit_args = dict(self.fit_args())
         fit_args.update(self.extract_supported_fit_args(kwargs))
-        history = pack.fit(Xd, Yd, **_clean_fit_args(fit_args))
+        early_stopping = _parse_early_stopping(fit_args.get("callbacks"))
+        history = pack.fit(
+            Xd, Yd, early_stopping=early_stopping,
+            **_clean_fit_args(fit_args),
+        )
         self.adopt_pack_result(
             spec, pack.state_for_model(0), _history_for_model(history, 0)
         )

@@ -50,7 +50,7 @@ from ..machine.model.anomaly.diff import (
     DiffBasedAnomalyDetector,
     DiffBasedKFCVAnomalyDetector,
 )
-from ..machine.model.models import KerasBaseEstimator
+from ..machine.model.models import _parse_early_stopping, KerasBaseEstimator
 from ..util import disk_registry
 
 logger = logging.getLogger(__name__)
@@ -766,8 +766,13 @@ def _metric_all_tags(metric, yt: np.ndarray, yp: np.ndarray):
 
 
 def _engine_fit_args(fit_args: Dict[str, Any]) -> Dict[str, Any]:
-    out = {}
+    out: Dict[str, Any] = {}
     for k in ("epochs", "batch_size", "shuffle", "verbose"):
         if k in fit_args:
             out[k] = fit_args[k]
+    # identical callbacks are part of the group key (fit_args), so the
+    # lockstep all-models-stalled semantics of pack.fit applies per group
+    es = _parse_early_stopping(fit_args.get("callbacks"))
+    if es is not None:
+        out["early_stopping"] = es
     return out

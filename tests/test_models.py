@@ -438,3 +438,41 @@ sklearn.pipeline.Pipeline:
     assert isinstance(pipe, Pipeline)
     pipe.fit(X, y)
     assert len(pipe.predict(X)) == len(y)
+
+
+def test_early_stopping_callback_honored():
+    """An EarlyStopping callback config actually stops training
+    (reference test_model.py::test_keras_autoencoder_fits_callbacks —
+    there the callback is passed to keras; here the engine implements
+    the patience semantics on train loss)."""
+    from gordo_amd.machine.model.models import (
+        KerasAutoEncoder,
+        _parse_early_stopping,
+    )
+
+    assert _parse_early_stopping(None) is None
+    assert _parse_early_stopping(
+        [{"tensorflow.keras.callbacks.EarlyStopping": {
+            "monitor": "val_loss", "patience": 10}}]
+    ) == {"patience": 10, "min_delta": 0.0}
+
+    # a huge min_delta means "never improving" -> stops after
+    # patience+1 epochs instead of running all 50
+    X = np.random.RandomState(0).random((64, 8))
+    model = KerasAutoEncoder(
+        kind="feedforward_hourglass",
+        epochs=50,
+        batch_size=32,
+        callbacks=[{"tensorflow.keras.callbacks.EarlyStopping": {
+            "monitor": "loss", "patience": 2, "min_delta": 1e9}}],
+    )
+    model.fit(X)
+    history = model.get_metadata()["history"]
+    assert len(history["loss"]) == 3  # epoch 0 sets best; 2 stalls; stop
+
+    # without the callback, all epochs run
+    model2 = KerasAutoEncoder(
+        kind="feedforward_hourglass", epochs=5, batch_size=32
+    )
+    model2.fit(X)
+    assert len(model2.get_metadata()["history"]["loss"]) == 5
