@@ -158,6 +158,11 @@ class BasePack:
     def train_batch(self, Xb, Tb) -> torch.Tensor:
         raise NotImplementedError
 
+    def eval_batch(self, Xb, Tb) -> torch.Tensor:
+        """Forward-only per-model loss for a gathered batch (the
+        validation pass of fit)."""
+        raise NotImplementedError
+
     # -- common ----------------------------------------------------------
     @property
     def compute_dtype(self):
@@ -307,6 +312,7 @@ class BasePack:
         shuffle: bool = True,
         verbose: int = 0,
         early_stopping: Optional[Dict[str, Any]] = None,
+        validation_split: float = 0.0,
         **_,
     ) -> Dict[str, list]:
         """
@@ -318,6 +324,18 @@ class BasePack:
         G, N = X.shape[0], X.shape[1]
         adam = self.spec.adam_params
         history: Dict[str, list] = {"loss": [], "accuracy": []}
+        n_all = self._n_samples(N)
+        # keras validation_split semantics: the LAST fraction of samples
+        # (in input order, before shuffling) is held out; history gains
+        # val_loss/val_accuracy and EarlyStopping can monitor val_loss
+        n_train = n_all
+        if validation_split and validation_split > 0.0:
+            n_train = int(n_all * (1.0 - float(validation_split)))
+            n_train = max(1, min(n_train, n_all))
+            if n_train < n_all:
+                history["val_loss"] = []
+                history["val_accuracy"] = []
+        n_val = n_all - n_train
         # EarlyStopping (keras semantics, restore_best_weights=False):
         # training ends when EVERY model in the pack has gone `patience`
         # epochs without improving its train loss by `min_delta` (models
@@ -329,16 +347,16 @@ class BasePack:
             es_min_delta = float(early_stopping.get("min_delta", 0.0))
             es_best = np.full(G, np.inf)
             es_wait = np.zeros(G, dtype=np.int64)
-        n_batches = max(1, math.ceil(self._n_samples(N) / batch_size))
+        n_batches = max(1, math.ceil(n_train / batch_size))
         gens = [torch.Generator().manual_seed(int(s) & 0x7FFFFFFF) for s in self.seeds]
         for epoch in range(epochs):
             if shuffle:
                 perm = torch.stack(
-                    [torch.randperm(self._n_samples(N), generator=g) for g in gens]
+                    [torch.randperm(n_train, generator=g) for g in gens]
                 ).to(self.device)
             else:
                 perm = (
-                    torch.arange(self._n_samples(N), device=self.device)
+                    torch.arange(n_train, device=self.device)
                     .unsqueeze(0)
                     .expand(G, -1)
                 )
@@ -357,13 +375,34 @@ class BasePack:
             epoch_loss = (epoch_loss / max(samples_seen, 1)).cpu().tolist()
             history["loss"].append(epoch_loss)
             history["accuracy"].append([0.0] * G)
+            if n_val > 0:
+                val_loss = torch.zeros(G, dtype=torch.float32, device=self.device)
+                vseen = 0
+                for vs_ in range(n_train, n_all, batch_size):
+                    vidx = (
+                        torch.arange(
+                            vs_, min(vs_ + batch_size, n_all), device=self.device
+                        ).unsqueeze(0).expand(G, -1)
+                    )
+                    Xvb, Tvb = self._gather_batch(X, Y, vidx)
+                    val_loss += self.eval_batch(Xvb, Tvb) * vidx.shape[1]
+                    vseen += vidx.shape[1]
+                val_losses = (val_loss / max(vseen, 1)).cpu().tolist()
+                history["val_loss"].append(val_losses)
+                history["val_accuracy"].append([0.0] * G)
             if verbose:
                 logger.info(
                     "epoch %d/%d mean-loss=%.6g", epoch + 1, epochs,
                     float(np.mean(epoch_loss)),
                 )
             if es_patience is not None:
-                losses = np.asarray(epoch_loss)
+                monitor_val = (
+                    str((early_stopping or {}).get("monitor", "loss"))
+                    .startswith("val") and n_val > 0
+                )
+                losses = np.asarray(
+                    history["val_loss"][-1] if monitor_val else epoch_loss
+                )
                 improved = losses < (es_best - es_min_delta)
                 es_best = np.minimum(es_best, losses)
                 es_wait = np.where(improved, 0, es_wait + 1)
@@ -417,6 +456,10 @@ class DensePack(BasePack):
 
     def predict(self, X: torch.Tensor) -> torch.Tensor:
         return self.forward(self._to_compute(X))[-1]
+
+    def eval_batch(self, Xb, Tb) -> torch.Tensor:
+        out = self.forward(Xb)[-1].float()
+        return ((out - Tb.float()) ** 2).mean(dim=(1, 2))
 
     def train_batch(self, Xb, Tb) -> torch.Tensor:
         acts = self.forward(Xb)
@@ -572,6 +615,10 @@ class LSTMPack(BasePack):
     def predict_windows(self, Xw: torch.Tensor) -> torch.Tensor:
         y, _ = self._forward_seq(self._to_compute(Xw), keep=False)
         return y
+
+    def eval_batch(self, Xw, Tb) -> torch.Tensor:
+        y = self.predict_windows(Xw).float()
+        return ((y - Tb.float()) ** 2).mean(dim=(1, 2))
 
     def _train_batch_fused(self, Xw, Tb) -> torch.Tensor:
         G, B, T, _ = Xw.shape

@@ -176,6 +176,7 @@ class KerasBaseEstimator(BaseEstimator, GordoBase):
         early_stopping = _parse_early_stopping(fit_args.get("callbacks"))
         history = pack.fit(
             Xd, Yd, early_stopping=early_stopping,
+            validation_split=float(fit_args.get("validation_split") or 0.0),
             **_clean_fit_args(fit_args),
         )
         self.adopt_pack_result(
@@ -280,13 +281,8 @@ def _parse_early_stopping(callbacks) -> Optional[Dict[str, Any]]:
             if key.rsplit(".", 1)[-1] == "EarlyStopping":
                 params = cb[key] or {}
                 monitor = params.get("monitor", "val_loss")
-                if str(monitor).startswith("val"):
-                    logger.warning(
-                        "EarlyStopping monitor %r: the engine trains "
-                        "without a validation split; monitoring train "
-                        "loss instead", monitor,
-                    )
                 return {
+                    "monitor": str(monitor),
                     "patience": int(params.get("patience", 0)),
                     "min_delta": float(params.get("min_delta", 0.0)),
                 }
@@ -480,6 +476,7 @@ class KerasLSTMBaseEstimator(KerasBaseEstimator, TransformerMixin, metaclass=abc
         early_stopping = _parse_early_stopping(fit_args.get("callbacks"))
         history = pack.fit(
             Xd, Yd, early_stopping=early_stopping,
+            validation_split=float(fit_args.get("validation_split") or 0.0),
             **_clean_fit_args(fit_args),
         )
         self.adopt_pack_result(

@@ -770,6 +770,8 @@ def _engine_fit_args(fit_args: Dict[str, Any]) -> Dict[str, Any]:
     for k in ("epochs", "batch_size", "shuffle", "verbose"):
         if k in fit_args:
             out[k] = fit_args[k]
+    if fit_args.get("validation_split"):
+        out["validation_split"] = float(fit_args["validation_split"])
     # identical callbacks are part of the group key (fit_args), so the
     # lockstep all-models-stalled semantics of pack.fit applies per group
     es = _parse_early_stopping(fit_args.get("callbacks"))

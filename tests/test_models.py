@@ -454,7 +454,7 @@ def test_early_stopping_callback_honored():
     assert _parse_early_stopping(
         [{"tensorflow.keras.callbacks.EarlyStopping": {
             "monitor": "val_loss", "patience": 10}}]
-    ) == {"patience": 10, "min_delta": 0.0}
+    ) == {"monitor": "val_loss", "patience": 10, "min_delta": 0.0}
 
     # a huge min_delta means "never improving" -> stops after
     # patience+1 epochs instead of running all 50
@@ -476,3 +476,28 @@ def test_early_stopping_callback_honored():
     )
     model2.fit(X)
     assert len(model2.get_metadata()["history"]["loss"]) == 5
+
+
+def test_validation_split_history():
+    """validation_split holds out the last fraction and reports
+    val_loss per epoch; EarlyStopping can monitor it."""
+    from gordo_amd.machine.model.models import KerasAutoEncoder
+
+    X = np.random.RandomState(0).random((100, 8))
+    model = KerasAutoEncoder(
+        kind="feedforward_hourglass", epochs=4, batch_size=32,
+        validation_split=0.2,
+    )
+    model.fit(X)
+    hist = model.get_metadata()["history"]
+    assert len(hist["val_loss"]) == 4
+    assert all(np.isfinite(v) for v in hist["val_loss"])
+
+    stopper = KerasAutoEncoder(
+        kind="feedforward_hourglass", epochs=50, batch_size=32,
+        validation_split=0.2,
+        callbacks=[{"tensorflow.keras.callbacks.EarlyStopping": {
+            "monitor": "val_loss", "patience": 1, "min_delta": 1e9}}],
+    )
+    stopper.fit(X)
+    assert len(stopper.get_metadata()["history"]["val_loss"]) == 2
