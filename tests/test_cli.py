@@ -517,3 +517,28 @@ def test_run_metrics_server_app():
     assert c.get("/healthcheck").status_code == 200
     resp = c.get("/metrics")
     assert resp.status_code == 200
+
+
+def test_build_insufficient_data_exit_code(tmp_path):
+    """InsufficientDataError maps to exit code 80 (reference
+    cli.py:26-39 exit-code table)."""
+    cfg = dict(
+        MACHINE_JSON,
+        dataset=dict(
+            MACHINE_JSON["dataset"],
+            # a 30-minute range at 10-min resolution -> 3 rows, below
+            # the configured threshold
+            train_end_date="2019-01-01T00:30:00+00:00",
+            n_samples_threshold=100,
+        ),
+    )
+    report_file = tmp_path / "report.json"
+    result = CliRunner().invoke(
+        gordo,
+        ["build", json.dumps(cfg), str(tmp_path / "out"),
+         "--exceptions-reporter-file", str(report_file)],
+    )
+    assert result.exit_code == 80, result.output
+    assert json.loads(report_file.read_text())["type"] == (
+        "InsufficientDataError"
+    )
