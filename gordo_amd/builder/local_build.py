@@ -7,7 +7,6 @@ from __future__ import annotations
 
 from typing import Iterable, Optional, Tuple
 
-from .build_model import ModelBuilder
 from .utils import create_model_builder
 from ..workflow.workflow_generator.workflow_generator import get_dict_from_yaml
 from ..workflow.config_elements.normalized_config import NormalizedConfig

@@ -9,7 +9,7 @@ lstm_autoencoder.py) produce these specs; the packed engine
 from __future__ import annotations
 
 from dataclasses import dataclass, field, asdict
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, List
 
 
 @dataclass

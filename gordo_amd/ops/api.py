@@ -14,7 +14,7 @@ from typing import Optional, Tuple
 import torch
 
 from . import reference as ref
-from .reference import (  # re-export
+from .reference import (  # noqa — ACT_* re-exported via gordo_amd.ops
     ACT_LINEAR,
     ACT_TANH,
     ACT_RELU,

@@ -23,7 +23,7 @@ from functools import lru_cache
 from typing import List, Union
 
 import dateutil.parser
-import numpy as np
+import numpy as np  # noqa — doctest namespace (dataframe_to_dict)
 import pandas as pd
 import pyarrow as pa
 import pyarrow.parquet as pq
