@@ -73,6 +73,32 @@ def test_fleet_build_two_ranks(tmp_path):
         assert (d / "model.pkl").is_file()
         assert (d / "metadata.json").is_file()
 
+    # sharded (2-rank) outputs == single-process outputs, machine for
+    # machine ("1-GPU vs 8-GPU build-output equivalence —
+    # same pickles/metadata modulo timings"); per-machine seeds make
+    # this exact regardless of which rank built which machine
+    from gordo_amd.parallel import PackedFleetBuilder
+    from gordo_amd.workflow import NormalizedConfig
+    from gordo_amd.workflow.workflow_generator import get_dict_from_yaml
+
+    norm = NormalizedConfig(
+        get_dict_from_yaml(str(cfg_path)), project_name="fleet-proj"
+    )
+    solo = dict(PackedFleetBuilder(norm.machines, save_models=False).build_all())
+    for i in range(4):
+        meta = json.loads((out_dir / f"fleet-m-{i}" / "metadata.json").read_text())
+        dist_scores = meta["metadata"]["build_metadata"]["model"][
+            "cross_validation"]["scores"]
+        solo_scores = (
+            solo[f"fleet-m-{i}"].metadata.build_metadata.model
+            .cross_validation.scores
+        )
+        assert set(dist_scores) == set(solo_scores)
+        for key in solo_scores:
+            assert dist_scores[key]["fold-mean"] == pytest.approx(
+                solo_scores[key]["fold-mean"], rel=1e-5, abs=1e-7
+            ), key
+
 
 def test_shard_machines_balanced():
     from gordo_amd.parallel import shard_machines
