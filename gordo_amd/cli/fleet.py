@@ -44,6 +44,9 @@ def fleet_cli():
 @click.option("--replace-cache", is_flag=True)
 @click.option("--status-file", type=click.Path(), default=None,
               help="Write per-machine build status JSON here (rank 0)")
+@click.option("--requeue/--no-requeue", default=True,
+              help="After a distributed run, rebuild failed or missing "
+                   "machines in-process (rank-crash recovery)")
 def fleet_build(
     machine_config: str,
     project_name: str,
@@ -52,6 +55,7 @@ def fleet_build(
     gpus: int,
     replace_cache: bool,
     status_file: Optional[str],
+    requeue: bool = True,
 ):
     """Build every Machine in the config across the node's GPUs."""
     import torch
@@ -78,14 +82,98 @@ def fleet_build(
             )
             if model_register_dir:
                 env["GORDO_FLEET_MODEL_REGISTER_DIR"] = model_register_dir
-            if status_file:
-                env["GORDO_FLEET_STATUS_FILE"] = status_file
-            raise SystemExit(subprocess.call(cmd, env=env))
+            if status_file is None:
+                status_file = os.path.join(output_dir, ".fleet-status.json")
+            env["GORDO_FLEET_STATUS_FILE"] = status_file
+            os.makedirs(output_dir, exist_ok=True)
+            rc = subprocess.call(cmd, env=env)
+            if requeue:
+                n_failed = requeue_failed(
+                    machine_config, project_name, output_dir,
+                    model_register_dir, replace_cache, status_file,
+                    distributed_rc=rc,
+                )
+                raise SystemExit(0 if n_failed == 0 else 1)
+            raise SystemExit(rc)
 
     _run_fleet_build(
         machine_config, project_name, output_dir, model_register_dir,
         replace_cache, status_file,
     )
+
+
+def requeue_failed(
+    machine_config: str,
+    project_name: str,
+    output_dir: str,
+    model_register_dir: Optional[str],
+    replace_cache: bool,
+    status_file: str,
+    distributed_rc: int = 0,
+) -> int:
+    """Elastic recovery after a distributed fleet run:
+    machines that failed, or never produced a model dir because their
+    rank died (torchrun tears the whole job down on a rank crash),
+    are rebuilt in-process on one device. Returns the number of
+    machines still failed afterwards and rewrites ``status_file`` with
+    the merged outcome."""
+    from ..parallel import PackedFleetBuilder
+    from ..workflow import NormalizedConfig
+    from ..workflow.workflow_generator import get_dict_from_yaml
+
+    config = get_dict_from_yaml(machine_config)
+    all_names = [m.get("name") for m in config.get("machines", [])]
+
+    prior_status: dict = {}
+    summary: dict = {}
+    if os.path.exists(status_file):
+        try:
+            with open(status_file) as f:
+                summary = json.load(f)
+            prior_status = dict(summary.get("status", {}))
+        except (ValueError, OSError):
+            summary = {}
+
+    def built(name: str) -> bool:
+        return os.path.isfile(os.path.join(output_dir, name, "model.pkl"))
+
+    retry = [
+        n for n in all_names
+        if prior_status.get(n) is not None or not built(n)
+    ]
+    if not retry:
+        return 0
+    logger.warning(
+        "Requeueing %d machine(s) after distributed run (rc=%d): %s",
+        len(retry), distributed_rc, retry,
+    )
+    sub_config = dict(config)
+    sub_config["machines"] = [
+        m for m in config.get("machines", []) if m.get("name") in retry
+    ]
+    norm = NormalizedConfig(sub_config, project_name=project_name)
+    builder = PackedFleetBuilder(
+        norm.machines,
+        output_dir=output_dir,
+        model_register_dir=model_register_dir,
+        replace_cache=replace_cache,
+    )
+    for name, res in builder.build_all():
+        prior_status[name] = (
+            None if not isinstance(res, BaseException) else repr(res)
+        )
+    n_ok = sum(1 for n in all_names if prior_status.get(n) is None)
+    summary.update(
+        project=project_name,
+        n_machines=len(all_names),
+        n_ok=n_ok,
+        n_failed=len(all_names) - n_ok,
+        requeued=retry,
+        status=prior_status,
+    )
+    with open(status_file, "w") as f:
+        json.dump(summary, f, indent=2, default=str)
+    return len(all_names) - n_ok
 
 
 def _run_fleet_build(

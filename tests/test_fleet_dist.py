@@ -131,3 +131,45 @@ def test_shard_machines_balanced():
         names = [machines[i].name for i in shard]
         assert sum(n.startswith("a-") for n in names) == 3
         assert sum(n.startswith("b-") for n in names) == 1
+
+
+def test_requeue_failed_rebuilds_missing(tmp_path):
+    """Rank-crash recovery: a machine with no model dir (its rank died)
+    and a machine recorded as failed are both rebuilt in-process; the
+    merged status reflects the recovery."""
+    from gordo_amd.cli.fleet import requeue_failed
+
+    config = CONFIG.format(
+        machines="".join(MACHINE_TMPL.format(i=i) for i in range(3))
+    )
+    cfg_path = tmp_path / "cfg.yml"
+    cfg_path.write_text(config)
+    out_dir = tmp_path / "models"
+    status_file = tmp_path / "status.json"
+
+    # simulate a partial run: machine 0 built fine, machine 1's output
+    # is missing (rank death), machine 2 recorded a failure
+    from gordo_amd.parallel import PackedFleetBuilder
+    from gordo_amd.workflow import NormalizedConfig
+    from gordo_amd.workflow.workflow_generator import get_dict_from_yaml
+
+    norm = NormalizedConfig(
+        get_dict_from_yaml(str(cfg_path)), project_name="fleet-proj"
+    )
+    PackedFleetBuilder(
+        [norm.machines[0]], output_dir=str(out_dir)
+    ).build_all()
+    status_file.write_text(json.dumps({
+        "status": {"fleet-m-0": None, "fleet-m-2": "RuntimeError('rank died')"}
+    }))
+
+    n_failed = requeue_failed(
+        str(cfg_path), "fleet-proj", str(out_dir), None, False,
+        str(status_file), distributed_rc=1,
+    )
+    assert n_failed == 0
+    summary = json.loads(status_file.read_text())
+    assert summary["n_ok"] == 3
+    assert sorted(summary["requeued"]) == ["fleet-m-1", "fleet-m-2"]
+    for i in range(3):
+        assert (out_dir / f"fleet-m-{i}" / "model.pkl").is_file()
