@@ -133,7 +133,47 @@ class GordoServerPrometheusMetrics:
         label_values.extend(self.main_label_values(req, resp))
         return label_values
 
+    def prepare_gpu_gauges(self):
+        """Per-GPU serving gauges (SURVEY §5.3/§5.5 'new framework'):
+        models resident in the serving LRU and HBM bytes allocated —
+        the capacity signals for a 288 GB MI355X serving thousands of
+        models."""
+        from prometheus_client import Gauge
+
+        models_cached = Gauge(
+            "gordo_server_models_cached",
+            "Models resident in the serving LRU cache",
+            registry=self.registry,
+        )
+
+        def _cache_size() -> float:
+            from .. import utils as server_utils
+
+            return float(server_utils.load_model.cache_info().currsize)
+
+        models_cached.set_function(_cache_size)
+
+        gpu_mem = Gauge(
+            "gordo_server_gpu_memory_allocated_bytes",
+            "torch HBM bytes allocated on the serving device",
+            registry=self.registry,
+        )
+
+        def _gpu_mem() -> float:
+            try:
+                import torch
+
+                if torch.cuda.is_available():
+                    return float(torch.cuda.memory_allocated())
+            except ImportError:
+                pass
+            return 0.0
+
+        gpu_mem.set_function(_gpu_mem)
+
     def prepare_app(self, app: Flask):
+        self.prepare_gpu_gauges()
+
         @app.before_request
         def _start_prometheus():
             g.prometheus_metrics = self

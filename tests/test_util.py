@@ -271,3 +271,22 @@ def test_fix_resource_limits_edges():
         fix_resource_limits(
             {"requests": {"memory": "1M"}, "limits": {"memory": 3}}
         )
+
+
+def test_prometheus_gpu_gauges():
+    from flask import Flask
+    from prometheus_client.registry import CollectorRegistry
+
+    from gordo_amd.server.prometheus import GordoServerPrometheusMetrics
+
+    app = Flask("gauge-test")
+    registry = CollectorRegistry()
+    GordoServerPrometheusMetrics(
+        args_labels=[], info={"version": "1"}, registry=registry
+    ).prepare_app(app)
+    cached = registry.get_sample_value("gordo_server_models_cached")
+    gpu_mem = registry.get_sample_value(
+        "gordo_server_gpu_memory_allocated_bytes"
+    )
+    assert cached is not None and cached >= 0
+    assert gpu_mem == 0.0  # no GPU in the CPU test lane
