@@ -137,7 +137,13 @@ def get_download_model():
 @base_blueprint.route("/gordo/v0/<gordo_project>/models", methods=["GET"])
 def get_model_list(gordo_project: str):
     try:
-        available_models = os.listdir(g.collection_dir)
+        # models are directories; stray files in the collection dir
+        # (e.g. a fleet status JSON) are not models
+        available_models = [
+            name
+            for name in os.listdir(g.collection_dir)
+            if os.path.isdir(os.path.join(g.collection_dir, name))
+        ]
     except FileNotFoundError:
         available_models = []
     return jsonify({"models": available_models})

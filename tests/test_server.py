@@ -477,3 +477,15 @@ def test_frame_json_response_nonadjacent_families():
     data = json.loads(resp.get_data())["data"]
     assert data["a"] == {"x": {"i": 1.0}, "y": {"i": 3.0}}
     assert data["b"] == {"x": {"i": 2.0}}
+
+
+def test_models_list_ignores_files(api_client, gordo_project,
+                                   model_collection_directory):
+    stray = os.path.join(model_collection_directory, ".fleet-status.json")
+    with open(stray, "w") as f:
+        f.write("{}")
+    try:
+        resp = api_client.get(f"/gordo/v0/{gordo_project}/models")
+        assert ".fleet-status.json" not in resp.json["models"]
+    finally:
+        os.unlink(stray)
