@@ -366,7 +366,9 @@ def frame_json_response(context: dict, frame: pd.DataFrame, status: int = 200):
                 [str(c[1]) for c in frame.columns],
                 frame.values,
             )
-        except ValueError:
+        except (ValueError, TypeError, RuntimeError):
+            # non-ascii keys / non-string index entries: the python
+            # codec below handles every shape
             data = None
         if data is not None:
             rest = _json.dumps(context, default=str).encode()
