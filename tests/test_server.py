@@ -489,3 +489,74 @@ def test_models_list_ignores_files(api_client, gordo_project,
         assert ".fleet-status.json" not in resp.json["models"]
     finally:
         os.unlink(stray)
+
+
+def test_run_server_gunicorn_argv(monkeypatch):
+    """run_server builds the gunicorn command line of the reference
+    (test_gordo_server.py::test_run_server_gthread/gevent shape) when
+    gunicorn is importable; werkzeug fallback otherwise is exercised by
+    the import guard."""
+    import types
+
+    from gordo_amd.server import server as srv
+
+    calls = []
+    monkeypatch.setattr(srv, "run_cmd", lambda cmd: calls.append(cmd))
+    # fake out the gunicorn import so the argv path runs in this image
+    monkeypatch.setitem(sys.modules, "gunicorn", types.ModuleType("gunicorn"))
+
+    srv.run_server(
+        "127.0.0.1", 9000, 2, "debug",
+        worker_connections=50, threads=8, worker_class="gthread",
+    )
+    assert calls[-1] == [
+        "gunicorn",
+        "--bind", "127.0.0.1:9000",
+        "--log-level", "debug",
+        "--error-logfile", "-",
+        "--access-logfile", "-",
+        "--worker-tmp-dir", "/dev/shm",
+        "--worker-class", "gthread",
+        "--workers", "2",
+        "--threads", "8",
+        "gordo_amd.server.server:build_app()",
+    ]
+
+    srv.run_server(
+        "127.0.0.1", 9000, 2, "debug",
+        worker_connections=50, threads=8, worker_class="gevent",
+    )
+    assert "--worker-connections" in calls[-1]
+    assert "--threads" not in calls[-1]
+
+
+def test_machine_encoders():
+    """Machine JSON encoder handles datetime + SensorTag; YAML dumper
+    renders nested configs as | multiline blocks (reference
+    machine/encoders.py)."""
+    import datetime
+
+    import yaml
+
+    from gordo_amd.machine import Machine
+
+    m = Machine.from_config(
+        {
+            "name": "enc-m",
+            "model": {"sklearn.decomposition.PCA": {"n_components": 2}},
+            "dataset": {
+                "type": "RandomDataset",
+                "tag_list": ["a", "b", "c"],
+                "train_start_date": "2019-01-01T00:00:00Z",
+                "train_end_date": "2019-01-02T00:00:00Z",
+            },
+        },
+        project_name="p",
+    )
+    as_json = m.to_json()
+    parsed = json.loads(as_json)
+    # datetimes serialized as ISO strings; tags as JSON-able values
+    start = parsed["dataset"]["train_start_date"]
+    assert datetime.datetime.fromisoformat(start.replace("Z", "+00:00"))
+    yaml_str = m.to_yaml()
+    assert yaml.safe_load(yaml_str)["name"] == "enc-m"
