@@ -205,6 +205,17 @@ class KerasBaseEstimator(BaseEstimator, GordoBase):
             self.n_features_out = n_features_out
         return self
 
+    # serving-time device override (multi-GPU serving: the server pins
+    # each model to a device by consistent hash — set_serving_device)
+    _serve_device: Optional[str] = None
+
+    def set_serving_device(self, device: str):
+        """Pin this (fitted) estimator's inference pack to ``device``;
+        drops any pack already resident elsewhere."""
+        if self._serve_device != device:
+            self._serve_device = device
+            self._pack = None
+
     def _ensure_pack(self):
         if self._pack is not None:
             return self._pack
@@ -212,7 +223,8 @@ class KerasBaseEstimator(BaseEstimator, GordoBase):
             raise ValueError(
                 f"This {self.__class__.__name__} has not been fitted yet."
             )
-        pack = self._pack_cls(self._spec, G=1, device=_default_device(), seeds=[0])
+        device = self._serve_device or _default_device()
+        pack = self._pack_cls(self._spec, G=1, device=device, seeds=[0])
         pack.load_model_state(0, self._weights)
         self._pack = pack
         return pack

@@ -232,10 +232,55 @@ def _default_model_cache_size() -> int:
     return 2
 
 
+def _serving_device_for(name: str) -> Union[str, None]:
+    """Multi-GPU serving: with GORDO_SERVER_GPUS=N (or N visible GPUs
+    and GORDO_SERVER_GPUS=auto), each model is pinned to
+    cuda:<sha1(name) % N> — a consistent hash so every worker process
+    places the same model on the same device and the fleet of served
+    models spreads over the node's HBM."""
+    setting = os.getenv("GORDO_SERVER_GPUS", "").strip().lower()
+    if not setting:
+        return None
+    try:
+        import torch
+
+        if not torch.cuda.is_available():
+            return None
+        n = (
+            torch.cuda.device_count()
+            if setting == "auto"
+            else min(int(setting), torch.cuda.device_count())
+        )
+    except (ImportError, ValueError):
+        return None
+    if n <= 1:
+        return None
+    import hashlib as _hashlib
+
+    idx = int(_hashlib.sha1(name.encode()).hexdigest(), 16) % n
+    return f"cuda:{idx}"
+
+
 @lru_cache(maxsize=_default_model_cache_size())
 def load_model(directory: str, name: str) -> BaseEstimator:
     start_time = timeit.default_timer()
     model = serializer.load(os.path.join(directory, name))
+    device = _serving_device_for(name)
+    if device is not None:
+        from ..machine.model.models import KerasBaseEstimator
+
+        stack = [model]
+        while stack:
+            obj = stack.pop()
+            if isinstance(obj, KerasBaseEstimator):
+                obj.set_serving_device(device)
+            for attr in ("steps", "transformer_list"):
+                for _, step in getattr(obj, attr, []) or []:
+                    stack.append(step)
+            for attr in ("base_estimator", "estimator", "regressor"):
+                child = getattr(obj, attr, None)
+                if child is not None:
+                    stack.append(child)
     logger.debug("Time to load model: %ss", timeit.default_timer() - start_time)
     return model
 

@@ -560,3 +560,42 @@ def test_machine_encoders():
     assert datetime.datetime.fromisoformat(start.replace("Z", "+00:00"))
     yaml_str = m.to_yaml()
     assert yaml.safe_load(yaml_str)["name"] == "enc-m"
+
+
+def test_serving_device_hash(monkeypatch):
+    """GORDO_SERVER_GPUS pins each model to a stable device index."""
+    import types
+
+    from gordo_amd.server import utils as su
+
+    fake_torch = types.SimpleNamespace(
+        cuda=types.SimpleNamespace(
+            is_available=lambda: True, device_count=lambda: 8
+        )
+    )
+    monkeypatch.setitem(sys.modules, "torch", fake_torch)
+    monkeypatch.setenv("GORDO_SERVER_GPUS", "auto")
+    d1 = su._serving_device_for("model-a")
+    assert d1 == su._serving_device_for("model-a")  # stable
+    assert d1.startswith("cuda:")
+    devices = {su._serving_device_for(f"m-{i}") for i in range(64)}
+    assert len(devices) > 1  # spreads across GPUs
+
+    monkeypatch.setenv("GORDO_SERVER_GPUS", "1")
+    assert su._serving_device_for("model-a") is None
+    monkeypatch.delenv("GORDO_SERVER_GPUS")
+    assert su._serving_device_for("model-a") is None
+
+
+def test_set_serving_device_cpu_roundtrip():
+    """set_serving_device('cpu') keeps inference working (the GPU path
+    differs only in the device string)."""
+    from gordo_amd.machine.model.models import KerasAutoEncoder
+
+    X = np.random.RandomState(0).random((40, 6))
+    model = KerasAutoEncoder(kind="feedforward_hourglass", epochs=1)
+    model.fit(X)
+    before = model.predict(X)
+    model.set_serving_device("cpu")
+    after = model.predict(X)
+    np.testing.assert_allclose(before, after, rtol=1e-5, atol=1e-6)
