@@ -241,6 +241,12 @@ class DiffBasedAnomalyDetector(AnomalyDetectorBase):
     # rows below which the H2D round-trip outweighs the fused kernel
     _DEVICE_SCORE_MIN_ROWS = 512
 
+    def set_serving_device(self, device: str):
+        """Pin the fused scoring kernel (and, via the server's object
+        walk, the nested estimator) to ``device`` for multi-GPU
+        serving."""
+        self._serve_device = device
+
     def _fused_device_scores(self, data: pd.DataFrame, y: pd.DataFrame):
         """Serving hot path: one fused HIP kernel (ops.anomaly_score,
         kernel K9) computes every residual column family when a GPU is
@@ -260,23 +266,24 @@ class DiffBasedAnomalyDetector(AnomalyDetectorBase):
                 or not hasattr(self.scaler, "min_")
             ):
                 return None
+            device = getattr(self, "_serve_device", None) or "cuda"
             out = torch.as_tensor(
                 np.ascontiguousarray(
                     data["model-output"].to_numpy(dtype=np.float32)
                 ),
-                device="cuda",
+                device=device,
             )
             yt = torch.as_tensor(
                 np.ascontiguousarray(
                     y.to_numpy(dtype=np.float32)[-len(data):, :]
                 ),
-                device="cuda",
+                device=device,
             )
             scale = torch.as_tensor(
-                np.asarray(self.scaler.scale_, dtype=np.float32), device="cuda"
+                np.asarray(self.scaler.scale_, dtype=np.float32), device=device
             )
             minv = torch.as_tensor(
-                np.asarray(self.scaler.min_, dtype=np.float32), device="cuda"
+                np.asarray(self.scaler.min_, dtype=np.float32), device=device
             )
             ts, tots, tu, totu, _, _ = ops.anomaly_score(
                 out, yt, scale, minv, None, 1.0

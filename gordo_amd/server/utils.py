@@ -267,12 +267,10 @@ def load_model(directory: str, name: str) -> BaseEstimator:
     model = serializer.load(os.path.join(directory, name))
     device = _serving_device_for(name)
     if device is not None:
-        from ..machine.model.models import KerasBaseEstimator
-
         stack = [model]
         while stack:
             obj = stack.pop()
-            if isinstance(obj, KerasBaseEstimator):
+            if hasattr(obj, "set_serving_device"):
                 obj.set_serving_device(device)
             for attr in ("steps", "transformer_list"):
                 for _, step in getattr(obj, attr, []) or []:
