@@ -183,23 +183,31 @@ class DiffBasedAnomalyDetector(AnomalyDetectorBase):
             scaled_mse = self._scaled_mse_per_timestep(split_model, y_true, y_pred)
             mae = self._absolute_error(y_true, y_pred)
 
-            aggregate_threshold_fold = scaled_mse.rolling(6).min().max()
+            aggregate_threshold_fold = model_utils.trail_min_max(
+                scaled_mse.to_numpy(), 6
+            )
             self.aggregate_thresholds_per_fold_[f"fold-{i}"] = aggregate_threshold_fold
 
-            tag_thresholds_fold = mae.rolling(6).min().max()
-            tag_thresholds_fold.name = f"fold-{i}"
+            tag_thresholds_fold = pd.Series(
+                model_utils.trail_min_max(mae.to_numpy(), 6),
+                index=mae.columns,
+                name=f"fold-{i}",
+            )
             self.feature_thresholds_per_fold_ = pd.concat(
                 [self.feature_thresholds_per_fold_, tag_thresholds_fold.to_frame().T]
             )
 
             if self.window is not None:
                 smooth_aggregate_threshold_fold = (
-                    scaled_mse.rolling(self.window).min().max()
+                    model_utils.trail_min_max(scaled_mse.to_numpy(), self.window)
                 )
                 self.smooth_aggregate_thresholds_per_fold_[f"fold-{i}"] = (
                     smooth_aggregate_threshold_fold
                 )
-                smooth_tag_thresholds_fold = mae.rolling(self.window).min().max()
+                smooth_tag_thresholds_fold = pd.Series(
+                    model_utils.trail_min_max(mae.to_numpy(), self.window),
+                    index=mae.columns,
+                )
                 smooth_tag_thresholds_fold.name = f"fold-{i}"
                 self.smooth_feature_thresholds_per_fold_ = pd.concat(
                     [

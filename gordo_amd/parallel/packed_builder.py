@@ -51,6 +51,7 @@ from ..machine.model.anomaly.diff import (
     DiffBasedKFCVAnomalyDetector,
 )
 from ..machine.model.models import _parse_early_stopping, KerasBaseEstimator
+from ..machine.model.utils import trail_min_max as _trail_min_max
 from ..util import disk_registry
 
 logger = logging.getLogger(__name__)
@@ -717,22 +718,6 @@ class PackedFleetBuilder:
                 dataset_meta=p.dataset_meta,
             ),
         )
-
-
-def _trail_min_max(x: np.ndarray, w: int):
-    """``pd.Series/DataFrame(x).rolling(w).min().max()`` in O(n) C time
-    (scipy minimum_filter1d; the pandas rolling kernel is O(n*w) and was
-    ~40% of a 125-machine build step at window 144). origin=(w-1)//2
-    turns the centered filter into the trailing window; the first w-1
-    positions (NaN under pandas) are sliced off before the max, which is
-    exactly pandas' NaN-skipping max. Verified equivalent in
-    tests/test_packed.py::test_trail_min_max_matches_pandas."""
-    from scipy.ndimage import minimum_filter1d
-
-    if x.shape[0] < w:
-        return np.nan if x.ndim == 1 else np.full(x.shape[1], np.nan)
-    mf = minimum_filter1d(x, size=w, axis=0, mode="nearest", origin=(w - 1) // 2)
-    return mf[w - 1:].max(axis=0)
 
 
 def _metric_all_tags(metric, yt: np.ndarray, yp: np.ndarray):
