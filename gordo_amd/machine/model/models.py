@@ -303,9 +303,12 @@ def _parse_early_stopping(callbacks) -> Optional[Dict[str, Any]]:
 
 def _clean_fit_args(args: Dict[str, Any]) -> Dict[str, Any]:
     out = dict(args)
+    # callbacks/validation_split have engine analogs and are passed
+    # explicitly by fit(); the rest have no effect in the pack engine —
+    # say so instead of silently eating them
+    for k in ("callbacks", "validation_split"):
+        out.pop(k, None)
     for k in (
-        "callbacks",
-        "validation_split",
         "class_weight",
         "initial_epoch",
         "steps_per_epoch",
@@ -314,7 +317,10 @@ def _clean_fit_args(args: Dict[str, Any]) -> Dict[str, Any]:
         "workers",
         "use_multiprocessing",
     ):
-        out.pop(k, None)
+        if out.pop(k, None):
+            logger.warning(
+                "fit arg %r is not supported by the pack engine; ignoring", k
+            )
     return out
 
 
