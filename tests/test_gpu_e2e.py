@@ -160,3 +160,53 @@ def test_packed_fleet_builder_on_gpu(tmp_path):
     )
     frame = model.anomaly(X, X)
     assert "total-anomaly-confidence" in {c[0] for c in frame.columns}
+
+
+def test_early_stopping_and_validation_split_on_gpu():
+    """The fit-loop additions (EarlyStopping patience, validation_split
+    holdout + val_loss) behave on the HIP engine exactly as on CPU."""
+    import numpy as np
+
+    from gordo_amd.machine.model.models import KerasAutoEncoder
+
+    X = np.random.RandomState(0).random((256, 12))
+    model = KerasAutoEncoder(
+        kind="feedforward_hourglass", epochs=40, batch_size=64,
+        validation_split=0.25,
+        callbacks=[{"tensorflow.keras.callbacks.EarlyStopping": {
+            "monitor": "val_loss", "patience": 2, "min_delta": 1e9}}],
+    )
+    model.fit(X)
+    hist = model.get_metadata()["history"]
+    assert len(hist["loss"]) == 3  # stopped by patience, not epochs
+    assert len(hist["val_loss"]) == 3
+    assert all(np.isfinite(v) for v in hist["val_loss"])
+    out = model.predict(X)
+    assert out.shape == X.shape
+
+
+def test_serving_response_encoder_on_gpu_box():
+    """The C++ response encoder .so built on the CPU host loads and is
+    byte-identical on the GPU box too (it travels with the snapshot)."""
+    import json
+
+    import numpy as np
+    import pandas as pd
+
+    from gordo_amd.server import _gordo_fastjson as fj
+    from gordo_amd.server import utils as su
+
+    df = pd.DataFrame(
+        np.random.default_rng(0).random((50, 8)),
+        columns=pd.MultiIndex.from_product(
+            (("model-input", "model-output"), [f"t{i}" for i in range(4)])
+        ),
+        index=pd.date_range("2020-01-01", periods=50, freq="10min"),
+    )
+    fast = fj.encode_frame(
+        df.index.astype(str).tolist(),
+        [c[0] for c in df.columns],
+        [c[1] for c in df.columns],
+        df.values,
+    )
+    assert fast == json.dumps(su.dataframe_to_dict(df)).encode()
