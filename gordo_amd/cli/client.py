@@ -63,10 +63,22 @@ def client_cli(ctx, project, host, port, scheme, parallelism, metadata):
     default=None,
     help="Save prediction CSVs (gzip) here, one per target",
 )
+@click.option(
+    "--forward-to-disk",
+    type=click.Path(),
+    default=None,
+    help="Forward each target's predictions as parquet into this dir "
+         "(the ForwardPredictionsToDisk forwarder)",
+)
 @click.pass_context
-def predict_cmd(ctx, start, end, target, output_dir):
+def predict_cmd(ctx, start, end, target, output_dir, forward_to_disk):
     """Anomaly predictions over [START, END) for each target."""
+    from ..client.forwarders import ForwardPredictionsToDisk
+
     client = make_client(ctx.obj)
+    forwarder = (
+        ForwardPredictionsToDisk(forward_to_disk) if forward_to_disk else None
+    )
     results = client.predict(
         start, end, targets=list(target) or None
     )
@@ -77,6 +89,8 @@ def predict_cmd(ctx, start, end, target, output_dir):
             for err in errors:
                 click.echo(f"{name}: {err}", err=True)
             continue
+        if forwarder is not None:
+            forwarder.forward_predictions(frame, name)
         if output_dir:
             path = os.path.join(output_dir, f"{name}.csv.gz")
             frame.to_csv(path, compression="gzip")

@@ -236,12 +236,15 @@ def test_client_cli_predict(cli_env, gordo_name, tmp_path):
 
     from gordo_amd.cli.cli import gordo
 
+    fwd_dir = tmp_path / "fwd"
     out = CliRunner().invoke(
         gordo,
         ["client", "--project", cli_env, "predict",
          "2019-01-01T00:00:00Z", "2019-01-01T06:00:00Z",
-         "--target", gordo_name, "--output-dir", str(tmp_path)],
+         "--target", gordo_name, "--output-dir", str(tmp_path),
+         "--forward-to-disk", str(fwd_dir)],
     )
     assert out.exit_code == 0, out.output
     saved = list(tmp_path.glob("*.csv.gz"))
     assert len(saved) == 1 and gordo_name in saved[0].name
+    assert (fwd_dir / f"{gordo_name}.parquet").is_file()
