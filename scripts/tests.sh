@@ -12,5 +12,7 @@ case "$LANE" in
   gpu)     exec python -m pytest tests/ -q -m gpu ;;
   bench)   exec python -m pytest benchmarks/ -q -s ;;
   doctest) exec python -m pytest tests/test_doctests.py -q ;;
+  lint)    exec python -m pytest tests/test_formatting.py -q ;;
+  sanitize) exec bash scripts/gpu_sanitize.sh ;;
   *) echo "unknown lane: $LANE" >&2; exit 2 ;;
 esac
