@@ -128,8 +128,19 @@ def lstm_seq_fwd_v3(xW, Wh):
 
 
 def lstm_seq_bwd(dSeq, gacts, cs, Wh, last_only):
-    """Fused on-device LSTM backward (BPTT) scan (GPU only, H<=64)."""
-    return _require_hip().lstm_seq_bwd(dSeq, gacts, cs, Wh, last_only)
+    """Fused on-device LSTM backward (BPTT) scan (GPU only, H<=64).
+    ``GORDO_LSTM_V3=1`` selects the pipelined v3 kernel."""
+    import os as _os
+
+    ext = _require_hip()
+    if _os.environ.get("GORDO_LSTM_V3") == "1":
+        return ext.lstm_seq_bwd_v3(dSeq, gacts, cs, Wh, last_only)
+    return ext.lstm_seq_bwd(dSeq, gacts, cs, Wh, last_only)
+
+
+def lstm_seq_bwd_v3(dSeq, gacts, cs, Wh, last_only):
+    """The pipelined backward scan, directly (for A/B tests)."""
+    return _require_hip().lstm_seq_bwd_v3(dSeq, gacts, cs, Wh, last_only)
 
 
 def lstm_seq_available(H: int) -> bool:

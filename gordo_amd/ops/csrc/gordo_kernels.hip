@@ -682,6 +682,9 @@ std::vector<torch::Tensor> anomaly_score(
 namespace gordo_lstm {
 std::vector<torch::Tensor> lstm_seq_fwd(torch::Tensor xW, torch::Tensor Wh);
 std::vector<torch::Tensor> lstm_seq_fwd_v3(torch::Tensor xW, torch::Tensor Wh);
+torch::Tensor lstm_seq_bwd_v3(torch::Tensor dSeq, torch::Tensor gacts,
+                              torch::Tensor cs, torch::Tensor Wh,
+                              bool last_only);
 torch::Tensor lstm_seq_bwd(torch::Tensor dSeq, torch::Tensor gacts,
                            torch::Tensor cs, torch::Tensor Wh,
                            bool last_only);
@@ -690,6 +693,8 @@ torch::Tensor lstm_seq_bwd(torch::Tensor dSeq, torch::Tensor gacts,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("anomaly_score", &gordo_anomaly::anomaly_score,
           "fused DiffBased anomaly scoring (serving hot path)");
+  mod.def("lstm_seq_bwd_v3", &gordo_lstm::lstm_seq_bwd_v3,
+          "pipelined fused LSTM backward scan (GORDO_LSTM_V3 opt-in)");
   mod.def("lstm_seq_fwd_v3", &gordo_lstm::lstm_seq_fwd_v3,
           "pipelined fused LSTM forward scan (GORDO_LSTM_V3 opt-in)");
   mod.def("lstm_seq_fwd", &gordo_lstm::lstm_seq_fwd,

@@ -316,6 +316,196 @@ __global__ __launch_bounds__(256) void lstm_seq_bwd_kernel(
 
 
 // ===========================================================================
+// v3 backward scan — v1 structure plus a software-pipelined prefetch of
+// the NEXT (earlier) timestep's gate activations and cell states: those
+// loads depend only on t, so they issue before the MFMA dh-carry phase
+// and their HBM latency hides under the matrix work + barrier. The bwd
+// workgroup is LDS-bound to ~2 waves/SIMD, so hardware occupancy can't
+// hide the 7 loads per element the gate-backward phase makes — the
+// prefetch buffer (2 packed uints + 2 floats per element slot, <=64
+// VGPRs at the 64-row tile, within the 128-VGPR/2-wave budget) does it
+// in software. dSeq stays un-prefetched: in the flagship autoencoder
+// path (last_only) it is read only at t=T-1. DORMANT until
+// GPU-validated: dispatched via GORDO_LSTM_V3=1.
+// ===========================================================================
+
+template <int ROWS>
+__global__ __launch_bounds__(256) void lstm_seq_bwd_v3_kernel(
+    const bf16* __restrict__ dSeq, const bf16* __restrict__ gacts,
+    const float* __restrict__ cs, const bf16* __restrict__ Wh,
+    bf16* __restrict__ dG, int B, int T, int H, int ldg, int last_only) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int H4 = 4 * H;
+  bf16* WhN = reinterpret_cast<bf16*>(smem);               // [H][ldg] native
+  bf16* dgS = WhN + (size_t)H * ldg;                       // [ROWS][ldg]
+  bf16* dhS = dgS + (size_t)ROWS * ldg;                    // [ROWS][LDK]
+  float* dcS = reinterpret_cast<float*>(dhS + (size_t)ROWS * LDK);  // [ROWS][H]
+
+  const int g = blockIdx.x / ((B + ROWS - 1) / ROWS);
+  const int r0 = (blockIdx.x % ((B + ROWS - 1) / ROWS)) * ROWS;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int l15 = lane & 15;
+  const int kslot = lane >> 4;
+
+  const bf16* Whg = Wh + (size_t)g * H * H4;
+  const bf16* gag = gacts + ((size_t)g * B + r0) * T * H4;
+  const float* csg = cs + ((size_t)g * B + r0) * T * H;
+  const bf16* dSg = last_only ? dSeq + ((size_t)g * B + r0) * H
+                              : dSeq + ((size_t)g * B + r0) * T * H;
+  bf16* dGg = dG + ((size_t)g * B + r0) * T * H4;
+  const int rows_here = min(ROWS, B - r0);
+
+  for (int i = tid; i < H * ldg; i += 256) {
+    int h = i / ldg, n = i % ldg;
+    WhN[i] = (n < H4) ? Whg[(size_t)h * H4 + n] : lf2bf(0.f);
+  }
+  for (int i = tid; i < ROWS * LDK; i += 256) dhS[i] = lf2bf(0.f);
+  for (int i = tid; i < ROWS * H; i += 256) dcS[i] = 0.f;
+  for (int i = tid; i < ROWS * ldg; i += 256) dgS[i] = lf2bf(0.f);
+  __syncthreads();
+
+  // element slots this thread owns in the strided e-loop (H <= 64).
+  // The prefetch buffer is capped at 8 slots (32 VGPRs): the full
+  // 16-slot buffer at the 64-row tile pushed the kernel to 360
+  // combined V+AGPRs -> 1 wave/SIMD, LOSING the LDS-bound 2-wave
+  // occupancy. 8 slots cover the whole e-loop at the 32-row tile and
+  // ~70% of it at 64 rows/H=42; tail slots load v1-style.
+  constexpr int MAXE = (ROWS * 64 + 255) / 256;
+  constexpr int PF = MAXE > 8 ? 8 : MAXE;
+  unsigned int p_if[PF];   // gacts i,f packed
+  unsigned int p_go[PF];   // gacts g,o packed
+  float p_cc[PF];
+  float p_cp[PF];
+
+  auto prefetch_step = [&](int t) {
+    #pragma unroll
+    for (int s = 0; s < PF; ++s) {
+      int e = tid + s * 256;
+      unsigned int v_if = 0x3F003F00u;  // two bf16(0.5)
+      unsigned int v_go = 0x3F000000u;  // low: bf16(0)=g, high: bf16(0.5)=o
+      float cc = 0.f, cp = 0.f;
+      if (e < ROWS * H) {
+        int row = e / H, hh = e % H;
+        if (row < rows_here) {
+          size_t gbase = ((size_t)row * T + t) * H4;
+          v_if = (unsigned int)bf16_bits(gag[gbase + hh]) |
+                 ((unsigned int)bf16_bits(gag[gbase + H + hh]) << 16);
+          v_go = (unsigned int)bf16_bits(gag[gbase + 2 * H + hh]) |
+                 ((unsigned int)bf16_bits(gag[gbase + 3 * H + hh]) << 16);
+          size_t cbase = ((size_t)row * T + t) * H + hh;
+          cc = csg[cbase];
+          cp = (t > 0) ? csg[cbase - H] : 0.f;
+        }
+      }
+      p_if[s] = v_if;
+      p_go[s] = v_go;
+      p_cc[s] = cc;
+      p_cp[s] = cp;
+    }
+  };
+
+  prefetch_step(T - 1);
+
+  for (int t = T - 1; t >= 0; --t) {
+    // ---- fused gate backward: prefetched slots (register-indexed,
+    // unrolled), then the tail slots with v1-style direct loads ----
+    auto gate_bwd = [&](int e, int t_, float i_g, float f_g, float g_g,
+                        float o_g, float cc, float cp) {
+      int row = e / H, hh = e % H;
+      float dh = lbf2f(dhS[row * LDK + hh]);
+      if (row < rows_here) {
+        if (last_only) {
+          if (t_ == T - 1) dh += lbf2f(dSg[(size_t)row * H + hh]);
+        } else {
+          dh += lbf2f(dSg[((size_t)row * T + t_) * H + hh]);
+        }
+      }
+      float tc = fast_tanhf_(cc);
+      float dc = dcS[row * H + hh] + dh * o_g * (1.f - tc * tc);
+      float di = dc * g_g;
+      float df = dc * cp;
+      float dg = dc * i_g;
+      float do_ = dh * tc;
+      dcS[row * H + hh] = dc * f_g;
+      float vi = di * i_g * (1.f - i_g);
+      float vf = df * f_g * (1.f - f_g);
+      float vg = dg * (1.f - g_g * g_g);
+      float vo = do_ * o_g * (1.f - o_g);
+      dgS[row * ldg + hh] = lf2bf(vi);
+      dgS[row * ldg + H + hh] = lf2bf(vf);
+      dgS[row * ldg + 2 * H + hh] = lf2bf(vg);
+      dgS[row * ldg + 3 * H + hh] = lf2bf(vo);
+      if (row < rows_here) {
+        size_t gbase = ((size_t)row * T + t_) * H4;
+        dGg[gbase + hh] = lf2bf(vi);
+        dGg[gbase + H + hh] = lf2bf(vf);
+        dGg[gbase + 2 * H + hh] = lf2bf(vg);
+        dGg[gbase + 3 * H + hh] = lf2bf(vo);
+      }
+    };
+    #pragma unroll
+    for (int s = 0; s < PF; ++s) {
+      int e = tid + s * 256;
+      if (e >= ROWS * H) break;
+      gate_bwd(e, t,
+               lbf2f(bits_bf16((unsigned short)p_if[s])),
+               lbf2f(bits_bf16((unsigned short)(p_if[s] >> 16))),
+               lbf2f(bits_bf16((unsigned short)p_go[s])),
+               lbf2f(bits_bf16((unsigned short)(p_go[s] >> 16))),
+               p_cc[s], p_cp[s]);
+    }
+    for (int e = tid + PF * 256; e < ROWS * H; e += 256) {
+      int row = e / H, hh = e % H;
+      float i_g = 0.5f, f_g = 0.5f, g_g = 0.f, o_g = 0.5f, cc = 0.f,
+            cp = 0.f;
+      if (row < rows_here) {
+        size_t gbase = ((size_t)row * T + t) * H4;
+        i_g = lbf2f(gag[gbase + hh]);
+        f_g = lbf2f(gag[gbase + H + hh]);
+        g_g = lbf2f(gag[gbase + 2 * H + hh]);
+        o_g = lbf2f(gag[gbase + 3 * H + hh]);
+        size_t cbase = ((size_t)row * T + t) * H + hh;
+        cc = csg[cbase];
+        cp = (t > 0) ? csg[cbase - H] : 0.f;
+      }
+      gate_bwd(e, t, i_g, f_g, g_g, o_g, cc, cp);
+    }
+    // issue the next (earlier) step's loads before the MFMA phase
+    if (t > 0) prefetch_step(t - 1);
+    __syncthreads();
+
+    // ---- dh_carry = dgates @ Wh^T (MFMA): out [ROWS][H] ----
+    if (wid < ROWS / 16) {
+      f32x4 acc[4] = {};
+      for (int kk = 0; kk < H4; kk += 32) {
+        bf16x8 a = *reinterpret_cast<const bf16x8*>(
+            &dgS[(wid * 16 + l15) * ldg + kk + kslot * 8]);
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+          int col = fn * 16 + l15;  // h index
+          bf16x8 b = *reinterpret_cast<const bf16x8*>(
+              &WhN[(size_t)min(col, H - 1) * ldg + kk + kslot * 8]);
+          acc[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, b, acc[fn], 0, 0, 0);
+        }
+      }
+      #pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int col = fn * 16 + l15;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = wid * 16 + (lane >> 4) * 4 + r;
+          if (col < H) dhS[row * LDK + col] = lf2bf(acc[fn][r]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// ===========================================================================
 // v3 forward scan — v1 structure plus a software-pipelined double
 // buffer for the x-side gate tile: while the pointwise phase of step t
 // runs (LDS + global stores, no dependence on xW[t+1]), the loads of
@@ -807,6 +997,43 @@ std::vector<torch::Tensor> lstm_seq_fwd_v3(torch::Tensor xW,
                        cs.data_ptr<float>(), (bf16*)gacts.data_ptr(), B, T, H,
                        ldg);
   return {hs, cs, gacts};
+}
+
+
+torch::Tensor lstm_seq_bwd_v3(torch::Tensor dSeq, torch::Tensor gacts,
+                              torch::Tensor cs, torch::Tensor Wh,
+                              bool last_only) {
+  // identical contract to lstm_seq_bwd; pipelined kernel, always the
+  // barriered layout (A/B-tested against v1 on the same shapes)
+  TORCH_CHECK(gacts.is_cuda() && gacts.dim() == 4, "gacts must be [G,B,T,4H]");
+  auto dc = dSeq.to(torch::kBFloat16).contiguous();
+  auto gc = gacts.to(torch::kBFloat16).contiguous();
+  auto cc = cs.to(torch::kFloat32).contiguous();
+  auto Whc = Wh.to(torch::kBFloat16).contiguous();
+  int G = gc.size(0), B = gc.size(1), T = gc.size(2), H4 = gc.size(3);
+  int H = H4 / 4;
+  TORCH_CHECK(H <= 64, "lstm_seq_bwd_v3 supports H <= 64");
+  int ldg = pad_ldg(H4);
+  auto dG = torch::empty_like(gc);
+  int rows = pick_rows(G, B);
+  size_t lds = (size_t)H * ldg * 2 + (size_t)rows * ldg * 2 +
+               (size_t)rows * LDK * 2 + (size_t)rows * H * 4;
+  TORCH_CHECK(lds <= 160 * 1024, "LDS budget exceeded");
+  int blocks = G * ((B + rows - 1) / rows);
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  if (rows == 64)
+    hipLaunchKernelGGL(lstm_seq_bwd_v3_kernel<64>, dim3(blocks), dim3(256),
+                       lds, stream, (const bf16*)dc.data_ptr(),
+                       (const bf16*)gc.data_ptr(), cc.data_ptr<float>(),
+                       (const bf16*)Whc.data_ptr(), (bf16*)dG.data_ptr(), B,
+                       T, H, ldg, last_only ? 1 : 0);
+  else
+    hipLaunchKernelGGL(lstm_seq_bwd_v3_kernel<32>, dim3(blocks), dim3(256),
+                       lds, stream, (const bf16*)dc.data_ptr(),
+                       (const bf16*)gc.data_ptr(), cc.data_ptr<float>(),
+                       (const bf16*)Whc.data_ptr(), (bf16*)dG.data_ptr(), B,
+                       T, H, ldg, last_only ? 1 : 0);
+  return dG;
 }
 
 torch::Tensor lstm_seq_bwd(torch::Tensor dSeq, torch::Tensor gacts,

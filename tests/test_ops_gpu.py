@@ -390,3 +390,27 @@ def test_lstm_seq_v3_matches_v1(G, B, T, H):
     assert torch.equal(v1_hs, v3_hs)
     assert torch.equal(v1_cs, v3_cs)
     assert torch.equal(v1_ga, v3_ga)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("G,B,T,H,last_only", [
+    (2, 64, 12, 38, False),
+    (1, 100, 20, 42, True),
+    (3, 32, 144, 25, True),
+    (4, 130, 50, 60, False),
+])
+def test_lstm_seq_bwd_v3_matches_v1(G, B, T, H, last_only):
+    """The pipelined v3 backward scan is bit-identical to v1 — same
+    math, the gate/cell loads are just double-buffered."""
+    require_hip()
+    H4 = 4 * H
+    xW = to_dev_bf16(_rand(G, B, T, H4, seed=50))
+    Wh = to_dev_bf16(_rand(G, H, H4, seed=51) * 0.3)
+    hs, cs, ga = ops.lstm_seq_fwd(xW, Wh)
+    if last_only:
+        dSeq = to_dev_bf16(_rand(G, B, H, seed=52))
+    else:
+        dSeq = to_dev_bf16(_rand(G, B, T, H, seed=52))
+    v1 = ops.lstm_seq_bwd(dSeq, ga, cs, Wh, last_only)
+    v3 = ops.lstm_seq_bwd_v3(dSeq, ga, cs, Wh, last_only)
+    assert torch.equal(v1, v3)
