@@ -15,10 +15,18 @@ torch.cuda.init()
 G,B,T,H = 8, 512, 144, 42
 xW = torch.randn(G,B,T,4*H, device="cuda", dtype=torch.bfloat16)
 Wh = torch.randn(G,H,4*H, device="cuda", dtype=torch.bfloat16)*0.1
-for name, fn in (("v1", ops.lstm_seq_fwd), ("v3", ops.lstm_seq_fwd_v3)):
+for name, fn in (("fwd v1", ops.lstm_seq_fwd), ("fwd v3", ops.lstm_seq_fwd_v3)):
     for _ in range(3): fn(xW, Wh)
     torch.cuda.synchronize(); t0=time.perf_counter()
     for _ in range(20): fn(xW, Wh)
+    torch.cuda.synchronize()
+    print(name, f"{(time.perf_counter()-t0)/20*1e3:.3f} ms")
+hs, cs, ga = ops.lstm_seq_fwd(xW, Wh)
+dSeq = torch.randn(G,B,H, device="cuda", dtype=torch.bfloat16)
+for name, fn in (("bwd v1", ops.lstm_seq_bwd), ("bwd v3", ops.lstm_seq_bwd_v3)):
+    for _ in range(3): fn(dSeq, ga, cs, Wh, True)
+    torch.cuda.synchronize(); t0=time.perf_counter()
+    for _ in range(20): fn(dSeq, ga, cs, Wh, True)
     torch.cuda.synchronize()
     print(name, f"{(time.perf_counter()-t0)/20*1e3:.3f} ms")
 PY
