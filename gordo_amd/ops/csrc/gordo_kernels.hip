@@ -143,10 +143,18 @@ __global__ __launch_bounds__(256) void grouped_gemm_kernel(
       int kk = (tid & 3) * 8;
       int gm = m0 + row;
       bf16 v[8];
-      #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        int gk = k0 + kk + e;
-        v[e] = (gm < M && gk < K) ? Ag[(size_t)gm * K + gk] : f2bf(0.f);
+      if ((K & 7) == 0 && gm < M && k0 + kk + 8 <= K) {
+        // K 8-aligned => every row's 8-element k-slice is one 16-byte
+        // naturally-aligned global load (guide G13; the scalar loop
+        // below is the unaligned-K fallback)
+        *reinterpret_cast<bf16x8*>(v) = *reinterpret_cast<const bf16x8*>(
+            &Ag[(size_t)gm * K + k0 + kk]);
+      } else {
+        #pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          int gk = k0 + kk + e;
+          v[e] = (gm < M && gk < K) ? Ag[(size_t)gm * K + gk] : f2bf(0.f);
+        }
       }
       lds_store8(&As[row * LDT + kk], v);
     }
@@ -169,10 +177,15 @@ __global__ __launch_bounds__(256) void grouped_gemm_kernel(
       int kk = (tid & 3) * 8;
       int gn = n0 + n;
       bf16 v[8];
-      #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        int gk = k0 + kk + e;
-        v[e] = (gn < N && gk < K) ? Bg[(size_t)gn * K + gk] : f2bf(0.f);
+      if ((K & 7) == 0 && gn < N && k0 + kk + 8 <= K) {
+        *reinterpret_cast<bf16x8*>(v) = *reinterpret_cast<const bf16x8*>(
+            &Bg[(size_t)gn * K + k0 + kk]);
+      } else {
+        #pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          int gk = k0 + kk + e;
+          v[e] = (gn < N && gk < K) ? Bg[(size_t)gn * K + gk] : f2bf(0.f);
+        }
       }
       lds_store8(&Bs[n * LDT + kk], v);
     }
