@@ -542,3 +542,43 @@ def test_build_insufficient_data_exit_code(tmp_path):
     assert json.loads(report_file.read_text())["type"] == (
         "InsufficientDataError"
     )
+
+
+def test_workflow_all_options_combination(tmp_path):
+    """Everything on at once renders to valid YAML: gpu-fleet + KEDA +
+    security contexts + resource labels + custom envs + revision."""
+    cfg = _wf_config(
+        tmp_path,
+        n=3,
+        globals_yaml="""
+globals:
+  runtime: |
+    pod_security_context:
+      runAsUser: 1000
+    security_context:
+      runAsNonRoot: true
+    builder:
+      resources:
+        requests: {memory: 1000, cpu: 100}
+        limits: {memory: 2000, cpu: 200}
+""",
+    )
+    result = CliRunner().invoke(
+        gordo,
+        ["workflow", "generate", "--machine-config", str(cfg),
+         "--project-name", "wf-all",
+         "--project-revision", "1600000000000",
+         "--gpu-fleet", "--n-gpus", "8",
+         "--keda-enabled",
+         "--builder-retries", "3",
+         "--resource-labels", '{"team": "mlops"}',
+         "--custom-model-builder-envs",
+         '[{"name": "A", "value": "1"}, {"name": "B", "value": "2"}]'],
+    )
+    assert result.exit_code == 0, result.output
+    docs = list(yaml.safe_load_all(result.output))
+    assert docs[0]["kind"] == "Workflow"
+    out = result.output
+    for frag in ("fleet-build", "keda.sh/v1alpha1", "runAsUser: 1000",
+                 "runAsNonRoot: true", "team: mlops", "1600000000000"):
+        assert frag in out, frag
