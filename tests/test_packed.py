@@ -422,3 +422,50 @@ def test_packed_diffbased_thresholds_match_modelbuilder(tmp_path):
             atol=1e-4,
             err_msg=key,
         )
+
+
+def test_packed_build_with_callbacks_and_val_split(tmp_path):
+    """Machines configuring EarlyStopping + validation_split build
+    through the packed path (fit args carry into the engine group)."""
+    from gordo_amd.parallel import PackedFleetBuilder
+    from gordo_amd.workflow import NormalizedConfig
+
+    cfg = {
+        "machines": [
+            {
+                "name": f"es-m-{i}",
+                "dataset": {
+                    "type": "SineWaveDataset",
+                    "tag_list": ["a", "b", "c"],
+                    "train_start_date": "2019-01-01T00:00:00+00:00",
+                    "train_end_date": "2019-01-02T00:00:00+00:00",
+                },
+                "model": {
+                    "gordo_amd.machine.model.models.KerasAutoEncoder": {
+                        "kind": "feedforward_hourglass",
+                        "epochs": 30,
+                        "validation_split": 0.2,
+                        "callbacks": [{
+                            "tensorflow.keras.callbacks.EarlyStopping": {
+                                "monitor": "val_loss", "patience": 1,
+                                "min_delta": 1e9,
+                            }
+                        }],
+                    }
+                },
+            }
+            for i in range(2)
+        ]
+    }
+    norm = NormalizedConfig(cfg, project_name="p")
+    results = dict(
+        PackedFleetBuilder(norm.machines, save_models=False).build_all()
+    )
+    assert all(not isinstance(v, BaseException) for v in results.values()), {
+        k: repr(v) for k, v in results.items()
+    }
+    hist = results["es-m-0"].metadata.build_metadata.model.model_meta[
+        "history"]
+    # stopped after 2 epochs (patience 1, never "improving"), not 30
+    assert len(hist["loss"]) == 2
+    assert "val_loss" in hist and len(hist["val_loss"]) == 2
