@@ -35,6 +35,9 @@ def main():
     ap.add_argument("--endpoint", default="anomaly",
                     choices=["anomaly", "prediction", "both"])
     ap.add_argument("--rows", type=int, default=ROWS)
+    ap.add_argument("--format", default="json", choices=["json", "parquet"],
+                    help="request/response content type (parquet bypasses "
+                         "the JSON codec entirely)")
     ap.add_argument("--n-models", type=int, default=1,
                     help="serve N distinct models round-robin (the "
                          "many-model LRU/HBM-residency story)")
@@ -116,6 +119,11 @@ def main():
         X = pd.DataFrame(np.random.random((args.rows, N_TAGS)),
                          columns=sensors)
         payload = {"X": dataframe_to_dict(X), "y": dataframe_to_dict(X)}
+        parquet_blob = None
+        if args.format == "parquet":
+            from gordo_amd.server.utils import dataframe_into_parquet_bytes
+
+            parquet_blob = dataframe_into_parquet_bytes(X)
         results = {}
         for endpoint in (
             ["anomaly", "prediction"] if args.endpoint == "both"
@@ -131,9 +139,21 @@ def main():
                 for name in model_names
             ]
             url = urls[0]
+            def post(c, u):
+                if args.format == "parquet":
+                    import io
+
+                    return c.post(
+                        u + "?format=parquet",
+                        data={"X": (io.BytesIO(parquet_blob), "X"),
+                              "y": (io.BytesIO(parquet_blob), "y")},
+                        content_type="multipart/form-data",
+                    )
+                return c.post(u, json=payload)
+
             # warmup + correctness
             client = app.test_client()
-            resp = client.post(url, json=payload)
+            resp = post(client, url)
             assert resp.status_code == 200, resp.data[:300]
 
             latencies = []
@@ -149,7 +169,7 @@ def main():
                         counter["n"] += 1
                         my_url = urls[counter["n"] % len(urls)]
                     t0 = time.perf_counter()
-                    r = c.post(my_url, json=payload)
+                    r = post(c, my_url)
                     dt = time.perf_counter() - t0
                     assert r.status_code == 200
                     with lock:
@@ -171,6 +191,7 @@ def main():
                 "mean_latency_ms": statistics.mean(latencies) * 1000,
                 "p50_latency_ms": statistics.median(latencies) * 1000,
                 "rounds": len(latencies),
+                "format": args.format,
                 "threads": args.threads,
                 "payload_rows": args.rows,
                 "n_tags": N_TAGS,
