@@ -34,6 +34,8 @@ PY
 # 3) bench with the CPU-phase fixes (thresholds O(n), frame fast path)
 timeout 1200 python bench.py --gpus 1 --steps 3 --warmup 1 2>&1 | tail -2
 
-# 4) serving with the C++ JSON encoder
+# 4) serving with the C++ JSON encoder, then the codec-free parquet mode
 timeout 600 python scripts/bench_serving.py --rounds 60 --threads 8 \
   --endpoint both 2>/dev/null | tail -1
+timeout 600 python scripts/bench_serving.py --rounds 60 --threads 8 \
+  --endpoint both --format parquet 2>/dev/null | tail -1
