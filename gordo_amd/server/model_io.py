@@ -9,8 +9,14 @@ logger = logging.getLogger(__name__)
 
 
 def get_model_output(model, X: np.ndarray) -> np.ndarray:
-    """Predict, falling back to transform."""
+    """Predict, falling back to transform. With GORDO_SERVE_BATCH=1,
+    concurrent predictions for the same model coalesce through the
+    micro-batcher (server/batcher.py)."""
     try:
+        from . import batcher
+
+        if batcher.enabled() and hasattr(model, "predict"):
+            return batcher.batched_predict(model, X)
         return model.predict(X)
     except AttributeError:
         try:
