@@ -41,6 +41,9 @@ def main():
     ap.add_argument("--n-models", type=int, default=1,
                     help="serve N distinct models round-robin (the "
                          "many-model LRU/HBM-residency story)")
+    ap.add_argument("--serve-batch", action="store_true",
+                    help="enable the per-model micro-batcher "
+                         "(GORDO_SERVE_BATCH=1)")
     ap.add_argument("--direct", action="store_true",
                     help="measure model.anomaly() directly (no HTTP/JSON): "
                          "the batched inference engine path, BASELINE "
@@ -111,6 +114,8 @@ def main():
             }))
             return
 
+        if args.serve_batch:
+            os.environ["GORDO_SERVE_BATCH"] = "1"
         from gordo_amd.server.server import build_app
 
         app = build_app()

@@ -39,3 +39,5 @@ timeout 600 python scripts/bench_serving.py --rounds 60 --threads 8 \
   --endpoint both 2>/dev/null | tail -1
 timeout 600 python scripts/bench_serving.py --rounds 60 --threads 8 \
   --endpoint both --format parquet 2>/dev/null | tail -1
+timeout 600 python scripts/bench_serving.py --rounds 60 --threads 8 \
+  --endpoint both --serve-batch 2>/dev/null | tail -1
