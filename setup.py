@@ -74,6 +74,7 @@ setup(
     package_data={
         "gordo_amd.workflow.workflow_generator": ["resources/*.template"],
         "gordo_amd.ops": ["csrc/*"],
+        "gordo_amd.server": ["csrc/*"],
     },
     python_requires=">=3.10",
     entry_points={"console_scripts": ["gordo=gordo_amd.cli:gordo"]},
