@@ -102,7 +102,15 @@ def trail_min_max(x, w: int):
     centered filter into the trailing window; the first w-1 positions
     (NaN under pandas) are sliced off before the max — exactly pandas'
     NaN-skipping max. Exact-equivalence tested in
-    tests/test_packed.py::test_trail_min_max_matches_pandas."""
+    tests/test_packed.py::test_trail_min_max_matches_pandas.
+
+    >>> import numpy as np, pandas as pd
+    >>> a = np.array([5.0, 1.0, 4.0, 2.0, 8.0, 0.5])
+    >>> float(trail_min_max(a, 3))
+    2.0
+    >>> float(pd.Series(a).rolling(3).min().max())
+    2.0
+    """
     import numpy as np
     from scipy.ndimage import minimum_filter1d
 

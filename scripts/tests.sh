@@ -9,6 +9,7 @@ set -e
 LANE="${1:-cpu}"
 case "$LANE" in
   cpu)     exec python -m pytest tests/ -q -m "not gpu" ;;
+  cpu-par)  exec python -m pytest tests/ -q -m "not gpu" -n auto ;;
   gpu)     exec python -m pytest tests/ -q -m gpu ;;
   bench)   exec python -m pytest benchmarks/ -q -s ;;
   doctest) exec python -m pytest tests/test_doctests.py -q ;;
