@@ -45,6 +45,16 @@ from ..util import disk_registry
 logger = logging.getLogger(__name__)
 
 
+def _has_predict(model) -> bool:
+    """hasattr(model, "predict") without sklearn-1.7's unfitted-Pipeline
+    FutureWarning (accessing .predict on an unfitted Pipeline warns)."""
+    import warnings
+
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore", FutureWarning)
+        return hasattr(model, "predict")
+
+
 class ModelBuilder:
     def __init__(self, machine: Machine):
         self.machine = machine
@@ -131,7 +141,7 @@ class ModelBuilder:
             metrics_list = self.metrics_from_list(
                 self.machine.evaluation.get("metrics")
             )
-            if hasattr(model, "predict"):
+            if _has_predict(model):
                 start = time.time()
                 scaler = self.machine.evaluation.get("scoring_scaler")
                 metrics_dict = self.build_metrics_dict(metrics_list, y, scaler=scaler)
@@ -282,7 +292,7 @@ class ModelBuilder:
     @staticmethod
     def _determine_offset(model: BaseEstimator, X) -> int:
         X = getattr(X, "values", X)
-        out = model.predict(X) if hasattr(model, "predict") else model.transform(X)
+        out = model.predict(X) if _has_predict(model) else model.transform(X)
         return len(X) - len(out)
 
     @staticmethod

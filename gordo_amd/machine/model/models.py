@@ -216,6 +216,13 @@ class KerasBaseEstimator(BaseEstimator, GordoBase):
             self._serve_device = device
             self._pack = None
 
+    def __sklearn_is_fitted__(self) -> bool:
+        """Let sklearn's check_is_fitted see the engine-backed fitted
+        state (no trailing-underscore attrs here, so without this a
+        fitted Pipeline ending in this estimator warns in sklearn 1.7
+        and would raise in 1.8)."""
+        return self._weights is not None or self._pack is not None
+
     def _ensure_pack(self):
         if self._pack is not None:
             return self._pack
