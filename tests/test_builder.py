@@ -196,7 +196,10 @@ def test_local_build(config_str):
         assert machine.project_name == "local-build"
 
 
-class _ScalingRegressor:
+from sklearn.base import BaseEstimator as _SkBase
+
+
+class _ScalingRegressor(_SkBase):
     """Predicts X * multiplier — used to probe per-feature error scaling."""
 
     def __init__(self, multiplier):
