@@ -2,15 +2,11 @@ import logging
 import os
 import sys
 import tempfile
-import warnings
 
 import numpy as np
 import pytest
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
-
-warnings.filterwarnings("ignore", category=UserWarning, module="sklearn")
-warnings.filterwarnings("ignore", category=FutureWarning, module="sklearn")
 
 SENSORS = [f"tag-{i}" for i in range(4)]
 GORDO_NAME = "machine-1"
