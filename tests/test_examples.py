@@ -13,7 +13,8 @@ EXAMPLES = os.path.join(
 
 @pytest.mark.timeout(300)
 @pytest.mark.parametrize(
-    "script", ["minimal_build_and_serve.py", "fleet_and_client.py"]
+    "script", ["minimal_build_and_serve.py", "fleet_and_client.py",
+               "training_controls.py"]
 )
 def test_example_runs(script):
     proc = subprocess.run(
