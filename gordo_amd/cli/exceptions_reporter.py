@@ -54,6 +54,7 @@ class ExceptionsReporter:
     ) -> List[Tuple[Type[Exception], int]]:
         """Sort so subclasses are found before their bases
         (inheritance-aware, reference :62-77)."""
+        exceptions = list(exceptions)  # may be a generator; reused 3x
         inheritance_levels: Dict[Type[BaseException], int] = Counter()
         for exc, _ in exceptions:
             for e, _ in exceptions:
