@@ -10,10 +10,14 @@ def normalize_sensor_tags(
     build_dataset_metadata: Dict[str, Any],
     tag_list: List,
     asset: Optional[str] = None,
+    **_extra_fields,
 ) -> List[SensorTag]:
     """
     Resolve full SensorTags for ``tag_list`` using tag metadata recorded
     in the dataset build metadata (spec: gordo/utils.py:15-50).
+    Extra default-tag fields beyond ``asset`` (the reference's
+    gordo-core tags carry more) are accepted and ignored — SensorTag
+    here keeps only name/asset.
     """
     tags_meta: Dict[str, Dict[str, Any]] = (
         (build_dataset_metadata or {})
