@@ -599,3 +599,39 @@ def test_set_serving_device_cpu_roundtrip():
     model.set_serving_device("cpu")
     after = model.predict(X)
     np.testing.assert_allclose(before, after, rtol=1e-5, atol=1e-6)
+
+
+def test_prediction_model_error_returns_400(
+    flask_app, gordo_project, model_collection_directory, sensors
+):
+    """A model that raises during predict/transform maps to a 400 with
+    an error body (reference base.py:30-113 error branches)."""
+    import shutil
+
+    from sklearn.preprocessing import FunctionTransformer
+
+    from gordo_amd import serializer
+
+    src = os.path.join(model_collection_directory, "machine-1")
+    broken = os.path.join(model_collection_directory, "broken-model")
+    shutil.copytree(src, broken)
+    # transform = matrix inverse of a non-square frame -> LinAlgError
+    # (a ValueError subclass), exercising the 400 branch
+    serializer.dump(FunctionTransformer(np.linalg.inv), broken)
+
+    client = flask_app.test_client()
+    X = pd.DataFrame(
+        np.random.default_rng(0).random((10, len(sensors))),
+        columns=sensors,
+        index=pd.date_range("2019-01-01", periods=10, freq="10min", tz="UTC"),
+    )
+    try:
+        resp = client.post(
+            f"/gordo/v0/{gordo_project}/broken-model/prediction",
+            json={"X": server_utils.dataframe_to_dict(X)},
+        )
+        assert resp.status_code == 400
+        assert "error" in resp.json
+    finally:
+        shutil.rmtree(broken)
+        server_utils.load_model.cache_clear()
