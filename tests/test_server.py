@@ -635,3 +635,40 @@ def test_prediction_model_error_returns_400(
     finally:
         shutil.rmtree(broken)
         server_utils.load_model.cache_clear()
+
+
+def test_anomaly_on_non_detector_422(
+    flask_app, gordo_project, model_collection_directory, sensors
+):
+    """/anomaly/prediction on a model without .anomaly -> 422
+    (reference anomaly.py:51-55)."""
+    import shutil
+
+    from sklearn.preprocessing import StandardScaler
+
+    from gordo_amd import serializer
+
+    src = os.path.join(model_collection_directory, "machine-1")
+    plain = os.path.join(model_collection_directory, "plain-model")
+    shutil.copytree(src, plain)
+    scaler = StandardScaler()
+    scaler.fit(np.random.default_rng(0).random((20, len(sensors))))
+    serializer.dump(scaler, plain)
+
+    client = flask_app.test_client()
+    X = pd.DataFrame(
+        np.random.default_rng(1).random((10, len(sensors))),
+        columns=sensors,
+        index=pd.date_range("2019-01-01", periods=10, freq="10min", tz="UTC"),
+    )
+    body = {"X": server_utils.dataframe_to_dict(X),
+            "y": server_utils.dataframe_to_dict(X)}
+    try:
+        resp = client.post(
+            f"/gordo/v0/{gordo_project}/plain-model/anomaly/prediction",
+            json=body,
+        )
+        assert resp.status_code == 422
+    finally:
+        shutil.rmtree(plain)
+        server_utils.load_model.cache_clear()
