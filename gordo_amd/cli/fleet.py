@@ -66,11 +66,12 @@ def fleet_build(
             gpus = torch.cuda.device_count() if torch.cuda.is_available() else 1
         if gpus > 1:
             # relaunch under torch.distributed.run, one rank per GPU
+            # standalone rendezvous picks a free port (a fixed port can
+            # collide with another torchrun on a shared box)
             cmd = [
                 sys.executable, "-m", "torch.distributed.run",
                 "--nnodes=1", f"--nproc-per-node={gpus}",
-                "--master-addr", "127.0.0.1",
-                "--master-port", os.environ.get("MASTER_PORT", "29517"),
+                "--standalone", "--local-addr", "127.0.0.1",
                 "-m", "gordo_amd.cli.fleet_worker",
             ]
             env = dict(os.environ)

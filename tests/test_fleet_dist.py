@@ -173,3 +173,31 @@ def test_requeue_failed_rebuilds_missing(tmp_path):
     assert sorted(summary["requeued"]) == ["fleet-m-1", "fleet-m-2"]
     for i in range(3):
         assert (out_dir / f"fleet-m-{i}" / "model.pkl").is_file()
+
+
+@pytest.mark.timeout(300)
+def test_fleet_cli_multi_rank_with_requeue(tmp_path):
+    """`gordo fleet build --gpus 2` end to end through the CLI: torchrun
+    relaunch, default status file, requeue no-op on success."""
+    config = CONFIG.format(
+        machines="".join(MACHINE_TMPL.format(i=i) for i in range(2))
+    )
+    cfg_path = tmp_path / "cfg.yml"
+    cfg_path.write_text(config)
+    out_dir = tmp_path / "models"
+
+    env = dict(os.environ)
+    env.update(GORDO_DIST_BACKEND="gloo", CUDA_VISIBLE_DEVICES="")
+    proc = subprocess.run(
+        [sys.executable, "-m", "gordo_amd", "fleet", "build",
+         "--machine-config", str(cfg_path),
+         "--project-name", "cli-fleet",
+         "--output-dir", str(out_dir),
+         "--gpus", "2"],
+        env=env, capture_output=True, text=True, timeout=280,
+    )
+    assert proc.returncode == 0, proc.stderr[-2500:]
+    summary = json.loads((out_dir / ".fleet-status.json").read_text())
+    assert summary["n_ok"] == 2, summary
+    for i in range(2):
+        assert (out_dir / f"fleet-m-{i}" / "model.pkl").is_file()
