@@ -1,9 +1,7 @@
-import logging
 import os
 import sys
 import tempfile
 
-import numpy as np
 import pytest
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))

@@ -2,7 +2,6 @@
 import threading
 
 import numpy as np
-import pytest
 
 from gordo_amd.server.batcher import MicroBatcher
 

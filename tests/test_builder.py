@@ -391,7 +391,6 @@ def test_cv_mode_build_only(tmp_path):
     """build_only: model saved, no CV run, scores empty (reference
     tests/gordo/cli/test_cli.py::test_build_cv_mode_build_only)."""
     import json
-    import os
 
     machine = make_machine(
         model=SKLEARN_MODEL, evaluation={"cv_mode": "build_only"}

@@ -1,5 +1,4 @@
 import json
-import os
 
 import pytest
 import yaml
@@ -7,7 +6,7 @@ from click.testing import CliRunner
 
 from gordo_amd.cli import gordo
 from gordo_amd.cli.cli import expand_model, get_all_score_strings
-from gordo_amd.cli.exceptions_reporter import ExceptionsReporter, ReportLevel
+from gordo_amd.cli.exceptions_reporter import ExceptionsReporter
 
 
 MACHINE_JSON = {

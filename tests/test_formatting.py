@@ -6,7 +6,6 @@ imports outside __init__ re-export modules)."""
 import ast
 import pathlib
 
-import pytest
 
 PKG = pathlib.Path(__file__).resolve().parent.parent / "gordo_amd"
 

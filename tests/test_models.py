@@ -12,7 +12,6 @@ from gordo_amd.machine.model import (
 from gordo_amd.machine.model.factories import (
     feedforward_hourglass,
     feedforward_model,
-    lstm_hourglass,
     lstm_model,
 )
 from gordo_amd.machine.model.factories.utils import hourglass_calc_dims
