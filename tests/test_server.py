@@ -672,3 +672,23 @@ def test_anomaly_on_non_detector_422(
     finally:
         shutil.rmtree(plain)
         server_utils.load_model.cache_clear()
+
+
+def test_frame_json_response_without_extension(monkeypatch):
+    """Python-codec fallback when the C++ encoder is unavailable."""
+    from flask import Flask
+
+    from gordo_amd.server import utils as su
+
+    monkeypatch.setattr(su, "_gordo_fastjson", None)
+    df = pd.DataFrame(
+        [[1.0, 2.0]],
+        columns=pd.MultiIndex.from_tuples([("a", "x"), ("a", "y")]),
+        index=["i"],
+    )
+    app = Flask("t2")
+    with app.test_request_context():
+        resp = su.frame_json_response({"extra": "1"}, df)
+    body = json.loads(resp.get_data())
+    assert body["data"]["a"] == {"x": {"i": 1.0}, "y": {"i": 2.0}}
+    assert body["extra"] == "1"
