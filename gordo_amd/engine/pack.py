@@ -221,7 +221,8 @@ class BasePack:
     # destructor intermittently aborts the process with "The graph
     # should be registered to the state" (HIPGeneratorImpl.cpp:158)
     # when graphs are garbage-collected in long multi-pack processes.
-    _graph_enabled = os.environ.get("GORDO_HIPGRAPH", "0") == "1"
+    _graph_enabled = True  # env re-checked per step; instance sets
+    # this False permanently after a capture failure
     # hip stream-capture mode is process-global: captures from two
     # threads at once crash the process. One capture at a time.
     _graph_capture_lock = __import__("threading").Lock()
@@ -230,7 +231,14 @@ class BasePack:
         """Replay (capturing on first use) the train_batch graph for
         this batch shape. Returns the live loss tensor, or None when
         capture is unavailable for this configuration."""
-        if not (self._graph_enabled and self.device.type == "cuda"):
+        if not (
+            self._graph_enabled
+            and os.environ.get("GORDO_HIPGRAPH", "0") == "1"
+            and self.device.type == "cuda"
+        ):
+            # _graph_enabled False (set per-instance after a capture
+            # failure) wins; otherwise the env is re-read so flipping
+            # GORDO_HIPGRAPH at runtime takes effect
             return None
         key = (tuple(Xb.shape), tuple(Tb.shape))
         cached = getattr(self, "_graph_cache", None)
