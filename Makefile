@@ -15,5 +15,11 @@ bench:
 serve-bench:
 	python scripts/bench_serving.py --rounds 200 --threads 8 --endpoint both
 
+lint:
+	python -m pytest tests/test_formatting.py -q
+
+sanitize:
+	bash scripts/gpu_sanitize.sh
+
 docker:
 	docker build -t gordo-amd .
