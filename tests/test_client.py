@@ -248,3 +248,17 @@ def test_client_cli_predict(cli_env, gordo_name, tmp_path):
     saved = list(tmp_path.glob("*.csv.gz"))
     assert len(saved) == 1 and gordo_name in saved[0].name
     assert (fwd_dir / f"{gordo_name}.parquet").is_file()
+
+
+def test_client_predict_all_targets(client, gordo_name, second_gordo_name):
+    """predict() without explicit targets covers every served model."""
+    import dateutil.parser
+
+    start = dateutil.parser.isoparse("2019-01-01T00:00:00Z")
+    end = dateutil.parser.isoparse("2019-01-01T06:00:00Z")
+    results = client.predict(start, end)
+    names = {name for name, _, _ in results}
+    assert {gordo_name, second_gordo_name} <= names
+    for name, frame, errors in results:
+        assert not errors, (name, errors)
+        assert len(frame) > 0
