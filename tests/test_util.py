@@ -290,3 +290,15 @@ def test_prometheus_gpu_gauges():
     )
     assert cached is not None and cached >= 0
     assert gpu_mem == 0.0  # no GPU in the CPU test lane
+
+
+def test_capture_args_records_init_params():
+    from gordo_amd.util.utils import capture_args
+
+    class Thing:
+        @capture_args
+        def __init__(self, a, b=2, **kw):
+            pass
+
+    t = Thing(1, b=3, extra="x")
+    assert t._params == {"a": 1, "b": 3, "extra": "x"}
