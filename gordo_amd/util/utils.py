@@ -25,9 +25,12 @@ def capture_args(init):
         bound.apply_defaults()
         params = dict(bound.arguments)
         params.pop("self", None)
-        if "kwargs" in params and isinstance(params["kwargs"], dict):
-            extra = params.pop("kwargs")
-            params.update(extra)
+        # flatten the VAR_KEYWORD parameter whatever its name
+        for pname, p in sig.parameters.items():
+            if p.kind is inspect.Parameter.VAR_KEYWORD and pname in params:
+                extra = params.pop(pname)
+                if isinstance(extra, dict):
+                    params.update(extra)
         self._params = params
         return init(self, *args, **kwargs)
 
