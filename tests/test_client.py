@@ -262,3 +262,16 @@ def test_client_predict_all_targets(client, gordo_name, second_gordo_name):
     for name, frame, errors in results:
         assert not errors, (name, errors)
         assert len(frame) > 0
+
+
+def test_client_cli_metadata_all_targets(cli_env, gordo_name,
+                                         second_gordo_name):
+    from click.testing import CliRunner
+
+    from gordo_amd.cli.cli import gordo
+
+    out = CliRunner().invoke(
+        gordo, ["client", "--project", cli_env, "metadata"]
+    )
+    assert out.exit_code == 0, out.output
+    assert gordo_name in out.output and second_gordo_name in out.output
