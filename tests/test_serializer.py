@@ -188,3 +188,16 @@ def test_into_definition_captures_kwargs():
     params = step[key]
     assert params["n_components"] == 3
     assert params["whiten"] is True
+
+
+def test_info_json_checksum_content(tmp_path):
+    """info.json carries a checksum over the pickled model (the
+    reference's cache-integrity breadcrumb)."""
+    from sklearn.preprocessing import MinMaxScaler
+
+    serializer.dump(
+        MinMaxScaler(), str(tmp_path),
+        info={"checksum": "abc123"},
+    )
+    info = serializer.load_info(str(tmp_path))
+    assert info == {"checksum": "abc123"}
