@@ -22,6 +22,10 @@ import click
 
 logger = logging.getLogger(__name__)
 
+# default status-file name when the caller gives none; lives inside the
+# output dir and is ignored by the server's /models listing (dirs only)
+FLEET_STATUS_BASENAME = ".fleet-status.json"
+
 
 @click.group("fleet")
 def fleet_cli():
@@ -84,7 +88,7 @@ def fleet_build(
             if model_register_dir:
                 env["GORDO_FLEET_MODEL_REGISTER_DIR"] = model_register_dir
             if status_file is None:
-                status_file = os.path.join(output_dir, ".fleet-status.json")
+                status_file = os.path.join(output_dir, FLEET_STATUS_BASENAME)
             env["GORDO_FLEET_STATUS_FILE"] = status_file
             os.makedirs(output_dir, exist_ok=True)
             rc = subprocess.call(cmd, env=env)
