@@ -253,3 +253,24 @@ def test_sinewave_load_frame_bit_identical():
     assert list(X1.columns) == list(X2.columns)
     assert (X1.index == X2.index).all()
     assert (y1.values == y2.values).all()
+
+
+def test_machine_equality_and_repr():
+    from gordo_amd.machine import Machine
+
+    cfg = {
+        "name": "eq-m",
+        "model": {"sklearn.decomposition.PCA": {"n_components": 2}},
+        "dataset": {
+            "type": "RandomDataset",
+            "tag_list": ["a", "b", "c"],
+            "train_start_date": "2019-01-01T00:00:00Z",
+            "train_end_date": "2019-01-02T00:00:00Z",
+        },
+    }
+    m1 = Machine.from_config(cfg, project_name="p")
+    m2 = Machine.from_config(cfg, project_name="p")
+    assert m1 == m2
+    m3 = Machine.from_config({**cfg, "name": "eq-n"}, project_name="p")
+    assert m1 != m3
+    assert "eq-m" in repr(m1) or "eq-m" in str(m1.to_dict())
