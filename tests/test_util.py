@@ -302,3 +302,10 @@ def test_capture_args_records_init_params():
 
     t = Thing(1, b=3, extra="x")
     assert t._params == {"a": 1, "b": 3, "extra": "x"}
+
+
+def test_replace_all_non_ascii_chars_more():
+    from gordo_amd.util.text import replace_all_non_ascii_chars
+
+    assert replace_all_non_ascii_chars("køl-æble", "-") == "k-l--ble"
+    assert replace_all_non_ascii_chars("plain", "_") == "plain"
