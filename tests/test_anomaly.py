@@ -221,3 +221,21 @@ def test_diff_detector_metadata_lifecycle(mode):
             assert key in md, key
     if mode == "kfcv":
         assert "threshold-percentile" in md
+
+
+def test_anomaly_frequency_parameter(Xy):
+    """anomaly(..., frequency=) sets the `end` column spacing (the
+    serving path passes the dataset resolution)."""
+    import pandas as pd
+
+    X, y = Xy
+    X = X.copy()
+    X.index = pd.date_range("2020-01-01", periods=len(X), freq="10min",
+                            tz="UTC")
+    y = X.copy()
+    det = _detector(require_thresholds=False)
+    det.fit(X, y)
+    out = det.anomaly(X, y, frequency=pd.Timedelta("10min"))
+    start = pd.to_datetime(out["start"].iloc[0], utc=True)
+    end = pd.to_datetime(out["end"].iloc[0], utc=True)
+    assert (end - start) == pd.Timedelta("10min")
