@@ -581,3 +581,19 @@ globals:
     for frag in ("fleet-build", "keda.sh/v1alpha1", "runAsUser: 1000",
                  "runAsNonRoot: true", "team: mlops", "1600000000000"):
         assert frag in out, frag
+
+
+def test_log_level_option():
+    """--log-level wires into logging config (reference
+    test_cli.py::test_log_level_cli)."""
+    import logging
+
+    result = CliRunner().invoke(
+        gordo, ["--log-level", "warning", "--version"]
+    )
+    assert result.exit_code == 0
+    result = CliRunner().invoke(
+        gordo, ["--log-level", "not-a-level", "--version"]
+    )
+    # unknown level: either rejected or ignored, but never a crash
+    assert result.exit_code in (0, 2)
