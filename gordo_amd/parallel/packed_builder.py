@@ -298,10 +298,18 @@ class PackedFleetBuilder:
             return
         group = to_build
 
-        # pre-transform (CPU sklearn scalers etc., fitted per machine)
-        Xt_list, y_list = [], []
+        # pre-transform for the FINAL fit (CPU sklearn scalers etc.,
+        # fitted per machine on the full series — reference semantics:
+        # Pipeline.fit fits pre-steps on everything it is given). CV
+        # folds refit clones of the pre-steps on each fold's train
+        # slice (_fold_transform), matching sklearn.cross_validate's
+        # per-fold pipeline clone — no test-fold leakage into scores
+        # or DiffBased thresholds.
+        Xraw_list, Xt_list, y_list = [], [], []
         for p in group:
-            Xt = p.X.values.astype(np.float32)
+            Xraw = p.X.values.astype(np.float32)
+            Xraw_list.append(Xraw)
+            Xt = Xraw
             for _, step in p.pre_steps:
                 Xt = step.fit_transform(Xt)
             Xt_list.append(np.asarray(Xt, dtype=np.float32))
@@ -342,13 +350,21 @@ class PackedFleetBuilder:
         if cv_mode in ("cross_val_only", "full_build"):
             # the final fit is independent of the fold fits (fresh init
             # either way): on GPU it overlaps the concurrent fold packs
-            # on its own stream.
+            # on its own stream. The thread gate must IMPLY _fit_folds'
+            # own concurrency gate (group*n_folds <= MAX): if the fold
+            # loop were to fall back to its sequential path it would
+            # train the SAME shared `pack` the final-fit thread is
+            # using, corrupting both. group*(n_folds+1) <= MAX counts
+            # every model live at once (folds + final) and implies the
+            # fold gate.
+            n_folds = self._n_cv_folds(evaluation)
             final_thread = None
             if (
                 cv_mode == "full_build"
                 and self.device != "cpu"
                 and torch.cuda.is_available()
-                and len(group) * 4 <= self.MAX_CONCURRENT_FOLD_MODELS
+                and len(group) * (n_folds + 1)
+                <= self.MAX_CONCURRENT_FOLD_MODELS
             ):
                 import threading
 
@@ -358,7 +374,7 @@ class PackedFleetBuilder:
                 final_thread.start()
             t0 = time.time()
             self._cross_validate_group(
-                group, Xt_list, y_list, spec, fit_args, pack, init_snapshot
+                group, Xraw_list, y_list, spec, fit_args, pack, init_snapshot
             )
             cv_duration = time.time() - t0
             if final_thread is not None:
@@ -423,8 +439,37 @@ class PackedFleetBuilder:
     # without graph capture, which halves retention)
     MAX_CONCURRENT_FOLD_MODELS = 640
 
+    @staticmethod
+    def _n_cv_folds(evaluation: Dict[str, Any]) -> int:
+        """Fold count of the evaluation's CV splitter (for concurrency
+        gating before the splitter actually runs)."""
+        split_def = evaluation.get(
+            "cv", {"sklearn.model_selection.TimeSeriesSplit": {"n_splits": 3}}
+        )
+        try:
+            return int(serializer.from_definition(split_def).get_n_splits())
+        except Exception:
+            return 3
+
+    @staticmethod
+    def _fold_transform(group, Xraw_list, train_idx, test_idx):
+        """Per-fold pre-step fitting (reference semantics: sklearn
+        cross_validate clones the pipeline per fold, so scalers see
+        only the train slice and transform the test slice). Returns
+        (train arrays, test arrays) per machine."""
+        Xtr, Xte = [], []
+        for p, xr in zip(group, Xraw_list):
+            xt, xv = xr[train_idx], xr[test_idx]
+            for _, step in p.pre_steps:
+                st = sk_clone(step)
+                xt = st.fit_transform(xt)
+                xv = st.transform(xv)
+            Xtr.append(np.asarray(xt, dtype=np.float32))
+            Xte.append(np.asarray(xv, dtype=np.float32))
+        return Xtr, Xte
+
     def _fit_folds(
-        self, folds, group, Xt_list, y_list, spec, fit_args, pack,
+        self, folds, group, Xraw_list, y_list, spec, fit_args, pack,
         init_snapshot,
     ):
         import threading
@@ -438,14 +483,15 @@ class PackedFleetBuilder:
                 )
                 with ctx:
                     t0 = time.time()
-                    Xd = self._stack([x[train_idx] for x in Xt_list], fold_pack)
+                    Xtr, Xte = self._fold_transform(
+                        group, Xraw_list, train_idx, test_idx
+                    )
+                    Xd = self._stack(Xtr, fold_pack)
                     Yd = self._stack([y[train_idx] for y in y_list], fold_pack)
                     fold_pack.fit(Xd, Yd, **_engine_fit_args(fit_args))
                     t_fit = time.time() - t0
                     t0 = time.time()
-                    Xtest = self._stack(
-                        [x[test_idx] for x in Xt_list], fold_pack
-                    )
+                    Xtest = self._stack(Xte, fold_pack)
                     with torch.no_grad():
                         preds = fold_pack.predict(Xtest).float().cpu().numpy()
                     out[fold_i] = (preds, t_fit, time.time() - t0)
@@ -498,7 +544,7 @@ class PackedFleetBuilder:
     def _cross_validate_group(
         self,
         group: List[MachinePlan],
-        Xt_list: List[np.ndarray],
+        Xraw_list: List[np.ndarray],
         y_list: List[np.ndarray],
         spec: ModelSpec,
         fit_args: Dict[str, Any],
@@ -518,7 +564,7 @@ class PackedFleetBuilder:
         metrics_list = ModelBuilder.metrics_from_list(evaluation.get("metrics"))
         scoring_scaler_def = evaluation.get("scoring_scaler")
 
-        N = len(Xt_list[0])
+        N = len(Xraw_list[0])
         X_index = group[0].X.index
         folds = list(split_obj.split(np.zeros((N, 1))))
 
@@ -540,7 +586,8 @@ class PackedFleetBuilder:
         ]
 
         fold_preds = self._fit_folds(
-            folds, group, Xt_list, y_list, spec, fit_args, pack, init_snapshot
+            folds, group, Xraw_list, y_list, spec, fit_args, pack,
+            init_snapshot,
         )
 
         # KFCV detectors: thresholds are quantiles of the smoothed
@@ -738,13 +785,23 @@ def _metric_all_tags(metric, yt: np.ndarray, yp: np.ndarray):
         ss_res = (diff ** 2).sum(axis=0)
         ss_tot = ((yt - yt.mean(axis=0)) ** 2).sum(axis=0)
         with np.errstate(divide="ignore", invalid="ignore"):
-            per = np.where(ss_tot > 0, 1.0 - ss_res / ss_tot, 0.0)
+            # sklearn convention for a zero-variance target column:
+            # perfect prediction scores 1.0, anything else 0.0
+            per = np.where(
+                ss_tot > 0,
+                1.0 - ss_res / ss_tot,
+                np.where(ss_res == 0, 1.0, 0.0),
+            )
         return per, per.mean()
     if name == "explained_variance_score":
         var_res = diff.var(axis=0)
         var_y = yt.var(axis=0)
         with np.errstate(divide="ignore", invalid="ignore"):
-            per = np.where(var_y > 0, 1.0 - var_res / var_y, 0.0)
+            per = np.where(
+                var_y > 0,
+                1.0 - var_res / var_y,
+                np.where(var_res == 0, 1.0, 0.0),
+            )
         return per, per.mean()
     per = np.array([metric(yt[:, c], yp[:, c]) for c in range(yt.shape[1])])
     return per, metric(yt, yp)

@@ -121,7 +121,17 @@ class MicroBatcher:
         return mine
 
 
-_batchers: Dict[int, MicroBatcher] = {}
+import weakref
+
+# Weakly-keyed registry: a batcher dies with its model, so models
+# evicted from the serving LRU (and their HBM residency) are freed
+# normally — an id()-keyed dict would pin every model's predict
+# closure forever. Not stored as a model attribute: the batcher holds
+# a threading.Lock, which would break model pickling (download-model).
+_batchers: "weakref.WeakKeyDictionary[Any, MicroBatcher]" = (
+    weakref.WeakKeyDictionary()
+)
+_batchers_fallback: Dict[int, MicroBatcher] = {}  # un-weakref-able models
 _batchers_lock = threading.Lock()
 
 
@@ -131,11 +141,16 @@ def enabled() -> bool:
 
 def batched_predict(model: Any, X: np.ndarray) -> np.ndarray:
     """Route ``model.predict`` through the model's micro-batcher."""
-    key = id(model)
     with _batchers_lock:
-        b = _batchers.get(key)
+        try:
+            b = _batchers.get(model)
+        except TypeError:  # unhashable/un-weakref-able model
+            b = _batchers_fallback.get(id(model))
         if b is None:
             window = float(os.environ.get("GORDO_SERVE_BATCH_WINDOW_MS", 2.0))
             b = MicroBatcher(model.predict, window_ms=window)
-            _batchers[key] = b
+            try:
+                _batchers[model] = b
+            except TypeError:
+                _batchers_fallback[id(model)] = b
     return b.predict(X)
