@@ -144,7 +144,14 @@ def lstm_seq_bwd_v3(dSeq, gacts, cs, Wh, last_only):
 
 
 def lstm_seq_available(H: int) -> bool:
-    return hip_available() and H <= 64
+    """Fused sequence-scan coverage: the LDS-resident kernels serve
+    H <= 64; the big-H kernels (Wh streamed from L2, 64-column tile
+    loop) serve 64 < H <= 256 when H % 8 == 0 — which covers the
+    reference's default LSTM dims (256, 128, 64), reference
+    gordo/machine/model/factories/lstm_autoencoder.py:112."""
+    if not hip_available():
+        return False
+    return H <= 64 or (H <= 256 and H % 8 == 0)
 
 
 def anomaly_score(out, y, scale, minv, feat_thr, agg_thr):

@@ -893,6 +893,250 @@ __global__ __launch_bounds__(256) void lstm_seq_bwd_v2_kernel(
   }
 }
 
+// ===========================================================================
+// Big-H scan kernels (64 < H <= 256, H % 8 == 0) — the reference's
+// DEFAULT LSTM dims are (256, 128, 64) (reference
+// gordo/machine/model/factories/lstm_autoencoder.py:112), which the
+// H<=64 kernels cannot serve: WhT for H=128 is 139 KB of LDS alone.
+// Design: Wh stays in GLOBAL memory (one model's Wh is 128 KB-512 KB —
+// L2-resident per XCD after the first timestep; B-fragments are
+// 16-byte k-contiguous loads), LDS holds only the per-tile h / c /
+// gate state, and each wave loops over 64-column tiles of the 4H gate
+// dimension. Compute per step is O(H^2) while the serial dependence
+// stays 1 step, so the latency-boundness of the small-H scans fades:
+// at H=256 each workgroup issues 2048 MFMAs per timestep.
+// ===========================================================================
+
+template <int ROWS>
+__global__ __launch_bounds__(256) void lstm_seq_fwd_big_kernel(
+    const bf16* __restrict__ xW, const bf16* __restrict__ WhT_g,
+    bf16* __restrict__ hs, float* __restrict__ cs,
+    bf16* __restrict__ gacts, int B, int T, int H, int ldg, int Kp) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int H4 = 4 * H;
+  const int LDH = Kp + 8;  // padded h-row length (bank-conflict pad)
+  bf16* hS = reinterpret_cast<bf16*>(smem);               // [ROWS][LDH]
+  bf16* gS = hS + (size_t)ROWS * LDH;                     // [ROWS][ldg]
+  float* cS = reinterpret_cast<float*>(gS + (size_t)ROWS * ldg);  // [ROWS][H]
+
+  const int g = blockIdx.x / ((B + ROWS - 1) / ROWS);
+  const int r0 = (blockIdx.x % ((B + ROWS - 1) / ROWS)) * ROWS;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int l15 = lane & 15;
+  const int kslot = lane >> 4;
+
+  const bf16* WhTg = WhT_g + (size_t)g * H4 * Kp;
+  const bf16* xWg = xW + ((size_t)g * B + r0) * T * H4;
+  bf16* hsg = hs + ((size_t)g * B + r0) * T * H;
+  float* csg = cs + ((size_t)g * B + r0) * T * H;
+  bf16* gag = gacts + ((size_t)g * B + r0) * T * H4;
+  const int rows_here = min(ROWS, B - r0);
+
+  // zero h (incl. the [H, Kp) pad columns the MFMA K loop reads), c
+  for (int i = tid; i < ROWS * LDH; i += 256) hS[i] = lf2bf(0.f);
+  for (int i = tid; i < ROWS * H; i += 256) cS[i] = 0.f;
+  __syncthreads();
+
+  constexpr int FM = ROWS / 16;
+
+  for (int t = 0; t < T; ++t) {
+    // ---- gates = h @ Wh (MFMA, Wh streamed from L2) + xW_t ----
+    for (int wcol0 = wid * 64; wcol0 < H4; wcol0 += 256) {
+      // prefetch this tile's x-side gate values (register-resident;
+      // HBM latency hides under the MFMA loop)
+      bf16 xv[FM][4][4];
+      #pragma unroll
+      for (int fm = 0; fm < FM; ++fm) {
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+          int col = wcol0 + fn * 16 + l15;
+          #pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            int row = fm * 16 + kslot * 4 + r;
+            xv[fm][fn][r] =
+                (col < H4 && row < rows_here)
+                    ? xWg[((size_t)row * T + t) * H4 + col]
+                    : lf2bf(0.f);
+          }
+        }
+      }
+      f32x4 acc[FM][4] = {};
+      for (int kk = 0; kk < Kp; kk += 32) {
+        #pragma unroll
+        for (int fm = 0; fm < FM; ++fm) {
+          bf16x8 a = *reinterpret_cast<const bf16x8*>(
+              &hS[(fm * 16 + l15) * LDH + kk + kslot * 8]);
+          #pragma unroll
+          for (int fn = 0; fn < 4; ++fn) {
+            int col = wcol0 + fn * 16 + l15;
+            bf16x8 b = *reinterpret_cast<const bf16x8*>(
+                &WhTg[(size_t)min(col, H4 - 1) * Kp + kk + kslot * 8]);
+            acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a, b, acc[fm][fn], 0, 0, 0);
+          }
+        }
+      }
+      #pragma unroll
+      for (int fm = 0; fm < FM; ++fm) {
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+          int col = wcol0 + fn * 16 + l15;
+          if (col >= H4) continue;
+          #pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            int row = fm * 16 + kslot * 4 + r;
+            gS[row * ldg + col] =
+                lf2bf(acc[fm][fn][r] + lbf2f(xv[fm][fn][r]));
+          }
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- fused gate math + h/c update + outputs (identical to v1) ----
+    for (int e = tid; e < ROWS * H; e += 256) {
+      int row = e / H, hh = e % H;
+      const bf16* grow = &gS[row * ldg];
+      float i_g = sigmoidf_(lbf2f(grow[hh]));
+      float f_g = sigmoidf_(lbf2f(grow[H + hh]));
+      float g_g = fast_tanhf_(lbf2f(grow[2 * H + hh]));
+      float o_g = sigmoidf_(lbf2f(grow[3 * H + hh]));
+      float cc = f_g * cS[row * H + hh] + i_g * g_g;
+      float hv = o_g * fast_tanhf_(cc);
+      cS[row * H + hh] = cc;
+      hS[row * LDH + hh] = lf2bf(hv);
+      if (row < rows_here) {
+        size_t base = ((size_t)row * T + t) * H + hh;
+        hsg[base] = lf2bf(hv);
+        csg[base] = cc;
+        size_t gbase = ((size_t)row * T + t) * H4;
+        gag[gbase + hh] = lf2bf(i_g);
+        gag[gbase + H + hh] = lf2bf(f_g);
+        gag[gbase + 2 * H + hh] = lf2bf(g_g);
+        gag[gbase + 3 * H + hh] = lf2bf(o_g);
+      }
+    }
+    __syncthreads();
+  }
+}
+
+template <int ROWS>
+__global__ __launch_bounds__(256) void lstm_seq_bwd_big_kernel(
+    const bf16* __restrict__ dSeq, const bf16* __restrict__ gacts,
+    const float* __restrict__ cs, const bf16* __restrict__ WhN_g,
+    bf16* __restrict__ dG, int B, int T, int H, int ldg, int last_only) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int H4 = 4 * H;          // K length of the carry GEMM (32-mult)
+  const int LDH = ((H + 31) & ~31) + 8;
+  bf16* dgS = reinterpret_cast<bf16*>(smem);              // [ROWS][ldg]
+  bf16* dhS = dgS + (size_t)ROWS * ldg;                   // [ROWS][LDH]
+  float* dcS = reinterpret_cast<float*>(dhS + (size_t)ROWS * LDH);  // [ROWS][H]
+
+  const int g = blockIdx.x / ((B + ROWS - 1) / ROWS);
+  const int r0 = (blockIdx.x % ((B + ROWS - 1) / ROWS)) * ROWS;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int l15 = lane & 15;
+  const int kslot = lane >> 4;
+
+  const bf16* WhNg = WhN_g + (size_t)g * H * H4;  // native [H][4H]
+  const bf16* gag = gacts + ((size_t)g * B + r0) * T * H4;
+  const float* csg = cs + ((size_t)g * B + r0) * T * H;
+  const bf16* dSg = last_only ? dSeq + ((size_t)g * B + r0) * H
+                              : dSeq + ((size_t)g * B + r0) * T * H;
+  bf16* dGg = dG + ((size_t)g * B + r0) * T * H4;
+  const int rows_here = min(ROWS, B - r0);
+
+  for (int i = tid; i < ROWS * LDH; i += 256) dhS[i] = lf2bf(0.f);
+  for (int i = tid; i < ROWS * H; i += 256) dcS[i] = 0.f;
+  for (int i = tid; i < ROWS * ldg; i += 256) dgS[i] = lf2bf(0.f);
+  __syncthreads();
+
+  constexpr int FM = ROWS / 16;
+
+  for (int t = T - 1; t >= 0; --t) {
+    // ---- fused gate backward (identical to v1) ----
+    for (int e = tid; e < ROWS * H; e += 256) {
+      int row = e / H, hh = e % H;
+      float dh = lbf2f(dhS[row * LDH + hh]);
+      if (row < rows_here) {
+        if (last_only) {
+          if (t == T - 1) dh += lbf2f(dSg[(size_t)row * H + hh]);
+        } else {
+          dh += lbf2f(dSg[((size_t)row * T + t) * H + hh]);
+        }
+      }
+      size_t gbase = ((size_t)row * T + t) * H4;
+      size_t cbase = ((size_t)row * T + t) * H + hh;
+      float i_g = 0.5f, f_g = 0.5f, g_g = 0.f, o_g = 0.5f, cc = 0.f,
+            cp = 0.f;
+      if (row < rows_here) {
+        i_g = lbf2f(gag[gbase + hh]);
+        f_g = lbf2f(gag[gbase + H + hh]);
+        g_g = lbf2f(gag[gbase + 2 * H + hh]);
+        o_g = lbf2f(gag[gbase + 3 * H + hh]);
+        cc = csg[cbase];
+        cp = (t > 0) ? csg[cbase - H] : 0.f;
+      }
+      float tc = fast_tanhf_(cc);
+      float dc = dcS[row * H + hh] + dh * o_g * (1.f - tc * tc);
+      float di = dc * g_g;
+      float df = dc * cp;
+      float dg = dc * i_g;
+      float do_ = dh * tc;
+      dcS[row * H + hh] = dc * f_g;
+      float vi = di * i_g * (1.f - i_g);
+      float vf = df * f_g * (1.f - f_g);
+      float vg = dg * (1.f - g_g * g_g);
+      float vo = do_ * o_g * (1.f - o_g);
+      dgS[row * ldg + hh] = lf2bf(vi);
+      dgS[row * ldg + H + hh] = lf2bf(vf);
+      dgS[row * ldg + 2 * H + hh] = lf2bf(vg);
+      dgS[row * ldg + 3 * H + hh] = lf2bf(vo);
+      if (row < rows_here) {
+        dGg[gbase + hh] = lf2bf(vi);
+        dGg[gbase + H + hh] = lf2bf(vf);
+        dGg[gbase + 2 * H + hh] = lf2bf(vg);
+        dGg[gbase + 3 * H + hh] = lf2bf(vo);
+      }
+    }
+    __syncthreads();
+
+    // ---- dh_carry = dgates @ Wh^T (MFMA; WhN from L2): out [ROWS][H] ----
+    for (int col0 = wid * 64; col0 < H; col0 += 256) {
+      #pragma unroll
+      for (int fm = 0; fm < FM; ++fm) {
+        f32x4 acc[4] = {};
+        for (int kk = 0; kk < H4; kk += 32) {
+          bf16x8 a = *reinterpret_cast<const bf16x8*>(
+              &dgS[(fm * 16 + l15) * ldg + kk + kslot * 8]);
+          #pragma unroll
+          for (int fn = 0; fn < 4; ++fn) {
+            int col = col0 + fn * 16 + l15;  // h index
+            bf16x8 b = *reinterpret_cast<const bf16x8*>(
+                &WhNg[(size_t)min(col, H - 1) * H4 + kk + kslot * 8]);
+            acc[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a, b, acc[fn], 0, 0, 0);
+          }
+        }
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+          int col = col0 + fn * 16 + l15;
+          #pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            int row = fm * 16 + kslot * 4 + r;
+            if (col < H) dhS[row * LDH + col] = lf2bf(acc[fn][r]);
+          }
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
 // ---------------------------------------------------------------------------
 namespace gordo_lstm {
 
@@ -906,8 +1150,99 @@ inline int pick_rows(int G, int B) {
   return (G * ((B + 63) / 64) >= 512) ? 64 : 32;
 }
 
+// ---- big-H path (64 < H <= 256, H % 8 == 0) ----
+inline bool big_ok(int H) { return H > 64 && H <= 256 && H % 8 == 0; }
+
+inline size_t big_lds(int rows, int H, int ldg, int Kp) {
+  int LDH = Kp + 8;
+  return (size_t)rows * LDH * 2 + (size_t)rows * ldg * 2 +
+         (size_t)rows * H * 4;
+}
+
+inline int pick_rows_big(int B, int H, int ldg, int Kp) {
+  // 64-row tiles halve the per-step Wh L2 re-reads (each workgroup
+  // streams the whole Wh every timestep) — use them whenever the LDS
+  // fits AND the batch actually fills the tile.
+  if (B >= 48 && big_lds(64, H, ldg, Kp) <= 160 * 1024) return 64;
+  return 32;
+}
+
+std::vector<torch::Tensor> lstm_seq_fwd_big(torch::Tensor xW,
+                                            torch::Tensor Wh) {
+  TORCH_CHECK(xW.is_cuda() && xW.dim() == 4, "xW must be [G,B,T,4H] on GPU");
+  auto xc = xW.to(torch::kBFloat16).contiguous();
+  auto Whc = Wh.to(torch::kBFloat16).contiguous();
+  int G = xc.size(0), B = xc.size(1), T = xc.size(2), H4 = xc.size(3);
+  int H = H4 / 4;
+  TORCH_CHECK(big_ok(H), "lstm_seq_fwd_big needs 64 < H <= 256, H % 8 == 0");
+  int Kp = (H + 31) & ~31;
+  int ldg = pad_ldg(H4);
+  // pre-transpose Wh -> [G, 4H, Kp] (zero-padded K) once per launch so
+  // B-fragment loads are 16-byte k-contiguous
+  auto WhT = torch::zeros({G, H4, Kp},
+                          Whc.options().dtype(torch::kBFloat16));
+  WhT.narrow(2, 0, H).copy_(Whc.transpose(1, 2));
+  WhT = WhT.contiguous();
+  auto hs = torch::empty({G, B, T, H}, xc.options());
+  auto cs = torch::empty({G, B, T, H}, xc.options().dtype(torch::kFloat32));
+  auto gacts = torch::empty({G, B, T, H4}, xc.options());
+  int rows = pick_rows_big(B, H, ldg, Kp);
+  size_t lds = big_lds(rows, H, ldg, Kp);
+  TORCH_CHECK(lds <= 160 * 1024, "LDS budget exceeded (big fwd)");
+  int blocks = G * ((B + rows - 1) / rows);
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  if (rows == 64)
+    hipLaunchKernelGGL(lstm_seq_fwd_big_kernel<64>, dim3(blocks), dim3(256),
+                       lds, stream, (const bf16*)xc.data_ptr(),
+                       (const bf16*)WhT.data_ptr(), (bf16*)hs.data_ptr(),
+                       cs.data_ptr<float>(), (bf16*)gacts.data_ptr(), B, T, H,
+                       ldg, Kp);
+  else
+    hipLaunchKernelGGL(lstm_seq_fwd_big_kernel<32>, dim3(blocks), dim3(256),
+                       lds, stream, (const bf16*)xc.data_ptr(),
+                       (const bf16*)WhT.data_ptr(), (bf16*)hs.data_ptr(),
+                       cs.data_ptr<float>(), (bf16*)gacts.data_ptr(), B, T, H,
+                       ldg, Kp);
+  return {hs, cs, gacts};
+}
+
+torch::Tensor lstm_seq_bwd_big(torch::Tensor dSeq, torch::Tensor gacts,
+                               torch::Tensor cs, torch::Tensor Wh,
+                               bool last_only) {
+  TORCH_CHECK(gacts.is_cuda() && gacts.dim() == 4, "gacts must be [G,B,T,4H]");
+  auto dc = dSeq.to(torch::kBFloat16).contiguous();
+  auto gc = gacts.to(torch::kBFloat16).contiguous();
+  auto ccs = cs.to(torch::kFloat32).contiguous();
+  auto Whc = Wh.to(torch::kBFloat16).contiguous();
+  int G = gc.size(0), B = gc.size(1), T = gc.size(2), H4 = gc.size(3);
+  int H = H4 / 4;
+  TORCH_CHECK(big_ok(H), "lstm_seq_bwd_big needs 64 < H <= 256, H % 8 == 0");
+  int Kp = (H + 31) & ~31;
+  int ldg = pad_ldg(H4);
+  auto dG = torch::empty_like(gc);
+  int rows = pick_rows_big(B, H, ldg, Kp);
+  size_t lds = big_lds(rows, H, ldg, Kp);
+  TORCH_CHECK(lds <= 160 * 1024, "LDS budget exceeded (big bwd)");
+  int blocks = G * ((B + rows - 1) / rows);
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  if (rows == 64)
+    hipLaunchKernelGGL(lstm_seq_bwd_big_kernel<64>, dim3(blocks), dim3(256),
+                       lds, stream, (const bf16*)dc.data_ptr(),
+                       (const bf16*)gc.data_ptr(), ccs.data_ptr<float>(),
+                       (const bf16*)Whc.data_ptr(), (bf16*)dG.data_ptr(), B,
+                       T, H, ldg, last_only ? 1 : 0);
+  else
+    hipLaunchKernelGGL(lstm_seq_bwd_big_kernel<32>, dim3(blocks), dim3(256),
+                       lds, stream, (const bf16*)dc.data_ptr(),
+                       (const bf16*)gc.data_ptr(), ccs.data_ptr<float>(),
+                       (const bf16*)Whc.data_ptr(), (bf16*)dG.data_ptr(), B,
+                       T, H, ldg, last_only ? 1 : 0);
+  return dG;
+}
+
 std::vector<torch::Tensor> lstm_seq_fwd(torch::Tensor xW, torch::Tensor Wh) {
   TORCH_CHECK(xW.is_cuda() && xW.dim() == 4, "xW must be [G,B,T,4H] on GPU");
+  if (xW.size(3) / 4 > 64) return lstm_seq_fwd_big(xW, Wh);
   auto xc = xW.to(torch::kBFloat16).contiguous();
   auto Whc = Wh.to(torch::kBFloat16).contiguous();
   int G = xc.size(0), B = xc.size(1), T = xc.size(2), H4 = xc.size(3);
@@ -967,8 +1302,9 @@ std::vector<torch::Tensor> lstm_seq_fwd_v3(torch::Tensor xW,
                                             torch::Tensor Wh) {
   // identical contract to lstm_seq_fwd, pipelined kernel (always the
   // barriered layout, never the v2 path: v3 is measured against v1 on
-  // the same shapes)
+  // the same shapes). H > 64 routes to the big-H kernel (no v3 there).
   TORCH_CHECK(xW.is_cuda() && xW.dim() == 4, "xW must be [G,B,T,4H] on GPU");
+  if (xW.size(3) / 4 > 64) return lstm_seq_fwd_big(xW, Wh);
   auto xc = xW.to(torch::kBFloat16).contiguous();
   auto Whc = Wh.to(torch::kBFloat16).contiguous();
   int G = xc.size(0), B = xc.size(1), T = xc.size(2), H4 = xc.size(3);
@@ -1004,8 +1340,11 @@ torch::Tensor lstm_seq_bwd_v3(torch::Tensor dSeq, torch::Tensor gacts,
                               torch::Tensor cs, torch::Tensor Wh,
                               bool last_only) {
   // identical contract to lstm_seq_bwd; pipelined kernel, always the
-  // barriered layout (A/B-tested against v1 on the same shapes)
+  // barriered layout (A/B-tested against v1 on the same shapes).
+  // H > 64 routes to the big-H kernel (no v3 there).
   TORCH_CHECK(gacts.is_cuda() && gacts.dim() == 4, "gacts must be [G,B,T,4H]");
+  if (gacts.size(3) / 4 > 64)
+    return lstm_seq_bwd_big(dSeq, gacts, cs, Wh, last_only);
   auto dc = dSeq.to(torch::kBFloat16).contiguous();
   auto gc = gacts.to(torch::kBFloat16).contiguous();
   auto cc = cs.to(torch::kFloat32).contiguous();
@@ -1040,6 +1379,8 @@ torch::Tensor lstm_seq_bwd(torch::Tensor dSeq, torch::Tensor gacts,
                            torch::Tensor cs, torch::Tensor Wh,
                            bool last_only) {
   TORCH_CHECK(gacts.is_cuda() && gacts.dim() == 4, "gacts must be [G,B,T,4H]");
+  if (gacts.size(3) / 4 > 64)
+    return lstm_seq_bwd_big(dSeq, gacts, cs, Wh, last_only);
   auto dc = dSeq.to(torch::kBFloat16).contiguous();
   auto gc = gacts.to(torch::kBFloat16).contiguous();
   auto cc = cs.to(torch::kFloat32).contiguous();

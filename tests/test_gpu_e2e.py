@@ -84,6 +84,44 @@ def test_lstm_pack_gpu_matches_cpu():
     assert np.abs(out_cpu - out_gpu).mean() < 0.05
 
 
+def test_lstm_pack_default_dims_gpu_matches_cpu():
+    """The reference's DEFAULT LSTM dims (256, 128, 64) — reference
+    lstm_autoencoder.py:112 — exercise the big-H (H > 64) fused scan
+    kernels end to end through LSTMPack."""
+    require_hip()
+    from gordo_amd.engine.pack import LSTMPack
+    from gordo_amd.engine.spec import LayerSpec, ModelSpec
+
+    spec = ModelSpec(
+        model_type="lstm", n_features=10, n_features_out=10,
+        layers=[
+            LayerSpec(kind="lstm", units=256, return_sequences=True),
+            LayerSpec(kind="lstm", units=128, return_sequences=True),
+            LayerSpec(kind="lstm", units=64, return_sequences=False),
+            LayerSpec(kind="dense", units=10, activation="linear"),
+        ],
+        lookback_window=12,
+    )
+    rng = np.random.default_rng(7)
+    X = rng.random((1, 160, 10)).astype("float32")
+    Xt = torch.from_numpy(X)
+
+    gpu = LSTMPack(spec, G=1, device="cuda", seeds=[5])
+    assert gpu._use_fused(), "256/128/64 stack must take the fused path"
+    cpu = LSTMPack(spec, G=1, device="cpu", seeds=[5])
+    hist_cpu = cpu.fit(Xt, Xt.clone(), epochs=2, batch_size=64)
+    Xg = Xt.to("cuda", gpu.compute_dtype)
+    hist_gpu = gpu.fit(Xg, Xg.clone(), epochs=2, batch_size=64)
+    for e in range(2):
+        assert hist_gpu["loss"][e][0] == pytest.approx(
+            hist_cpu["loss"][e][0], rel=0.15, abs=5e-3
+        )
+    out_cpu = cpu.predict(Xt).float().numpy()
+    out_gpu = gpu.predict(Xt).float().cpu().numpy()
+    assert out_gpu.shape == out_cpu.shape
+    assert np.abs(out_cpu - out_gpu).mean() < 0.05
+
+
 def test_estimator_fit_on_gpu_uses_hip():
     require_hip()
     from gordo_amd.machine.model import KerasAutoEncoder

@@ -184,11 +184,28 @@ def test_lstm_pointwise_roundtrip():
     (3, 32, 144, 25, True),   # full lookback
 ])
 def test_lstm_seq_fused_vs_reference(G, B, T, H, last_only):
+    _check_lstm_seq(G, B, T, H, last_only, wh_scale=0.3)
+
+
+@pytest.mark.parametrize("G,B,T,H,last_only", [
+    (2, 48, 12, 72, False),    # 64-row tile path (B >= 48)
+    (1, 33, 20, 128, True),    # ragged row tile, reference default dim
+    (2, 64, 24, 256, True),    # largest supported H
+    (1, 20, 16, 96, False),    # 32-row tile path (B < 48)
+])
+def test_lstm_seq_big_vs_reference(G, B, T, H, last_only):
+    """Big-H (64 < H <= 256) scan kernels — Wh streamed from L2 —
+    vs the per-timestep fp32 oracle (reference default LSTM dims are
+    256/128/64: lstm_autoencoder.py:112)."""
+    _check_lstm_seq(G, B, T, H, last_only, wh_scale=2.4 / (H ** 0.5))
+
+
+def _check_lstm_seq(G, B, T, H, last_only, wh_scale):
     """Fused sequence-scan kernels vs the per-timestep fp32 oracle."""
     require_hip()
     H4 = 4 * H
     xW = _rand(G, B, T, H4, seed=30)
-    Wh = _rand(G, H, H4, seed=31) * 0.3
+    Wh = _rand(G, H, H4, seed=31) * wh_scale
 
     # reference forward (fp32, per-timestep)
     hs_ref = torch.empty(G, B, T, H)
