@@ -109,17 +109,27 @@ def adam_step(p, g, m, v, lr, beta1, beta2, eps, step, p_lp=None,
 
 
 def lstm_seq_fwd(xW, Wh):
-    """Fused on-device LSTM forward scan (GPU only, H<=64).
+    """Fused on-device LSTM sequence scan (GPU only; H<=64 LDS-resident
+    kernels, 64<H<=256 streamed-Wh kernels).
 
-    ``GORDO_LSTM_V3=1`` selects the software-pipelined v3 kernel
-    (double-buffered x-gate prefetch) — opt-in until GPU-validated
-    (ROADMAP round-2 lever #1)."""
+    The software-pipelined v3 kernel (double-buffered x-gate register
+    prefetch) is the DEFAULT for the barriered H<=64 layout since the
+    round-2 GPU A/B (fwd 1.11 vs 1.51 ms, bwd 0.89 vs 0.99 ms at the
+    bench shape G=8 B=512 T=144 H=42 — gpurun validate_r2). Set
+    ``GORDO_LSTM_V1=1`` to fall back to the unpipelined v1 scan.
+    H%16==0 still takes the barrier-free v2 path inside the extension;
+    H>64 routes to the big-H kernels in both entry points."""
     import os as _os
 
     ext = _require_hip()
-    if _os.environ.get("GORDO_LSTM_V3") == "1":
-        return ext.lstm_seq_fwd_v3(xW, Wh)
-    return ext.lstm_seq_fwd(xW, Wh)
+    H = xW.shape[-1] // 4
+    if (
+        _os.environ.get("GORDO_LSTM_V1") == "1"
+        or H > 64            # big-H dispatch lives in the v1 entry
+        or H % 16 == 0       # v2 barrier-free path (v1 entry dispatches)
+    ):
+        return ext.lstm_seq_fwd(xW, Wh)
+    return ext.lstm_seq_fwd_v3(xW, Wh)
 
 
 def lstm_seq_fwd_v3(xW, Wh):
@@ -128,14 +138,21 @@ def lstm_seq_fwd_v3(xW, Wh):
 
 
 def lstm_seq_bwd(dSeq, gacts, cs, Wh, last_only):
-    """Fused on-device LSTM backward (BPTT) scan (GPU only, H<=64).
-    ``GORDO_LSTM_V3=1`` selects the pipelined v3 kernel."""
+    """Fused on-device LSTM backward (BPTT) scan (GPU only). Same
+    version dispatch as ``lstm_seq_fwd``: pipelined v3 by default for
+    the barriered layout, v2 for H%16==0, big-H kernels for H>64;
+    ``GORDO_LSTM_V1=1`` forces the unpipelined v1 scan."""
     import os as _os
 
     ext = _require_hip()
-    if _os.environ.get("GORDO_LSTM_V3") == "1":
-        return ext.lstm_seq_bwd_v3(dSeq, gacts, cs, Wh, last_only)
-    return ext.lstm_seq_bwd(dSeq, gacts, cs, Wh, last_only)
+    H = gacts.shape[-1] // 4
+    if (
+        _os.environ.get("GORDO_LSTM_V1") == "1"
+        or H > 64
+        or H % 16 == 0
+    ):
+        return ext.lstm_seq_bwd(dSeq, gacts, cs, Wh, last_only)
+    return ext.lstm_seq_bwd_v3(dSeq, gacts, cs, Wh, last_only)
 
 
 def lstm_seq_bwd_v3(dSeq, gacts, cs, Wh, last_only):

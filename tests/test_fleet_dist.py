@@ -39,8 +39,14 @@ def test_fleet_build_two_ranks(tmp_path):
     env = dict(os.environ)
     env.update(
         GORDO_DIST_BACKEND="gloo",  # 2 CPU ranks even on a 1-GPU box
-        CUDA_VISIBLE_DEVICES="",    # children must not touch the GPU a
-                                    # parent test process may hold
+        # children must not touch the GPU a parent test process may
+        # hold: hide it at every layer (torch reads CUDA_, HIP reads
+        # HIP_, the ROCm runtime itself reads ROCR_ — CUDA_ alone
+        # still lets HSA enumerate the device in child processes,
+        # the suspected source of the round-1 GPU-box flake)
+        CUDA_VISIBLE_DEVICES="",
+        HIP_VISIBLE_DEVICES="",
+        ROCR_VISIBLE_DEVICES="",
         GORDO_FLEET_MACHINE_CONFIG=str(cfg_path),
         GORDO_FLEET_PROJECT_NAME="fleet-proj",
         GORDO_FLEET_OUTPUT_DIR=str(out_dir),
@@ -98,6 +104,61 @@ def test_fleet_build_two_ranks(tmp_path):
             assert dist_scores[key]["fold-mean"] == pytest.approx(
                 solo_scores[key]["fold-mean"], rel=1e-5, abs=1e-7
             ), key
+
+
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("world,n_machines", [
+    (4, 8),   # 2 machines per rank
+    (4, 3),   # MORE RANKS THAN MACHINES: one rank gets an empty shard
+    (8, 8),   # the full-node shape, 1 machine per rank
+])
+def test_fleet_build_multi_rank(tmp_path, world, n_machines):
+    """4- and 8-rank gloo fleet builds (VERDICT round-2 item #3): the
+    exact collective path the driver's 8-GPU RCCL run takes — scatter
+    of serialized shards (including empty ones), per-rank builds,
+    status gather — at world sizes beyond 2."""
+    config = CONFIG.format(
+        machines="".join(MACHINE_TMPL.format(i=i) for i in range(n_machines))
+    )
+    cfg_path = tmp_path / "cfg.yml"
+    cfg_path.write_text(config)
+    out_dir = tmp_path / "models"
+    status_file = tmp_path / "status.json"
+
+    env = dict(os.environ)
+    env.update(
+        GORDO_DIST_BACKEND="gloo",
+        CUDA_VISIBLE_DEVICES="",
+        HIP_VISIBLE_DEVICES="",
+        ROCR_VISIBLE_DEVICES="",
+        GORDO_FLEET_MACHINE_CONFIG=str(cfg_path),
+        GORDO_FLEET_PROJECT_NAME="fleet-proj",
+        GORDO_FLEET_OUTPUT_DIR=str(out_dir),
+        GORDO_FLEET_STATUS_FILE=str(status_file),
+        GORDO_FLEET_REPLACE_CACHE="0",
+    )
+    proc = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", f"--nproc-per-node={world}",
+            "--standalone", "--local-addr", "127.0.0.1",
+            "-m", "gordo_amd.cli.fleet_worker",
+        ],
+        env=env,
+        capture_output=True,
+        text=True,
+        timeout=280,
+    )
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    summary = json.loads(status_file.read_text())
+    debug = (summary, proc.stdout[-1500:], proc.stderr[-1500:])
+    assert summary["n_machines"] == n_machines, debug
+    assert summary["n_ok"] == n_machines, debug
+    assert summary["world_size"] == world, debug
+    for i in range(n_machines):
+        d = out_dir / f"fleet-m-{i}"
+        assert (d / "model.pkl").is_file(), debug
+        assert (d / "metadata.json").is_file(), debug
 
 
 def test_shard_machines_balanced():
@@ -187,7 +248,8 @@ def test_fleet_cli_multi_rank_with_requeue(tmp_path):
     out_dir = tmp_path / "models"
 
     env = dict(os.environ)
-    env.update(GORDO_DIST_BACKEND="gloo", CUDA_VISIBLE_DEVICES="")
+    env.update(GORDO_DIST_BACKEND="gloo", CUDA_VISIBLE_DEVICES="",
+               HIP_VISIBLE_DEVICES="", ROCR_VISIBLE_DEVICES="")
     proc = subprocess.run(
         [sys.executable, "-m", "gordo_amd", "fleet", "build",
          "--machine-config", str(cfg_path),

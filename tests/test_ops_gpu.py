@@ -393,16 +393,19 @@ def test_lstm_seq_v2_vs_reference(G, B, T, H, last_only):
     (4, 130, 50, 60),    # near the H cap, ragged rows (64 would hit
                           # the v2 dispatch in lstm_seq_fwd: H%16==0)
 ])
-def test_lstm_seq_v3_matches_v1(G, B, T, H):
-    """The pipelined v3 forward scan is numerically identical to the
-    shipping v1 kernel — same layout, same math, only the x-gate tile
-    is double-buffered. Exact equality expected (bit-identical inputs,
-    same op order)."""
+def test_lstm_seq_v3_matches_v1(G, B, T, H, monkeypatch):
+    """The pipelined v3 forward scan (the default since round 2) is
+    numerically identical to the v1 kernel — same layout, same math,
+    only the x-gate tile is double-buffered. Exact equality expected
+    (bit-identical inputs, same op order). GORDO_LSTM_V1 forces the
+    v1 arm through the public dispatch."""
     require_hip()
     H4 = 4 * H
     xW = to_dev_bf16(_rand(G, B, T, H4, seed=40))
     Wh = to_dev_bf16(_rand(G, H, H4, seed=41) * 0.3)
+    monkeypatch.setenv("GORDO_LSTM_V1", "1")
     v1_hs, v1_cs, v1_ga = ops.lstm_seq_fwd(xW, Wh)
+    monkeypatch.delenv("GORDO_LSTM_V1")
     v3_hs, v3_cs, v3_ga = ops.lstm_seq_fwd_v3(xW, Wh)
     assert torch.equal(v1_hs, v3_hs)
     assert torch.equal(v1_cs, v3_cs)
@@ -416,7 +419,7 @@ def test_lstm_seq_v3_matches_v1(G, B, T, H):
     (3, 32, 144, 25, True),
     (4, 130, 50, 60, False),
 ])
-def test_lstm_seq_bwd_v3_matches_v1(G, B, T, H, last_only):
+def test_lstm_seq_bwd_v3_matches_v1(G, B, T, H, last_only, monkeypatch):
     """The pipelined v3 backward scan is bit-identical to v1 — same
     math, the gate/cell loads are just double-buffered."""
     require_hip()
@@ -428,6 +431,8 @@ def test_lstm_seq_bwd_v3_matches_v1(G, B, T, H, last_only):
         dSeq = to_dev_bf16(_rand(G, B, H, seed=52))
     else:
         dSeq = to_dev_bf16(_rand(G, B, T, H, seed=52))
+    monkeypatch.setenv("GORDO_LSTM_V1", "1")
     v1 = ops.lstm_seq_bwd(dSeq, ga, cs, Wh, last_only)
+    monkeypatch.delenv("GORDO_LSTM_V1")
     v3 = ops.lstm_seq_bwd_v3(dSeq, ga, cs, Wh, last_only)
     assert torch.equal(v1, v3)
