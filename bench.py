@@ -134,7 +134,9 @@ def main():
 
     out_dir = tempfile.mkdtemp(prefix=f"gordo-bench-r{rank}-")
 
-    def one_step():
+    phase_acc: dict = {}
+
+    def one_step(timed: bool = False):
         builder = PackedFleetBuilder(
             my_machines, output_dir=out_dir, model_register_dir=None,
             device=device, save_models=True,
@@ -144,6 +146,9 @@ def main():
         if failed:
             errs = {n: repr(r) for n, r in results if isinstance(r, BaseException)}
             raise RuntimeError(f"bench build failures: {errs}")
+        if timed:
+            for k, v in builder.phase_times.items():
+                phase_acc[k] = phase_acc.get(k, 0.0) + v
 
     import torch.distributed as dist
 
@@ -159,7 +164,7 @@ def main():
     barrier_sync()
     t0 = time.time()
     for _ in range(args.steps):
-        one_step()
+        one_step(timed=True)
     barrier_sync()
     elapsed = time.time() - t0
 
@@ -174,6 +179,14 @@ def main():
         elapsed = float(t.item())
 
     shutil.rmtree(out_dir, ignore_errors=True)
+
+    if rank == 0 and args.verbose:
+        # per-step phase budget (seconds averaged over the timed steps;
+        # 'total' is the sequential wall — overlapped keys are informational)
+        budget = {k: round(v / args.steps, 3) for k, v in
+                  sorted(phase_acc.items())}
+        print(json.dumps({"phase_budget_s_per_step": budget}),
+              file=sys.stderr)
 
     if rank == 0:
         ms_per_step = elapsed * 1000.0 / args.steps

@@ -130,15 +130,27 @@ class PackedFleetBuilder:
         self.data_workers = data_workers
         self.save_models = save_models
         self.replace_cache = replace_cache
+        # wall-clock phase ledger for the build-step budget table
+        # (BASELINE.md): sequential wall segments; overlapped work
+        # (threaded final fit) is recorded under its own key.
+        self.phase_times: Dict[str, float] = {}
+
+    def _phase(self, key: str, dt: float):
+        self.phase_times[key] = self.phase_times.get(key, 0.0) + dt
 
     # ---- public ----------------------------------------------------------
     def build_all(self) -> List[Tuple[str, Any]]:
         """Build every machine. Returns [(name, Machine-with-metadata |
         exception)]. Models are saved under output_dir/<name>/ when
         save_models is set."""
+        t_all0 = time.time()
         plans = [MachinePlan(machine=m) for m in self.machines]
+        t0 = time.time()
         self._fetch_data(plans)
+        self._phase("fetch", time.time() - t0)
+        t0 = time.time()
         self._instantiate_models(plans)
+        self._phase("instantiate", time.time() - t0)
 
         packable = [p for p in plans if p.packable and p.error is None]
         fallback = [p for p in plans if not p.packable and p.error is None]
@@ -156,6 +168,7 @@ class PackedFleetBuilder:
                 for p in group:
                     p.error = e
 
+        t0 = time.time()
         for p in fallback:
             try:
                 logger.info("Fallback per-machine build: %s", p.machine.name)
@@ -174,6 +187,15 @@ class PackedFleetBuilder:
             except Exception as e:
                 logger.exception("Fallback build failed: %s", p.machine.name)
                 p.error = e
+
+        if fallback:
+            self._phase("fallback_builds", time.time() - t0)
+        self._phase("total", time.time() - t_all0)
+        if logger.isEnabledFor(logging.INFO):
+            budget = ", ".join(
+                f"{k}={v:.2f}s" for k, v in sorted(self.phase_times.items())
+            )
+            logger.info("build-step phase budget: %s", budget)
 
         results: List[Tuple[str, Any]] = []
         for p in plans:
@@ -283,6 +305,7 @@ class PackedFleetBuilder:
             [l.units for l in spec.layers], self.device,
         )
 
+        t_seg = time.time()
         # cache probe: skip machines already registered
         to_build: List[MachinePlan] = []
         for p in group:
@@ -297,6 +320,8 @@ class PackedFleetBuilder:
         if not to_build:
             return
         group = to_build
+        self._phase("cache_probe", time.time() - t_seg)
+        t_seg = time.time()
 
         # pre-transform for the FINAL fit (CPU sklearn scalers etc.,
         # fitted per machine on the full series — reference semantics:
@@ -314,6 +339,7 @@ class PackedFleetBuilder:
                 Xt = step.fit_transform(Xt)
             Xt_list.append(np.asarray(Xt, dtype=np.float32))
             y_list.append(p.y.values.astype(np.float32))
+        self._phase("pre_transform", time.time() - t_seg)
 
         cv_mode = str(evaluation.get("cv_mode", "full_build")).lower()
         cv_duration = None
@@ -339,10 +365,12 @@ class PackedFleetBuilder:
         # so fold fits reset to an init snapshot instead of re-creating;
         # the init itself is deterministic in (arch, per-machine seeds)
         # and memoized across builds.
+        t_seg = time.time()
         init_key = (spec.arch_key(), tuple(p.seed for p in group))
         cached_init = _INIT_CACHE.get(init_key)
         pack = self._make_pack(spec, group, init_p32=cached_init)
         init_snapshot = pack.store.p32.clone()
+        self._phase("pack_init", time.time() - t_seg)
         if cached_init is None:
             if len(_INIT_CACHE) >= 8:
                 _INIT_CACHE.pop(next(iter(_INIT_CACHE)))
@@ -377,9 +405,12 @@ class PackedFleetBuilder:
                 group, Xraw_list, y_list, spec, fit_args, pack, init_snapshot
             )
             cv_duration = time.time() - t0
+            self._phase("cv_wall", cv_duration)
             if final_thread is not None:
+                t_seg = time.time()
                 final_thread.join()
                 torch.cuda.synchronize()
+                self._phase("final_fit_join", time.time() - t_seg)
             if cv_mode == "cross_val_only":
                 for p in group:
                     self._finalize(p, None, cv_duration, final=False)
@@ -387,11 +418,15 @@ class PackedFleetBuilder:
 
         # final full fit (already done concurrently when possible)
         if "history" not in fit_state:
+            t_seg = time.time()
             final_fit(None)
+            self._phase("final_fit_serial", time.time() - t_seg)
         history = fit_state["history"]
         train_duration = fit_state["duration"]
+        self._phase("final_fit_gpu(overlapped)", train_duration)
 
         # per-machine adoption + detector finalization
+        t_seg = time.time()
         offset = len(Xt_list[0]) - pack._n_samples(len(Xt_list[0]))
         all_states = pack.states_for_all_models()
         for g_idx, p in enumerate(group):
@@ -410,7 +445,10 @@ class PackedFleetBuilder:
             self._finalize(p, offset, cv_duration, final=True,
                            train_duration=train_duration / len(group))
 
+        self._phase("adopt", time.time() - t_seg)
+
         # save + register (parallel: pickle+json dumps are I/O bound)
+        t_seg = time.time()
         if self.save_models and self.output_dir:
             def save_one(p):
                 out = os.path.join(self.output_dir, p.machine.name)
@@ -424,6 +462,7 @@ class PackedFleetBuilder:
 
             with concurrent.futures.ThreadPoolExecutor(8) as ex:
                 list(ex.map(save_one, group))
+        self._phase("save", time.time() - t_seg)
         pack.release_graphs()
         logger.info(
             "Packed build of %d machines done in %.2fs",
@@ -585,10 +624,13 @@ class PackedFleetBuilder:
             {} for _ in group
         ]
 
+        t_seg = time.time()
         fold_preds = self._fit_folds(
             folds, group, Xraw_list, y_list, spec, fit_args, pack,
             init_snapshot,
         )
+        self._phase("cv_fold_fits", time.time() - t_seg)
+        t_seg = time.time()
 
         # KFCV detectors: thresholds are quantiles of the smoothed
         # validation metric over the FULL reassembled prediction series
@@ -692,6 +734,8 @@ class PackedFleetBuilder:
                 "  fold %d: fit %.2fs, predict %.2fs, score+thresholds %.2fs",
                 fold_i, t_fit, t_pred, time.time() - t_s0,
             )
+
+        self._phase("cv_score_thresholds", time.time() - t_seg)
 
         # finalize KFCV thresholds over the reassembled series
         for g_idx in kfcv_pred:
