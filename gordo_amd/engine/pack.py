@@ -306,10 +306,71 @@ class BasePack:
     def release_graphs(self):
         """Drop captured graphs deterministically (outside any capture)
         instead of at arbitrary GC time."""
-        cached = getattr(self, "_graph_cache", None)
-        if cached:
-            with BasePack._graph_capture_lock:
-                cached.clear()
+        for attr in ("_graph_cache", "_pred_graph_cache"):
+            cached = getattr(self, attr, None)
+            if cached:
+                with BasePack._graph_capture_lock:
+                    cached.clear()
+
+    def predict_captured(self, X: torch.Tensor) -> torch.Tensor:
+        """hipGraph-replayed inference forward for static request shapes
+        (the serving hot path: north-star "batched anomaly inference on
+        hipGraph-captured steps"). Each distinct input shape captures
+        once and replays thereafter — one graph launch instead of one
+        host launch per layer. Opt-in via ``GORDO_SERVE_HIPGRAPH=1``
+        (the torch-ROCm graph-destructor abort makes capture opt-in
+        everywhere — ROADMAP); anything unexpected falls back to the
+        eager predict. Thread-safe: replay + readout under a per-pack
+        lock (the static buffers are shared)."""
+        if not (
+            os.environ.get("GORDO_SERVE_HIPGRAPH", "0") == "1"
+            and self.device.type == "cuda"
+            and self._graph_enabled
+        ):
+            return self.predict(X)
+        Xc = self._to_compute(X)
+        key = tuple(Xc.shape)
+        lock = getattr(self, "_pred_graph_lock", None)
+        if lock is None:
+            lock = self._pred_graph_lock = __import__("threading").Lock()
+        with lock:
+            cached = getattr(self, "_pred_graph_cache", None)
+            if cached is None:
+                cached = self._pred_graph_cache = {}
+            entry = cached.get(key)
+            if entry is None:
+                if len(cached) >= 8:  # bound distinct request shapes
+                    return self.predict(Xc)
+                try:
+                    import gc
+
+                    gc.collect()  # same destructor hazard as train capture
+                    with BasePack._graph_capture_lock:
+                        sx = Xc.clone()
+                        s = torch.cuda.Stream()
+                        s.wait_stream(torch.cuda.current_stream())
+                        with torch.cuda.stream(s), torch.no_grad():
+                            self.predict(sx)
+                        torch.cuda.current_stream().wait_stream(s)
+                        graph = torch.cuda.CUDAGraph()
+                        with torch.cuda.graph(
+                            graph, capture_error_mode="thread_local"
+                        ), torch.no_grad():
+                            out = self.predict(sx)
+                    entry = cached[key] = (graph, sx, out)
+                except Exception:
+                    logger.warning(
+                        "serving hipGraph capture failed; staying eager",
+                        exc_info=True,
+                    )
+                    self._graph_enabled = False
+                    return self.predict(Xc)
+            graph, sx, out = entry
+            sx.copy_(Xc)
+            graph.replay()
+            # clone before releasing the lock: the static out buffer is
+            # overwritten by the next replay
+            return out.clone()
 
     def fit(
         self,

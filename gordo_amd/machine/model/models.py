@@ -240,7 +240,7 @@ class KerasBaseEstimator(BaseEstimator, GordoBase):
         X = _as_2d_array(X)
         pack = self._ensure_pack()
         with torch.no_grad():
-            out = pack.predict(torch.from_numpy(X).unsqueeze(0))
+            out = pack.predict_captured(torch.from_numpy(X).unsqueeze(0))
         return out[0].float().cpu().numpy()
 
     # transform == predict so the estimator can sit mid-pipeline

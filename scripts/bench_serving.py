@@ -47,8 +47,18 @@ def main():
     ap.add_argument("--direct", action="store_true",
                     help="measure model.anomaly() directly (no HTTP/JSON): "
                          "the batched inference engine path, BASELINE "
-                         "config #5 shape")
+                         "config #5 shape (honors --n-models/--threads: "
+                         "N models resident, T concurrent streams)")
+    ap.add_argument("--hipgraph", action="store_true",
+                    help="hipGraph-capture the serving forward per "
+                         "request shape (GORDO_SERVE_HIPGRAPH=1)")
+    ap.add_argument("--profile-stages", action="store_true",
+                    help="single-thread per-stage timing of the anomaly "
+                         "endpoint work: JSON decode, model forward, "
+                         "score, response encode")
     args = ap.parse_args()
+    if args.hipgraph:
+        os.environ["GORDO_SERVE_HIPGRAPH"] = "1"
 
     import pandas as pd
     import torch
@@ -93,15 +103,39 @@ def main():
         if args.direct:
             import pandas as pd
 
-            model = serializer.load(os.path.join(collection, "serve-bench"))
+            names = ["serve-bench"] + [
+                f"serve-bench-{i}" for i in range(1, args.n_models)
+            ]
+            models = [serializer.load(os.path.join(collection, n))
+                      for n in names]
             X = pd.DataFrame(np.random.random((args.rows, N_TAGS)),
                              columns=sensors)
-            model.anomaly(X, X)  # warmup (captures/caches)
+            for m in models:
+                m.anomaly(X, X)  # warmup (captures/caches)
             if torch.cuda.is_available():
                 torch.cuda.synchronize()
+            counter = {"n": 0}
+            lock = threading.Lock()
+
+            def direct_worker():
+                while True:
+                    with lock:
+                        if counter["n"] >= args.rounds:
+                            return
+                        i = counter["n"]
+                        counter["n"] += 1
+                    models[i % len(models)].anomaly(X, X)
+
             t0 = time.time()
-            for _ in range(args.rounds):
-                model.anomaly(X, X)
+            if args.threads > 1:
+                ts = [threading.Thread(target=direct_worker)
+                      for _ in range(args.threads)]
+                for t in ts:
+                    t.start()
+                for t in ts:
+                    t.join()
+            else:
+                direct_worker()
             if torch.cuda.is_available():
                 torch.cuda.synchronize()
             dt = (time.time() - t0) / args.rounds
@@ -110,7 +144,80 @@ def main():
                 "rows_per_call": args.rows,
                 "ms_per_call": dt * 1000,
                 "rows_per_sec": args.rows / dt,
+                "n_models": args.n_models,
+                "threads": args.threads,
+                "hipgraph": bool(args.hipgraph),
                 "device": "cuda" if torch.cuda.is_available() else "cpu",
+            }))
+            return
+
+        if args.profile_stages:
+            import pandas as pd
+
+            from gordo_amd.server.utils import (
+                dataframe_from_dict,
+                dataframe_to_dict,
+            )
+
+            model = serializer.load(os.path.join(collection, "serve-bench"))
+            Xp = pd.DataFrame(np.random.random((args.rows, N_TAGS)),
+                              columns=sensors)
+            payload = {"X": dataframe_to_dict(Xp), "y": dataframe_to_dict(Xp)}
+            raw = json.dumps(payload).encode()
+            # one pass to warm everything
+            dec = json.loads(raw)
+            Xd = dataframe_from_dict(dec["X"])
+            yd = dataframe_from_dict(dec["y"])
+            frame = model.anomaly(Xd, yd)
+            from gordo_amd.server import utils as sutils
+
+            def encode_response(fr):
+                # the server's real encode path: C++ fastjson when the
+                # frame qualifies, python codec otherwise
+                # (server/utils.py frame_json_response)
+                fj = sutils._gordo_fastjson
+                if fj is not None:
+                    index = fr.index.astype(str)
+                    return b'{"data": ' + fj.encode_frame(
+                        index.tolist(),
+                        [str(c[0]) for c in fr.columns],
+                        [str(c[1]) for c in fr.columns],
+                        fr.values,
+                    ) + b"}"
+                return json.dumps(
+                    {"data": sutils.dataframe_to_dict(fr)}
+                ).encode()
+
+            encode_response(frame)
+            stages = {k: 0.0 for k in
+                      ("json_decode", "df_from_dict", "model_anomaly",
+                       "encode_response")}
+            n = args.rounds
+            for _ in range(n):
+                t0 = time.perf_counter()
+                dec = json.loads(raw)
+                t1 = time.perf_counter()
+                Xd = dataframe_from_dict(dec["X"])
+                yd = dataframe_from_dict(dec["y"])
+                t2 = time.perf_counter()
+                frame = model.anomaly(Xd, yd)
+                if torch.cuda.is_available():
+                    torch.cuda.synchronize()
+                t3 = time.perf_counter()
+                encode_response(frame)
+                t4 = time.perf_counter()
+                stages["json_decode"] += t1 - t0
+                stages["df_from_dict"] += t2 - t1
+                stages["model_anomaly"] += t3 - t2
+                stages["encode_response"] += t4 - t3
+            print(json.dumps({
+                "metric": "anomaly endpoint stage budget (ms/request)",
+                "rows": args.rows,
+                "hipgraph": bool(args.hipgraph),
+                "device": "cuda" if torch.cuda.is_available() else "cpu",
+                "stages_ms": {k: round(v / n * 1000, 3)
+                              for k, v in stages.items()},
+                "sum_ms": round(sum(stages.values()) / n * 1000, 3),
             }))
             return
 
