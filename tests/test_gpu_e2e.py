@@ -122,6 +122,32 @@ def test_lstm_pack_default_dims_gpu_matches_cpu():
     assert np.abs(out_cpu - out_gpu).mean() < 0.05
 
 
+def test_predict_captured_matches_eager(monkeypatch):
+    """hipGraph-captured serving forward (GORDO_SERVE_HIPGRAPH=1)
+    replays bit-identically to the eager forward, across shapes and
+    repeated replays."""
+    require_hip()
+    from gordo_amd.machine.model import KerasAutoEncoder
+
+    X = np.random.default_rng(11).random((300, 20)).astype("float32")
+    model = KerasAutoEncoder(kind="feedforward_hourglass", epochs=2,
+                             batch_size=64)
+    model.fit(X)
+    eager = model.predict(X[:100])
+    eager2 = model.predict(X[:50])
+    monkeypatch.setenv("GORDO_SERVE_HIPGRAPH", "1")
+    cap_first = model.predict(X[:100])     # capture
+    cap_replay = model.predict(X[100:200])  # replay, new data
+    cap_other = model.predict(X[:50])      # second shape bucket
+    assert np.array_equal(cap_first, eager)
+    assert np.array_equal(cap_other, eager2)
+    monkeypatch.delenv("GORDO_SERVE_HIPGRAPH")
+    assert np.array_equal(cap_replay, model.predict(X[100:200]))
+    pack = model._pack
+    assert len(pack._pred_graph_cache) == 2
+    pack.release_graphs()
+
+
 def test_estimator_fit_on_gpu_uses_hip():
     require_hip()
     from gordo_amd.machine.model import KerasAutoEncoder
