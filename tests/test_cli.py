@@ -278,7 +278,8 @@ machines:
             "--project-name", "wf-proj",
             "--project-revision", "1577000000000",
             "--builder-retries", "7",
-            "--keda-enabled",
+            "--with-keda", "--ml-server-hpa-type", "keda",
+            "--prometheus-server-address", "http://prometheus:9090",
             "--docker-registry", "my.registry",
             "--docker-repository", "my-repo",
             "--resource-labels", '{"team": "x"}',
@@ -568,7 +569,8 @@ globals:
          "--project-name", "wf-all",
          "--project-revision", "1600000000000",
          "--gpu-fleet", "--n-gpus", "8",
-         "--keda-enabled",
+         "--with-keda", "--ml-server-hpa-type", "keda",
+         "--prometheus-server-address", "http://prometheus:9090",
          "--builder-retries", "3",
          "--resource-labels", '{"team": "mlops"}',
          "--custom-model-builder-envs",
@@ -597,3 +599,184 @@ def test_log_level_option():
     )
     # unknown level: either rejected or ignored, but never a crash
     assert result.exit_code in (0, 2)
+
+
+_WF_CFG = """
+machines:
+  - name: wf-g-m
+    dataset: |
+      type: RandomDataset
+      tag_list: [a, b]
+      train_start_date: '2019-01-01T00:00:00+00:00'
+      train_end_date: '2019-01-02T00:00:00+00:00'
+    model: |
+      sklearn.decomposition.PCA:
+        n_components: 2
+"""
+
+
+def _render_wf(tmp_path, *extra):
+    cfg = tmp_path / "cfg.yml"
+    cfg.write_text(_WF_CFG)
+    result = CliRunner().invoke(
+        gordo,
+        ["workflow", "generate", "--machine-config", str(cfg),
+         "--project-name", "wf-g", "--project-revision", "7", *extra],
+    )
+    return result
+
+
+def test_workflow_stateful_services_rendered(tmp_path):
+    """Influx/grafana/postgres stateful services render when influx is
+    enabled (the default globals enable it — reference template
+    :186-708; VERDICT round-2 item #6)."""
+    result = _render_wf(tmp_path)
+    assert result.exit_code == 0, result.output
+    out = result.output
+    docs = list(yaml.safe_load_all(out))
+    names = {t["name"] for t in docs[0]["spec"]["templates"]}
+    assert {"gordo-influx", "gordo-influx-statefulset",
+            "gordo-influx-service", "gordo-grafana", "gordo-postgres",
+            "gordo-postgres-statefulset", "gordo-model",
+            "apply-with-retries", "cleanup-old-revisions",
+            "gordo-client", "gordo-client-waiter"} <= names
+    assert "gordo-influx-wf-g" in out
+    assert "gordo-postgres-wf-g" in out
+    assert "gordo-grafana-wf-g" in out
+    # statefulsets gate on readiness like the reference resource steps
+    assert "successCondition: status.readyReplicas > 0" in out
+
+
+def test_workflow_stateful_services_disabled(tmp_path):
+    """runtime.influx.enable false drops influx/grafana/postgres AND
+    the client backfill (reference: client wiring follows influx)."""
+    cfg = tmp_path / "cfg.yml"
+    cfg.write_text(_WF_CFG + """
+globals:
+  runtime: |
+    influx:
+      enable: false
+""")
+    result = CliRunner().invoke(
+        gordo,
+        ["workflow", "generate", "--machine-config", str(cfg),
+         "--project-name", "wf-g"],
+    )
+    assert result.exit_code == 0, result.output
+    docs = list(yaml.safe_load_all(result.output))
+    names = {t["name"] for t in docs[0]["spec"]["templates"]}
+    assert "gordo-influx" not in names
+    assert "gordo-postgres" not in names
+    assert "gordo-client" not in names
+
+
+def test_workflow_hpa_default_and_keda(tmp_path):
+    """Default HPA type is k8s_cpu (HorizontalPodAutoscaler with the
+    CPU target); keda type renders the ScaledObject and requires
+    --with-keda + --prometheus-server-address (reference
+    workflow_generator.py:37-42, 271-301)."""
+    result = _render_wf(tmp_path)
+    assert result.exit_code == 0
+    out = result.output
+    assert "HorizontalPodAutoscaler" in out
+    assert "targetCPUUtilizationPercentage: 50" in out
+    assert "keda.sh/v1alpha1" not in out
+
+    r2 = _render_wf(tmp_path, "--ml-server-hpa-type", "keda")
+    assert r2.exit_code != 0  # --with-keda required
+
+    r3 = _render_wf(tmp_path, "--ml-server-hpa-type", "keda", "--with-keda")
+    assert r3.exit_code != 0  # prometheus address required
+
+    r4 = _render_wf(
+        tmp_path, "--ml-server-hpa-type", "keda", "--with-keda",
+        "--prometheus-server-address", "http://prom:9090",
+        "--keda-prometheus-threshold", "2.5",
+    )
+    assert r4.exit_code == 0, r4.output
+    assert "keda.sh/v1alpha1" in r4.output
+    assert 'threshold: "2.5"' in r4.output
+    assert "HorizontalPodAutoscaler" not in r4.output
+    # the KEDA query's project placeholder resolves to the project name
+    assert 'project=~"wf-g"' in r4.output
+
+    r5 = _render_wf(tmp_path, "--ml-server-hpa-type", "none")
+    assert r5.exit_code == 0
+    assert "HorizontalPodAutoscaler" not in r5.output
+    assert "keda.sh/v1alpha1" not in r5.output
+
+
+def test_workflow_owner_references_and_namespace(tmp_path):
+    ref = ('[{"uid": "u1", "name": "own", "kind": "Gordo", '
+           '"apiVersion": "v1", "blockOwnerDeletion": true}]')
+    result = _render_wf(
+        tmp_path, "--owner-references", ref, "--namespace", "prod-ns"
+    )
+    assert result.exit_code == 0, result.output
+    out = result.output
+    assert '"uid": "u1"' in out
+    assert "namespace: prod-ns" in out
+    docs = list(yaml.safe_load_all(out))
+    assert docs[0]["metadata"]["ownerReferences"][0]["name"] == "own"
+    # invalid: missing required keys
+    bad = _render_wf(tmp_path, "--owner-references", '[{"uid": "u1"}]')
+    assert bad.exit_code != 0
+
+
+def test_workflow_retry_and_server_knobs(tmp_path):
+    result = _render_wf(
+        tmp_path,
+        "--retry-backoff-duration", "33s",
+        "--retry-backoff-factor", "4",
+        "--gordo-server-workers", "3",
+        "--gordo-server-threads", "12",
+        "--gordo-server-probe-timeout", "17",
+        "--gordo-server-readiness-initial-delay", "9",
+        "--gordo-server-liveness-initial-delay", "500",
+        "--server-termination-grace-period", "120",
+        "--server-target-cpu-utilization-percentage", "70",
+    )
+    assert result.exit_code == 0, result.output
+    out = result.output
+    assert 'duration: "33s"' in out
+    assert "factor: 4" in out
+    assert "GORDO_SERVER_WORKERS" in out and '"3"' in out
+    assert "GORDO_SERVER_THREADS" in out and '"12"' in out
+    assert "timeoutSeconds: 17" in out
+    assert "initialDelaySeconds: 9" in out
+    assert "initialDelaySeconds: 500" in out
+    assert "terminationGracePeriodSeconds: 120" in out
+    assert "targetCPUUtilizationPercentage: 70" in out
+
+
+def test_workflow_without_prometheus_and_labels(tmp_path):
+    result = _render_wf(
+        tmp_path,
+        "--without-prometheus",
+        "--model-builder-labels", '{"mb": "lab1"}',
+        "--server-labels", '{"srv": "lab2"}',
+        "--model-builder-class", "my.mod.Builder",
+        "--argo-binary", "argo3",
+    )
+    assert result.exit_code == 0, result.output
+    out = result.output
+    assert "run-metrics-server" not in out       # sidecar dropped
+    assert "mb: lab1" in out
+    assert "srv: lab2" in out
+    assert "MODEL_BUILDER_CLASS" in out and "my.mod.Builder" in out
+    assert "argo3 list" in out
+    # argo binary name is validated
+    bad = _render_wf(tmp_path, "--argo-binary", "rm -rf /")
+    assert bad.exit_code != 0
+
+
+def test_workflow_model_crd_per_machine(tmp_path):
+    """Each machine gets a Model CRD apply task (reference template
+    :1012) carrying its config, plus the old-revision cleanup step."""
+    result = _render_wf(tmp_path)
+    assert result.exit_code == 0
+    out = result.output
+    assert "model-crd-wf-g-m" in out
+    assert "kind: Model" in out
+    assert "equinor.com/v1" in out
+    assert "project-revision!=7" in out  # cleanup selector
