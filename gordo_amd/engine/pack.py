@@ -587,19 +587,33 @@ class LSTMPack(BasePack):
             self.store.views["Wd"][g].copy_(_glorot_uniform((fin, fout), gen))
 
     # ---- sequence forward/backward ------------------------------------
-    def _use_fused(self) -> bool:
+    def _use_fused(self, B: Optional[int] = None) -> bool:
         """The fused on-device sequence-scan kernels (lstm_seq.hip) run
-        the whole T-step recurrence in one launch per layer with Wh
-        resident in LDS; used on GPU whenever every layer fits the
-        kernel's H<=64 geometry."""
+        the whole T-step recurrence in one launch per layer (Wh in LDS
+        for H<=64, streamed from L2 for 64<H<=256). Used on GPU when
+        every layer has a fused kernel AND — for big-H stacks — the
+        scan grid fills the chip: the round-2 GPU A/B at dims
+        256/128/64 measured fused 14% SLOWER at G=16 (128 workgroups on
+        256 CUs) and 7% faster at G=64 (512 WGs), so stacks with H>64
+        take the fused path only when G*ceil(B/32) >= 256 (the
+        per-timestep grouped-GEMM fallback splits the gate columns over
+        many more workgroups and wins in the underfilled regime)."""
         if self.device.type != "cuda":
             return False
-        return all(ops.lstm_seq_available(H) for _fin, H, _rs in self.lstm_meta)
+        if not all(
+            ops.lstm_seq_available(H) for _fin, H, _rs in self.lstm_meta
+        ):
+            return False
+        if any(H > 64 for _fin, H, _rs in self.lstm_meta):
+            rows = B if B is not None else 256
+            if self.G * ((rows + 31) // 32) < 256:
+                return False
+        return True
 
     def _forward_seq(self, Xw: torch.Tensor, keep: bool):
         """Xw: [G, B, T, F]. Returns (y, cache)."""
         G, B, T, _ = Xw.shape
-        if self._use_fused():
+        if self._use_fused(B):
             return self._forward_seq_fused(Xw, keep)
         seq = Xw
         cache = []
@@ -743,7 +757,7 @@ class LSTMPack(BasePack):
         return loss
 
     def train_batch(self, Xw, Tb) -> torch.Tensor:
-        if self._use_fused():
+        if self._use_fused(Xw.shape[1]):
             return self._train_batch_fused(Xw, Tb)
         G, B, T, _ = Xw.shape
         y, cache = self._forward_seq(Xw, keep=True)
