@@ -147,15 +147,105 @@ class SineWaveDataProvider(DataProvider):
 
 class InfluxDataProvider(DataProvider):
     """
-    Placeholder for the reference's InfluxDB-backed provider. There is
-    no network in this environment; constructing it succeeds (configs
-    parse) but loading data raises.
+    InfluxDB-backed provider over the plain 1.x HTTP API (the
+    ``influxdb`` client package is not installed in this image, so this
+    speaks ``GET /query`` JSON directly via ``requests`` — behavioral
+    spec: gordo-core's InfluxDataProvider: one series per tag from
+    ``SELECT "<value_name>" FROM "<measurement>" WHERE tag = ...``).
+
+    ``uri`` accepts ``[user:pass@]host:port/dbname`` (the gordo
+    convention) or a full ``http://host:port`` plus ``database=``.
     """
 
-    def load_series(self, train_start_date, train_end_date, tag_list, resolution="10T"):
-        raise RuntimeError(
-            "InfluxDataProvider has no reachable InfluxDB in this environment"
+    def __init__(
+        self,
+        uri: Optional[str] = None,
+        api_key: Optional[str] = None,
+        api_key_header: str = "Ocp-Apim-Subscription-Key",
+        database: Optional[str] = None,
+        measurement: str = "resampled",
+        value_name: str = "Value",
+        timeout: float = 30.0,
+        **kwargs,
+    ):
+        super().__init__(
+            uri=uri, api_key=api_key, api_key_header=api_key_header,
+            database=database, measurement=measurement,
+            value_name=value_name, **kwargs,
         )
+        self.base_url, self.auth, db = parse_influx_uri(uri)
+        self.database = database or db
+        self.api_key = api_key
+        self.api_key_header = api_key_header
+        self.measurement = measurement
+        self.value_name = value_name
+        self.timeout = timeout
+
+    def _query(self, q: str):
+        import requests
+
+        headers = {}
+        if self.api_key:
+            headers[self.api_key_header] = self.api_key
+        resp = requests.get(
+            f"{self.base_url}/query",
+            params={"db": self.database, "q": q, "epoch": "ns"},
+            headers=headers,
+            auth=self.auth,
+            timeout=self.timeout,
+        )
+        resp.raise_for_status()
+        payload = resp.json()
+        results = payload.get("results", [])
+        if results and "error" in results[0]:
+            raise RuntimeError(f"influx query error: {results[0]['error']}")
+        return results
+
+    def load_series(self, train_start_date, train_end_date, tag_list,
+                    resolution="10T"):
+        for tag in tag_list:
+            q = (
+                f'SELECT "{self.value_name}" FROM "{self.measurement}" '
+                f"WHERE (\"tag\" = '{tag.name}') "
+                f"AND time >= '{pd.Timestamp(train_start_date).isoformat()}' "
+                f"AND time <= '{pd.Timestamp(train_end_date).isoformat()}'"
+            )
+            results = self._query(q)
+            series_list = (results[0] or {}).get("series") if results else None
+            if not series_list:
+                yield pd.Series(dtype="float64", name=tag.name)
+                continue
+            cols = series_list[0]["columns"]
+            vals = series_list[0]["values"]
+            ti = cols.index("time")
+            vi = cols.index(self.value_name) if self.value_name in cols else 1
+            index = pd.to_datetime(
+                [row[ti] for row in vals], utc=True, unit="ns"
+            )
+            yield pd.Series(
+                [row[vi] for row in vals], index=index, name=tag.name,
+                dtype="float64",
+            )
+
+
+def parse_influx_uri(uri: Optional[str]):
+    """``[user:pass@]host[:port][/db]`` or full http(s) URL ->
+    (base_url, auth-tuple-or-None, db-or-None)."""
+    if not uri:
+        return "http://localhost:8086", None, None
+    auth = None
+    rest = uri
+    scheme = "http"
+    if "://" in rest:
+        scheme, rest = rest.split("://", 1)
+    if "@" in rest:
+        creds, rest = rest.rsplit("@", 1)
+        user, _, pw = creds.partition(":")
+        auth = (user, pw)
+    db = None
+    if "/" in rest:
+        rest, db = rest.split("/", 1)
+    return f"{scheme}://{rest}", auth, (db or None)
 
 
 _PROVIDER_REGISTRY: Dict[str, type] = {
