@@ -213,12 +213,22 @@ def test_serializer_callbacks_and_params():
     assert params["n"] == 3
 
 
-def test_influx_provider_gated():
-    from gordo_amd.core import InfluxDataProvider
+def test_influx_provider_unreachable_fails_loudly():
+    """The (now real, HTTP-backed) influx provider fails loudly when
+    the store is unreachable — never silently empty. Live round-trip
+    coverage: tests/test_influx_http.py."""
+    import pandas as pd
 
-    p = InfluxDataProvider()
-    with pytest.raises(RuntimeError):
-        list(p.load_series(None, None, []))
+    from gordo_amd.core import InfluxDataProvider
+    from gordo_amd.core.sensor_tag import SensorTag
+
+    p = InfluxDataProvider(uri="127.0.0.1:1/none", timeout=0.5)
+    with pytest.raises(Exception):
+        list(p.load_series(
+            pd.Timestamp("2019-01-01", tz="UTC"),
+            pd.Timestamp("2019-01-02", tz="UTC"),
+            [SensorTag("t")],
+        ))
 
 
 def test_engine_spec_roundtrip():
