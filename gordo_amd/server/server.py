@@ -209,9 +209,88 @@ def run_server(
             cmd += ["--config", config_module]
         cmd.append(server_app)
         run_cmd(cmd)
+    elif workers and workers > 1:
+        logger.warning(
+            "gunicorn is not installed; serving with a prefork werkzeug "
+            "pool (%d worker processes x threads)", workers,
+        )
+        _run_prefork(host, port, workers)
     else:
         logger.warning(
             "gunicorn is not installed; serving with werkzeug (threaded)"
         )
         app = build_app()
         app.run(host=host, port=port, threaded=True, debug=False)
+
+
+def _run_prefork(host: str, port: int, workers: int):
+    """Multi-process serving without gunicorn: one shared listening
+    socket, N forked werkzeug workers accepting from it, parent
+    supervises and restarts dead workers (the reference's gunicorn
+    master/worker model, server.py:240-304 — worker processes are what
+    scale the GIL-bound request path; each worker holds its own model
+    LRU exactly like a gunicorn worker would).
+
+    Fork happens BEFORE any CUDA/HIP initialization in this process —
+    each worker lazily creates its own HIP context at first model load.
+    """
+    import errno
+    import signal
+    import socket
+
+    from werkzeug.serving import make_server
+
+    sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+    sock.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+    sock.bind((host, port))
+    sock.listen(1024)
+    sock.set_inheritable(True)
+
+    def spawn() -> int:
+        pid = os.fork()
+        if pid == 0:  # worker
+            code = 0
+            try:
+                signal.signal(signal.SIGTERM, signal.SIG_DFL)
+                signal.signal(signal.SIGINT, signal.SIG_DFL)
+                app = build_app()
+                srv = make_server(
+                    host, port, app, threaded=True, fd=sock.fileno()
+                )
+                srv.serve_forever()
+            except BaseException:
+                logger.exception("serving worker died")
+                code = 1
+            finally:
+                os._exit(code)
+        return pid
+
+    pids = {spawn() for _ in range(workers)}
+    stopping = {"flag": False}
+
+    def on_term(signum, frame):
+        stopping["flag"] = True
+        for p in list(pids):
+            try:
+                os.kill(p, signal.SIGTERM)
+            except ProcessLookupError:
+                pass
+
+    signal.signal(signal.SIGTERM, on_term)
+    signal.signal(signal.SIGINT, on_term)
+    try:
+        while pids:
+            try:
+                pid, _status = os.wait()
+            except OSError as e:
+                if e.errno == errno.EINTR:
+                    continue
+                raise
+            except ChildProcessError:
+                break
+            pids.discard(pid)
+            if not stopping["flag"]:
+                logger.warning("serving worker %d exited; restarting", pid)
+                pids.add(spawn())
+    finally:
+        sock.close()

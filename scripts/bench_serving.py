@@ -52,6 +52,10 @@ def main():
     ap.add_argument("--hipgraph", action="store_true",
                     help="hipGraph-capture the serving forward per "
                          "request shape (GORDO_SERVE_HIPGRAPH=1)")
+    ap.add_argument("--http-workers", type=int, default=0,
+                    help="serve over REAL HTTP through the prefork "
+                         "werkzeug pool with N worker processes "
+                         "(0 = in-process Flask test client)")
     ap.add_argument("--profile-stages", action="store_true",
                     help="single-thread per-stage timing of the anomaly "
                          "endpoint work: JSON decode, model forward, "
@@ -223,6 +227,11 @@ def main():
 
         if args.serve_batch:
             os.environ["GORDO_SERVE_BATCH"] = "1"
+
+        if args.http_workers > 0:
+            _run_http_bench(args, collection, sensors)
+            return
+
         from gordo_amd.server.server import build_app
 
         app = build_app()
@@ -312,6 +321,116 @@ def main():
             }
         print(json.dumps({"metric": "ml_server predictions/sec",
                           "results": results}))
+
+
+def _run_http_bench(args, collection, sensors):
+    """Real-HTTP benchmark against the prefork multi-process server:
+    the number the reference's gunicorn-worker deployment would see."""
+    import socket
+    import signal
+    import statistics
+    import subprocess
+    import urllib.request
+
+    import pandas as pd
+
+    from gordo_amd.server.utils import dataframe_to_dict
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    env = dict(os.environ)
+    env["MODEL_COLLECTION_DIR"] = collection
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "gordo_amd", "run-server",
+         "--host", "127.0.0.1", "--port", str(port),
+         "--workers", str(args.http_workers)],
+        env=env,
+    )
+    try:
+        X = pd.DataFrame(np.random.random((args.rows, N_TAGS)),
+                         columns=sensors)
+        payload = json.dumps(
+            {"X": dataframe_to_dict(X), "y": dataframe_to_dict(X)}
+        ).encode()
+        results = {}
+        endpoints = (["anomaly", "prediction"] if args.endpoint == "both"
+                     else [args.endpoint])
+        model_names = ["serve-bench"] + [
+            f"serve-bench-{i}" for i in range(1, args.n_models)
+        ]
+        for endpoint in endpoints:
+            urls = [
+                f"http://127.0.0.1:{port}/gordo/v0/bench/{name}/"
+                + ("anomaly/prediction" if endpoint == "anomaly"
+                   else "prediction")
+                for name in model_names
+            ]
+
+            def post(u):
+                req = urllib.request.Request(
+                    u, data=payload,
+                    headers={"Content-Type": "application/json"},
+                )
+                with urllib.request.urlopen(req, timeout=60) as r:
+                    assert r.status == 200
+                    r.read()
+
+            deadline = time.time() + 120
+            while True:  # wait for workers up + model load
+                try:
+                    post(urls[0])
+                    break
+                except Exception:
+                    if time.time() > deadline:
+                        raise
+                    time.sleep(0.5)
+            latencies = []
+            lock = threading.Lock()
+            counter = {"n": 0}
+
+            def worker():
+                while True:
+                    with lock:
+                        if counter["n"] >= args.rounds:
+                            return
+                        counter["n"] += 1
+                        my_url = urls[counter["n"] % len(urls)]
+                    t0 = time.perf_counter()
+                    post(my_url)
+                    with lock:
+                        latencies.append(time.perf_counter() - t0)
+
+            t0 = time.time()
+            ts = [threading.Thread(target=worker)
+                  for _ in range(args.threads)]
+            for t in ts:
+                t.start()
+            for t in ts:
+                t.join()
+            elapsed = time.time() - t0
+            rps = len(latencies) / elapsed
+            results[endpoint] = {
+                "requests_per_sec": rps,
+                "predictions_per_sec": rps * args.rows,
+                "mean_latency_ms": statistics.mean(latencies) * 1000,
+                "p50_latency_ms": statistics.median(latencies) * 1000,
+                "rounds": len(latencies),
+                "http_workers": args.http_workers,
+                "threads": args.threads,
+                "payload_rows": args.rows,
+                "n_models": args.n_models,
+            }
+        print(json.dumps({
+            "metric": "ml_server predictions/sec (real HTTP, prefork)",
+            "results": results,
+        }))
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        try:
+            proc.wait(timeout=20)
+        except subprocess.TimeoutExpired:
+            proc.kill()
 
 
 if __name__ == "__main__":

@@ -692,3 +692,73 @@ def test_frame_json_response_without_extension(monkeypatch):
     body = json.loads(resp.get_data())
     assert body["data"]["a"] == {"x": {"i": 1.0}, "y": {"i": 2.0}}
     assert body["extra"] == "1"
+
+
+@pytest.mark.timeout(120)
+def test_run_server_prefork_workers(tmp_path):
+    """Without gunicorn, --workers N serves through a prefork werkzeug
+    pool: N worker processes share one listening socket (the
+    reference's gunicorn master/worker model, server.py:240-304), all
+    answer requests, and the pool restarts a killed worker."""
+    import signal
+    import socket
+    import subprocess
+    import sys
+    import time
+    import urllib.request
+
+    import psutil
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    collection = tmp_path / "collection"
+    collection.mkdir()
+    env = dict(os.environ)
+    env.update(
+        MODEL_COLLECTION_DIR=str(collection),
+        CUDA_VISIBLE_DEVICES="", HIP_VISIBLE_DEVICES="",
+        ROCR_VISIBLE_DEVICES="",
+    )
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "gordo_amd", "run-server",
+         "--host", "127.0.0.1", "--port", str(port), "--workers", "2"],
+        env=env, stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL,
+    )
+    try:
+        url = f"http://127.0.0.1:{port}/healthcheck"
+        deadline = time.time() + 60
+        while True:
+            try:
+                with urllib.request.urlopen(url, timeout=2) as r:
+                    assert r.status == 200
+                    break
+            except Exception:
+                if time.time() > deadline:
+                    raise
+                time.sleep(0.3)
+        parent = psutil.Process(proc.pid)
+        kids = parent.children()
+        assert len(kids) == 2, kids
+        # requests keep flowing with both workers up
+        for _ in range(4):
+            with urllib.request.urlopen(url, timeout=5) as r:
+                assert r.status == 200
+        # supervision: kill one worker; the master respawns it
+        kids[0].send_signal(signal.SIGKILL)
+        deadline = time.time() + 30
+        while True:
+            alive = [c for c in parent.children() if c.is_running()]
+            if len(alive) == 2 and kids[0].pid not in [c.pid for c in alive]:
+                break
+            assert time.time() < deadline, alive
+            time.sleep(0.3)
+        with urllib.request.urlopen(url, timeout=5) as r:
+            assert r.status == 200
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        try:
+            proc.wait(timeout=20)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+            proc.wait(timeout=10)
