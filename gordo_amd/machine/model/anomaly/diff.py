@@ -341,25 +341,39 @@ class DiffBasedAnomalyDetector(AnomalyDetectorBase):
             )
             totu = np.square(tu).mean(axis=1)
 
-        tag_anomaly_scaled = pd.DataFrame(
-            ts,
-            index=data.index,
-            columns=pd.MultiIndex.from_product(
-                (("tag-anomaly-scaled",), y.columns.tolist())
-            ),
-        )
-        data = data.join(tag_anomaly_scaled)
-        data["total-anomaly-scaled"] = tots
+        # assemble every column family in ONE concat: sequential
+        # MultiIndex .join()s re-index the growing frame per family and
+        # were the dominant pandas cost of the serving request
+        # (column names/order identical to the reference's join chain,
+        # reference diff.py:356-444)
+        y_cols = y.columns.tolist()
+        index = data.index
 
-        unscaled_abs_diff = pd.DataFrame(
-            data=tu,
-            index=data.index,
-            columns=pd.MultiIndex.from_product(
-                (("tag-anomaly-unscaled",), y.columns.tolist())
-            ),
-        )
-        data = data.join(unscaled_abs_diff)
-        data["total-anomaly-unscaled"] = totu
+        def fam(name, values, cols=y_cols):
+            return pd.DataFrame(
+                values,
+                index=index,
+                columns=pd.MultiIndex.from_product(((name,), cols)),
+            )
+
+        def fam1(name, values):
+            # scalar families carry an empty sub-level, exactly like
+            # `frame[name] = values` on a 2-level frame
+            return pd.DataFrame(
+                np.asarray(values).reshape(-1, 1),
+                index=index,
+                columns=pd.MultiIndex.from_tuples([(name, "")]),
+            )
+
+        tag_anomaly_scaled = fam("tag-anomaly-scaled", ts)
+        unscaled_abs_diff = fam("tag-anomaly-unscaled", tu)
+        blocks = [
+            data,
+            tag_anomaly_scaled,
+            fam1("total-anomaly-scaled", tots),
+            unscaled_abs_diff,
+            fam1("total-anomaly-unscaled", totu),
+        ]
 
         if self.window is not None and self.smoothing_method is not None:
             smooth_tag_anomaly_scaled = self._smoothing(tag_anomaly_scaled)
@@ -368,9 +382,12 @@ class DiffBasedAnomalyDetector(AnomalyDetectorBase):
                     ["smooth-tag-anomaly-scaled"], level=0
                 )
             )
-            data = data.join(smooth_tag_anomaly_scaled)
-            data["smooth-total-anomaly-scaled"] = self._smoothing(
-                data["total-anomaly-scaled"]
+            blocks.append(smooth_tag_anomaly_scaled)
+            blocks.append(
+                fam1(
+                    "smooth-total-anomaly-scaled",
+                    self._smoothing(pd.Series(tots, index=index)),
+                )
             )
             smooth_tag_anomaly_unscaled = self._smoothing(unscaled_abs_diff)
             smooth_tag_anomaly_unscaled.columns = (
@@ -378,27 +395,32 @@ class DiffBasedAnomalyDetector(AnomalyDetectorBase):
                     ["smooth-tag-anomaly-unscaled"], level=0
                 )
             )
-            data = data.join(smooth_tag_anomaly_unscaled)
-            data["smooth-total-anomaly-unscaled"] = self._smoothing(
-                data["total-anomaly-unscaled"]
+            blocks.append(smooth_tag_anomaly_unscaled)
+            blocks.append(
+                fam1(
+                    "smooth-total-anomaly-unscaled",
+                    self._smoothing(pd.Series(totu, index=index)),
+                )
             )
 
         if hasattr(self, "feature_thresholds_"):
-            confidence = unscaled_abs_diff.values / self.feature_thresholds_.values
-            data = data.join(
-                pd.DataFrame(
+            confidence = tu / self.feature_thresholds_.values
+            blocks.append(
+                fam(
+                    "anomaly-confidence",
                     confidence,
-                    index=unscaled_abs_diff.index,
-                    columns=pd.MultiIndex.from_product(
-                        (("anomaly-confidence",), data["model-output"].columns)
-                    ),
+                    data["model-output"].columns,
                 )
             )
 
         if hasattr(self, "aggregate_threshold_"):
-            data["total-anomaly-confidence"] = (
-                data["total-anomaly-scaled"] / self.aggregate_threshold_
+            blocks.append(
+                fam1(
+                    "total-anomaly-confidence",
+                    tots / self.aggregate_threshold_,
+                )
             )
+        data = pd.concat(blocks, axis=1, copy=False)
 
         if self.require_thresholds and not any(
             hasattr(self, attr)

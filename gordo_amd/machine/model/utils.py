@@ -72,11 +72,13 @@ def make_base_dataframe(
         starts = [None] * n_out
         ends = [None] * n_out
 
-    data = pd.DataFrame(
-        {("start", ""): starts, ("end", ""): ends},
-        columns=pd.MultiIndex.from_product((("start", "end"), ("",))),
-        index=normalized_index,
-    )
+    blocks = [
+        pd.DataFrame(
+            {("start", ""): starts, ("end", ""): ends},
+            columns=pd.MultiIndex.from_product((("start", "end"), ("",))),
+            index=normalized_index,
+        )
+    ]
 
     for name, values, _tags in (
         ("model-input", model_input, tags),
@@ -89,10 +91,12 @@ def make_base_dataframe(
         else:
             sub_names = [str(i) for i in range(values.shape[1])]
         columns = pd.MultiIndex.from_tuples((name, s) for s in sub_names)
-        data = data.join(
+        blocks.append(
             pd.DataFrame(values, columns=columns, index=normalized_index)
         )
-    return data
+    # one concat, not sequential joins (each join re-indexes the frame;
+    # this is the serving hot path — reference server flow SURVEY §3.2)
+    return pd.concat(blocks, axis=1, copy=False)
 
 
 def trail_min_max(x, w: int):
