@@ -12,6 +12,7 @@ the model output is resident (kernel K9 of SURVEY.md §2.3).
 """
 from __future__ import annotations
 
+import functools
 import logging
 from datetime import timedelta
 from typing import Optional, Union
@@ -353,7 +354,7 @@ class DiffBasedAnomalyDetector(AnomalyDetectorBase):
             return pd.DataFrame(
                 values,
                 index=index,
-                columns=pd.MultiIndex.from_product(((name,), cols)),
+                columns=_mi_product(name, tuple(cols)),
             )
 
         def fam1(name, values):
@@ -362,7 +363,7 @@ class DiffBasedAnomalyDetector(AnomalyDetectorBase):
             return pd.DataFrame(
                 np.asarray(values).reshape(-1, 1),
                 index=index,
-                columns=pd.MultiIndex.from_tuples([(name, "")]),
+                columns=_mi_product(name, ("",)),
             )
 
         tag_anomaly_scaled = fam("tag-anomaly-scaled", ts)
@@ -432,6 +433,14 @@ class DiffBasedAnomalyDetector(AnomalyDetectorBase):
                 "these thresholds before calling `.anomaly`"
             )
         return data
+
+
+@functools.lru_cache(maxsize=1024)
+def _mi_product(name: str, cols: tuple) -> pd.MultiIndex:
+    """Column MultiIndexes are static per (family, tag list) — building
+    them per request cost ~8 ms/call of the serving path (pandas
+    factorize); MultiIndex objects are immutable, safe to share."""
+    return pd.MultiIndex.from_product(((name,), list(cols)))
 
 
 class DiffBasedKFCVAnomalyDetector(DiffBasedAnomalyDetector):

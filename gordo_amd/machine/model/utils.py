@@ -36,6 +36,16 @@ def _tag_name(tag) -> str:
     return tag.name if isinstance(tag, SensorTag) else str(tag)
 
 
+_START_END_COLS = pd.MultiIndex.from_product((("start", "end"), ("",)))
+
+
+@functools.lru_cache(maxsize=1024)
+def _mi_tuples(name: str, sub_names: tuple) -> pd.MultiIndex:
+    """Static per (family, tag list) — cached; ~ms-scale pandas
+    factorize per construction on the serving path."""
+    return pd.MultiIndex.from_tuples((name, s) for s in sub_names)
+
+
 def make_base_dataframe(
     tags: Union[List[SensorTag], List[str]],
     model_input: np.ndarray,
@@ -75,7 +85,7 @@ def make_base_dataframe(
     blocks = [
         pd.DataFrame(
             {("start", ""): starts, ("end", ""): ends},
-            columns=pd.MultiIndex.from_product((("start", "end"), ("",))),
+            columns=_START_END_COLS,
             index=normalized_index,
         )
     ]
@@ -90,7 +100,7 @@ def make_base_dataframe(
             sub_names = [_tag_name(t) for t in _tags]
         else:
             sub_names = [str(i) for i in range(values.shape[1])]
-        columns = pd.MultiIndex.from_tuples((name, s) for s in sub_names)
+        columns = _mi_tuples(name, tuple(sub_names))
         blocks.append(
             pd.DataFrame(values, columns=columns, index=normalized_index)
         )
