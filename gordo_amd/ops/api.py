@@ -178,6 +178,43 @@ def anomaly_score(out, y, scale, minv, feat_thr, agg_thr):
                                         float(agg_thr))
 
 
+def trail_min_max(X, w: int):
+    """Per-row ``rolling(w).min().max()`` (K10). GPU tensors run the
+    HIP kernel; CPU falls back to the scipy O(n) path
+    (machine/model/utils.trail_min_max)."""
+    if _on_gpu(X):
+        return _require_hip().trail_min_max(X, int(w))
+    from ..machine.model.utils import trail_min_max as cpu_tmm
+    import numpy as _np
+
+    return torch.as_tensor(
+        _np.atleast_1d(cpu_tmm(X.numpy().T, int(w))), dtype=torch.float32
+    )
+
+
+def windowed_quantile(X, w: int, q: float):
+    """Per-row rolling(w).quantile(q) with pandas min_periods=w NaN
+    semantics (K11; q=0.5 == the smm rolling median)."""
+    if _on_gpu(X):
+        return _require_hip().windowed_quantile(X, int(w), float(q))
+    import pandas as _pd
+
+    df = _pd.DataFrame(X.numpy().T)
+    out = df.rolling(int(w)).quantile(float(q)).to_numpy().T[:, int(w) - 1:]
+    return torch.as_tensor(out, dtype=torch.float32)
+
+
+def row_quantile(X, q: float):
+    """Per-row NaN-dropping linear-interpolated quantile (K12)."""
+    if _on_gpu(X):
+        return _require_hip().row_quantile(X, float(q))
+    import numpy as _np
+
+    return torch.as_tensor(
+        _np.nanquantile(X.numpy(), float(q), axis=1), dtype=torch.float32
+    )
+
+
 def lstm_pointwise_fwd(gates, c_prev):
     if _on_gpu(gates):
         return _require_hip().lstm_pointwise_fwd(gates, c_prev)

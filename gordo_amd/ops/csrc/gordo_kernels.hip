@@ -703,7 +703,21 @@ torch::Tensor lstm_seq_bwd(torch::Tensor dSeq, torch::Tensor gacts,
                            bool last_only);
 }  // namespace gordo_lstm
 
+// device threshold/statistics kernels (thresholds.hip: K10-K12)
+namespace gordo_thresholds {
+torch::Tensor trail_min_max(torch::Tensor X, int64_t w);
+torch::Tensor windowed_quantile(torch::Tensor X, int64_t w, double q);
+torch::Tensor row_quantile(torch::Tensor X, double q);
+}  // namespace gordo_thresholds
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("trail_min_max", &gordo_thresholds::trail_min_max,
+          "rolling(w).min().max() per row (K10)");
+  mod.def("windowed_quantile", &gordo_thresholds::windowed_quantile,
+          "rolling(w).quantile(q) per row — q=0.5 is the smm smoothing "
+          "(K11)");
+  mod.def("row_quantile", &gordo_thresholds::row_quantile,
+          "NaN-dropping linear-interpolated quantile per row (K12)");
   mod.def("anomaly_score", &gordo_anomaly::anomaly_score,
           "fused DiffBased anomaly scoring (serving hot path)");
   mod.def("lstm_seq_bwd_v3", &gordo_lstm::lstm_seq_bwd_v3,

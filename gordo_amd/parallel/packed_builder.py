@@ -130,6 +130,10 @@ class PackedFleetBuilder:
         self.data_workers = data_workers
         self.save_models = save_models
         self.replace_cache = replace_cache
+        # async model-save pool: serialization of a finished group
+        # overlaps the next group's GPU fits (joined in build_all)
+        self._save_pool = concurrent.futures.ThreadPoolExecutor(8)
+        self._save_futures: List[Tuple[MachinePlan, Any]] = []
         # wall-clock phase ledger for the build-step budget table
         # (BASELINE.md): sequential wall segments; overlapped work
         # (threaded final fit) is recorded under its own key.
@@ -190,6 +194,19 @@ class PackedFleetBuilder:
 
         if fallback:
             self._phase("fallback_builds", time.time() - t0)
+        # join outstanding async saves; failures surface per machine
+        t0 = time.time()
+        if self._save_futures:
+            for p, f in self._save_futures:
+                try:
+                    f.result()
+                except Exception as e:
+                    logger.exception(
+                        "async model save failed: %s", p.machine.name
+                    )
+                    p.error = e
+            self._save_futures.clear()
+        self._phase("save_join", time.time() - t0)
         self._phase("total", time.time() - t_all0)
         if logger.isEnabledFor(logging.INFO):
             budget = ", ".join(
@@ -447,8 +464,10 @@ class PackedFleetBuilder:
 
         self._phase("adopt", time.time() - t_seg)
 
-        # save + register (parallel: pickle+json dumps are I/O bound)
-        t_seg = time.time()
+        # save + register — ASYNC: pickle+json dumps are I/O/CPU bound
+        # and independent of the next group's GPU work, so they run on
+        # the builder-wide save pool and are joined at the end of
+        # build_all (phase ledger records the non-overlapped tail)
         if self.save_models and self.output_dir:
             def save_one(p):
                 out = os.path.join(self.output_dir, p.machine.name)
@@ -460,9 +479,10 @@ class PackedFleetBuilder:
                         out,
                     )
 
-            with concurrent.futures.ThreadPoolExecutor(8) as ex:
-                list(ex.map(save_one, group))
-        self._phase("save", time.time() - t_seg)
+            pool = self._save_pool
+            self._save_futures.extend(
+                (p, pool.submit(save_one, p)) for p in group
+            )
         pack.release_graphs()
         logger.info(
             "Packed build of %d machines done in %.2fs",
@@ -507,6 +527,81 @@ class PackedFleetBuilder:
             Xte.append(np.asarray(xv, dtype=np.float32))
         return Xtr, Xte
 
+    def _device_fold_thresholds(
+        self, group, fold_pack, Yd_train, y_list, test_idx, preds_t
+    ):
+        """Batched DEVICE computation of the DiffBased per-fold
+        thresholds (kernels K10/K11 of SURVEY.md §2.3) while the fold
+        predictions are still resident: per-machine MinMaxScaler from
+        the train y, scaled-MSE/MAE residuals, and
+        rolling(w).min().max() over the whole pack in three kernel
+        launches — replacing the per-machine pandas/scipy loop of the
+        scoring phase. Returns None on CPU or when no plain DiffBased
+        detector is in the group (KFCV keeps its reassembled-series
+        CPU path)."""
+        if fold_pack.device.type != "cuda":
+            return None
+        wants = [
+            g_idx for g_idx, p in enumerate(group)
+            if p.detector is not None
+            and not isinstance(p.detector, DiffBasedKFCVAnomalyDetector)
+        ]
+        if not wants:
+            return None
+        try:
+            with torch.no_grad():
+                G = len(group)
+                n_out = preds_t.shape[1]
+                pred = preds_t.float()
+                Yte = self._stack(
+                    [y[test_idx] for y in y_list], fold_pack
+                ).float()[:, -n_out:, :]
+                ytr = Yd_train.float()
+                ymin = ytr.amin(dim=1)          # [G, F]
+                yrange = ytr.amax(dim=1) - ymin
+                # sklearn MinMaxScaler handle_zeros_in_scale: scale=1
+                scale = torch.where(
+                    yrange == 0, torch.ones_like(yrange), 1.0 / yrange
+                ).unsqueeze(1)                   # [G, 1, F]
+                diff_scaled = (pred - Yte) * scale
+                scaled_mse = (diff_scaled ** 2).mean(dim=2)   # [G, n]
+                mae = (Yte - pred).abs()                      # [G, n, F]
+                F = mae.shape[2]
+                mae_rows = mae.transpose(1, 2).reshape(G * F, n_out)
+                agg = ops.trail_min_max(scaled_mse, 6).cpu().numpy()
+                tag = (
+                    ops.trail_min_max(mae_rows.contiguous(), 6)
+                    .cpu().numpy().reshape(G, F)
+                )
+                windows = sorted(
+                    {
+                        int(group[g].detector.window)
+                        for g in wants
+                        if getattr(group[g].detector, "window", None)
+                        is not None
+                    }
+                )
+                smooth = {}
+                for w in windows:
+                    if n_out < w:
+                        smooth[w] = (
+                            np.full(G, np.nan),
+                            np.full((G, F), np.nan),
+                        )
+                        continue
+                    smooth[w] = (
+                        ops.trail_min_max(scaled_mse, w).cpu().numpy(),
+                        ops.trail_min_max(mae_rows.contiguous(), w)
+                        .cpu().numpy().reshape(G, F),
+                    )
+            return {"agg": agg, "tag": tag, "smooth": smooth}
+        except Exception:
+            logger.warning(
+                "device threshold path failed; falling back to CPU",
+                exc_info=True,
+            )
+            return None
+
     def _fit_folds(
         self, folds, group, Xraw_list, y_list, spec, fit_args, pack,
         init_snapshot,
@@ -532,8 +627,12 @@ class PackedFleetBuilder:
                     t0 = time.time()
                     Xtest = self._stack(Xte, fold_pack)
                     with torch.no_grad():
-                        preds = fold_pack.predict(Xtest).float().cpu().numpy()
-                    out[fold_i] = (preds, t_fit, time.time() - t0)
+                        preds_t = fold_pack.predict(Xtest)
+                        dev_thr = self._device_fold_thresholds(
+                            group, fold_pack, Yd, y_list, test_idx, preds_t
+                        )
+                        preds = preds_t.float().cpu().numpy()
+                    out[fold_i] = (preds, t_fit, time.time() - t0, dev_thr)
             except Exception as e:  # surface via the caller
                 out[fold_i] = e
 
@@ -648,7 +747,7 @@ class PackedFleetBuilder:
 
         for fold_i, (train_idx, test_idx) in enumerate(folds):
             t_f0 = time.time()
-            preds, t_fit, t_pred = fold_preds[fold_i]
+            preds, t_fit, t_pred, dev_thr = fold_preds[fold_i]
             t_s0 = time.time()
 
             for g_idx, p in enumerate(group):
@@ -689,17 +788,25 @@ class PackedFleetBuilder:
                     ).mean(axis=1)
                 elif p.detector is not None:
                     det = p.detector
-                    fold_scaler = sk_clone(det.scaler)
-                    fold_scaler.fit(y_list[g_idx][train_idx])
-                    scaled_mse = (
-                        (fold_scaler.transform(y_pred)
-                         - fold_scaler.transform(y_true)) ** 2
-                    ).mean(axis=1)
-                    mae = np.abs(np.asarray(y_true) - np.asarray(y_pred))
-                    agg_thr = _trail_min_max(scaled_mse, 6)
-                    tag_thr = pd.Series(
-                        _trail_min_max(mae, 6), name=f"fold-{fold_i}"
-                    )
+                    if dev_thr is not None:
+                        # K10 device path: thresholds came back with the
+                        # fold predictions (one batched kernel per stat)
+                        agg_thr = float(dev_thr["agg"][g_idx])
+                        tag_thr = pd.Series(
+                            dev_thr["tag"][g_idx], name=f"fold-{fold_i}"
+                        )
+                    else:
+                        fold_scaler = sk_clone(det.scaler)
+                        fold_scaler.fit(y_list[g_idx][train_idx])
+                        scaled_mse = (
+                            (fold_scaler.transform(y_pred)
+                             - fold_scaler.transform(y_true)) ** 2
+                        ).mean(axis=1)
+                        mae = np.abs(np.asarray(y_true) - np.asarray(y_pred))
+                        agg_thr = _trail_min_max(scaled_mse, 6)
+                        tag_thr = pd.Series(
+                            _trail_min_max(mae, 6), name=f"fold-{fold_i}"
+                        )
                     if not hasattr(det, "aggregate_thresholds_per_fold_"):
                         det.aggregate_thresholds_per_fold_ = {}
                         det.feature_thresholds_per_fold_ = pd.DataFrame()
@@ -714,11 +821,18 @@ class PackedFleetBuilder:
                     # smoothing window is per machine, not per group
                     window = getattr(det, "window", None)
                     if window is not None:
-                        s_agg = _trail_min_max(scaled_mse, window)
-                        s_tag = pd.Series(
-                            _trail_min_max(mae, window),
-                            name=f"fold-{fold_i}",
-                        )
+                        if dev_thr is not None:
+                            s_agg_v, s_tag_v = dev_thr["smooth"][int(window)]
+                            s_agg = float(s_agg_v[g_idx])
+                            s_tag = pd.Series(
+                                s_tag_v[g_idx], name=f"fold-{fold_i}"
+                            )
+                        else:
+                            s_agg = _trail_min_max(scaled_mse, window)
+                            s_tag = pd.Series(
+                                _trail_min_max(mae, window),
+                                name=f"fold-{fold_i}",
+                            )
                         det.smooth_aggregate_thresholds_per_fold_[
                             f"fold-{fold_i}"
                         ] = s_agg
