@@ -534,6 +534,45 @@ class DiffBasedKFCVAnomalyDetector(DiffBasedAnomalyDetector):
         return self._calculate_threshold(self._absolute_error(y_true, y_pred))
 
     def _calculate_threshold(self, validation_metric):
+        dev = self._device_threshold(validation_metric)
+        if dev is not None:
+            return dev
         return self._smoothing(validation_metric).quantile(
             self.threshold_percentile
         )
+
+    def _device_threshold(self, validation_metric):
+        """K11+K12 device path: rolling(window).median() then
+        NaN-dropping quantile, both as HIP kernels when a GPU is
+        present and the smoothing is the default smm. Matches the
+        pandas pipeline at fp32 (min_periods=window NaN semantics in
+        the windowed kernel; linear interpolation in both). Returns a
+        Series for DataFrame input, a float for Series input, or None
+        to take the pandas path."""
+        try:
+            import torch
+
+            from .... import ops
+
+            if (
+                self.smoothing_method != "smm"
+                or not torch.cuda.is_available()
+                or not ops.hip_available()
+                or len(validation_metric) < max(self.window, 64)
+                or len(validation_metric) > 16384
+            ):
+                return None
+            arr = np.asarray(validation_metric, dtype=np.float32)
+            cols = arr.reshape(len(arr), -1).T  # [R, N]
+            dev = torch.as_tensor(np.ascontiguousarray(cols),
+                                  device="cuda")
+            sm = ops.windowed_quantile(dev, int(self.window), 0.5)
+            q = ops.row_quantile(sm, float(self.threshold_percentile))
+            out = q.cpu().numpy().astype(np.float64)
+            if isinstance(validation_metric, pd.DataFrame):
+                return pd.Series(out, index=validation_metric.columns,
+                                 name=self.threshold_percentile)
+            return float(out[0])
+        except Exception:
+            logger.debug("device threshold unavailable", exc_info=True)
+            return None

@@ -436,3 +436,98 @@ def test_lstm_seq_bwd_v3_matches_v1(G, B, T, H, last_only, monkeypatch):
     monkeypatch.delenv("GORDO_LSTM_V1")
     v3 = ops.lstm_seq_bwd_v3(dSeq, ga, cs, Wh, last_only)
     assert torch.equal(v1, v3)
+
+
+def test_trail_min_max_kernel_vs_pandas():
+    """K10 device kernel vs pandas rolling(w).min().max()."""
+    require_hip()
+    import pandas as pd
+
+    rng = np.random.default_rng(5)
+    X = rng.random((37, 4320)).astype("float32") * 10
+    for w in (6, 144):
+        got = ops.trail_min_max(torch.tensor(X, device="cuda"), w)
+        want = pd.DataFrame(X.T).rolling(w).min().max(axis=0).to_numpy()
+        np.testing.assert_allclose(
+            got.cpu().numpy(), want, rtol=1e-6, atol=1e-6
+        )
+    # short series -> NaN (pandas: all-NaN rolling -> max NaN)
+    short = torch.tensor(X[:2, :4], device="cuda")
+    assert torch.isnan(ops.trail_min_max(short, 6)).all()
+
+
+def test_windowed_quantile_kernel_vs_pandas():
+    """K11 device kernel (q=0.5 == pandas rolling median,
+    min_periods=w NaN semantics)."""
+    require_hip()
+    import pandas as pd
+
+    rng = np.random.default_rng(6)
+    X = rng.random((9, 700)).astype("float32")
+    X[3, 100:110] = np.nan  # NaN patch -> NaN windows
+    for w, q in ((144, 0.5), (7, 0.5), (12, 0.25)):
+        got = ops.windowed_quantile(
+            torch.tensor(X, device="cuda"), w, q
+        ).cpu().numpy()
+        want = (
+            pd.DataFrame(X.T).rolling(w).quantile(q)
+            .to_numpy().T[:, w - 1:]
+        )
+        np.testing.assert_allclose(got, want, rtol=2e-6, atol=2e-6)
+
+
+def test_row_quantile_kernel_vs_numpy():
+    """K12 device kernel vs NaN-dropping linear-interp quantile."""
+    require_hip()
+    rng = np.random.default_rng(7)
+    X = rng.random((21, 4177)).astype("float32")
+    X[2, :50] = np.nan
+    X[20, ::3] = np.nan
+    for q in (0.99, 0.5, 0.1):
+        got = ops.row_quantile(torch.tensor(X, device="cuda"), q)
+        want = np.nanquantile(X.astype("float64"), q, axis=1)
+        np.testing.assert_allclose(
+            got.cpu().numpy(), want, rtol=2e-5, atol=2e-6
+        )
+
+
+def test_device_fold_thresholds_match_cpu():
+    """The batched device threshold path of the packed builder (K10
+    over scaled-MSE/MAE with per-machine MinMax scaling) equals the
+    CPU pandas/scipy computation."""
+    require_hip()
+    from sklearn.preprocessing import MinMaxScaler
+
+    from gordo_amd.machine.model.utils import trail_min_max as cpu_tmm
+
+    rng = np.random.default_rng(8)
+    G, n_tr, n_te, F = 5, 400, 150, 7
+    ytr = rng.random((G, n_tr, F)).astype("float32")
+    yte = rng.random((G, n_te, F)).astype("float32")
+    pred = rng.random((G, n_te, F)).astype("float32")
+
+    ytr_d = torch.tensor(ytr, device="cuda")
+    yte_d = torch.tensor(yte, device="cuda").float()
+    pred_d = torch.tensor(pred, device="cuda").float()
+    ymin = ytr_d.amin(dim=1)
+    yrange = ytr_d.amax(dim=1) - ymin
+    scale = torch.where(
+        yrange == 0, torch.ones_like(yrange), 1.0 / yrange
+    ).unsqueeze(1)
+    scaled_mse = (((pred_d - yte_d) * scale) ** 2).mean(dim=2)
+    mae = (yte_d - pred_d).abs()
+    agg = ops.trail_min_max(scaled_mse, 6).cpu().numpy()
+    tag = (
+        ops.trail_min_max(
+            mae.transpose(1, 2).reshape(G * F, n_te).contiguous(), 6
+        ).cpu().numpy().reshape(G, F)
+    )
+    for g in range(G):
+        sc = MinMaxScaler().fit(ytr[g])
+        sm = ((sc.transform(pred[g]) - sc.transform(yte[g])) ** 2).mean(
+            axis=1
+        )
+        np.testing.assert_allclose(agg[g], cpu_tmm(sm, 6), rtol=1e-4)
+        np.testing.assert_allclose(
+            tag[g], cpu_tmm(np.abs(yte[g] - pred[g]), 6), rtol=1e-4
+        )
