@@ -336,6 +336,11 @@ def _run_http_bench(args, collection, sensors):
 
     from gordo_amd.server.utils import dataframe_to_dict
 
+    # GPU boxes may set http_proxy; 127.0.0.1 must never route there
+    opener = urllib.request.build_opener(
+        urllib.request.ProxyHandler({})
+    )
+
     with socket.socket() as s:
         s.bind(("127.0.0.1", 0))
         port = s.getsockname()[1]
@@ -372,19 +377,28 @@ def _run_http_bench(args, collection, sensors):
                     u, data=payload,
                     headers={"Content-Type": "application/json"},
                 )
-                with urllib.request.urlopen(req, timeout=60) as r:
+                with opener.open(req, timeout=60) as r:
                     assert r.status == 200
                     r.read()
 
-            deadline = time.time() + 120
+            deadline = time.time() + 180
+            last_err = None
             while True:  # wait for workers up + model load
                 try:
                     post(urls[0])
                     break
-                except Exception:
+                except Exception as e:
+                    last_err = e
                     if time.time() > deadline:
+                        print(json.dumps({
+                            "metric": "ml_server predictions/sec "
+                                      "(real HTTP, prefork)",
+                            "error": repr(last_err),
+                        }))
                         raise
                     time.sleep(0.5)
+            for u in urls:
+                post(u)  # warm every model
             latencies = []
             lock = threading.Lock()
             counter = {"n": 0}
