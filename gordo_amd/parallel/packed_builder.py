@@ -553,9 +553,11 @@ class PackedFleetBuilder:
                 G = len(group)
                 n_out = preds_t.shape[1]
                 pred = preds_t.float()
-                Yte = self._stack(
-                    [y[test_idx] for y in y_list], fold_pack
-                ).float()[:, -n_out:, :]
+                # y stays fp32 on device (the bf16 compute mirror would
+                # round the residual statistics ~0.4% vs the CPU path)
+                Yte = torch.from_numpy(
+                    np.stack([y[test_idx] for y in y_list])
+                ).to(fold_pack.device, torch.float32)[:, -n_out:, :]
                 ytr = Yd_train.float()
                 ymin = ytr.amin(dim=1)          # [G, F]
                 yrange = ytr.amax(dim=1) - ymin
