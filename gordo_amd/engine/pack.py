@@ -728,15 +728,12 @@ class LSTMPack(BasePack):
                 lc["gacts"], lc["cs"], Wh, last_only,
             ).view(G, B * T, 4 * H)
             hs = lc["hs"]
-            h_prev_all = torch.cat(
-                [torch.zeros_like(hs[:, :, :1]), hs[:, :, :-1]], dim=2
-            )
             dWx, dbl = ops.grouped_linear_wgrad(
                 lc["seq_in"].reshape(G, B * T, fin), dG_flat
             )
-            dWh, _ = ops.grouped_linear_wgrad(
-                h_prev_all.reshape(G, B * T, H), dG_flat
-            )
+            # dWh reads h_{t-1} via in-kernel shifted addressing — no
+            # h_prev_all concat (was a full [G,B,T,H] copy per layer)
+            dWh, _ = ops.grouped_linear_wgrad_hprev(hs, dG_flat, T)
             self.store.gviews[f"Wx{li}"].copy_(dWx)
             self.store.gviews[f"Wh{li}"].copy_(dWh)
             self.store.gviews[f"bl{li}"].copy_(dbl)
@@ -744,7 +741,7 @@ class LSTMPack(BasePack):
             # allocator reuses it for the next layer's dSeq/transients
             # (whole-batch caches are the memory ceiling at G~100+).
             cache[li] = None
-            del lc, hs, h_prev_all
+            del lc, hs
             prev_dSeq = dSeq
             if li > 0:
                 dSeq = ops.grouped_linear_bwd_data(dG_flat, Wx).view(
@@ -804,16 +801,11 @@ class LSTMPack(BasePack):
                 dG[:, :, t] = dgates
                 dh_carry = ops.grouped_linear_bwd_data(dgates, Wh)
             # batched weight grads over all (B, T) rows
-            h_prev_all = torch.cat(
-                [torch.zeros_like(hs[:, :, :1]), hs[:, :, :-1]], dim=2
-            )
             dG_flat = dG.view(G, B * T, 4 * H)
             dWx, dbl = ops.grouped_linear_wgrad(
                 lc["seq_in"].reshape(G, B * T, fin), dG_flat
             )
-            dWh, _ = ops.grouped_linear_wgrad(
-                h_prev_all.reshape(G, B * T, H), dG_flat
-            )
+            dWh, _ = ops.grouped_linear_wgrad_hprev(hs, dG_flat, T)
             self.store.gviews[f"Wx{li}"].copy_(dWx)
             self.store.gviews[f"Wh{li}"].copy_(dWh)
             self.store.gviews[f"bl{li}"].copy_(dbl)

@@ -84,6 +84,20 @@ def grouped_linear_wgrad(X, dZ) -> Tuple[torch.Tensor, torch.Tensor]:
     return ref.grouped_linear_wgrad(X, dZ)
 
 
+def grouped_linear_wgrad_hprev(hs, dZ, T: int):
+    """dWh = h_prev^T @ dG over flattened (b, t) rows, where
+    h_prev(b, t) = hs[b, t-1] and zero at t == 0 — the in-kernel
+    shifted addressing removes the h_prev_all concat from BPTT
+    (a full [G,B,T,H] copy per layer per batch)."""
+    if _on_gpu(hs):
+        return _require_hip().grouped_linear_wgrad_hprev(hs, dZ, int(T))
+    G, B, T_, H = hs.shape
+    h_prev = torch.cat(
+        [torch.zeros_like(hs[:, :, :1]), hs[:, :, :-1]], dim=2
+    ).reshape(G, B * T_, H)
+    return ref.grouped_linear_wgrad(h_prev, dZ)
+
+
 def grouped_gemm_acc(A, B, C):
     if _on_gpu(A):
         return _require_hip().grouped_gemm_acc(A, B, C)

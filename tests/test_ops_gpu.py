@@ -531,3 +531,24 @@ def test_device_fold_thresholds_match_cpu():
         np.testing.assert_allclose(
             tag[g], cpu_tmm(np.abs(yte[g] - pred[g]), 6), rtol=1e-4
         )
+
+
+@pytest.mark.parametrize("G,B,T,H,N4", [(3, 16, 12, 42, 168),
+                                        (1, 9, 7, 25, 100)])
+def test_grouped_wgrad_hprev_matches_concat(G, B, T, H, N4):
+    """The shifted-addressing recurrent wgrad (no h_prev_all concat)
+    equals wgrad over the explicitly concatenated h_prev rows."""
+    require_hip()
+    hs = _rand(G, B, T, H, seed=60)
+    dG = _rand(G, B * T, N4, seed=61)
+    h_prev = torch.cat(
+        [torch.zeros_like(hs[:, :, :1]), hs[:, :, :-1]], dim=2
+    ).reshape(G, B * T, H)
+    want_W, want_b = ops.grouped_linear_wgrad(
+        to_dev_bf16(h_prev), to_dev_bf16(dG)
+    )
+    got_W, got_b = ops.grouped_linear_wgrad_hprev(
+        to_dev_bf16(hs), to_dev_bf16(dG), T
+    )
+    torch.testing.assert_close(got_W, want_W, rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(got_b, want_b, rtol=1e-5, atol=1e-5)
