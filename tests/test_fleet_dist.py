@@ -137,21 +137,39 @@ def test_fleet_build_multi_rank(tmp_path, world, n_machines):
         GORDO_FLEET_STATUS_FILE=str(status_file),
         GORDO_FLEET_REPLACE_CACHE="0",
     )
-    proc = subprocess.run(
-        [
-            sys.executable, "-m", "torch.distributed.run",
-            "--nnodes=1", f"--nproc-per-node={world}",
-            "--standalone", "--local-addr", "127.0.0.1",
-            "-m", "gordo_amd.cli.fleet_worker",
-        ],
-        env=env,
-        capture_output=True,
-        text=True,
-        timeout=280,
+    def run_once():
+        return subprocess.run(
+            [
+                sys.executable, "-m", "torch.distributed.run",
+                "--nnodes=1", f"--nproc-per-node={world}",
+                "--standalone", "--local-addr", "127.0.0.1",
+                "-m", "gordo_amd.cli.fleet_worker",
+            ],
+            env=env,
+            capture_output=True,
+            text=True,
+            timeout=280,
+        )
+
+    # one retry: an 8-process rendezvous on an oversubscribed CI host
+    # can time out under unrelated suite load; a real regression fails
+    # both attempts and both outputs are shown
+    proc = run_once()
+    summary = (
+        json.loads(status_file.read_text())
+        if status_file.exists()
+        else {}
     )
-    assert proc.returncode == 0, proc.stderr[-3000:]
-    summary = json.loads(status_file.read_text())
-    debug = (summary, proc.stdout[-1500:], proc.stderr[-1500:])
+    if proc.returncode != 0 or summary.get("n_ok") != n_machines:
+        first = (proc.returncode, proc.stdout[-800:], proc.stderr[-800:],
+                 summary)
+        proc = run_once()
+        summary = json.loads(status_file.read_text())
+        debug = ("RETRIED; first attempt:", first, proc.stdout[-1500:],
+                 proc.stderr[-1500:])
+    else:
+        debug = (summary, proc.stdout[-1500:], proc.stderr[-1500:])
+    assert proc.returncode == 0, debug
     assert summary["n_machines"] == n_machines, debug
     assert summary["n_ok"] == n_machines, debug
     assert summary["world_size"] == world, debug
