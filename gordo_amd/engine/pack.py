@@ -832,15 +832,13 @@ class LSTMPack(BasePack):
 
     def _gather_batch(self, X, Y, idx):
         """idx: [G, B] window start indices. Builds [G,B,T,F] window
-        tensors and [G,B,Fo] targets straight from the series."""
+        tensors (K7 HIP gather kernel on GPU) and [G,B,Fo] targets
+        straight from the series."""
         G, N, F = X.shape
         Fo = Y.shape[2]
         B = idx.shape[1]
         T = self.lookback
-        rows = idx.unsqueeze(-1) + torch.arange(T, device=X.device)  # [G,B,T]
-        Xw = X.gather(
-            1, rows.reshape(G, B * T, 1).expand(G, B * T, F)
-        ).view(G, B, T, F)
+        Xw = ops.window_gather(X, idx, T)
         trow = idx + (T - 1 + self.lookahead)
         Tb = Y.gather(1, trow.unsqueeze(-1).expand(G, B, Fo))
         return Xw, Tb

@@ -98,6 +98,20 @@ def grouped_linear_wgrad_hprev(hs, dZ, T: int):
     return ref.grouped_linear_wgrad(h_prev, dZ)
 
 
+def window_gather(X, idx, T: int):
+    """K7 sliding-window featurizer: gather [G,B,T,F] lookback windows
+    from the resident series [G,N,F] (device analog of
+    create_keras_timeseriesgenerator, reference models.py:713-793)."""
+    if _on_gpu(X):
+        return _require_hip().window_gather(X, idx, int(T))
+    G, N, F = X.shape
+    B = idx.shape[1]
+    rows = idx.unsqueeze(-1).to(torch.long) + torch.arange(T)
+    return X.gather(
+        1, rows.reshape(G, B * T, 1).expand(G, B * T, F)
+    ).view(G, B, T, F)
+
+
 def grouped_gemm_acc(A, B, C):
     if _on_gpu(A):
         return _require_hip().grouped_gemm_acc(A, B, C)

@@ -358,6 +358,33 @@ __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
 // ---------------------------------------------------------------------------
 
 // dZ = (dA + l1*sign(Y)) * act'(Y)
+// ---------------------------------------------------------------------------
+// K7 — sliding-window featurizer: gather lookback
+// windows [G, B, T, F] straight out of the resident series [G, N, F].
+// One workgroup per (g, window); lanes stream the T*F window elements
+// with 16-byte loads when the feature row is 8-aligned (bf16 x 8).
+// The device analog of create_keras_timeseriesgenerator
+// (reference gordo/machine/model/models.py:713-793).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void window_gather_kernel(
+    const bf16* __restrict__ X, const int* __restrict__ idx,
+    bf16* __restrict__ out, int N, int F, int B, int T) {
+  const int g = blockIdx.x / B;
+  const int b = blockIdx.x % B;
+  const int start = idx[(size_t)g * B + b];
+  const bf16* src = X + ((size_t)g * N + start) * F;
+  bf16* dst = out + (((size_t)g * B + b) * T) * F;
+  const size_t n = (size_t)T * F;
+  const int tid = threadIdx.x;
+  if ((F & 7) == 0) {
+    const bf16x8* s8 = reinterpret_cast<const bf16x8*>(src);
+    bf16x8* d8 = reinterpret_cast<bf16x8*>(dst);
+    for (size_t i = tid; i < n / 8; i += 256) d8[i] = s8[i];
+  } else {
+    for (size_t i = tid; i < n; i += 256) dst[i] = src[i];
+  }
+}
+
 __global__ void act_l1_bwd_kernel(const bf16* __restrict__ dA,
                                   const bf16* __restrict__ Y,
                                   bf16* __restrict__ dZ, size_t n, int act,
@@ -627,6 +654,22 @@ torch::Tensor act_l1_bwd(torch::Tensor dA, torch::Tensor Y, int64_t act,
   return dZ;
 }
 
+torch::Tensor window_gather(torch::Tensor X, torch::Tensor idx,
+                            int64_t T) {
+  // X: [G, N, F] (bf16/fp-castable), idx: [G, B] int32 window starts.
+  CHECK_GPU(X);
+  auto Xc = to_bf16c(X);
+  auto ic = idx.to(torch::kInt32).contiguous();
+  int G = Xc.size(0), N = Xc.size(1), F = Xc.size(2);
+  int B = ic.size(1);
+  auto out = torch::empty({G, B, T, F}, Xc.options());
+  hipLaunchKernelGGL(window_gather_kernel, dim3((unsigned)G * B),
+                     dim3(256), 0, cur_stream(),
+                     (const bf16*)Xc.data_ptr(), ic.data_ptr<int>(),
+                     (bf16*)out.data_ptr(), N, F, B, (int)T);
+  return out;
+}
+
 std::vector<torch::Tensor> mse_bwd(torch::Tensor Y, torch::Tensor T) {
   CHECK_GPU(Y);
   auto Yc = to_bf16c(Y);
@@ -766,6 +809,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "dW = X^T@dZ, db = colsum(dZ) per group (MFMA)");
   mod.def("grouped_linear_wgrad_hprev", &grouped_linear_wgrad_hprev,
           "dWh = h_prev^T@dG with in-kernel t-1 shift (no concat)");
+  mod.def("window_gather", &window_gather,
+          "K7 sliding-window featurizer: [G,N,F] -> [G,B,T,F]");
   mod.def("grouped_gemm_acc", &grouped_gemm_acc, "C += A@B per group (MFMA)");
   mod.def("act_l1_bwd", &act_l1_bwd, "fused activation+L1 backward");
   mod.def("mse_bwd", &mse_bwd, "fused per-model MSE loss + grad");

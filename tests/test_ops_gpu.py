@@ -552,3 +552,19 @@ def test_grouped_wgrad_hprev_matches_concat(G, B, T, H, N4):
     )
     torch.testing.assert_close(got_W, want_W, rtol=1e-5, atol=1e-5)
     torch.testing.assert_close(got_b, want_b, rtol=1e-5, atol=1e-5)
+
+
+@pytest.mark.parametrize("F", [50, 7])  # 16B-vector path and scalar path
+def test_window_gather_matches_torch(F):
+    """K7 dedicated featurizer kernel vs the torch gather reference."""
+    require_hip()
+    G, N, B, T = 3, 500, 40, 144
+    X = to_dev_bf16(_rand(G, N, F, seed=70))
+    idx = torch.randint(0, N - T + 1, (G, B), device="cuda",
+                        dtype=torch.int32)
+    got = ops.window_gather(X, idx, T)
+    rows = idx.long().unsqueeze(-1) + torch.arange(T, device="cuda")
+    want = X.gather(
+        1, rows.reshape(G, B * T, 1).expand(G, B * T, F)
+    ).view(G, B, T, F)
+    assert torch.equal(got, want)
