@@ -225,72 +225,70 @@ def run_server(
 
 def _run_prefork(host: str, port: int, workers: int):
     """Multi-process serving without gunicorn: one shared listening
-    socket, N forked werkzeug workers accepting from it, parent
-    supervises and restarts dead workers (the reference's gunicorn
-    master/worker model, server.py:240-304 — worker processes are what
-    scale the GIL-bound request path; each worker holds its own model
-    LRU exactly like a gunicorn worker would).
+    socket, N worker processes accepting from it, parent supervises
+    and restarts dead workers (the reference's gunicorn master/worker
+    model, server.py:240-304 — worker processes are what scale the
+    GIL-bound request path; each worker holds its own model LRU exactly
+    like a gunicorn worker would).
 
-    Fork happens BEFORE any CUDA/HIP initialization in this process —
-    each worker lazily creates its own HIP context at first model load.
+    Workers are EXEC'd (fresh interpreters inheriting the listener fd),
+    not forked: torch poisons CUDA in forked children once the parent
+    has touched the CUDA runtime ("Cannot re-initialize CUDA in forked
+    subprocess" — observed on the round-2 GPU lease), and exec'd
+    workers each own a clean HIP context.
     """
-    import errno
     import signal
     import socket
-
-    from werkzeug.serving import make_server
+    import subprocess
+    import sys
+    import time as _time
 
     sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
     sock.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
     sock.bind((host, port))
     sock.listen(1024)
     sock.set_inheritable(True)
+    fd = sock.fileno()
 
-    def spawn() -> int:
-        pid = os.fork()
-        if pid == 0:  # worker
-            code = 0
-            try:
-                signal.signal(signal.SIGTERM, signal.SIG_DFL)
-                signal.signal(signal.SIGINT, signal.SIG_DFL)
-                app = build_app()
-                srv = make_server(
-                    host, port, app, threaded=True, fd=sock.fileno()
-                )
-                srv.serve_forever()
-            except BaseException:
-                logger.exception("serving worker died")
-                code = 1
-            finally:
-                os._exit(code)
-        return pid
+    def spawn():
+        env = dict(os.environ)
+        env["GORDO_SERVER_FD"] = str(fd)
+        env["GORDO_SERVER_HOST"] = host
+        env["GORDO_SERVER_PORT"] = str(port)
+        return subprocess.Popen(
+            [sys.executable, "-m", "gordo_amd.server.worker"],
+            env=env,
+            pass_fds=(fd,),
+        )
 
-    pids = {spawn() for _ in range(workers)}
+    procs = [spawn() for _ in range(workers)]
     stopping = {"flag": False}
 
     def on_term(signum, frame):
         stopping["flag"] = True
-        for p in list(pids):
-            try:
-                os.kill(p, signal.SIGTERM)
-            except ProcessLookupError:
-                pass
+        for p in procs:
+            if p.poll() is None:
+                p.terminate()
 
     signal.signal(signal.SIGTERM, on_term)
     signal.signal(signal.SIGINT, on_term)
     try:
-        while pids:
-            try:
-                pid, _status = os.wait()
-            except OSError as e:
-                if e.errno == errno.EINTR:
-                    continue
-                raise
-            except ChildProcessError:
-                break
-            pids.discard(pid)
-            if not stopping["flag"]:
-                logger.warning("serving worker %d exited; restarting", pid)
-                pids.add(spawn())
+        while True:
+            if stopping["flag"]:
+                for p in procs:
+                    try:
+                        p.wait(timeout=20)
+                    except subprocess.TimeoutExpired:
+                        p.kill()
+                return
+            for i, p in enumerate(procs):
+                rc = p.poll()
+                if rc is not None and not stopping["flag"]:
+                    logger.warning(
+                        "serving worker %d exited rc=%s; restarting",
+                        p.pid, rc,
+                    )
+                    procs[i] = spawn()
+            _time.sleep(0.5)
     finally:
         sock.close()
