@@ -1146,8 +1146,18 @@ namespace gordo_lstm {
 inline int pad_ldg(int h4) { return ((h4 + 31) & ~31) + 8; }
 
 inline int pick_rows(int G, int B) {
-  // 32-row tiles when 64-row tiles would underfill the 256 CUs
-  return (G * ((B + 63) / 64) >= 512) ? 64 : 32;
+  // smallest row tile whose grid reaches ~2 workgroups per CU: the
+  // scans are latency-bound chains, so co-resident workgroups on one
+  // CU overlap each other's stalls (WhT is staged once per launch, so
+  // smaller tiles do not re-read Wh). GORDO_LSTM_ROWS forces a tile
+  // for A/B measurement.
+  if (const char* e = getenv("GORDO_LSTM_ROWS")) {
+    int r = atoi(e);
+    if (r == 16 || r == 32 || r == 64) return r;
+  }
+  if (G * ((B + 63) / 64) >= 512) return 64;
+  if (G * ((B + 31) / 32) >= 512) return 32;
+  return 16;
 }
 
 // ---- big-H path (64 < H <= 256, H % 8 == 0) ----
@@ -1289,8 +1299,14 @@ std::vector<torch::Tensor> lstm_seq_fwd(torch::Tensor xW, torch::Tensor Wh) {
                        (const bf16*)Whc.data_ptr(), (bf16*)hs.data_ptr(),
                        cs.data_ptr<float>(), (bf16*)gacts.data_ptr(), B, T, H,
                        ldg);
-  else
+  else if (rows == 32)
     hipLaunchKernelGGL(lstm_seq_fwd_kernel<32>, dim3(blocks), dim3(256), lds,
+                       stream, (const bf16*)xc.data_ptr(),
+                       (const bf16*)Whc.data_ptr(), (bf16*)hs.data_ptr(),
+                       cs.data_ptr<float>(), (bf16*)gacts.data_ptr(), B, T, H,
+                       ldg);
+  else
+    hipLaunchKernelGGL(lstm_seq_fwd_kernel<16>, dim3(blocks), dim3(256), lds,
                        stream, (const bf16*)xc.data_ptr(),
                        (const bf16*)Whc.data_ptr(), (bf16*)hs.data_ptr(),
                        cs.data_ptr<float>(), (bf16*)gacts.data_ptr(), B, T, H,
@@ -1326,8 +1342,14 @@ std::vector<torch::Tensor> lstm_seq_fwd_v3(torch::Tensor xW,
                        (const bf16*)Whc.data_ptr(), (bf16*)hs.data_ptr(),
                        cs.data_ptr<float>(), (bf16*)gacts.data_ptr(), B, T, H,
                        ldg);
-  else
+  else if (rows == 32)
     hipLaunchKernelGGL(lstm_seq_fwd_v3_kernel<32>, dim3(blocks), dim3(256),
+                       lds, stream, (const bf16*)xc.data_ptr(),
+                       (const bf16*)Whc.data_ptr(), (bf16*)hs.data_ptr(),
+                       cs.data_ptr<float>(), (bf16*)gacts.data_ptr(), B, T, H,
+                       ldg);
+  else
+    hipLaunchKernelGGL(lstm_seq_fwd_v3_kernel<16>, dim3(blocks), dim3(256),
                        lds, stream, (const bf16*)xc.data_ptr(),
                        (const bf16*)Whc.data_ptr(), (bf16*)hs.data_ptr(),
                        cs.data_ptr<float>(), (bf16*)gacts.data_ptr(), B, T, H,
@@ -1366,8 +1388,14 @@ torch::Tensor lstm_seq_bwd_v3(torch::Tensor dSeq, torch::Tensor gacts,
                        (const bf16*)gc.data_ptr(), cc.data_ptr<float>(),
                        (const bf16*)Whc.data_ptr(), (bf16*)dG.data_ptr(), B,
                        T, H, ldg, last_only ? 1 : 0);
-  else
+  else if (rows == 32)
     hipLaunchKernelGGL(lstm_seq_bwd_v3_kernel<32>, dim3(blocks), dim3(256),
+                       lds, stream, (const bf16*)dc.data_ptr(),
+                       (const bf16*)gc.data_ptr(), cc.data_ptr<float>(),
+                       (const bf16*)Whc.data_ptr(), (bf16*)dG.data_ptr(), B,
+                       T, H, ldg, last_only ? 1 : 0);
+  else
+    hipLaunchKernelGGL(lstm_seq_bwd_v3_kernel<16>, dim3(blocks), dim3(256),
                        lds, stream, (const bf16*)dc.data_ptr(),
                        (const bf16*)gc.data_ptr(), cc.data_ptr<float>(),
                        (const bf16*)Whc.data_ptr(), (bf16*)dG.data_ptr(), B,
@@ -1427,8 +1455,14 @@ torch::Tensor lstm_seq_bwd(torch::Tensor dSeq, torch::Tensor gacts,
                        (const bf16*)gc.data_ptr(), cc.data_ptr<float>(),
                        (const bf16*)Whc.data_ptr(), (bf16*)dG.data_ptr(), B,
                        T, H, ldg, last_only ? 1 : 0);
-  else
+  else if (rows == 32)
     hipLaunchKernelGGL(lstm_seq_bwd_kernel<32>, dim3(blocks), dim3(256), lds,
+                       stream, (const bf16*)dc.data_ptr(),
+                       (const bf16*)gc.data_ptr(), cc.data_ptr<float>(),
+                       (const bf16*)Whc.data_ptr(), (bf16*)dG.data_ptr(), B,
+                       T, H, ldg, last_only ? 1 : 0);
+  else
+    hipLaunchKernelGGL(lstm_seq_bwd_kernel<16>, dim3(blocks), dim3(256), lds,
                        stream, (const bf16*)dc.data_ptr(),
                        (const bf16*)gc.data_ptr(), cc.data_ptr<float>(),
                        (const bf16*)Whc.data_ptr(), (bf16*)dG.data_ptr(), B,

@@ -568,3 +568,23 @@ def test_window_gather_matches_torch(F):
         1, rows.reshape(G, B * T, 1).expand(G, B * T, F)
     ).view(G, B, T, F)
     assert torch.equal(got, want)
+
+
+def test_lstm_scan_row_tiles_bit_equal(monkeypatch):
+    """Row-tile choice (16/32/64) is a pure scheduling knob: per-row
+    math is identical, so outputs are bit-equal across tiles."""
+    require_hip()
+    G, B, T, H = 2, 96, 20, 42
+    xW = to_dev_bf16(_rand(G, B, T, 4 * H, seed=80))
+    Wh = to_dev_bf16(_rand(G, H, 4 * H, seed=81) * 0.3)
+    outs = {}
+    for rows in ("16", "32", "64"):
+        monkeypatch.setenv("GORDO_LSTM_ROWS", rows)
+        hs, cs, ga = ops.lstm_seq_fwd(xW, Wh)
+        dSeq = to_dev_bf16(_rand(G, B, H, seed=82))
+        dG = ops.lstm_seq_bwd(dSeq, ga, cs, Wh, True)
+        outs[rows] = (hs, cs, ga, dG)
+    monkeypatch.delenv("GORDO_LSTM_ROWS")
+    for rows in ("32", "64"):
+        for a, b in zip(outs["16"], outs[rows]):
+            assert torch.equal(a, b), rows
