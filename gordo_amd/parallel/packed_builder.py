@@ -197,14 +197,13 @@ class PackedFleetBuilder:
         # join outstanding async saves; failures surface per machine
         t0 = time.time()
         if self._save_futures:
-            for p, f in self._save_futures:
+            for plans_of_group, f in self._save_futures:
                 try:
                     f.result()
                 except Exception as e:
-                    logger.exception(
-                        "async model save failed: %s", p.machine.name
-                    )
-                    p.error = e
+                    logger.exception("async adopt/save failed")
+                    for p in plans_of_group:
+                        p.error = e
             self._save_futures.clear()
         self._phase("save_join", time.time() - t0)
         self._phase("total", time.time() - t_all0)
@@ -442,50 +441,55 @@ class PackedFleetBuilder:
         train_duration = fit_state["duration"]
         self._phase("final_fit_gpu(overlapped)", train_duration)
 
-        # per-machine adoption + detector finalization
-        t_seg = time.time()
+        # per-machine adoption + detector finalization + save — ASYNC:
+        # one D2H of the flat parameter buffer then pure CPU work
+        # (numpy slicing, metadata assembly, pickle+json dumps), all
+        # independent of the NEXT group's GPU fits; runs on the
+        # builder-wide pool and is joined at the end of build_all.
         offset = len(Xt_list[0]) - pack._n_samples(len(Xt_list[0]))
-        all_states = pack.states_for_all_models()
-        for g_idx, p in enumerate(group):
-            hist = {
-                k: [float(ep[g_idx]) for ep in v] for k, v in history.items()
-            }
-            p.keras_est.adopt_pack_result(
-                spec,
-                all_states[g_idx],
-                hist,
-                n_features=Xt_list[g_idx].shape[1],
-                n_features_out=y_list[g_idx].shape[1],
-            )
-            if p.detector is not None:
-                p.detector.scaler.fit(p.y)
-            self._finalize(p, offset, cv_duration, final=True,
-                           train_duration=train_duration / len(group))
+        n_feat = [x.shape[1] for x in Xt_list]
+        n_feat_out = [y.shape[1] for y in y_list]
 
-        self._phase("adopt", time.time() - t_seg)
+        def adopt_and_save(group=group, pack=pack, history=history,
+                           train_duration=train_duration,
+                           cv_duration=cv_duration, spec=spec):
+            t0 = time.time()
+            all_states = pack.states_for_all_models()
+            for g_idx, p in enumerate(group):
+                hist = {
+                    k: [float(ep[g_idx]) for ep in v]
+                    for k, v in history.items()
+                }
+                p.keras_est.adopt_pack_result(
+                    spec,
+                    all_states[g_idx],
+                    hist,
+                    n_features=n_feat[g_idx],
+                    n_features_out=n_feat_out[g_idx],
+                )
+                if p.detector is not None:
+                    p.detector.scaler.fit(p.y)
+                self._finalize(p, offset, cv_duration, final=True,
+                               train_duration=train_duration / len(group))
+                if self.save_models and self.output_dir:
+                    out = os.path.join(self.output_dir, p.machine.name)
+                    ModelBuilder._save_model(p.model, p.machine, out)
+                    if self.model_register_dir:
+                        disk_registry.write_key(
+                            self.model_register_dir,
+                            ModelBuilder(p.machine).cache_key,
+                            out,
+                        )
+            pack.release_graphs()
+            self._phase("adopt_save(overlapped)", time.time() - t0)
 
-        # save + register — ASYNC: pickle+json dumps are I/O/CPU bound
-        # and independent of the next group's GPU work, so they run on
-        # the builder-wide save pool and are joined at the end of
-        # build_all (phase ledger records the non-overlapped tail)
-        if self.save_models and self.output_dir:
-            def save_one(p):
-                out = os.path.join(self.output_dir, p.machine.name)
-                ModelBuilder._save_model(p.model, p.machine, out)
-                if self.model_register_dir:
-                    disk_registry.write_key(
-                        self.model_register_dir,
-                        ModelBuilder(p.machine).cache_key,
-                        out,
-                    )
-
-            pool = self._save_pool
-            self._save_futures.extend(
-                (p, pool.submit(save_one, p)) for p in group
-            )
-        pack.release_graphs()
+        # one future per GROUP: plans are only touched by this closure
+        # until the join, so no cross-thread aliasing
+        self._save_futures.append(
+            (list(group), self._save_pool.submit(adopt_and_save))
+        )
         logger.info(
-            "Packed build of %d machines done in %.2fs",
+            "Packed build of %d machines dispatched in %.2fs",
             len(group), time.time() - t0_all,
         )
 
