@@ -77,7 +77,15 @@ setup(
         "gordo_amd.server": ["csrc/*"],
     },
     python_requires=">=3.10",
-    entry_points={"console_scripts": ["gordo=gordo_amd.cli:gordo"]},
+    entry_points={
+        "console_scripts": [
+            "gordo=gordo_amd.cli:gordo",
+            # the Argo client pods invoke `gordo-client ... predict`
+            # (reference template :1375; gordo-client ships it as its
+            # own package) — alias onto the in-repo client CLI
+            "gordo-client=gordo_amd.cli.client:client_cli",
+        ]
+    },
     ext_modules=ext_modules,
     cmdclass=cmdclass,
 )
