@@ -751,22 +751,64 @@ class PackedFleetBuilder:
             for g_idx in kfcv_pred
         }
 
+        # batched scoring-scaler transform (K13 vectorization across
+        # the pack): all MinMaxScaler-style scalers apply as one affine
+        # op on the stacked [G, n, F] arrays; any other scaler type
+        # keeps the per-machine path.
+        batch_affine = all(
+            sc is not None and hasattr(sc, "scale_") and hasattr(sc, "min_")
+            for sc in scoring_scalers
+        ) and len(group) > 1
+        if batch_affine:
+            sc_scale = np.stack(
+                [np.asarray(sc.scale_) for sc in scoring_scalers]
+            )[:, None, :]
+            sc_min = np.stack(
+                [np.asarray(sc.min_) for sc in scoring_scalers]
+            )[:, None, :]
+
         for fold_i, (train_idx, test_idx) in enumerate(folds):
             t_f0 = time.time()
             preds, t_fit, t_pred, dev_thr = fold_preds[fold_i]
             t_s0 = time.time()
+
+            yt_b = yp_b = None
+            if batch_affine:
+                n_out = preds.shape[1]
+                y_true_b = np.stack(
+                    [y[test_idx][-n_out:] for y in y_list]
+                )
+                yt_b = y_true_b * sc_scale + sc_min
+                yp_b = preds * sc_scale + sc_min
+                metric_cache = {
+                    metric.__name__: _metric_all_tags_batched(
+                        metric, yt_b, yp_b
+                    )
+                    for metric in metrics_list
+                }
 
             for g_idx, p in enumerate(group):
                 y_true_full = y_list[g_idx][test_idx]
                 y_pred = preds[g_idx]
                 y_true = y_true_full[-len(y_pred):]
                 sc = scoring_scalers[g_idx]
-                yt = sc.transform(y_true) if sc is not None else y_true
-                yp = sc.transform(y_pred) if sc is not None else y_pred
+                if batch_affine:
+                    yt, yp = yt_b[g_idx], yp_b[g_idx]
+                else:
+                    yt = sc.transform(y_true) if sc is not None else y_true
+                    yp = sc.transform(y_pred) if sc is not None else y_pred
                 tags = [t.name for t in p.machine.dataset.target_tag_list]
                 for metric in metrics_list:
                     mname = metric.__name__.replace("_", "-")
-                    per_tag, agg = _metric_all_tags(metric, yt, yp)
+                    if batch_affine:
+                        cached = metric_cache.get(metric.__name__)
+                        per_tag, agg = (
+                            (cached[0][g_idx], cached[1][g_idx])
+                            if cached is not None
+                            else _metric_all_tags(metric, yt, yp)
+                        )
+                    else:
+                        per_tag, agg = _metric_all_tags(metric, yt, yp)
                     for col, tag in enumerate(tags):
                         key = f'{mname}-{tag.replace(" ", "-")}'
                         per_machine_scores[g_idx].setdefault(key, []).append(
@@ -929,6 +971,41 @@ class PackedFleetBuilder:
                 dataset_meta=p.dataset_meta,
             ),
         )
+
+
+def _metric_all_tags_batched(metric, yt: np.ndarray, yp: np.ndarray):
+    """Batched [G, n, F] version of _metric_all_tags for the four
+    default metrics; returns (per_tag [G, F], agg [G]) or None for
+    unknown metrics (callers fall back per machine)."""
+    name = getattr(metric, "__name__", "")
+    diff = yp - yt
+    if name == "mean_squared_error":
+        per = (diff ** 2).mean(axis=1)
+        return per, per.mean(axis=1)
+    if name == "mean_absolute_error":
+        per = np.abs(diff).mean(axis=1)
+        return per, per.mean(axis=1)
+    if name == "r2_score":
+        ss_res = (diff ** 2).sum(axis=1)
+        ss_tot = ((yt - yt.mean(axis=1, keepdims=True)) ** 2).sum(axis=1)
+        with np.errstate(divide="ignore", invalid="ignore"):
+            per = np.where(
+                ss_tot > 0,
+                1.0 - ss_res / ss_tot,
+                np.where(ss_res == 0, 1.0, 0.0),
+            )
+        return per, per.mean(axis=1)
+    if name == "explained_variance_score":
+        var_res = diff.var(axis=1)
+        var_y = yt.var(axis=1)
+        with np.errstate(divide="ignore", invalid="ignore"):
+            per = np.where(
+                var_y > 0,
+                1.0 - var_res / var_y,
+                np.where(var_res == 0, 1.0, 0.0),
+            )
+        return per, per.mean(axis=1)
+    return None
 
 
 def _metric_all_tags(metric, yt: np.ndarray, yp: np.ndarray):
