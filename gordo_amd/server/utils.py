@@ -20,7 +20,7 @@ import timeit
 import zlib
 from datetime import datetime
 from functools import lru_cache
-from typing import List, Union
+from typing import List, Union, Optional
 
 import dateutil.parser
 import numpy as np  # noqa — doctest namespace (dataframe_to_dict)
@@ -103,6 +103,9 @@ def dataframe_from_dict(data: dict) -> pd.DataFrame:
     >>> dataframe_from_dict(serialized).shape
     (2, 2)
     """
+    fast = _dataframe_from_dict_fast(data)
+    if fast is not None:
+        return fast
     if isinstance(data, dict) and any(
         isinstance(val, dict) for val in data.values()
     ):
@@ -124,6 +127,52 @@ def dataframe_from_dict(data: dict) -> pd.DataFrame:
         df.index = df.index.map(int)
     df.sort_index(inplace=True)
     return df
+
+
+def _dataframe_from_dict_fast(data) -> Optional[pd.DataFrame]:
+    """Serving-path fast lane for the common request shape
+    {col: {index_key: float}} with every column sharing one key set
+    (what dataframe_to_dict emits): one numpy fill + one vectorized
+    index parse instead of pandas dict assembly + per-key
+    dateutil.isoparse. Returns None for ANY other shape — the exact
+    reference-semantics path below handles those."""
+    try:
+        if not isinstance(data, dict) or not data:
+            return None
+        cols = list(data)
+        first = data[cols[0]]
+        if not isinstance(first, dict) or not first:
+            return None
+        idx_keys = list(first)
+        n = len(idx_keys)
+        if not isinstance(idx_keys[0], str):
+            return None
+        col_arrays = {}
+        for c in cols:
+            d = data[c]
+            if not isinstance(d, dict) or len(d) != n:
+                return None
+            if list(d) != idx_keys:
+                return None
+            arr = np.asarray(list(d.values()))
+            # per-column dtype like DataFrame.from_dict (int columns
+            # stay int64 — the values echo back into the response)
+            if arr.dtype.kind not in "if":
+                return None
+            col_arrays[c] = arr
+        try:
+            index = pd.to_datetime(idx_keys, format="ISO8601", utc=False)
+        except (ValueError, TypeError):
+            try:
+                index = pd.Index([int(k) for k in idx_keys])
+            except (ValueError, TypeError):
+                return None
+        df = pd.DataFrame(col_arrays, index=index)
+        if not df.index.is_monotonic_increasing:
+            df.sort_index(inplace=True)
+        return df
+    except Exception:
+        return None
 
 
 def parse_iso_datetime(datetime_str: str) -> datetime:

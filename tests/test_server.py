@@ -762,3 +762,49 @@ def test_run_server_prefork_workers(tmp_path):
         except subprocess.TimeoutExpired:
             proc.kill()
             proc.wait(timeout=10)
+
+
+def test_dataframe_from_dict_fast_lane_equivalence():
+    """The serving-path fast decoder produces exactly what the
+    reference-semantics path produces for every shape it accepts, and
+    falls back for everything else."""
+    import numpy as np
+    import pandas as pd
+
+    from gordo_amd.server import utils as su
+
+    def legacy(data):
+        # force the reference path by bypassing the fast lane
+        import unittest.mock as mock
+
+        with mock.patch.object(su, "_dataframe_from_dict_fast",
+                               lambda d: None):
+            return su.dataframe_from_dict(data)
+
+    idx = pd.date_range("2020-01-01", periods=5, freq="10min", tz="UTC")
+    frame = pd.DataFrame(
+        np.random.random((5, 3)), index=idx, columns=["a", "b", "c"]
+    )
+    payloads = [
+        su.dataframe_to_dict(frame),                       # tz-aware
+        su.dataframe_to_dict(frame.tz_localize(None)),     # naive
+        {"x": {"0": 1, "1": 2}, "y": {"0": 3, "1": 4}},    # int index+vals
+        {"x": {"1": 0.5, "0": 1.5}},                       # unsorted
+        # mixed dtypes per column (ints stay int64)
+        {"i": {"2020-01-01": 1, "2020-01-02": 2},
+         "f": {"2020-01-01": 0.5, "2020-01-02": 1.5}},
+    ]
+    for payload in payloads:
+        fast = su.dataframe_from_dict(payload)
+        ref = legacy(payload)
+        pd.testing.assert_frame_equal(fast, ref)
+
+    # shapes the fast lane must refuse (falls back, still correct)
+    odd = {"x": {"0": 1.0}, "y": {"1": 2.0}}  # mismatched keys
+    assert su._dataframe_from_dict_fast(odd) is None
+    assert su.dataframe_from_dict(odd) is not None
+    nested = {
+        "top": {"sub": {"2019-01-01": 0.0, "2019-01-02": 1.0}}
+    }  # 2-level (anomaly-style) dict
+    assert su._dataframe_from_dict_fast(nested) is None
+    assert su.dataframe_from_dict(nested).shape == (2, 1)
