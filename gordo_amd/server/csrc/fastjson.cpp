@@ -206,6 +206,152 @@ py::bytes encode_frame(py::sequence index, py::sequence tops,
     return py::bytes(out);
 }
 
+// ---------------------------------------------------------------------
+// Request decoder: the STRICT fast lane for the serving POST payload
+// {"X": {col: {key: number}}, "y": {...}} that dataframe_to_dict
+// emits. Any deviation (escapes in strings, nested deeper, extra
+// top-level keys, non-number values other than null, column key-set
+// mismatch) returns None and the caller takes the stdlib-json path —
+// correctness never depends on this parser accepting a payload.
+struct Cursor {
+    const char* p;
+    const char* end;
+    bool fail = false;
+    void ws() { while (p < end && (*p == ' ' || *p == '\t' || *p == '\n' || *p == '\r')) p++; }
+    bool lit(char c) { ws(); if (p < end && *p == c) { p++; return true; } fail = true; return false; }
+    bool peek(char c) { ws(); return p < end && *p == c; }
+};
+
+// parse a JSON string WITHOUT escapes; empty-view + fail on escapes
+static bool parse_plain_string(Cursor& c, std::string& out) {
+    if (!c.lit('"')) return false;
+    const char* s = c.p;
+    while (c.p < c.end && *c.p != '"') {
+        if (*c.p == '\\') { c.fail = true; return false; }
+        c.p++;
+    }
+    if (c.p >= c.end) { c.fail = true; return false; }
+    out.assign(s, c.p - s);
+    c.p++;  // closing quote
+    return true;
+}
+
+struct NumTok { double v; bool is_int; bool is_nan; };
+
+static bool parse_number(Cursor& c, NumTok& t) {
+    c.ws();
+    if (c.p + 4 <= c.end && std::strncmp(c.p, "null", 4) == 0) {
+        c.p += 4; t.v = std::nan(""); t.is_int = false; t.is_nan = true;
+        return true;
+    }
+    if (c.p + 3 <= c.end && std::strncmp(c.p, "NaN", 3) == 0) {
+        c.p += 3; t.v = std::nan(""); t.is_int = false; t.is_nan = true;
+        return true;
+    }
+    const char* s = c.p;
+    char* endp = nullptr;
+    t.v = std::strtod(s, &endp);
+    if (endp == s || endp > c.end) { c.fail = true; return false; }
+    t.is_int = true; t.is_nan = false;
+    for (const char* q = s; q < endp; ++q)
+        if (*q == '.' || *q == 'e' || *q == 'E') { t.is_int = false; break; }
+    c.p = endp;
+    return true;
+}
+
+// one frame: {col: {key: number, ...}, ...} -> (cols, keys, object of
+// per-column numpy arrays). Returns false -> fallback.
+static bool parse_frame(Cursor& c, py::list& cols, py::list& keys,
+                        py::list& arrays) {
+    if (!c.lit('{')) return false;
+    std::vector<std::string> idx_keys;
+    bool first_col = true;
+    if (c.peek('}')) { c.fail = true; return false; }  // empty: fallback
+    while (true) {
+        std::string col;
+        if (!parse_plain_string(c, col)) return false;
+        if (!c.lit(':')) return false;
+        if (!c.lit('{')) return false;
+        std::vector<double> vals;
+        bool all_int = true;
+        size_t ki = 0;
+        if (c.peek('}')) { c.fail = true; return false; }
+        while (true) {
+            std::string key;
+            if (!parse_plain_string(c, key)) return false;
+            if (first_col) {
+                idx_keys.push_back(key);
+            } else {
+                if (ki >= idx_keys.size() || idx_keys[ki] != key) {
+                    c.fail = true; return false;
+                }
+            }
+            ki++;
+            if (!c.lit(':')) return false;
+            NumTok t;
+            if (!parse_number(c, t)) return false;
+            if (!t.is_int) all_int = false;
+            vals.push_back(t.v);
+            if (c.peek(',')) { c.lit(','); continue; }
+            break;
+        }
+        if (!c.lit('}')) return false;
+        if (!first_col && ki != idx_keys.size()) { c.fail = true; return false; }
+        if (first_col) {
+            for (auto& k : idx_keys) keys.append(py::str(k));
+        }
+        cols.append(py::str(col));
+        const py::ssize_t n = (py::ssize_t)vals.size();
+        if (all_int) {
+            py::array_t<long long> a(n);
+            auto w = a.mutable_unchecked<1>();
+            for (py::ssize_t i = 0; i < n; i++)
+                w(i) = (long long)vals[(size_t)i];
+            arrays.append(a);
+        } else {
+            py::array_t<double> a(n);
+            std::memcpy(a.mutable_data(), vals.data(), n * sizeof(double));
+            arrays.append(a);
+        }
+        first_col = false;
+        if (c.peek(',')) { c.lit(','); continue; }
+        break;
+    }
+    if (!c.lit('}')) return false;
+    return true;
+}
+
+py::object decode_request(py::bytes payload) {
+    std::string buf = payload;  // copy; payloads are ~100s of KB
+    Cursor c{buf.data(), buf.data() + buf.size()};
+    py::dict out;
+    if (!c.lit('{')) return py::none();
+    if (c.peek('}')) return py::none();
+    while (true) {
+        std::string key;
+        if (!parse_plain_string(c, key)) return py::none();
+        if (key != "X" && key != "y") return py::none();  // unknown: fallback
+        if (!c.lit(':')) return py::none();
+        c.ws();
+        if (key == "y" && c.p + 4 <= c.end &&
+            std::strncmp(c.p, "null", 4) == 0) {
+            c.p += 4;
+            out[py::str(key)] = py::none();
+        } else {
+            py::list cols, keys, arrays;
+            if (!parse_frame(c, cols, keys, arrays)) return py::none();
+            out[py::str(key)] = py::make_tuple(cols, keys, arrays);
+        }
+        if (c.peek(',')) { c.lit(','); continue; }
+        break;
+    }
+    if (!c.lit('}')) return py::none();
+    c.ws();
+    if (c.p != c.end) return py::none();
+    if (!out.contains("X")) return py::none();
+    return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(_gordo_fastjson, mod) {
@@ -213,4 +359,8 @@ PYBIND11_MODULE(_gordo_fastjson, mod) {
     mod.def("encode_frame", &encode_frame, py::arg("index"),
             py::arg("tops"), py::arg("subs"), py::arg("values"),
             "Encode a 2-level-column response frame to JSON bytes");
+    mod.def("decode_request", &decode_request, py::arg("payload"),
+            "Strict fast-lane decode of the {X: {col: {key: num}}} "
+            "request payload; None on any deviation (caller falls "
+            "back to stdlib json)");
 }

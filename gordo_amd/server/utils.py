@@ -129,6 +129,53 @@ def dataframe_from_dict(data: dict) -> pd.DataFrame:
     return df
 
 
+def _decode_request_fast(payload: bytes):
+    """C++ strict-lane decode of the JSON POST body (fastjson
+    decode_request): one pass over the bytes straight into per-column
+    numpy arrays, skipping json.loads + python dict assembly. Returns
+    {"X": DataFrame, "y": DataFrame|None} or None for any payload the
+    strict parser refuses (the stdlib path handles those)."""
+    if _gordo_fastjson is None or not hasattr(
+        _gordo_fastjson, "decode_request"
+    ):
+        return None
+    try:
+        raw = _gordo_fastjson.decode_request(payload)
+    except Exception:
+        return None
+    if raw is None:
+        return None
+    out = {}
+    for key, triple in raw.items():
+        if triple is None:
+            out[key] = None
+            continue
+        cols, keys, arrays = triple
+        df = _frame_from_columns(dict(zip(cols, arrays)), keys)
+        if df is None:
+            return None
+        out[key] = df
+    if "X" not in out:
+        return None
+    return out
+
+
+def _frame_from_columns(col_arrays, idx_keys):
+    """Shared tail of the fast decode lanes: vectorized index parse +
+    frame assembly with per-column dtypes; None -> caller falls back."""
+    try:
+        index = pd.to_datetime(idx_keys, format="ISO8601", utc=False)
+    except (ValueError, TypeError):
+        try:
+            index = pd.Index([int(k) for k in idx_keys])
+        except (ValueError, TypeError):
+            return None
+    df = pd.DataFrame(col_arrays, index=index)
+    if not df.index.is_monotonic_increasing:
+        df.sort_index(inplace=True)
+    return df
+
+
 def _dataframe_from_dict_fast(data) -> Optional[pd.DataFrame]:
     """Serving-path fast lane for the common request shape
     {col: {index_key: float}} with every column sharing one key set
@@ -160,17 +207,7 @@ def _dataframe_from_dict_fast(data) -> Optional[pd.DataFrame]:
             if arr.dtype.kind not in "if":
                 return None
             col_arrays[c] = arr
-        try:
-            index = pd.to_datetime(idx_keys, format="ISO8601", utc=False)
-        except (ValueError, TypeError):
-            try:
-                index = pd.Index([int(k) for k in idx_keys])
-            except (ValueError, TypeError):
-                return None
-        df = pd.DataFrame(col_arrays, index=index)
-        if not df.index.is_monotonic_increasing:
-            df.sort_index(inplace=True)
-        return df
+        return _frame_from_columns(col_arrays, idx_keys)
     except Exception:
         return None
 
@@ -231,14 +268,19 @@ def extract_X_y(method):
                 f"Cannot extract X and y from '{request.method}' request."
             )
         if request.is_json:
-            if "X" not in (request.json or {}):
-                return make_response(
-                    (jsonify(message='Cannot predict without "X"'), 400)
-                )
-            X = dataframe_from_dict(request.json["X"])
-            y = request.json.get("y")
-            if y is not None:
-                y = dataframe_from_dict(y)
+            decoded = _decode_request_fast(request.get_data(cache=True))
+            if decoded is not None:
+                X = decoded["X"]
+                y = decoded.get("y")
+            else:
+                if "X" not in (request.json or {}):
+                    return make_response(
+                        (jsonify(message='Cannot predict without "X"'), 400)
+                    )
+                X = dataframe_from_dict(request.json["X"])
+                y = request.json.get("y")
+                if y is not None:
+                    y = dataframe_from_dict(y)
         else:
             if "X" not in request.files:
                 return make_response(

@@ -808,3 +808,56 @@ def test_dataframe_from_dict_fast_lane_equivalence():
     }  # 2-level (anomaly-style) dict
     assert su._dataframe_from_dict_fast(nested) is None
     assert su.dataframe_from_dict(nested).shape == (2, 1)
+
+
+def test_decode_request_fast_equivalence():
+    """The C++ strict-lane request decoder matches json.loads +
+    dataframe_from_dict for everything it accepts, and refuses
+    anything irregular."""
+    import json as _json
+
+    import numpy as np
+    import pandas as pd
+    import pytest as _pytest
+
+    from gordo_amd.server import utils as su
+
+    fj = su._gordo_fastjson
+    if fj is None or not hasattr(fj, "decode_request"):
+        _pytest.skip("fastjson not built")
+
+    idx = pd.date_range("2020-01-01", periods=7, freq="10min", tz="UTC")
+    X = pd.DataFrame(np.random.random((7, 4)),
+                     index=idx, columns=["t 1", "t2", "t3", "t4"])
+    X.iloc[0, 1] = np.nan
+    for payload in (
+        {"X": su.dataframe_to_dict(X)},
+        {"X": su.dataframe_to_dict(X), "y": su.dataframe_to_dict(X)},
+        {"X": su.dataframe_to_dict(X), "y": None},
+        {"X": {"a": {"0": 1, "1": 2}}},  # int index + int values
+    ):
+        raw = _json.dumps(payload).encode()
+        fast = su._decode_request_fast(raw)
+        assert fast is not None, payload.keys()
+        want_X = su.dataframe_from_dict(payload["X"])
+        pd.testing.assert_frame_equal(fast["X"], want_X)
+        if payload.get("y") is not None:
+            pd.testing.assert_frame_equal(
+                fast["y"], su.dataframe_from_dict(payload["y"])
+            )
+    # NaN round trip: stdlib emits NaN token, decoder must accept it
+    raw = _json.dumps({"X": su.dataframe_to_dict(X)}).encode()
+    assert b"NaN" in raw
+    fast = su._decode_request_fast(raw)
+    assert np.isnan(fast["X"].iloc[0, 1])
+
+    # refused shapes -> None (stdlib path takes over)
+    for bad in (
+        b'{"X": {"a": {"k\\u0041": 1}}}',          # escaped key
+        b'{"X": {"a": {"0": 1}}, "extra": 2}',     # unknown top key
+        b'{"X": {"a": {"0": 1}, "b": {"1": 1}}}',  # key-set mismatch
+        b'{"X": {"a": {"0": "s"}}}',               # non-number value
+        b'{"X": []}',                              # wrong shape
+        b'not json',
+    ):
+        assert su._decode_request_fast(bad) is None
