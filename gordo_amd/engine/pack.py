@@ -36,6 +36,26 @@ def _compute_dtype(device: torch.device) -> torch.dtype:
     return torch.bfloat16 if device.type == "cuda" else torch.float32
 
 
+def _pad8(n: int) -> int:
+    return (int(n) + 7) & ~7
+
+
+def pad_enabled(device) -> bool:
+    """Feature/hidden dims are zero-padded to multiples of 8 on GPU
+    (GORDO_PAD8=0 disables): bf16x8 = 16 bytes unlocks vectorized
+    global staging loads in every grouped GEMM (guide G13 — scalar
+    bf16 staging is a 2-2.5x tax), and padded hidden units are
+    PROVABLY inert: zero weights + zero bias give g=tanh(0)=0 so
+    c=f*0+i*0=0 and h=o*tanh(0)=0 forever; all their gradients are
+    exactly 0 (dc=dh=0 chains), so Adam keeps their weights at 0.
+    Only the MSE mean needs the REAL element count (ops.mse_bwd
+    real_n). CPU packs stay unpadded (they are the fp32 oracle)."""
+    return (
+        torch.device(device).type == "cuda"
+        and os.environ.get("GORDO_PAD8", "1") != "0"
+    )
+
+
 def _glorot_uniform(shape, gen):
     fan_in, fan_out = shape[-2], shape[-1]
     limit = math.sqrt(6.0 / (fan_in + fan_out))
@@ -60,6 +80,7 @@ class _ParamStore:
         self.device = device
         self.compute_dtype = _compute_dtype(device)
         self._shapes: List[Tuple[str, Tuple[int, ...]]] = []
+        self.logical: Dict[str, Tuple[int, ...]] = {}
         self._offsets: Dict[str, Tuple[int, int]] = {}
         self._total = 0
         self.views: Dict[str, torch.Tensor] = {}
@@ -67,10 +88,12 @@ class _ParamStore:
         self.gviews: Dict[str, torch.Tensor] = {}
         self.step_count = 0
 
-    def declare(self, name: str, shape: Tuple[int, ...]):
+    def declare(self, name: str, shape: Tuple[int, ...],
+                logical: Optional[Tuple[int, ...]] = None):
         n = int(np.prod(shape))
         self._offsets[name] = (self._total, n)
         self._shapes.append((name, shape))
+        self.logical[name] = tuple(logical) if logical else tuple(shape)
         self._total += n
 
     def allocate(self):
@@ -133,6 +156,11 @@ class BasePack:
             ("cuda" if torch.cuda.is_available() else "cpu")
         )
         self.store = _ParamStore(self.device)
+        self.pad8 = pad_enabled(self.device)
+        self._pad = _pad8 if self.pad8 else (lambda n: int(n))
+        # per-param custom logical<->physical converters (LSTM gate
+        # interleave); default is plain dim slicing by store.logical
+        self._extractors: Dict[str, Any] = {}
         self.seeds = list(seeds) if seeds is not None else list(range(G))
         assert len(self.seeds) == G
         self._declare_params()
@@ -171,12 +199,48 @@ class BasePack:
     def _to_compute(self, t: torch.Tensor) -> torch.Tensor:
         return t.to(device=self.device, dtype=self.compute_dtype)
 
+    def _pad_features(self, t: torch.Tensor, logical_f: int) -> torch.Tensor:
+        """Zero-pad the last (feature) dim to the physical width."""
+        pf = self._pad(logical_f)
+        if t.shape[-1] == pf:
+            return t
+        assert t.shape[-1] == logical_f, (t.shape, logical_f)
+        return torch.nn.functional.pad(t, (0, pf - logical_f))
+
     def param_names(self) -> List[str]:
         return [name for name, _ in self.store._shapes]
 
+    def _extract_logical(self, name: str, arr: np.ndarray) -> np.ndarray:
+        """Physical [per-model] array -> logical (serialized) layout."""
+        ex = self._extractors.get(name)
+        if ex is not None:
+            return ex[0](arr)
+        logical = self.store.logical[name][1:]
+        if arr.shape == tuple(logical):
+            return arr.copy()
+        return arr[tuple(slice(0, d) for d in logical)].copy()
+
+    def _insert_logical(self, name: str, view: torch.Tensor,
+                        arr: np.ndarray):
+        """Write a logical array into the physical [per-model] view
+        (pad regions stay zero)."""
+        ex = self._extractors.get(name)
+        t = torch.from_numpy(np.ascontiguousarray(arr))
+        if ex is not None:
+            ex[1](view, t)
+            return
+        logical = self.store.logical[name][1:]
+        if tuple(view.shape) == tuple(logical):
+            view.copy_(t)
+            return
+        view.zero_()
+        view[tuple(slice(0, d) for d in logical)].copy_(t)
+
     def state_for_model(self, g: int) -> Dict[str, np.ndarray]:
         return {
-            name: self.store.views[name][g].detach().cpu().numpy().copy()
+            name: self._extract_logical(
+                name, self.store.views[name][g].detach().cpu().numpy()
+            )
             for name in self.param_names()
         }
 
@@ -190,12 +254,14 @@ class BasePack:
             off, n = self.store._offsets[name]
             arr = flat[off : off + n].reshape(shape)
             for g in range(self.G):
-                out[g][name] = arr[g].copy()
+                out[g][name] = self._extract_logical(name, arr[g])
         return out
 
     def load_model_state(self, g: int, state: Dict[str, np.ndarray]):
         for name, arr in state.items():
-            self.store.views[name][g].copy_(torch.from_numpy(np.asarray(arr)))
+            self._insert_logical(
+                name, self.store.views[name][g], np.asarray(arr)
+            )
         self.store.sync_lp()
 
     def optimizer_state_for_model(self, g: int) -> Dict[str, np.ndarray]:
@@ -390,6 +456,9 @@ class BasePack:
         per-model per-epoch values: {"loss": [[...G...] per epoch],
         "accuracy": ...}.
         """
+        if self.pad8:
+            X = self._pad_features(X, self._logical_in)
+            Y = self._pad_features(Y, self._logical_out)
         G, N = X.shape[0], X.shape[1]
         adam = self.spec.adam_params
         history: Dict[str, list] = {"loss": [], "accuracy": []}
@@ -500,16 +569,32 @@ class DensePack(BasePack):
 
     def _declare_params(self):
         dims = self.spec.dense_dims()
-        self.layer_meta = dims
-        for i, (fin, fout, _act, _l1) in enumerate(dims):
-            self.store.declare(f"W{i}", (self.G, fin, fout))
-            self.store.declare(f"b{i}", (self.G, fout))
+        self.layer_meta_logical = dims
+        p = self._pad
+        # physical layer dims padded to 8 (pad units provably inert —
+        # see pad_enabled); activations between layers inherit the
+        # padded widths so every grouped GEMM stages 16-byte vectors
+        self.layer_meta = [
+            (p(fin), p(fout), act, l1) for fin, fout, act, l1 in dims
+        ]
+        self._logical_in = dims[0][0]
+        self._logical_out = dims[-1][1]
+        for i, (fin, fout, _act, _l1) in enumerate(self.layer_meta):
+            lfin, lfout = dims[i][0], dims[i][1]
+            self.store.declare(f"W{i}", (self.G, fin, fout),
+                               logical=(self.G, lfin, lfout))
+            self.store.declare(f"b{i}", (self.G, fout),
+                               logical=(self.G, lfout))
 
     def _init_weights(self):
         for g in range(self.G):
             gen = torch.Generator().manual_seed(int(self.seeds[g]) & 0x7FFFFFFF)
-            for i, (fin, fout, _act, _l1) in enumerate(self.layer_meta):
-                self.store.views[f"W{i}"][g].copy_(_glorot_uniform((fin, fout), gen))
+            for i, (lfin, lfout, _act, _l1) in enumerate(
+                self.layer_meta_logical
+            ):
+                self.store.views[f"W{i}"][g][:lfin, :lfout].copy_(
+                    _glorot_uniform((lfin, lfout), gen)
+                )
                 # biases stay zero
 
     def forward(self, X: torch.Tensor) -> List[torch.Tensor]:
@@ -524,15 +609,24 @@ class DensePack(BasePack):
         return acts
 
     def predict(self, X: torch.Tensor) -> torch.Tensor:
-        return self.forward(self._to_compute(X))[-1]
+        Xc = self._to_compute(X)
+        if self.pad8:
+            Xc = self._pad_features(Xc, self._logical_in)
+        out = self.forward(Xc)[-1]
+        return out[..., : self._logical_out]
 
     def eval_batch(self, Xb, Tb) -> torch.Tensor:
+        # pad diffs are exactly 0; divide by the REAL element count
         out = self.forward(Xb)[-1].float()
-        return ((out - Tb.float()) ** 2).mean(dim=(1, 2))
+        n = Xb.shape[1] * self._logical_out
+        return ((out - Tb.float()) ** 2).sum(dim=(1, 2)) / n
 
     def train_batch(self, Xb, Tb) -> torch.Tensor:
         acts = self.forward(Xb)
-        loss, dA = ops.mse_bwd(acts[-1], Tb.to(acts[-1].dtype))
+        loss, dA = ops.mse_bwd(
+            acts[-1], Tb.to(acts[-1].dtype),
+            Xb.shape[1] * self._logical_out,
+        )
         for i in range(len(self.layer_meta) - 1, -1, -1):
             _fin, _fout, act, l1 = self.layer_meta[i]
             dZ = ops.act_l1_bwd(dA, acts[i + 1], act, l1)
@@ -554,37 +648,104 @@ class LSTMPack(BasePack):
     analog of create_keras_timeseriesgenerator, reference
     models.py:713-793)."""
 
+    @staticmethod
+    def _gate_converters(H, pH, row_l):
+        """Logical<->physical converters for gate-interleaved LSTM
+        params: physical gate blocks are pH wide (padded), logical H —
+        so slicing alone can't extract them."""
+        def extract(arr):
+            if arr.ndim == 1:
+                return np.concatenate(
+                    [arr[gb * pH : gb * pH + H] for gb in range(4)]
+                ).copy()
+            r = arr[:row_l] if row_l else arr
+            return np.concatenate(
+                [r[:, gb * pH : gb * pH + H] for gb in range(4)], axis=1
+            ).copy()
+
+        def insert(view, t):
+            view.zero_()
+            if view.dim() == 1:
+                for gb in range(4):
+                    view[gb * pH : gb * pH + H].copy_(
+                        t[gb * H : (gb + 1) * H]
+                    )
+            else:
+                rl = row_l if row_l else view.shape[0]
+                for gb in range(4):
+                    view[:rl, gb * pH : gb * pH + H].copy_(
+                        t[:, gb * H : (gb + 1) * H]
+                    )
+
+        return extract, insert
+
     def _declare_params(self):
         spec = self.spec
-        self.lstm_meta = []  # (fin, H, return_sequences)
-        fin = spec.n_features
+        p = self._pad
+        self.lstm_meta = []          # PHYSICAL (fin, H, return_sequences)
+        self.lstm_meta_logical = []  # logical dims for serialization
+        fin_l = spec.n_features
         lstm_layers = [l for l in spec.layers if l.kind == "lstm"]
         dense_layers = [l for l in spec.layers if l.kind == "dense"]
         assert len(dense_layers) == 1, "LSTM spec needs exactly one output dense layer"
+        self._logical_in = fin_l
         for i, layer in enumerate(lstm_layers):
-            H = layer.units
-            self.store.declare(f"Wx{i}", (self.G, fin, 4 * H))
-            self.store.declare(f"Wh{i}", (self.G, H, 4 * H))
-            self.store.declare(f"bl{i}", (self.G, 4 * H))
+            H_l = layer.units
+            fin, H = p(fin_l), p(H_l)
+            self.store.declare(f"Wx{i}", (self.G, fin, 4 * H),
+                               logical=(self.G, fin_l, 4 * H_l))
+            self.store.declare(f"Wh{i}", (self.G, H, 4 * H),
+                               logical=(self.G, H_l, 4 * H_l))
+            self.store.declare(f"bl{i}", (self.G, 4 * H),
+                               logical=(self.G, 4 * H_l))
+            if H != H_l or fin != fin_l:
+                self._extractors[f"Wx{i}"] = self._gate_converters(
+                    H_l, H, fin_l
+                )
+                self._extractors[f"Wh{i}"] = self._gate_converters(
+                    H_l, H, H_l
+                )
+                self._extractors[f"bl{i}"] = self._gate_converters(
+                    H_l, H, None
+                )
             self.lstm_meta.append((fin, H, layer.return_sequences))
-            fin = H
+            self.lstm_meta_logical.append((fin_l, H_l, layer.return_sequences))
+            fin_l = H_l
         out_layer = dense_layers[0]
-        self.dense_meta = (fin, out_layer.units, out_layer.activation)
-        self.store.declare("Wd", (self.G, fin, out_layer.units))
-        self.store.declare("bd", (self.G, out_layer.units))
+        self._logical_out = out_layer.units
+        self.dense_meta = (p(fin_l), p(out_layer.units), out_layer.activation)
+        self.dense_meta_logical = (fin_l, out_layer.units, out_layer.activation)
+        self.store.declare("Wd", (self.G, p(fin_l), p(out_layer.units)),
+                           logical=(self.G, fin_l, out_layer.units))
+        self.store.declare("bd", (self.G, p(out_layer.units)),
+                           logical=(self.G, out_layer.units))
 
     def _init_weights(self):
+        # draw the LOGICAL matrices in the same RNG order as the
+        # unpadded CPU pack (seed determinism across devices), then
+        # scatter into the padded gate blocks
         for g in range(self.G):
             gen = torch.Generator().manual_seed(int(self.seeds[g]) & 0x7FFFFFFF)
-            for i, (fin, H, _rs) in enumerate(self.lstm_meta):
-                self.store.views[f"Wx{i}"][g].copy_(_glorot_uniform((fin, 4 * H), gen))
-                wh = torch.cat([_orthogonal((H, H), gen) for _ in range(4)], dim=1)
-                self.store.views[f"Wh{i}"][g].copy_(wh)
-                b = torch.zeros(4 * H)
-                b[H : 2 * H] = 1.0  # unit forget-gate bias (Keras default)
-                self.store.views[f"bl{i}"][g].copy_(b)
-            fin, fout, _act = self.dense_meta
-            self.store.views["Wd"][g].copy_(_glorot_uniform((fin, fout), gen))
+            for i, (fin_l, H_l, _rs) in enumerate(self.lstm_meta_logical):
+                _pfin, pH, _ = self.lstm_meta[i]
+                wx = _glorot_uniform((fin_l, 4 * H_l), gen)
+                wh = torch.cat(
+                    [_orthogonal((H_l, H_l), gen) for _ in range(4)], dim=1
+                )
+                b = torch.zeros(4 * H_l)
+                b[H_l : 2 * H_l] = 1.0  # unit forget-gate bias (Keras)
+                self._insert_logical(
+                    f"Wx{i}", self.store.views[f"Wx{i}"][g], wx.numpy()
+                )
+                self._insert_logical(
+                    f"Wh{i}", self.store.views[f"Wh{i}"][g], wh.numpy()
+                )
+                self._insert_logical(
+                    f"bl{i}", self.store.views[f"bl{i}"][g], b.numpy()
+                )
+            fin_l, fout_l, _act = self.dense_meta_logical
+            wd = _glorot_uniform((fin_l, fout_l), gen)
+            self.store.views["Wd"][g][:fin_l, :fout_l].copy_(wd)
 
     # ---- sequence forward/backward ------------------------------------
     def _use_fused(self, B: Optional[int] = None) -> bool:
@@ -696,17 +857,24 @@ class LSTMPack(BasePack):
         return y, cache
 
     def predict_windows(self, Xw: torch.Tensor) -> torch.Tensor:
-        y, _ = self._forward_seq(self._to_compute(Xw), keep=False)
-        return y
+        Xc = self._to_compute(Xw)
+        if self.pad8 and Xc.shape[-1] == self._logical_in:
+            Xc = self._pad_features(Xc, self._logical_in)
+        y, _ = self._forward_seq(Xc, keep=False)
+        return y[..., : self._logical_out]
 
     def eval_batch(self, Xw, Tb) -> torch.Tensor:
         y = self.predict_windows(Xw).float()
-        return ((y - Tb.float()) ** 2).mean(dim=(1, 2))
+        return (
+            (y - Tb.float()[..., : self._logical_out]) ** 2
+        ).mean(dim=(1, 2))
 
     def _train_batch_fused(self, Xw, Tb) -> torch.Tensor:
         G, B, T, _ = Xw.shape
         y, cache = self._forward_seq_fused(Xw, keep=True)
-        loss, dY = ops.mse_bwd(y, Tb.to(y.dtype))
+        loss, dY = ops.mse_bwd(
+            y, Tb.to(y.dtype), B * self._logical_out
+        )
 
         fin, fout, act = self.dense_meta
         head = cache[-1]
@@ -758,7 +926,9 @@ class LSTMPack(BasePack):
             return self._train_batch_fused(Xw, Tb)
         G, B, T, _ = Xw.shape
         y, cache = self._forward_seq(Xw, keep=True)
-        loss, dY = ops.mse_bwd(y, Tb.to(y.dtype))
+        loss, dY = ops.mse_bwd(
+            y, Tb.to(y.dtype), B * self._logical_out
+        )
 
         # output dense layer backward
         fin, fout, act = self.dense_meta
@@ -848,6 +1018,8 @@ class LSTMPack(BasePack):
         (offset-aligned like the reference: output starts at row
         lookback-1+lookahead)."""
         X = self._to_compute(X)
+        if self.pad8:
+            X = self._pad_features(X, self._logical_in)
         G, N, F = X.shape
         nw = self.n_windows(N)
         outs = []

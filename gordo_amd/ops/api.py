@@ -118,10 +118,13 @@ def grouped_gemm_acc(A, B, C):
     return ref.grouped_gemm_acc(A, B, C)
 
 
-def mse_bwd(Y, T) -> Tuple[torch.Tensor, torch.Tensor]:
+def mse_bwd(Y, T, real_n: int = -1) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Per-model MSE loss + grad. ``real_n``: divisor for the mean
+    when the feature dim carries zero padding (pad diffs are exactly 0,
+    so only the normalization needs the real count)."""
     if _on_gpu(Y):
-        return _require_hip().mse_bwd(Y, T)
-    return ref.mse_bwd(Y, T)
+        return _require_hip().mse_bwd(Y, T, int(real_n))
+    return ref.mse_bwd(Y, T, real_n)
 
 
 def adam_step(p, g, m, v, lr, beta1, beta2, eps, step, p_lp=None,

@@ -400,15 +400,19 @@ __global__ void act_l1_bwd_kernel(const bf16* __restrict__ dA,
 }
 
 // loss[g] = mean((Y-T)^2) over (B*F); dY = 2(Y-T)/(B*F)
+// real_n: divisor for the mean — the count of REAL elements when the
+// feature dim is zero-padded to 8 (pad diffs are exactly 0, so only
+// the normalization changes; see engine/pack.py padding notes).
 __global__ void mse_bwd_kernel(const bf16* __restrict__ Y,
                                const bf16* __restrict__ T,
                                bf16* __restrict__ dY,
-                               float* __restrict__ loss, int per_g) {
+                               float* __restrict__ loss, int per_g,
+                               int real_n) {
   int g = blockIdx.y;
   const bf16* Yg = Y + (size_t)g * per_g;
   const bf16* Tg = T + (size_t)g * per_g;
   bf16* dYg = dY + (size_t)g * per_g;
-  float inv_n = 1.f / (float)per_g;
+  float inv_n = 1.f / (float)real_n;
   float local = 0.f;
   for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < (size_t)per_g;
        i += (size_t)gridDim.x * blockDim.x) {
@@ -670,19 +674,21 @@ torch::Tensor window_gather(torch::Tensor X, torch::Tensor idx,
   return out;
 }
 
-std::vector<torch::Tensor> mse_bwd(torch::Tensor Y, torch::Tensor T) {
+std::vector<torch::Tensor> mse_bwd(torch::Tensor Y, torch::Tensor T,
+                                   int64_t real_n) {
   CHECK_GPU(Y);
   auto Yc = to_bf16c(Y);
   auto Tc = to_bf16c(T);
   int G = Yc.size(0);
   int64_t per_g = Yc.numel() / G;
+  if (real_n <= 0) real_n = per_g;
   auto dY = torch::empty_like(Yc);
   auto loss = torch::zeros({G}, Yc.options().dtype(torch::kFloat32));
   int blocks = (int)std::min<int64_t>((per_g + 255) / 256, 512);
   hipLaunchKernelGGL(mse_bwd_kernel, dim3(blocks, G), dim3(256), 0,
                      cur_stream(), (const bf16*)Yc.data_ptr(),
                      (const bf16*)Tc.data_ptr(), (bf16*)dY.data_ptr(),
-                     loss.data_ptr<float>(), (int)per_g);
+                     loss.data_ptr<float>(), (int)per_g, (int)real_n);
   return {loss, dY};
 }
 
@@ -813,7 +819,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "K7 sliding-window featurizer: [G,N,F] -> [G,B,T,F]");
   mod.def("grouped_gemm_acc", &grouped_gemm_acc, "C += A@B per group (MFMA)");
   mod.def("act_l1_bwd", &act_l1_bwd, "fused activation+L1 backward");
-  mod.def("mse_bwd", &mse_bwd, "fused per-model MSE loss + grad");
+  mod.def("mse_bwd", &mse_bwd, py::arg("Y"), py::arg("T"),
+          py::arg("real_n") = -1,
+          "fused per-model MSE loss + grad (real_n: divisor when the "
+          "feature dim is zero-padded)");
   mod.def("adam_step", &adam_step, "fused Adam + bf16 mirror refresh");
   mod.def("lstm_pointwise_fwd", &lstm_pointwise_fwd, "LSTM cell fwd");
   mod.def("lstm_pointwise_bwd", &lstm_pointwise_bwd, "LSTM cell bwd");

@@ -104,14 +104,16 @@ def grouped_gemm_acc(A: torch.Tensor, B: torch.Tensor, C: torch.Tensor):
 
 
 def mse_bwd(
-    Y: torch.Tensor, T: torch.Tensor
+    Y: torch.Tensor, T: torch.Tensor, real_n: int = -1
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Per-model MSE loss and its gradient.
 
     loss[g] = mean((Y-T)^2) over (B, F); dY = 2 (Y-T) / (B*F).
+    ``real_n`` overrides the divisor when the feature dim carries zero
+    padding (pad diffs are exactly 0; only the mean's count changes).
     """
     diff = Y - T
-    n = Y.shape[1] * Y.shape[2]
+    n = Y.shape[1] * Y.shape[2] if real_n <= 0 else real_n
     loss = (diff * diff).sum(dim=(1, 2)) / n
     dY = diff * (2.0 / n)
     return loss, dY

@@ -382,7 +382,16 @@ class PackedFleetBuilder:
         # the init itself is deterministic in (arch, per-machine seeds)
         # and memoized across builds.
         t_seg = time.time()
-        init_key = (spec.arch_key(), tuple(p.seed for p in group))
+        from ..engine.pack import pad_enabled
+
+        # the snapshot layout depends on whether this device pads dims
+        # to 8 (GPU) — a CPU (unpadded) snapshot must never be reused
+        # for a GPU pack in the same process
+        init_key = (
+            spec.arch_key(),
+            tuple(p.seed for p in group),
+            pad_enabled(self.device),
+        )
         cached_init = _INIT_CACHE.get(init_key)
         pack = self._make_pack(spec, group, init_p32=cached_init)
         init_snapshot = pack.store.p32.clone()

@@ -521,3 +521,77 @@ def test_early_stopping_group_semantics_and_drift():
         assert hist["loss"][e][1] == pytest.approx(
             hist_b["loss"][e][0], rel=1e-5, abs=1e-8
         )
+
+
+def test_pad8_packs_match_unpadded(monkeypatch):
+    """Feature/hidden padding to 8 (the GPU staging-vectorization
+    layout) is numerically inert: forced on CPU, a padded pack matches
+    the unpadded pack on losses, predictions and the serialized
+    (logical) state — pad units provably stay zero."""
+    from gordo_amd.engine import pack as packmod
+
+    rng = np.random.default_rng(3)
+
+    # dense: odd dims everywhere
+    dspec = dense_spec(n_features=5, units=(4, 3, 4))
+    X = torch.tensor(rng.random((2, 200, 5)).astype("float32"))
+    plain = DensePack(dspec, G=2, device="cpu", seeds=[1, 2])
+    monkeypatch.setattr(packmod, "pad_enabled", lambda d: True)
+    padded = DensePack(dspec, G=2, device="cpu", seeds=[1, 2])
+    assert padded.store.views["W0"].shape[1:] == (8, 8)
+    # pad regions start zero and the logical state matches exactly
+    s0, s1 = plain.state_for_model(0), padded.state_for_model(0)
+    for k in s0:
+        np.testing.assert_array_equal(s0[k], s1[k])
+    h_plain = plain.fit(X, X.clone(), epochs=3, batch_size=64,
+                        shuffle=False)
+    h_pad = padded.fit(X, X.clone(), epochs=3, batch_size=64,
+                       shuffle=False)
+    for e in range(3):
+        for g in range(2):
+            assert h_pad["loss"][e][g] == pytest.approx(
+                h_plain["loss"][e][g], rel=1e-5, abs=1e-9
+            )
+    out_plain = plain.predict(X[:, :50]).float().numpy()
+    out_pad = padded.predict(X[:, :50]).float().numpy()
+    assert out_pad.shape == out_plain.shape
+    np.testing.assert_allclose(out_pad, out_plain, rtol=1e-4, atol=1e-6)
+    # pad weight regions stayed EXACTLY zero through training
+    W0 = padded.store.views["W0"][0].numpy()
+    assert (W0[5:, :] == 0).all() and (W0[:, 4:] == 0).all()
+    # trained logical state round-trips into a fresh unpadded pack
+    fresh = DensePack(dspec, G=1, device="cpu", seeds=[9])
+    monkeypatch.undo()
+    fresh.load_model_state(0, padded.state_for_model(0))
+    np.testing.assert_allclose(
+        fresh.predict(X[:1, :50]).numpy(), out_pad[:1], rtol=1e-4,
+        atol=1e-6,
+    )
+
+
+def test_pad8_lstm_pack_matches_unpadded(monkeypatch):
+    from gordo_amd.engine import pack as packmod
+
+    rng = np.random.default_rng(4)
+    spec = lstm_spec(n_features=4, H=6, lookback=5)
+    X = torch.tensor(rng.random((1, 120, 4)).astype("float32"))
+    plain = LSTMPack(spec, G=1, device="cpu", seeds=[7])
+    monkeypatch.setattr(packmod, "pad_enabled", lambda d: True)
+    padded = LSTMPack(spec, G=1, device="cpu", seeds=[7])
+    assert padded.store.views["Wh0"].shape[1:] == (8, 32)
+    s0, s1 = plain.state_for_model(0), padded.state_for_model(0)
+    for k in s0:
+        np.testing.assert_array_equal(s0[k], s1[k])
+    hp = plain.fit(X, X.clone(), epochs=2, batch_size=32, shuffle=False)
+    hq = padded.fit(X, X.clone(), epochs=2, batch_size=32, shuffle=False)
+    for e in range(2):
+        assert hq["loss"][e][0] == pytest.approx(
+            hp["loss"][e][0], rel=1e-5, abs=1e-9
+        )
+    np.testing.assert_allclose(
+        padded.predict(X).numpy(), plain.predict(X).numpy(),
+        rtol=1e-4, atol=1e-6,
+    )
+    # forget-gate pad bias stayed zero; real forget bias trained from 1
+    bl = padded.store.views["bl0"][0].numpy()
+    assert (bl[8 + 6 : 16] == 0).all()
