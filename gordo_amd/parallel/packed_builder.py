@@ -346,16 +346,24 @@ class PackedFleetBuilder:
         # slice (_fold_transform), matching sklearn.cross_validate's
         # per-fold pipeline clone — no test-fold leakage into scores
         # or DiffBased thresholds.
-        Xraw_list, Xt_list, y_list = [], [], []
+        Xraw_list, y_list = [], []
         for p in group:
-            Xraw = p.X.values.astype(np.float32)
-            Xraw_list.append(Xraw)
-            Xt = Xraw
-            for _, step in p.pre_steps:
-                Xt = step.fit_transform(Xt)
-            Xt_list.append(np.asarray(Xt, dtype=np.float32))
+            Xraw_list.append(p.X.values.astype(np.float32))
             y_list.append(p.y.values.astype(np.float32))
-        self._phase("pre_transform", time.time() - t_seg)
+
+        def full_pre_transform():
+            # runs INSIDE the final-fit closure so the per-machine
+            # sklearn fits overlap the concurrent fold fits on GPU
+            t0 = time.time()
+            out = []
+            for p, Xraw in zip(group, Xraw_list):
+                Xt = Xraw
+                for _, step in p.pre_steps:
+                    Xt = step.fit_transform(Xt)
+                out.append(np.asarray(Xt, dtype=np.float32))
+            self._phase("pre_transform(overlapped)", time.time() - t0)
+            return out
+        self._phase("data_prep", time.time() - t_seg)
 
         cv_mode = str(evaluation.get("cv_mode", "full_build")).lower()
         cv_duration = None
@@ -368,6 +376,8 @@ class PackedFleetBuilder:
                 else _nullcontext()
             )
             with ctx:
+                Xt_list = full_pre_transform()
+                fit_state["Xt_list"] = Xt_list
                 self._reset_pack(pack, init_snapshot)
                 Xd = self._stack(Xt_list, pack)
                 Yd = self._stack(y_list, pack)
@@ -455,8 +465,9 @@ class PackedFleetBuilder:
         # (numpy slicing, metadata assembly, pickle+json dumps), all
         # independent of the NEXT group's GPU fits; runs on the
         # builder-wide pool and is joined at the end of build_all.
-        offset = len(Xt_list[0]) - pack._n_samples(len(Xt_list[0]))
-        n_feat = [x.shape[1] for x in Xt_list]
+        n_rows = len(Xraw_list[0])
+        offset = n_rows - pack._n_samples(n_rows)
+        n_feat = [x.shape[1] for x in fit_state["Xt_list"]]
         n_feat_out = [y.shape[1] for y in y_list]
 
         def adopt_and_save(group=group, pack=pack, history=history,
