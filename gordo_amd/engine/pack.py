@@ -825,25 +825,40 @@ class LSTMPack(BasePack):
         return y, cache
 
     def _forward_seq_fused(self, Xw: torch.Tensor, keep: bool):
-        """GPU path: one big x-side GEMM + one fused sequence-scan
-        kernel per layer."""
+        """GPU path: one fused scan per layer. Layers fitting the v4
+        geometry compute the x-side gate GEMM INSIDE the scan (no xW
+        HBM round trip — the fleet's scans are bandwidth-bound) and
+        skip the cs/gacts stores on inference passes; other layers use
+        the two-step x-GEMM + scan."""
+        import os as _os
+
         G, B, T, _ = Xw.shape
+        v4_on = _os.environ.get("GORDO_LSTM_V4", "1") != "0"
         seq = Xw
         cache = []
         for li, (fin, H, rs) in enumerate(self.lstm_meta):
             Wx = self.store.cviews[f"Wx{li}"]
             Wh = self.store.cviews[f"Wh{li}"]
             b = self.store.views[f"bl{li}"]
-            xW = ops.grouped_linear_fwd(
-                seq.reshape(G, B * T, fin), Wx, b, "linear"
-            ).view(G, B, T, 4 * H)
-            hs, cs, gacts = ops.lstm_seq_fwd(xW, Wh)
+            if v4_on and ops.lstm_v4_available(H, fin):
+                out = ops.lstm_seq_fwd_fused(seq, Wx, Wh, b,
+                                             store_aux=keep)
+                hs = out[0]
+                cs = out[1] if keep else None
+                gacts = out[2] if keep else None
+            else:
+                xW = ops.grouped_linear_fwd(
+                    seq.reshape(G, B * T, fin), Wx, b, "linear"
+                ).view(G, B, T, 4 * H)
+                hs, cs, gacts = ops.lstm_seq_fwd(xW, Wh)
+                if not keep:
+                    cs = gacts = None
             cache.append(
                 dict(
                     seq_in=seq if keep else None,
                     hs=hs,
-                    cs=cs if keep else None,
-                    gacts=gacts if keep else None,
+                    cs=cs,
+                    gacts=gacts,
                 )
             )
             seq = hs

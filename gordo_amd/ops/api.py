@@ -163,6 +163,45 @@ def lstm_seq_fwd(xW, Wh):
     return ext.lstm_seq_fwd_v3(xW, Wh)
 
 
+def lstm_v4_available(H: int, F: int) -> bool:
+    """v4 fused-xW scan geometry: H <= 64, H%8==0, F%8==0, F<=128
+    (the pad8 engine layout satisfies the alignment)."""
+    return hip_available() and H <= 64 and H % 8 == 0 and F % 8 == 0 \
+        and F <= 128
+
+
+def lstm_seq_fwd_fused(xseq, Wx, Wh, bias, store_aux=True):
+    """v4 scan: the x-side gate GEMM runs INSIDE the recurrence over an
+    LDS-resident [Wh ; Wx] tile — no xW round trip through HBM (the
+    fleet's scans are bandwidth-bound). ``store_aux=False`` skips the
+    cs/gacts stores for inference. GORDO_LSTM_V4=0 falls back to the
+    two-step path at the engine level."""
+    if _on_gpu(xseq):
+        return _require_hip().lstm_seq_fwd_fused(
+            xseq, Wx, Wh, bias, bool(store_aux)
+        )
+    # CPU oracle: explicit xW then the per-timestep reference scan
+    G, B, T, F = xseq.shape
+    H = Wh.shape[1]
+    gates_all = ref.grouped_linear_fwd(
+        xseq.reshape(G, B * T, F), Wx, bias, "linear"
+    ).view(G, B, T, 4 * H)
+    h = torch.zeros(G, B, H, dtype=xseq.dtype)
+    c = torch.zeros(G, B, H, dtype=torch.float32)
+    hs = torch.empty(G, B, T, H, dtype=xseq.dtype)
+    cs = torch.empty(G, B, T, H, dtype=torch.float32)
+    ga = torch.empty(G, B, T, 4 * H, dtype=xseq.dtype)
+    for t in range(T):
+        gates = gates_all[:, :, t] + torch.bmm(h, Wh)
+        h, c, gact = ref.lstm_pointwise_fwd(gates, c)
+        hs[:, :, t] = h
+        cs[:, :, t] = c
+        ga[:, :, t] = gact
+    if store_aux:
+        return [hs, cs, ga]
+    return [hs]
+
+
 def lstm_seq_fwd_v3(xW, Wh):
     """The pipelined forward scan, directly (for A/B tests)."""
     return _require_hip().lstm_seq_fwd_v3(xW, Wh)

@@ -773,6 +773,11 @@ std::vector<torch::Tensor> anomaly_score(
 // fused LSTM sequence-scan entry points (lstm_seq.hip)
 namespace gordo_lstm {
 std::vector<torch::Tensor> lstm_seq_fwd(torch::Tensor xW, torch::Tensor Wh);
+std::vector<torch::Tensor> lstm_seq_fwd_fused(torch::Tensor xseq,
+                                              torch::Tensor Wx,
+                                              torch::Tensor Wh,
+                                              torch::Tensor bias,
+                                              bool store_aux);
 std::vector<torch::Tensor> lstm_seq_fwd_v3(torch::Tensor xW, torch::Tensor Wh);
 torch::Tensor lstm_seq_bwd_v3(torch::Tensor dSeq, torch::Tensor gacts,
                               torch::Tensor cs, torch::Tensor Wh,
@@ -805,6 +810,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "pipelined fused LSTM forward scan (GORDO_LSTM_V3 opt-in)");
   mod.def("lstm_seq_fwd", &gordo_lstm::lstm_seq_fwd,
           "fused LSTM forward sequence scan (Wh resident in LDS)");
+  mod.def("lstm_seq_fwd_fused", &gordo_lstm::lstm_seq_fwd_fused,
+          "v4 scan with the x-side gate GEMM fused in (halves fwd "
+          "HBM traffic); store_aux=false skips cs/gacts (inference)");
   mod.def("lstm_seq_bwd", &gordo_lstm::lstm_seq_bwd,
           "fused LSTM backward (BPTT) sequence scan");
   mod.def("grouped_linear_fwd", &grouped_linear_fwd,

@@ -1137,6 +1137,177 @@ __global__ __launch_bounds__(256) void lstm_seq_bwd_big_kernel(
   }
 }
 
+// ===========================================================================
+// v4 fused-xW forward scan — computes the x-side gate GEMM INSIDE the
+// recurrence instead of precomputing xW = x@Wx + b into a [G,B,T,4H]
+// HBM tensor: in the fleet the scans are HBM-BANDWIDTH bound (the xW
+// round trip is ~770 MB per layer pass at the bench shape), so folding
+// the two GEMMs into one K = [h | x_t] x [Wh ; Wx] MFMA loop over an
+// LDS-resident concatenated weight tile roughly halves forward-path
+// traffic and removes the separate grouped-GEMM launch. The x_{t+1}
+// row tile is register-prefetched during the pointwise phase (the v3
+// pattern) and written into the LDS K-buffer before the barrier.
+// STORE_AUX=false additionally skips the cs/gacts stores for
+// inference (predict/CV-validation passes need only hs).
+// Geometry: H <= 64, H % 8 == 0, F % 8 == 0 (the pad8 engine layout).
+// ===========================================================================
+template <int ROWS, bool STORE_AUX>
+__global__ __launch_bounds__(256) void lstm_seq_fwd_v4_kernel(
+    const bf16* __restrict__ xseq,   // [G, B, T, F]
+    const bf16* __restrict__ WcatT,  // [G, 4H, Kc] (Kc = H + F)
+    const bf16* __restrict__ bias,   // [G, 4H]
+    bf16* __restrict__ hs, float* __restrict__ cs,
+    bf16* __restrict__ gacts, int B, int T, int H, int F, int ldg) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int H4 = 4 * H;
+  const int Kc = H + F;          // both % 8 == 0
+  // pad the K row to the MFMA K step so the last fragment read stays
+  // in bounds (zeros staged beyond Kc)
+  const int LDKc = ((Kc + 31) & ~31) + 8;
+  bf16* WT = reinterpret_cast<bf16*>(smem);               // [H4][LDKc]
+  bf16* hxS = WT + (size_t)H4 * LDKc;                     // [ROWS][LDKc]
+  bf16* gS = hxS + (size_t)ROWS * LDKc;                   // [ROWS][ldg]
+  bf16* bS = gS + (size_t)ROWS * ldg;                     // [ldg]
+  float* cS = reinterpret_cast<float*>(bS + ldg);         // [ROWS][H]
+
+  const int nb = (B + ROWS - 1) / ROWS;
+  const int g = blockIdx.x / nb;
+  const int r0 = (blockIdx.x % nb) * ROWS;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int l15 = lane & 15;
+  const int kslot = lane >> 4;
+
+  const bf16* WTg = WcatT + (size_t)g * H4 * Kc;
+  const bf16* xg = xseq + ((size_t)g * B + r0) * T * F;
+  bf16* hsg = hs + ((size_t)g * B + r0) * T * H;
+  float* csg = cs + ((size_t)g * B + r0) * T * H;
+  bf16* gag = gacts + ((size_t)g * B + r0) * T * H4;
+  const int rows_here = min(ROWS, B - r0);
+
+  // stage the concatenated weight tile (zero-padded K tail) + bias
+  for (int i = tid; i < H4 * LDKc; i += 256) {
+    int n = i / LDKc, k = i % LDKc;
+    WT[i] = (k < Kc) ? WTg[(size_t)n * Kc + k] : lf2bf(0.f);
+  }
+  for (int i = tid; i < ldg; i += 256)
+    bS[i] = (i < H4) ? bias[(size_t)g * H4 + i] : lf2bf(0.f);
+  for (int i = tid; i < ROWS * LDKc; i += 256) hxS[i] = lf2bf(0.f);
+  for (int i = tid; i < ROWS * H; i += 256) cS[i] = 0.f;
+  // x_0 into the K-buffer's x section (16-byte lanes: F % 8 == 0)
+  for (int i = tid; i < ROWS * (F / 8); i += 256) {
+    int row = i / (F / 8), f8 = i % (F / 8);
+    bf16x8 v = {};
+    if (row < rows_here)
+      v = *reinterpret_cast<const bf16x8*>(
+          &xg[((size_t)row * T + 0) * F + f8 * 8]);
+    *reinterpret_cast<bf16x8*>(&hxS[row * LDKc + H + f8 * 8]) = v;
+  }
+  __syncthreads();
+
+  constexpr int FM = ROWS / 16;
+  const int wcol0 = wid * 64;
+  const bool wave_active = wcol0 < H4;
+  // per-thread x-prefetch slots (strided like the staging loop)
+  constexpr int NPF = (ROWS * 16 + 255) / 256;  // F/8 <= 16 at F<=128
+  bf16x8 xpf[NPF];
+
+  for (int t = 0; t < T; ++t) {
+    // ---- gates = [h | x_t] @ [Wh ; Wx] + b (MFMA over Kc) ----
+    if (wave_active) {
+      f32x4 acc[FM][4] = {};
+      for (int kk = 0; kk < Kc; kk += 32) {
+        #pragma unroll
+        for (int fm = 0; fm < FM; ++fm) {
+          bf16x8 a = *reinterpret_cast<const bf16x8*>(
+              &hxS[(fm * 16 + l15) * LDKc + kk + kslot * 8]);
+          #pragma unroll
+          for (int fn = 0; fn < 4; ++fn) {
+            int col = wcol0 + fn * 16 + l15;
+            bf16x8 b = *reinterpret_cast<const bf16x8*>(
+                &WT[(size_t)min(col, H4 - 1) * LDKc + kk + kslot * 8]);
+            acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a, b, acc[fm][fn], 0, 0, 0);
+          }
+        }
+      }
+      #pragma unroll
+      for (int fm = 0; fm < FM; ++fm) {
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+          int col = wcol0 + fn * 16 + l15;
+          if (col >= H4) continue;
+          float bv = lbf2f(bS[col]);
+          #pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            int row = fm * 16 + kslot * 4 + r;
+            gS[row * ldg + col] = lf2bf(acc[fm][fn][r] + bv);
+          }
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- register-prefetch x_{t+1} (no dependence on this step) ----
+    const int nx = ROWS * (F / 8);
+    if (t + 1 < T) {
+      #pragma unroll
+      for (int s = 0; s < NPF; ++s) {
+        int i = tid + s * 256;
+        if (i < nx) {
+          int row = i / (F / 8), f8 = i % (F / 8);
+          bf16x8 v = {};
+          if (row < rows_here)
+            v = *reinterpret_cast<const bf16x8*>(
+                &xg[((size_t)row * T + (t + 1)) * F + f8 * 8]);
+          xpf[s] = v;
+        }
+      }
+    }
+
+    // ---- fused gate math + h/c update + outputs ----
+    for (int e = tid; e < ROWS * H; e += 256) {
+      int row = e / H, hh = e % H;
+      const bf16* grow = &gS[row * ldg];
+      float i_g = sigmoidf_(lbf2f(grow[hh]));
+      float f_g = sigmoidf_(lbf2f(grow[H + hh]));
+      float g_g = fast_tanhf_(lbf2f(grow[2 * H + hh]));
+      float o_g = sigmoidf_(lbf2f(grow[3 * H + hh]));
+      float cc = f_g * cS[row * H + hh] + i_g * g_g;
+      float hv = o_g * fast_tanhf_(cc);
+      cS[row * H + hh] = cc;
+      hxS[row * LDKc + hh] = lf2bf(hv);
+      if (row < rows_here) {
+        size_t base = ((size_t)row * T + t) * H + hh;
+        hsg[base] = lf2bf(hv);
+        if (STORE_AUX) {
+          csg[base] = cc;
+          size_t gbase = ((size_t)row * T + t) * H4;
+          gag[gbase + hh] = lf2bf(i_g);
+          gag[gbase + H + hh] = lf2bf(f_g);
+          gag[gbase + 2 * H + hh] = lf2bf(g_g);
+          gag[gbase + 3 * H + hh] = lf2bf(o_g);
+        }
+      }
+    }
+    // write the prefetched x_{t+1} into the K-buffer before the
+    // barrier (its x section is not read by the pointwise phase)
+    if (t + 1 < T) {
+      #pragma unroll
+      for (int s = 0; s < NPF; ++s) {
+        int i = tid + s * 256;
+        if (i < nx) {
+          int row = i / (F / 8), f8 = i % (F / 8);
+          *reinterpret_cast<bf16x8*>(
+              &hxS[row * LDKc + H + f8 * 8]) = xpf[s];
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
 // ---------------------------------------------------------------------------
 namespace gordo_lstm {
 
@@ -1248,6 +1419,72 @@ torch::Tensor lstm_seq_bwd_big(torch::Tensor dSeq, torch::Tensor gacts,
                        (const bf16*)Whc.data_ptr(), (bf16*)dG.data_ptr(), B,
                        T, H, ldg, last_only ? 1 : 0);
   return dG;
+}
+
+inline bool v4_ok(int H, int F) {
+  return H <= 64 && H % 8 == 0 && F % 8 == 0 && F <= 128;
+}
+
+std::vector<torch::Tensor> lstm_seq_fwd_fused(torch::Tensor xseq,
+                                              torch::Tensor Wx,
+                                              torch::Tensor Wh,
+                                              torch::Tensor bias,
+                                              bool store_aux) {
+  // xseq [G,B,T,F]; Wx [G,F,4H]; Wh [G,H,4H]; bias [G,4H].
+  // Computes the x-side gate GEMM inside the scan (see kernel note).
+  TORCH_CHECK(xseq.is_cuda() && xseq.dim() == 4,
+              "xseq must be [G,B,T,F] on GPU");
+  auto xc = xseq.to(torch::kBFloat16).contiguous();
+  auto Wxc = Wx.to(torch::kBFloat16);
+  auto Whc = Wh.to(torch::kBFloat16);
+  auto bc = bias.to(torch::kBFloat16).contiguous();
+  int G = xc.size(0), B = xc.size(1), T = xc.size(2), F = xc.size(3);
+  int H = Whc.size(1), H4 = 4 * H;
+  TORCH_CHECK(v4_ok(H, F), "lstm_seq_fwd_fused geometry unsupported");
+  int Kc = H + F;
+  int LDKc = ((Kc + 31) & ~31) + 8;
+  int ldg = pad_ldg(H4);
+  // concatenated transposed weights [G, 4H, Kc] = [Wh ; Wx]^T
+  auto Wcat = torch::cat({Whc, Wxc}, 1).transpose(1, 2).contiguous();
+  auto hs = torch::empty({G, B, T, H}, xc.options());
+  auto cs = store_aux
+                ? torch::empty({G, B, T, H},
+                               xc.options().dtype(torch::kFloat32))
+                : torch::empty({1}, xc.options().dtype(torch::kFloat32));
+  auto gacts = store_aux ? torch::empty({G, B, T, H4}, xc.options())
+                         : torch::empty({1}, xc.options());
+  int rows = pick_rows(G, B);
+  auto lds_for = [&](int r) {
+    return (size_t)H4 * LDKc * 2 + (size_t)r * LDKc * 2 +
+           (size_t)r * ldg * 2 + (size_t)ldg * 2 + (size_t)r * H * 4;
+  };
+  while (rows > 16 && lds_for(rows) > 160 * 1024) rows /= 2;
+  size_t lds = lds_for(rows);
+  TORCH_CHECK(lds <= 160 * 1024, "LDS budget exceeded (v4 fwd)");
+  int blocks = G * ((B + rows - 1) / rows);
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  auto xp = (const bf16*)xc.data_ptr();
+  auto wp = (const bf16*)Wcat.data_ptr();
+  auto bp = (const bf16*)bc.data_ptr();
+  auto hp = (bf16*)hs.data_ptr();
+  auto cp = cs.data_ptr<float>();
+  auto gp = (bf16*)gacts.data_ptr();
+  #define V4_LAUNCH(R, S)                                              \
+    hipLaunchKernelGGL((lstm_seq_fwd_v4_kernel<R, S>), dim3(blocks),   \
+                       dim3(256), lds, stream, xp, wp, bp, hp, cp, gp, \
+                       B, T, H, F, ldg)
+  if (store_aux) {
+    if (rows == 64) V4_LAUNCH(64, true);
+    else if (rows == 32) V4_LAUNCH(32, true);
+    else V4_LAUNCH(16, true);
+  } else {
+    if (rows == 64) V4_LAUNCH(64, false);
+    else if (rows == 32) V4_LAUNCH(32, false);
+    else V4_LAUNCH(16, false);
+  }
+  #undef V4_LAUNCH
+  if (store_aux) return {hs, cs, gacts};
+  return {hs};
 }
 
 std::vector<torch::Tensor> lstm_seq_fwd(torch::Tensor xW, torch::Tensor Wh) {

@@ -588,3 +588,75 @@ def test_lstm_scan_row_tiles_bit_equal(monkeypatch):
     for rows in ("32", "64"):
         for a, b in zip(outs["16"], outs[rows]):
             assert torch.equal(a, b), rows
+
+
+@pytest.mark.parametrize("G,B,T,H,F", [
+    (2, 48, 20, 48, 56),    # bench-like padded dims
+    (1, 33, 12, 16, 8),     # minimal geometry, ragged rows
+    (2, 64, 24, 64, 128),   # caps
+])
+def test_lstm_seq_v4_fused_vs_twostep(G, B, T, H, F):
+    """The v4 fused-xW scan equals the two-step (x-GEMM then scan)
+    path within bf16 accumulation tolerance, and its inference mode
+    (store_aux=False) returns the same hs."""
+    require_hip()
+    x = to_dev_bf16(_rand(G, B, T, F, seed=90) * 0.5)
+    Wx = to_dev_bf16(_rand(G, F, 4 * H, seed=91) * 0.2)
+    Wh = to_dev_bf16(_rand(G, H, 4 * H, seed=92) * 0.2)
+    b = _rand(G, 4 * H, seed=93).cuda() * 0.1
+
+    xW = ops.grouped_linear_fwd(
+        x.reshape(G, B * T, F), Wx, b, "linear"
+    ).view(G, B, T, 4 * H)
+    hs1, cs1, ga1 = ops.lstm_seq_fwd(xW, Wh)
+
+    hs4, cs4, ga4 = ops.lstm_seq_fwd_fused(x, Wx, Wh, b, store_aux=True)
+    torch.testing.assert_close(
+        hs4.float(), hs1.float(), rtol=5e-2, atol=2e-2
+    )
+    torch.testing.assert_close(cs4, cs1, rtol=5e-2, atol=3e-2)
+    torch.testing.assert_close(
+        ga4.float(), ga1.float(), rtol=5e-2, atol=2e-2
+    )
+    (hs_inf,) = ops.lstm_seq_fwd_fused(x, Wx, Wh, b, store_aux=False)
+    assert torch.equal(hs_inf, hs4)
+
+    # CPU fp32 oracle through the same fused entry
+    want = ops.lstm_seq_fwd_fused(
+        x.float().cpu(), Wx.float().cpu(), Wh.float().cpu(),
+        b.float().cpu(), store_aux=True,
+    )
+    torch.testing.assert_close(
+        hs4.float().cpu(), want[0], rtol=6e-2, atol=3e-2
+    )
+
+
+def test_lstm_pack_v4_matches_twostep_end_to_end(monkeypatch):
+    """LSTMPack training with the v4 fused scans tracks the two-step
+    path (same pack, same seeds) within bf16 drift."""
+    require_hip()
+    from gordo_amd.engine.pack import LSTMPack
+    from gordo_amd.engine.spec import LayerSpec, ModelSpec
+
+    spec = ModelSpec(
+        model_type="lstm", n_features=10, n_features_out=10,
+        layers=[
+            LayerSpec(kind="lstm", units=12, return_sequences=True),
+            LayerSpec(kind="lstm", units=12, return_sequences=False),
+            LayerSpec(kind="dense", units=10, activation="linear"),
+        ],
+        lookback_window=16,
+    )
+    X = torch.tensor(
+        np.random.default_rng(5).random((1, 300, 10)).astype("float32")
+    )
+    losses = {}
+    for arm, env in (("v4", "1"), ("twostep", "0")):
+        monkeypatch.setenv("GORDO_LSTM_V4", env)
+        p = LSTMPack(spec, G=1, device="cuda", seeds=[3])
+        Xg = X.to("cuda", p.compute_dtype)
+        h = p.fit(Xg, Xg.clone(), epochs=2, batch_size=64, shuffle=False)
+        losses[arm] = [e[0] for e in h["loss"]]
+    monkeypatch.delenv("GORDO_LSTM_V4")
+    for a, b in zip(losses["v4"], losses["twostep"]):
+        assert a == pytest.approx(b, rel=0.05, abs=1e-3)
