@@ -275,9 +275,40 @@ __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
   f32x4 acc[2][2] = {};
   float db_part = 0.f;
 
+  // With pad8 layouts (K/N multiples of 8) the transpose staging flips
+  // to ROW-CONTIGUOUS 16-byte global loads + scalar LDS-transposed
+  // writes: the legacy path's per-element column gather costs 8
+  // separate L2 requests per thread (PMC: wgrad 56% wave-wait + 31%
+  // issue-stall). LDS write conflicts are bounded 2-way.
+  const bool vecA = ((K & 7) == 0) && ((reinterpret_cast<uintptr_t>(Ag)
+                                        & 15) == 0);
+  const bool vecZ = ((N & 7) == 0) && ((reinterpret_cast<uintptr_t>(Zg)
+                                        & 15) == 0);
   for (int m0 = m_begin; m0 < m_end; m0 += BK) {
     // stage A[m0..+32][k0..+64] transposed into As[k][m]
-    {
+    if (vecA) {
+      int m = tid >> 3;           // 0..31
+      int kk8 = (tid & 7) * 8;    // 0..56
+      int gm = m0 + m;
+      int gk0 = k0 + kk8;
+      bool ok = gm < M;
+      size_t row = (size_t)gm;
+      if (tshiftT > 0) {
+        ok = ok && (gm % tshiftT) > 0;  // t == 0 -> h_prev is zero
+        row = (size_t)gm - 1;
+      }
+      bf16 v[8];
+      if (ok && gk0 + 8 <= K) {
+        *reinterpret_cast<bf16x8*>(v) =
+            *reinterpret_cast<const bf16x8*>(&Ag[row * K + gk0]);
+      } else {
+        #pragma unroll
+        for (int e = 0; e < 8; ++e)
+          v[e] = (ok && gk0 + e < K) ? Ag[row * K + gk0 + e] : f2bf(0.f);
+      }
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) As[(kk8 + e) * LDT + m] = v[e];
+    } else {
       int k = tid >> 2;           // 0..63
       int mm = (tid & 3) * 8;     // 0..24
       int gk = k0 + k;
@@ -298,7 +329,34 @@ __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
     // stage dZ[m0..+32][n0..+64] transposed into Zs[n][m]; k-tile-0
     // blocks fold the bias-grad column sum into the same pass (what a
     // separate colsum kernel did with a latency-bound column walk).
-    {
+    if (vecZ) {
+      int m = tid >> 3;
+      int nn8 = (tid & 7) * 8;
+      int gm = m0 + m;
+      int gn0 = n0 + nn8;
+      bf16 v[8];
+      if (gm < M && gn0 + 8 <= N) {
+        *reinterpret_cast<bf16x8*>(v) =
+            *reinterpret_cast<const bf16x8*>(&Zg[(size_t)gm * N + gn0]);
+      } else {
+        #pragma unroll
+        for (int e = 0; e < 8; ++e)
+          v[e] = (gm < M && gn0 + e < N) ? Zg[(size_t)gm * N + gn0 + e]
+                                         : f2bf(0.f);
+      }
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) Zs[(nn8 + e) * LDT + m] = v[e];
+      if (k0 == 0) {
+        // bias-grad partials from LDS after the transpose (the legacy
+        // in-flight accumulation had per-n thread ownership)
+        __syncthreads();
+        int n = tid >> 2;
+        int mm = (tid & 3) * 8;
+        #pragma unroll
+        for (int e = 0; e < 8; ++e)
+          db_part += bf2f(Zs[n * LDT + mm + e]);
+      }
+    } else {
       int n = tid >> 2;
       int mm = (tid & 3) * 8;
       int gn = n0 + n;

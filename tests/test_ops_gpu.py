@@ -87,7 +87,13 @@ def test_grouped_linear_bwd_data(G, M, N, K):
 
 
 @pytest.mark.parametrize(
-    "G,M,K,N", [(2, 200, 50, 38), (1, 4176, 40, 160), (4, 37, 19, 21)]
+    "G,M,K,N", [
+        (2, 200, 50, 38),      # legacy scalar staging (K,N not 8-mult)
+        (1, 4176, 40, 160),    # vectorized A+Z staging (pad8 layout)
+        (4, 37, 19, 21),       # ragged tails, scalar path
+        (2, 333, 56, 48),      # vectorized with ragged M tail
+        (1, 512, 48, 21),      # vecA + legacy Z mix
+    ]
 )
 def test_grouped_linear_wgrad(G, M, K, N):
     require_hip()
