@@ -602,39 +602,43 @@ def test_lstm_scan_row_tiles_bit_equal(monkeypatch):
     (2, 64, 24, 64, 128),   # caps
 ])
 def test_lstm_seq_v4_fused_vs_twostep(G, B, T, H, F):
-    """The v4 fused-xW scan equals the two-step (x-GEMM then scan)
-    path within bf16 accumulation tolerance, and its inference mode
-    (store_aux=False) returns the same hs."""
+    """The v4 fused-xW scan tracks the fp32 oracle (its PRIMARY
+    correctness bar: v4 keeps the x-side contribution in the fp32
+    accumulator, while the two-step path rounds xW to bf16 first, so
+    v4-vs-twostep drift compounds over the recurrence and only a
+    loose sanity bound applies between them). Inference mode
+    (store_aux=False) is bit-identical to the training forward."""
     require_hip()
     x = to_dev_bf16(_rand(G, B, T, F, seed=90) * 0.5)
     Wx = to_dev_bf16(_rand(G, F, 4 * H, seed=91) * 0.2)
     Wh = to_dev_bf16(_rand(G, H, 4 * H, seed=92) * 0.2)
     b = _rand(G, 4 * H, seed=93).cuda() * 0.1
 
-    xW = ops.grouped_linear_fwd(
-        x.reshape(G, B * T, F), Wx, b, "linear"
-    ).view(G, B, T, 4 * H)
-    hs1, cs1, ga1 = ops.lstm_seq_fwd(xW, Wh)
-
     hs4, cs4, ga4 = ops.lstm_seq_fwd_fused(x, Wx, Wh, b, store_aux=True)
-    torch.testing.assert_close(
-        hs4.float(), hs1.float(), rtol=5e-2, atol=2e-2
-    )
-    torch.testing.assert_close(cs4, cs1, rtol=5e-2, atol=3e-2)
-    torch.testing.assert_close(
-        ga4.float(), ga1.float(), rtol=5e-2, atol=2e-2
-    )
     (hs_inf,) = ops.lstm_seq_fwd_fused(x, Wx, Wh, b, store_aux=False)
     assert torch.equal(hs_inf, hs4)
 
-    # CPU fp32 oracle through the same fused entry
-    want = ops.lstm_seq_fwd_fused(
+    # fp32 oracle through the same fused entry (primary)
+    want_hs, want_cs, want_ga = ops.lstm_seq_fwd_fused(
         x.float().cpu(), Wx.float().cpu(), Wh.float().cpu(),
         b.float().cpu(), store_aux=True,
     )
     torch.testing.assert_close(
-        hs4.float().cpu(), want[0], rtol=6e-2, atol=3e-2
+        hs4.float().cpu(), want_hs, rtol=6e-2, atol=3e-2
     )
+    torch.testing.assert_close(
+        cs4.cpu(), want_cs, rtol=6e-2, atol=5e-2
+    )
+    torch.testing.assert_close(
+        ga4.float().cpu(), want_ga, rtol=6e-2, atol=3e-2
+    )
+
+    # two-step path sanity: both approximate the same function
+    xW = ops.grouped_linear_fwd(
+        x.reshape(G, B * T, F), Wx, b, "linear"
+    ).view(G, B, T, 4 * H)
+    hs1, _cs1, _ga1 = ops.lstm_seq_fwd(xW, Wh)
+    assert (hs4.float() - hs1.float()).abs().mean().item() < 0.03
 
 
 def test_lstm_pack_v4_matches_twostep_end_to_end(monkeypatch):
