@@ -899,6 +899,9 @@ class LSTMPack(BasePack):
         self.store.gviews["bd"].copy_(dbd)
         dh_last = ops.grouped_linear_bwd_data(dZ, self.store.cviews["Wd"])
 
+        import os as _os
+
+        v5_on = _os.environ.get("GORDO_LSTM_V4", "1") != "0"
         dSeq = None
         for li in range(len(self.lstm_meta) - 1, -1, -1):
             fin, H, rs = self.lstm_meta[li]
@@ -906,10 +909,19 @@ class LSTMPack(BasePack):
             Wh = self.store.cviews[f"Wh{li}"]
             Wx = self.store.cviews[f"Wx{li}"]
             last_only = dSeq is None
-            dG_flat = ops.lstm_seq_bwd(
-                dh_last if last_only else dSeq,
-                lc["gacts"], lc["cs"], Wh, last_only,
-            ).view(G, B * T, 4 * H)
+            if li > 0 and v5_on and ops.lstm_v4_available(H, fin):
+                # v5: the dSeq GEMM rides inside the reverse scan
+                dG_flat, next_dSeq = ops.lstm_seq_bwd_fused(
+                    dh_last if last_only else dSeq,
+                    lc["gacts"], lc["cs"], Wh, Wx, last_only,
+                )
+                dG_flat = dG_flat.view(G, B * T, 4 * H)
+            else:
+                next_dSeq = None
+                dG_flat = ops.lstm_seq_bwd(
+                    dh_last if last_only else dSeq,
+                    lc["gacts"], lc["cs"], Wh, last_only,
+                ).view(G, B * T, 4 * H)
             hs = lc["hs"]
             dWx, dbl = ops.grouped_linear_wgrad(
                 lc["seq_in"].reshape(G, B * T, fin), dG_flat
@@ -927,8 +939,12 @@ class LSTMPack(BasePack):
             del lc, hs
             prev_dSeq = dSeq
             if li > 0:
-                dSeq = ops.grouped_linear_bwd_data(dG_flat, Wx).view(
-                    G, B, T, fin
+                dSeq = (
+                    next_dSeq
+                    if next_dSeq is not None
+                    else ops.grouped_linear_bwd_data(dG_flat, Wx).view(
+                        G, B, T, fin
+                    )
                 )
             del dG_flat, prev_dSeq
 

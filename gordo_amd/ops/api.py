@@ -225,6 +225,23 @@ def lstm_seq_bwd(dSeq, gacts, cs, Wh, last_only):
     return ext.lstm_seq_bwd_v3(dSeq, gacts, cs, Wh, last_only)
 
 
+def lstm_seq_bwd_fused(dSeq, gacts, cs, Wh, Wx, last_only):
+    """v5 reverse scan with the bwd-data GEMM fused in: returns
+    (dG, dX) where dX = dG @ Wx^T computed per step against an
+    LDS-resident WxT (no dG HBM re-read). Geometry: H<=64, F<=128."""
+    if _on_gpu(gacts):
+        out = _require_hip().lstm_seq_bwd_fused(
+            dSeq, gacts, cs, Wh, Wx, bool(last_only)
+        )
+        return out[0], out[1]
+    dG = lstm_seq_bwd(dSeq, gacts, cs, Wh, last_only)
+    G, B, T, H4 = gacts.shape
+    dX = ref.grouped_linear_bwd_data(
+        dG.view(G, B * T, H4), Wx
+    ).view(G, B, T, Wx.shape[1])
+    return dG, dX
+
+
 def lstm_seq_bwd_v3(dSeq, gacts, cs, Wh, last_only):
     """The pipelined backward scan, directly (for A/B tests)."""
     return _require_hip().lstm_seq_bwd_v3(dSeq, gacts, cs, Wh, last_only)

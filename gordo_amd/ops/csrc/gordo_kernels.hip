@@ -840,6 +840,9 @@ std::vector<torch::Tensor> lstm_seq_fwd_v3(torch::Tensor xW, torch::Tensor Wh);
 torch::Tensor lstm_seq_bwd_v3(torch::Tensor dSeq, torch::Tensor gacts,
                               torch::Tensor cs, torch::Tensor Wh,
                               bool last_only);
+std::vector<torch::Tensor> lstm_seq_bwd_fused(
+    torch::Tensor dSeq, torch::Tensor gacts, torch::Tensor cs,
+    torch::Tensor Wh, torch::Tensor Wx, bool last_only);
 torch::Tensor lstm_seq_bwd(torch::Tensor dSeq, torch::Tensor gacts,
                            torch::Tensor cs, torch::Tensor Wh,
                            bool last_only);
@@ -873,6 +876,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "HBM traffic); store_aux=false skips cs/gacts (inference)");
   mod.def("lstm_seq_bwd", &gordo_lstm::lstm_seq_bwd,
           "fused LSTM backward (BPTT) sequence scan");
+  mod.def("lstm_seq_bwd_fused", &gordo_lstm::lstm_seq_bwd_fused,
+          "v5 reverse scan with dSeq = dG@Wx^T fused in");
   mod.def("grouped_linear_fwd", &grouped_linear_fwd,
           "Y = act(X@W + b) per group (MFMA)");
   mod.def("grouped_linear_bwd_data", &grouped_linear_bwd_data,

@@ -316,6 +316,154 @@ __global__ __launch_bounds__(256) void lstm_seq_bwd_kernel(
 
 
 // ===========================================================================
+// v5 backward scan with FUSED dSeq — the dG consumer GEMM
+// dSeq_t = dG_t @ Wx^T runs inside the reverse scan (WxT LDS-resident,
+// the carry MFMA's column range widened from H to H+F), so the
+// 384 MB/layer dG re-read of the separate bwd-data GEMM disappears
+// (the fleet's BPTT is bandwidth-bound like the forward — see the v4
+// note). Structure is the v1 barriered scan; any H <= 64 (the v2/v3
+// variants keep the non-fused path for the bottom layer, which needs
+// no dSeq).
+// ===========================================================================
+template <int ROWS>
+__global__ __launch_bounds__(256) void lstm_seq_bwd_v5_kernel(
+    const bf16* __restrict__ dSeq, const bf16* __restrict__ gacts,
+    const float* __restrict__ cs, const bf16* __restrict__ Wh,
+    const bf16* __restrict__ Wx,   // [G, F, 4H] native
+    bf16* __restrict__ dG, bf16* __restrict__ dX,  // [G, B, T, F]
+    int B, int T, int H, int F, int ldg, int last_only) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int H4 = 4 * H;
+  bf16* WhN = reinterpret_cast<bf16*>(smem);               // [H][ldg]
+  bf16* WxN = WhN + (size_t)H * ldg;                       // [F][ldg]
+  bf16* dgS = WxN + (size_t)F * ldg;                       // [ROWS][ldg]
+  bf16* dhS = dgS + (size_t)ROWS * ldg;                    // [ROWS][LDK]
+  float* dcS = reinterpret_cast<float*>(dhS + (size_t)ROWS * LDK);
+
+  const int g = blockIdx.x / ((B + ROWS - 1) / ROWS);
+  const int r0 = (blockIdx.x % ((B + ROWS - 1) / ROWS)) * ROWS;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int l15 = lane & 15;
+  const int kslot = lane >> 4;
+
+  const bf16* Whg = Wh + (size_t)g * H * H4;
+  const bf16* Wxg = Wx + (size_t)g * F * H4;
+  const bf16* gag = gacts + ((size_t)g * B + r0) * T * H4;
+  const float* csg = cs + ((size_t)g * B + r0) * T * H;
+  const bf16* dSg = last_only ? dSeq + ((size_t)g * B + r0) * H
+                              : dSeq + ((size_t)g * B + r0) * T * H;
+  bf16* dGg = dG + ((size_t)g * B + r0) * T * H4;
+  bf16* dXg = dX + ((size_t)g * B + r0) * T * F;
+  const int rows_here = min(ROWS, B - r0);
+
+  for (int i = tid; i < H * ldg; i += 256) {
+    int h = i / ldg, n = i % ldg;
+    WhN[i] = (n < H4) ? Whg[(size_t)h * H4 + n] : lf2bf(0.f);
+  }
+  for (int i = tid; i < F * ldg; i += 256) {
+    int f = i / ldg, n = i % ldg;
+    WxN[i] = (n < H4) ? Wxg[(size_t)f * H4 + n] : lf2bf(0.f);
+  }
+  for (int i = tid; i < ROWS * LDK; i += 256) dhS[i] = lf2bf(0.f);
+  for (int i = tid; i < ROWS * H; i += 256) dcS[i] = 0.f;
+  for (int i = tid; i < ROWS * ldg; i += 256) dgS[i] = lf2bf(0.f);
+  __syncthreads();
+
+  constexpr int FM = ROWS / 16;
+  const int HF = H + F;
+
+  for (int t = T - 1; t >= 0; --t) {
+    // ---- fused gate backward (identical to v1) ----
+    for (int e = tid; e < ROWS * H; e += 256) {
+      int row = e / H, hh = e % H;
+      float dh = lbf2f(dhS[row * LDK + hh]);
+      if (row < rows_here) {
+        if (last_only) {
+          if (t == T - 1) dh += lbf2f(dSg[(size_t)row * H + hh]);
+        } else {
+          dh += lbf2f(dSg[((size_t)row * T + t) * H + hh]);
+        }
+      }
+      size_t gbase = ((size_t)row * T + t) * H4;
+      size_t cbase = ((size_t)row * T + t) * H + hh;
+      float i_g = 0.5f, f_g = 0.5f, g_g = 0.f, o_g = 0.5f, cc = 0.f,
+            cp = 0.f;
+      if (row < rows_here) {
+        i_g = lbf2f(gag[gbase + hh]);
+        f_g = lbf2f(gag[gbase + H + hh]);
+        g_g = lbf2f(gag[gbase + 2 * H + hh]);
+        o_g = lbf2f(gag[gbase + 3 * H + hh]);
+        cc = csg[cbase];
+        cp = (t > 0) ? csg[cbase - H] : 0.f;
+      }
+      float tc = fast_tanhf_(cc);
+      float dc = dcS[row * H + hh] + dh * o_g * (1.f - tc * tc);
+      float di = dc * g_g;
+      float df = dc * cp;
+      float dg = dc * i_g;
+      float do_ = dh * tc;
+      dcS[row * H + hh] = dc * f_g;
+      float vi = di * i_g * (1.f - i_g);
+      float vf = df * f_g * (1.f - f_g);
+      float vg = dg * (1.f - g_g * g_g);
+      float vo = do_ * o_g * (1.f - o_g);
+      dgS[row * ldg + hh] = lf2bf(vi);
+      dgS[row * ldg + H + hh] = lf2bf(vf);
+      dgS[row * ldg + 2 * H + hh] = lf2bf(vg);
+      dgS[row * ldg + 3 * H + hh] = lf2bf(vo);
+      if (row < rows_here) {
+        dGg[gbase + hh] = lf2bf(vi);
+        dGg[gbase + H + hh] = lf2bf(vf);
+        dGg[gbase + 2 * H + hh] = lf2bf(vg);
+        dGg[gbase + 3 * H + hh] = lf2bf(vo);
+      }
+    }
+    __syncthreads();
+
+    // ---- [dh_carry | dSeq_t] = dgates @ [Wh ; Wx]^T ----
+    for (int col0 = wid * 64; col0 < HF; col0 += 256) {
+      #pragma unroll
+      for (int fm = 0; fm < FM; ++fm) {
+        f32x4 acc[4] = {};
+        for (int kk = 0; kk < H4; kk += 32) {
+          bf16x8 a = *reinterpret_cast<const bf16x8*>(
+              &dgS[(fm * 16 + l15) * ldg + kk + kslot * 8]);
+          #pragma unroll
+          for (int fn = 0; fn < 4; ++fn) {
+            int col = col0 + fn * 16 + l15;
+            int cc2 = min(col, HF - 1);
+            const bf16* Brow =
+                (cc2 < H) ? &WhN[(size_t)cc2 * ldg]
+                          : &WxN[(size_t)(cc2 - H) * ldg];
+            bf16x8 b = *reinterpret_cast<const bf16x8*>(
+                &Brow[kk + kslot * 8]);
+            acc[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a, b, acc[fn], 0, 0, 0);
+          }
+        }
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+          int col = col0 + fn * 16 + l15;
+          #pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            int row = fm * 16 + kslot * 4 + r;
+            if (col < H) {
+              dhS[row * LDK + col] = lf2bf(acc[fn][r]);
+            } else if (col < HF && row < rows_here) {
+              dXg[((size_t)row * T + t) * F + (col - H)] =
+                  lf2bf(acc[fn][r]);
+            }
+          }
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// ===========================================================================
 // v3 backward scan — v1 structure plus a software-pipelined prefetch of
 // the NEXT (earlier) timestep's gate activations and cell states: those
 // loads depend only on t, so they issue before the MFMA dh-carry phase
@@ -1594,6 +1742,56 @@ std::vector<torch::Tensor> lstm_seq_fwd_v3(torch::Tensor xW,
   return {hs, cs, gacts};
 }
 
+
+std::vector<torch::Tensor> lstm_seq_bwd_fused(
+    torch::Tensor dSeq, torch::Tensor gacts, torch::Tensor cs,
+    torch::Tensor Wh, torch::Tensor Wx, bool last_only) {
+  // returns {dG, dX}: the reverse scan with dSeq_t = dG_t @ Wx^T
+  // fused in (see kernel note). Geometry: H <= 64, F <= 128.
+  TORCH_CHECK(gacts.is_cuda() && gacts.dim() == 4, "gacts must be [G,B,T,4H]");
+  auto dc = dSeq.to(torch::kBFloat16).contiguous();
+  auto gc = gacts.to(torch::kBFloat16).contiguous();
+  auto ccs = cs.to(torch::kFloat32).contiguous();
+  auto Whc = Wh.to(torch::kBFloat16).contiguous();
+  auto Wxc = Wx.to(torch::kBFloat16).contiguous();
+  int G = gc.size(0), B = gc.size(1), T = gc.size(2), H4 = gc.size(3);
+  int H = H4 / 4, F = Wxc.size(1);
+  TORCH_CHECK(H <= 64 && F <= 128, "lstm_seq_bwd_fused geometry");
+  int ldg = pad_ldg(H4);
+  auto dG = torch::empty_like(gc);
+  auto dX = torch::empty({G, B, T, F}, gc.options());
+  int rows = pick_rows(G, B);
+  auto lds_for = [&](int r) {
+    return (size_t)H * ldg * 2 + (size_t)F * ldg * 2 +
+           (size_t)r * ldg * 2 + (size_t)r * LDK * 2 + (size_t)r * H * 4;
+  };
+  while (rows > 16 && lds_for(rows) > 160 * 1024) rows /= 2;
+  size_t lds = lds_for(rows);
+  TORCH_CHECK(lds <= 160 * 1024, "LDS budget exceeded (v5 bwd)");
+  int blocks = G * ((B + rows - 1) / rows);
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  auto dp = (const bf16*)dc.data_ptr();
+  auto gp = (const bf16*)gc.data_ptr();
+  auto cp = ccs.data_ptr<float>();
+  auto whp = (const bf16*)Whc.data_ptr();
+  auto wxp = (const bf16*)Wxc.data_ptr();
+  auto dgp = (bf16*)dG.data_ptr();
+  auto dxp = (bf16*)dX.data_ptr();
+  int lo = last_only ? 1 : 0;
+  if (rows == 64)
+    hipLaunchKernelGGL(lstm_seq_bwd_v5_kernel<64>, dim3(blocks), dim3(256),
+                       lds, stream, dp, gp, cp, whp, wxp, dgp, dxp, B, T,
+                       H, F, ldg, lo);
+  else if (rows == 32)
+    hipLaunchKernelGGL(lstm_seq_bwd_v5_kernel<32>, dim3(blocks), dim3(256),
+                       lds, stream, dp, gp, cp, whp, wxp, dgp, dxp, B, T,
+                       H, F, ldg, lo);
+  else
+    hipLaunchKernelGGL(lstm_seq_bwd_v5_kernel<16>, dim3(blocks), dim3(256),
+                       lds, stream, dp, gp, cp, whp, wxp, dgp, dxp, B, T,
+                       H, F, ldg, lo);
+  return {dG, dX};
+}
 
 torch::Tensor lstm_seq_bwd_v3(torch::Tensor dSeq, torch::Tensor gacts,
                               torch::Tensor cs, torch::Tensor Wh,
