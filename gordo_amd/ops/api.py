@@ -184,7 +184,7 @@ def lstm_seq_fwd_fused(xseq, Wx, Wh, bias, store_aux=True):
     G, B, T, F = xseq.shape
     H = Wh.shape[1]
     gates_all = ref.grouped_linear_fwd(
-        xseq.reshape(G, B * T, F), Wx, bias, "linear"
+        xseq.reshape(G, B * T, F), Wx, bias, ACT_LINEAR
     ).view(G, B, T, 4 * H)
     h = torch.zeros(G, B, H, dtype=xseq.dtype)
     c = torch.zeros(G, B, H, dtype=torch.float32)
