@@ -153,11 +153,17 @@ def lstm_seq_fwd(xW, Wh):
     import os as _os
 
     ext = _require_hip()
+    G, B = xW.shape[0], xW.shape[1]
     H = xW.shape[-1] // 4
+    # v2 (barrier-free, 64-row per-wave blocks) only when its grid
+    # fills the chip: at fleet G its G*ceil(B/64) blocks underfill 256
+    # CUs and the row-tiled pipelined v3 wins (call-10 rocprof: v2 bwd
+    # at 60 WGs averaged 3x the v4 fwd per launch)
+    v2_fills = H % 16 == 0 and G * ((B + 63) // 64) >= 256
     if (
         _os.environ.get("GORDO_LSTM_V1") == "1"
         or H > 64            # big-H dispatch lives in the v1 entry
-        or H % 16 == 0       # v2 barrier-free path (v1 entry dispatches)
+        or v2_fills
     ):
         return ext.lstm_seq_fwd(xW, Wh)
     return ext.lstm_seq_fwd_v3(xW, Wh)
@@ -215,11 +221,13 @@ def lstm_seq_bwd(dSeq, gacts, cs, Wh, last_only):
     import os as _os
 
     ext = _require_hip()
+    G, B = gacts.shape[0], gacts.shape[1]
     H = gacts.shape[-1] // 4
+    v2_fills = H % 16 == 0 and G * ((B + 63) // 64) >= 256
     if (
         _os.environ.get("GORDO_LSTM_V1") == "1"
         or H > 64
-        or H % 16 == 0
+        or v2_fills
     ):
         return ext.lstm_seq_bwd(dSeq, gacts, cs, Wh, last_only)
     return ext.lstm_seq_bwd_v3(dSeq, gacts, cs, Wh, last_only)
