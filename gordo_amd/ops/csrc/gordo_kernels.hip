@@ -275,15 +275,16 @@ __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
   f32x4 acc[2][2] = {};
   float db_part = 0.f;
 
-  // With pad8 layouts (K/N multiples of 8) the transpose staging flips
-  // to ROW-CONTIGUOUS 16-byte global loads + scalar LDS-transposed
-  // writes: the legacy path's per-element column gather costs 8
-  // separate L2 requests per thread (PMC: wgrad 56% wave-wait + 31%
-  // issue-stall). LDS write conflicts are bounded 2-way.
-  const bool vecA = ((K & 7) == 0) && ((reinterpret_cast<uintptr_t>(Ag)
-                                        & 15) == 0);
-  const bool vecZ = ((N & 7) == 0) && ((reinterpret_cast<uintptr_t>(Zg)
-                                        & 15) == 0);
+  // Vectorized transpose staging (row-contiguous 16-byte loads +
+  // scalar LDS-transposed writes) MEASURED SLOWER than the legacy
+  // column gather at the fleet shape (0.288 vs 0.254 ms at
+  // M=36864 K=56 N=192 G=31 — gpurun r2_call10): the column gather's
+  // lane pattern already coalesces into 4x32B segments per
+  // instruction, while 8 ds_write_b16 per thread cost more LDS issue
+  // than the loads saved. Kept compiled-out for future re-tuning.
+  constexpr bool vecA = false;
+  constexpr bool vecZ = false;
+  (void)0;
   for (int m0 = m_begin; m0 < m_end; m0 += BK) {
     // stage A[m0..+32][k0..+64] transposed into As[k][m]
     if (vecA) {
