@@ -923,12 +923,11 @@ class LSTMPack(BasePack):
                     lc["gacts"], lc["cs"], Wh, last_only,
                 ).view(G, B * T, 4 * H)
             hs = lc["hs"]
-            dWx, dbl = ops.grouped_linear_wgrad(
-                lc["seq_in"].reshape(G, B * T, fin), dG_flat
+            # combined pass: dZ (dG) staged once for both weight grads;
+            # dWh reads h_{t-1} via in-kernel shifted addressing
+            dWx, dWh, dbl = ops.grouped_wgrad_xh(
+                lc["seq_in"].reshape(G, B * T, fin), hs, dG_flat, T
             )
-            # dWh reads h_{t-1} via in-kernel shifted addressing — no
-            # h_prev_all concat (was a full [G,B,T,H] copy per layer)
-            dWh, _ = ops.grouped_linear_wgrad_hprev(hs, dG_flat, T)
             self.store.gviews[f"Wx{li}"].copy_(dWx)
             self.store.gviews[f"Wh{li}"].copy_(dWh)
             self.store.gviews[f"bl{li}"].copy_(dbl)

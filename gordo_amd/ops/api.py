@@ -112,6 +112,18 @@ def window_gather(X, idx, T: int):
     ).view(G, B, T, F)
 
 
+def grouped_wgrad_xh(seq, hs, dZ, T: int):
+    """Combined recurrent weight-grad: (dWx, dWh, db) in one kernel
+    pass — dZ is staged once for both GEMMs (it is half of each
+    separate call's global traffic)."""
+    if _on_gpu(seq):
+        out = _require_hip().grouped_wgrad_xh(seq, hs, dZ, int(T))
+        return out[0], out[1], out[2]
+    dWx, db = ref.grouped_linear_wgrad(seq, dZ)
+    dWh, _ = grouped_linear_wgrad_hprev(hs, dZ, T)
+    return dWx, dWh, db
+
+
 def grouped_gemm_acc(A, B, C):
     if _on_gpu(A):
         return _require_hip().grouped_gemm_acc(A, B, C)

@@ -702,3 +702,20 @@ def test_lstm_seq_bwd_v5_fused_dseq(G, B, T, H, F, last_only):
     torch.testing.assert_close(
         dX5.float(), dX_ref.float(), rtol=5e-2, atol=2e-2
     )
+
+
+@pytest.mark.parametrize("G,B,T,F,H,N4", [(3, 16, 12, 50, 42, 168),
+                                          (1, 9, 7, 8, 16, 64)])
+def test_grouped_wgrad_xh_matches_separate(G, B, T, F, H, N4):
+    """The combined {dWx, dWh, db} pass equals the two separate wgrad
+    calls (one dZ staging instead of two)."""
+    require_hip()
+    seq = to_dev_bf16(_rand(G, B * T, F, seed=64))
+    hs = to_dev_bf16(_rand(G, B, T, H, seed=65))
+    dG = to_dev_bf16(_rand(G, B * T, N4, seed=66))
+    want_Wx, want_b = ops.grouped_linear_wgrad(seq, dG)
+    want_Wh, _ = ops.grouped_linear_wgrad_hprev(hs, dG, T)
+    got_Wx, got_Wh, got_b = ops.grouped_wgrad_xh(seq, hs, dG, T)
+    torch.testing.assert_close(got_Wx, want_Wx, rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(got_Wh, want_Wh, rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(got_b, want_b, rtol=1e-5, atol=1e-5)
