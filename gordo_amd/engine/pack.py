@@ -1002,10 +1002,9 @@ class LSTMPack(BasePack):
                 dh_carry = ops.grouped_linear_bwd_data(dgates, Wh)
             # batched weight grads over all (B, T) rows
             dG_flat = dG.view(G, B * T, 4 * H)
-            dWx, dbl = ops.grouped_linear_wgrad(
-                lc["seq_in"].reshape(G, B * T, fin), dG_flat
+            dWx, dWh, dbl = ops.grouped_wgrad_xh(
+                lc["seq_in"].reshape(G, B * T, fin), hs, dG_flat, T
             )
-            dWh, _ = ops.grouped_linear_wgrad_hprev(hs, dG_flat, T)
             self.store.gviews[f"Wx{li}"].copy_(dWx)
             self.store.gviews[f"Wh{li}"].copy_(dWh)
             self.store.gviews[f"bl{li}"].copy_(dbl)
