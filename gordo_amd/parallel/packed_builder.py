@@ -132,7 +132,9 @@ class PackedFleetBuilder:
         self.replace_cache = replace_cache
         # async model-save pool: serialization of a finished group
         # overlaps the next group's GPU fits (joined in build_all)
-        self._save_pool = concurrent.futures.ThreadPoolExecutor(8)
+        self._save_pool = concurrent.futures.ThreadPoolExecutor(
+            max(8, min(16, (os.cpu_count() or 8)))
+        )
         self._save_futures: List[Tuple[MachinePlan, Any]] = []
         # wall-clock phase ledger for the build-step budget table
         # (BASELINE.md): sequential wall segments; overlapped work
@@ -877,13 +879,11 @@ class PackedFleetBuilder:
                         )
                     if not hasattr(det, "aggregate_thresholds_per_fold_"):
                         det.aggregate_thresholds_per_fold_ = {}
-                        det.feature_thresholds_per_fold_ = pd.DataFrame()
+                        det._fold_tag_rows = []
                         det.smooth_aggregate_thresholds_per_fold_ = {}
-                        det.smooth_feature_thresholds_per_fold_ = pd.DataFrame()
+                        det._fold_smooth_tag_rows = []
                     det.aggregate_thresholds_per_fold_[f"fold-{fold_i}"] = agg_thr
-                    det.feature_thresholds_per_fold_ = pd.concat(
-                        [det.feature_thresholds_per_fold_, tag_thr.to_frame().T]
-                    )
+                    det._fold_tag_rows.append(tag_thr)
                     det.aggregate_threshold_ = agg_thr
                     det.feature_thresholds_ = tag_thr
                     # smoothing window is per machine, not per group
@@ -904,12 +904,7 @@ class PackedFleetBuilder:
                         det.smooth_aggregate_thresholds_per_fold_[
                             f"fold-{fold_i}"
                         ] = s_agg
-                        det.smooth_feature_thresholds_per_fold_ = pd.concat(
-                            [
-                                det.smooth_feature_thresholds_per_fold_,
-                                s_tag.to_frame().T,
-                            ]
-                        )
+                        det._fold_smooth_tag_rows.append(s_tag)
                         det.smooth_aggregate_threshold_ = s_agg
                         det.smooth_feature_thresholds_ = s_tag
             logger.info(
@@ -917,6 +912,19 @@ class PackedFleetBuilder:
                 fold_i, t_fit, t_pred, time.time() - t_s0,
             )
 
+        # one concat per machine instead of one per (machine, fold):
+        # growing a DataFrame row-by-row re-allocates every time
+        for p in group:
+            det = p.detector
+            rows = getattr(det, "_fold_tag_rows", None)
+            if rows is not None:
+                det.feature_thresholds_per_fold_ = pd.DataFrame(rows)
+                del det._fold_tag_rows
+                srows = det._fold_smooth_tag_rows
+                det.smooth_feature_thresholds_per_fold_ = (
+                    pd.DataFrame(srows) if srows else pd.DataFrame()
+                )
+                del det._fold_smooth_tag_rows
         self._phase("cv_score_thresholds", time.time() - t_seg)
 
         # finalize KFCV thresholds over the reassembled series
