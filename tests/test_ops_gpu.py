@@ -633,12 +633,17 @@ def test_lstm_seq_v4_fused_vs_twostep(G, B, T, H, F):
         ga4.float().cpu(), want_ga, rtol=6e-2, atol=3e-2
     )
 
-    # two-step path sanity: both approximate the same function
+    # two-step path sanity: v4 (fp32 x-accumulation) must track the
+    # fp32 oracle at least as well as the two-step path (which rounds
+    # xW to bf16 first) — an absolute bound between the two bf16 paths
+    # is brittle at large K
     xW = ops.grouped_linear_fwd(
         x.reshape(G, B * T, F), Wx, b, "linear"
     ).view(G, B, T, 4 * H)
     hs1, _cs1, _ga1 = ops.lstm_seq_fwd(xW, Wh)
-    assert (hs4.float() - hs1.float()).abs().mean().item() < 0.03
+    err4 = (hs4.float().cpu() - want_hs).abs().mean().item()
+    err1 = (hs1.float().cpu() - want_hs).abs().mean().item()
+    assert err4 <= err1 * 1.5 + 0.01, (err4, err1)
 
 
 def test_lstm_pack_v4_matches_twostep_end_to_end(monkeypatch):
