@@ -374,29 +374,58 @@ __global__ __launch_bounds__(256) void lstm_seq_bwd_v5_kernel(
   constexpr int FM = ROWS / 16;
   const int HF = H + F;
 
+  // v3-style register prefetch of the next (earlier) step's gate
+  // activations + cell states: their loads depend only on t, so they
+  // issue before the MFMA carry phase and the HBM latency hides under
+  // the matrix work (buffer capped at 8 slots — v3 note).
+  constexpr int MAXE = (ROWS * 64 + 255) / 256;
+  constexpr int PF = MAXE > 8 ? 8 : MAXE;
+  unsigned int p_if[PF];
+  unsigned int p_go[PF];
+  float p_cc[PF];
+  float p_cp[PF];
+
+  auto prefetch_step = [&](int t) {
+    #pragma unroll
+    for (int s2 = 0; s2 < PF; ++s2) {
+      int e = tid + s2 * 256;
+      unsigned int v_if = 0x3F003F00u;
+      unsigned int v_go = 0x3F000000u;
+      float cc = 0.f, cp = 0.f;
+      if (e < ROWS * H) {
+        int row = e / H, hh = e % H;
+        if (row < rows_here) {
+          size_t gbase = ((size_t)row * T + t) * H4;
+          v_if = (unsigned int)bf16_bits(gag[gbase + hh]) |
+                 ((unsigned int)bf16_bits(gag[gbase + H + hh]) << 16);
+          v_go = (unsigned int)bf16_bits(gag[gbase + 2 * H + hh]) |
+                 ((unsigned int)bf16_bits(gag[gbase + 3 * H + hh]) << 16);
+          size_t cbase = ((size_t)row * T + t) * H + hh;
+          cc = csg[cbase];
+          cp = (t > 0) ? csg[cbase - H] : 0.f;
+        }
+      }
+      p_if[s2] = v_if;
+      p_go[s2] = v_go;
+      p_cc[s2] = cc;
+      p_cp[s2] = cp;
+    }
+  };
+
+  prefetch_step(T - 1);
+
   for (int t = T - 1; t >= 0; --t) {
-    // ---- fused gate backward (identical to v1) ----
-    for (int e = tid; e < ROWS * H; e += 256) {
+    // ---- fused gate backward (prefetched slots + direct tail) ----
+    auto gate_bwd = [&](int e, int t_, float i_g, float f_g, float g_g,
+                        float o_g, float cc, float cp) {
       int row = e / H, hh = e % H;
       float dh = lbf2f(dhS[row * LDK + hh]);
       if (row < rows_here) {
         if (last_only) {
-          if (t == T - 1) dh += lbf2f(dSg[(size_t)row * H + hh]);
+          if (t_ == T - 1) dh += lbf2f(dSg[(size_t)row * H + hh]);
         } else {
-          dh += lbf2f(dSg[((size_t)row * T + t) * H + hh]);
+          dh += lbf2f(dSg[((size_t)row * T + t_) * H + hh]);
         }
-      }
-      size_t gbase = ((size_t)row * T + t) * H4;
-      size_t cbase = ((size_t)row * T + t) * H + hh;
-      float i_g = 0.5f, f_g = 0.5f, g_g = 0.f, o_g = 0.5f, cc = 0.f,
-            cp = 0.f;
-      if (row < rows_here) {
-        i_g = lbf2f(gag[gbase + hh]);
-        f_g = lbf2f(gag[gbase + H + hh]);
-        g_g = lbf2f(gag[gbase + 2 * H + hh]);
-        o_g = lbf2f(gag[gbase + 3 * H + hh]);
-        cc = csg[cbase];
-        cp = (t > 0) ? csg[cbase - H] : 0.f;
       }
       float tc = fast_tanhf_(cc);
       float dc = dcS[row * H + hh] + dh * o_g * (1.f - tc * tc);
@@ -414,12 +443,41 @@ __global__ __launch_bounds__(256) void lstm_seq_bwd_v5_kernel(
       dgS[row * ldg + 2 * H + hh] = lf2bf(vg);
       dgS[row * ldg + 3 * H + hh] = lf2bf(vo);
       if (row < rows_here) {
+        size_t gbase = ((size_t)row * T + t_) * H4;
         dGg[gbase + hh] = lf2bf(vi);
         dGg[gbase + H + hh] = lf2bf(vf);
         dGg[gbase + 2 * H + hh] = lf2bf(vg);
         dGg[gbase + 3 * H + hh] = lf2bf(vo);
       }
+    };
+    #pragma unroll
+    for (int s2 = 0; s2 < PF; ++s2) {
+      int e = tid + s2 * 256;
+      if (e >= ROWS * H) break;
+      gate_bwd(e, t,
+               lbf2f(bits_bf16((unsigned short)p_if[s2])),
+               lbf2f(bits_bf16((unsigned short)(p_if[s2] >> 16))),
+               lbf2f(bits_bf16((unsigned short)p_go[s2])),
+               lbf2f(bits_bf16((unsigned short)(p_go[s2] >> 16))),
+               p_cc[s2], p_cp[s2]);
     }
+    for (int e = tid + PF * 256; e < ROWS * H; e += 256) {
+      int row = e / H, hh = e % H;
+      float i_g = 0.5f, f_g = 0.5f, g_g = 0.f, o_g = 0.5f, cc = 0.f,
+            cp = 0.f;
+      if (row < rows_here) {
+        size_t gbase = ((size_t)row * T + t) * H4;
+        i_g = lbf2f(gag[gbase + hh]);
+        f_g = lbf2f(gag[gbase + H + hh]);
+        g_g = lbf2f(gag[gbase + 2 * H + hh]);
+        o_g = lbf2f(gag[gbase + 3 * H + hh]);
+        size_t cbase = ((size_t)row * T + t) * H + hh;
+        cc = csg[cbase];
+        cp = (t > 0) ? csg[cbase - H] : 0.f;
+      }
+      gate_bwd(e, t, i_g, f_g, g_g, o_g, cc, cp);
+    }
+    if (t > 0) prefetch_step(t - 1);
     __syncthreads();
 
     // ---- [dh_carry | dSeq_t] = dgates @ [Wh ; Wx]^T ----
