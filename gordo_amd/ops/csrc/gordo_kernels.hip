@@ -249,20 +249,14 @@ __global__ __launch_bounds__(256) void grouped_gemm_kernel(
 // sequence) and [K1, K) read A2 (stride K-K1, the h sequence with the
 // t-1 shift), producing [dWx ; dWh] in ONE pass so dZ is staged once
 // instead of twice (dZ staging is half of each call's global traffic).
-// A2/K1: optional SECOND A operand for the recurrent layer's combined
-// weight-grad — K columns [0, K1) read A (stride K1, the layer input
-// sequence) and [K1, K) read A2 (stride K-K1, the h sequence with the
-// t-1 shift), producing [dWx ; dWh] in ONE pass so dZ is staged once
-// instead of twice. The staging is SOFTWARE-PIPELINED (guide T14):
-// the next m-chunk's global loads issue BEFORE the MFMAs (their HBM
-// latency hides under the matrix work), the LDS writes land after,
-// double-buffered with ONE barrier per chunk.
 __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ dZ,
     float* __restrict__ dW, float* __restrict__ db, int M, int N, int K,
     int kt, int nt, int nblocks, int mchunk, int tshiftT,
     const bf16* __restrict__ A2, int K1) {
-  __shared__ bf16 sm[2][2 * BM * LDT];
+  __shared__ bf16 sm[2 * BM * LDT];
+  bf16* As = sm;             // At: [k 64][m 32+pad]
+  bf16* Zs = sm + BM * LDT;  // Zt: [n 64][m 32+pad]
 
   // split-M: blockIdx.y selects an M-chunk; partials atomicAdd into the
   // fp32 dW (the reduction dim M is B*T ~ 37k for LSTM wgrads — without
@@ -289,55 +283,108 @@ __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
   f32x4 acc[2][2] = {};
   float db_part = 0.f;
 
-  // thread's staging coordinates (transpose gather: the lane pattern
-  // coalesces into 4x32B segments per load instruction)
-  const int kA = tid >> 2;            // 0..63
-  const int mmA = (tid & 3) * 8;      // 0..24
-  const int gk = k0 + kA;
-  const bool part2 = A2g != nullptr && gk >= K1;
-  const int nZ = tid >> 2;
-  const int gn = n0 + nZ;
-
-  bf16 vA[8], vZ[8];
-  auto load_chunk = [&](int m0) {
-    #pragma unroll
-    for (int e = 0; e < 8; ++e) {
-      int gm = m0 + mmA + e;
-      bool ok = (gk < K && gm < M);
+  // Vectorized transpose staging (row-contiguous 16-byte loads +
+  // scalar LDS-transposed writes) MEASURED SLOWER than the legacy
+  // column gather at the fleet shape (0.288 vs 0.254 ms at
+  // M=36864 K=56 N=192 G=31 — gpurun r2_call10): the column gather's
+  // lane pattern already coalesces into 4x32B segments per
+  // instruction, while 8 ds_write_b16 per thread cost more LDS issue
+  // than the loads saved. Kept compiled-out for future re-tuning.
+  constexpr bool vecA = false;
+  constexpr bool vecZ = false;
+  (void)0;
+  for (int m0 = m_begin; m0 < m_end; m0 += BK) {
+    // stage A[m0..+32][k0..+64] transposed into As[k][m]
+    if (vecA) {
+      int m = tid >> 3;           // 0..31
+      int kk8 = (tid & 7) * 8;    // 0..56
+      int gm = m0 + m;
+      int gk0 = k0 + kk8;
+      bool ok = gm < M;
       size_t row = (size_t)gm;
-      if (tshiftT > 0 && (part2 || A2g == nullptr)) {
+      if (tshiftT > 0) {
         ok = ok && (gm % tshiftT) > 0;  // t == 0 -> h_prev is zero
         row = (size_t)gm - 1;
       }
-      vA[e] = !ok ? f2bf(0.f)
-              : part2 ? A2g[row * K2 + (gk - K1)]
-                      : Ag[row * K1 + gk];
+      bf16 v[8];
+      if (ok && gk0 + 8 <= K) {
+        *reinterpret_cast<bf16x8*>(v) =
+            *reinterpret_cast<const bf16x8*>(&Ag[row * K + gk0]);
+      } else {
+        #pragma unroll
+        for (int e = 0; e < 8; ++e)
+          v[e] = (ok && gk0 + e < K) ? Ag[row * K + gk0 + e] : f2bf(0.f);
+      }
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) As[(kk8 + e) * LDT + m] = v[e];
+    } else {
+      int k = tid >> 2;           // 0..63
+      int mm = (tid & 3) * 8;     // 0..24
+      int gk = k0 + k;
+      const bool part2 = A2g != nullptr && gk >= K1;
+      bf16 v[8];
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        int gm = m0 + mm + e;
+        bool ok = (gk < K && gm < M);
+        size_t row = (size_t)gm;
+        // the t-1 shift applies to the whole A when single-operand
+        // (the hprev entry), or to the A2 half in combined mode
+        if (tshiftT > 0 && (part2 || A2g == nullptr)) {
+          ok = ok && (gm % tshiftT) > 0;  // t == 0 -> h_prev is zero
+          row = (size_t)gm - 1;
+        }
+        v[e] = !ok ? f2bf(0.f)
+               : part2 ? A2g[row * K2 + (gk - K1)]
+                       : Ag[row * K1 + gk];
+      }
+      lds_store8(&As[k * LDT + mm], v);
     }
-    #pragma unroll
-    for (int e = 0; e < 8; ++e) {
-      int gm = m0 + mmA + e;
-      vZ[e] = (gn < N && gm < M) ? Zg[(size_t)gm * N + gn] : f2bf(0.f);
-      if (k0 == 0) db_part += bf2f(vZ[e]);
+    // stage dZ[m0..+32][n0..+64] transposed into Zs[n][m]; k-tile-0
+    // blocks fold the bias-grad column sum into the same pass (what a
+    // separate colsum kernel did with a latency-bound column walk).
+    if (vecZ) {
+      int m = tid >> 3;
+      int nn8 = (tid & 7) * 8;
+      int gm = m0 + m;
+      int gn0 = n0 + nn8;
+      bf16 v[8];
+      if (gm < M && gn0 + 8 <= N) {
+        *reinterpret_cast<bf16x8*>(v) =
+            *reinterpret_cast<const bf16x8*>(&Zg[(size_t)gm * N + gn0]);
+      } else {
+        #pragma unroll
+        for (int e = 0; e < 8; ++e)
+          v[e] = (gm < M && gn0 + e < N) ? Zg[(size_t)gm * N + gn0 + e]
+                                         : f2bf(0.f);
+      }
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) Zs[(nn8 + e) * LDT + m] = v[e];
+      if (k0 == 0) {
+        // bias-grad partials from LDS after the transpose (the legacy
+        // in-flight accumulation had per-n thread ownership)
+        __syncthreads();
+        int n = tid >> 2;
+        int mm = (tid & 3) * 8;
+        #pragma unroll
+        for (int e = 0; e < 8; ++e)
+          db_part += bf2f(Zs[n * LDT + mm + e]);
+      }
+    } else {
+      int n = tid >> 2;
+      int mm = (tid & 3) * 8;
+      int gn = n0 + n;
+      bf16 v[8];
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        int gm = m0 + mm + e;
+        v[e] = (gn < N && gm < M) ? Zg[(size_t)gm * N + gn] : f2bf(0.f);
+        if (k0 == 0) db_part += bf2f(v[e]);
+      }
+      lds_store8(&Zs[n * LDT + mm], v);
     }
-  };
-  auto write_chunk = [&](int buf) {
-    lds_store8(&sm[buf][kA * LDT + mmA], vA);
-    lds_store8(&sm[buf][BM * LDT + nZ * LDT + mmA], vZ);
-  };
+    __syncthreads();
 
-  int cur = 0;
-  if (m_begin < m_end) {
-    load_chunk(m_begin);
-    write_chunk(0);
-  }
-  __syncthreads();
-
-  for (int m0 = m_begin; m0 < m_end; m0 += BK) {
-    const bool has_next = (m0 + BK) < m_end;
-    if (has_next) load_chunk(m0 + BK);  // issue early (latency hides)
-
-    const bf16* As = sm[cur];
-    const bf16* Zs = sm[cur] + BM * LDT;
     #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
       bf16x8 a = *reinterpret_cast<const bf16x8*>(
@@ -350,12 +397,11 @@ __global__ __launch_bounds__(256) void grouped_wgrad_kernel(
             a, b, acc[fm][fn], 0, 0, 0);
       }
     }
-    if (has_next) write_chunk(cur ^ 1);  // other buffer: no read conflict
     __syncthreads();
-    cur ^= 1;
   }
 
   if (k0 == 0) {
+    int gn = n0 + (tid >> 2);
     if (gn < N) atomicAdd(&db[(size_t)g * N + gn], db_part);
   }
   const bool single_chunk = gridDim.y == 1;
