@@ -428,8 +428,7 @@ class PackedFleetBuilder:
                 cv_mode == "full_build"
                 and self.device != "cpu"
                 and torch.cuda.is_available()
-                and len(group) * (n_folds + 1)
-                <= self.MAX_CONCURRENT_FOLD_MODELS
+                and len(group) * (n_folds + 1) <= self._fold_cap(spec)
             ):
                 import threading
 
@@ -522,7 +521,15 @@ class PackedFleetBuilder:
     # 4 concurrent fits x 128-model packs ~ 80 GB of transient BPTT
     # caches at lookback 144 — comfortable in 288 GB (fold packs run
     # without graph capture, which halves retention)
-    MAX_CONCURRENT_FOLD_MODELS = 640
+    MAX_CONCURRENT_FOLD_MODELS = 640       # LSTM packs (BPTT caches)
+    MAX_CONCURRENT_FOLD_MODELS_DENSE = 8192  # dense models are ~100x lighter
+
+    def _fold_cap(self, spec) -> int:
+        return (
+            self.MAX_CONCURRENT_FOLD_MODELS
+            if spec.model_type == "lstm"
+            else self.MAX_CONCURRENT_FOLD_MODELS_DENSE
+        )
 
     @staticmethod
     def _n_cv_folds(evaluation: Dict[str, Any]) -> int:
@@ -667,7 +674,7 @@ class PackedFleetBuilder:
         concurrent = (
             self.device != "cpu"
             and torch.cuda.is_available()
-            and len(group) * len(folds) <= self.MAX_CONCURRENT_FOLD_MODELS
+            and len(group) * len(folds) <= self._fold_cap(spec)
         )
         out: Dict[int, Any] = {}
         if concurrent:
