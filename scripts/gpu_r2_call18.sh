@@ -1,0 +1,14 @@
+#!/bin/bash
+set -x
+cd "$GRAFT_REPO_ROOT" || cd /root/repo
+mkdir -p gpurun_out
+exec > >(tee gpurun_out/r2_call18.log) 2>&1
+for cap in 64 128 64; do
+  GORDO_MAX_PACK_LSTM=$cap timeout 1200 python bench.py --gpus 1 --steps 1 \
+    --warmup 1 --machines-per-gpu 1000 2>&1 | tail -1 | python -c "
+import json,sys
+d = json.loads(sys.stdin.read())
+print(f'cap=$cap: {d[\"value\"]:.0f} machines/hour ({d[\"ms_per_step\"]/1000:.1f} s/step)')"
+done
+# shard check at the new default cap (62 LSTM -> one group either way)
+timeout 900 python bench.py --gpus 1 --steps 2 --warmup 1 2>&1 | tail -1
