@@ -260,7 +260,7 @@ class PackedFleetBuilder:
     # model per batch (plus hipGraph retention); 128 LSTM models per
     # pack keeps a 288 GB GPU comfortable at lookback 144 while still
     # amortizing kernel launches. Dense models are ~100x lighter.
-    MAX_PACK_LSTM = int(os.environ.get("GORDO_MAX_PACK_LSTM", 64))
+    MAX_PACK_LSTM = int(os.environ.get("GORDO_MAX_PACK_LSTM", 128))
     MAX_PACK_DENSE = int(os.environ.get("GORDO_MAX_PACK_DENSE", 1024))
 
     def _group(self, plans: List[MachinePlan]) -> List[List[MachinePlan]]:
