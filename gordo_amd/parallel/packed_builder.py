@@ -409,7 +409,7 @@ class PackedFleetBuilder:
         init_snapshot = pack.store.p32.clone()
         self._phase("pack_init", time.time() - t_seg)
         if cached_init is None:
-            if len(_INIT_CACHE) >= 8:
+            if len(_INIT_CACHE) >= 32:  # fleet configs can have 9+ groups
                 _INIT_CACHE.pop(next(iter(_INIT_CACHE)))
             _INIT_CACHE[init_key] = init_snapshot.cpu()
         if cv_mode in ("cross_val_only", "full_build"):
