@@ -134,7 +134,12 @@ def _decode_request_fast(payload: bytes):
     decode_request): one pass over the bytes straight into per-column
     numpy arrays, skipping json.loads + python dict assembly. Returns
     {"X": DataFrame, "y": DataFrame|None} or None for any payload the
-    strict parser refuses (the stdlib path handles those)."""
+    strict parser refuses (the stdlib path handles those).
+
+    Note: JSON objects with DUPLICATE keys are undefined here (stdlib
+    keeps the last occurrence; the strict parser would see both and
+    fall back on the resulting key-set mismatch) — neither this
+    framework's client nor gordo-client can emit duplicates."""
     if _gordo_fastjson is None or not hasattr(
         _gordo_fastjson, "decode_request"
     ):
