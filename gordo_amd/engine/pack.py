@@ -199,6 +199,17 @@ class BasePack:
     def _to_compute(self, t: torch.Tensor) -> torch.Tensor:
         return t.to(device=self.device, dtype=self.compute_dtype)
 
+    def _pad_io(self, Xb, Tb):
+        """Defensive padding for DIRECT train_batch/eval_batch calls
+        (fit() pads the whole series up front; callers like profiling
+        harnesses or graph capture may pass logical-width tensors)."""
+        if self.pad8:
+            if Xb.shape[-1] == self._logical_in:
+                Xb = self._pad_features(Xb, self._logical_in)
+            if Tb is not None and Tb.shape[-1] == self._logical_out:
+                Tb = self._pad_features(Tb, self._logical_out)
+        return Xb, Tb
+
     def _pad_features(self, t: torch.Tensor, logical_f: int) -> torch.Tensor:
         """Zero-pad the last (feature) dim to the physical width."""
         pf = self._pad(logical_f)
@@ -617,11 +628,13 @@ class DensePack(BasePack):
 
     def eval_batch(self, Xb, Tb) -> torch.Tensor:
         # pad diffs are exactly 0; divide by the REAL element count
+        Xb, Tb = self._pad_io(Xb, Tb)
         out = self.forward(Xb)[-1].float()
         n = Xb.shape[1] * self._logical_out
         return ((out - Tb.float()) ** 2).sum(dim=(1, 2)) / n
 
     def train_batch(self, Xb, Tb) -> torch.Tensor:
+        Xb, Tb = self._pad_io(Xb, Tb)
         acts = self.forward(Xb)
         loss, dA = ops.mse_bwd(
             acts[-1], Tb.to(acts[-1].dtype),
@@ -879,6 +892,7 @@ class LSTMPack(BasePack):
         return y[..., : self._logical_out]
 
     def eval_batch(self, Xw, Tb) -> torch.Tensor:
+        Xw, Tb = self._pad_io(Xw, Tb)
         y = self.predict_windows(Xw).float()
         return (
             (y - Tb.float()[..., : self._logical_out]) ** 2
@@ -952,6 +966,7 @@ class LSTMPack(BasePack):
         return loss
 
     def train_batch(self, Xw, Tb) -> torch.Tensor:
+        Xw, Tb = self._pad_io(Xw, Tb)
         if self._use_fused(Xw.shape[1]):
             return self._train_batch_fused(Xw, Tb)
         G, B, T, _ = Xw.shape

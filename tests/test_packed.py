@@ -595,3 +595,27 @@ def test_pad8_lstm_pack_matches_unpadded(monkeypatch):
     # forget-gate pad bias stayed zero; real forget bias trained from 1
     bl = padded.store.views["bl0"][0].numpy()
     assert (bl[8 + 6 : 16] == 0).all()
+
+
+def test_pad8_direct_train_batch(monkeypatch):
+    """Direct train_batch/eval_batch calls with LOGICAL-width tensors
+    (profiling harnesses, graph capture) are padded defensively —
+    fit() pads the whole series, but direct callers may not
+    (regression: 'lstm_seq_fwd_fused geometry unsupported')."""
+    from gordo_amd.engine import pack as packmod
+
+    monkeypatch.setattr(packmod, "pad_enabled", lambda d: True)
+    spec = lstm_spec(n_features=4, H=6, lookback=5)
+    p = LSTMPack(spec, G=1, device="cpu", seeds=[1])
+    X = torch.rand(1, 40, 5, 4)   # [G, B, T, F] logical widths
+    Tb = torch.rand(1, 40, 4)
+    loss = p.train_batch(X, Tb)
+    assert torch.isfinite(loss).all()
+    ev = p.eval_batch(X, Tb)
+    assert torch.isfinite(ev).all()
+
+    dspec = dense_spec(n_features=5, units=(4, 3))
+    dp = packmod.DensePack(dspec, G=2, device="cpu", seeds=[1, 2])
+    Xb = torch.rand(2, 32, 5)
+    loss = dp.train_batch(Xb, Xb.clone())
+    assert torch.isfinite(loss).all()
