@@ -136,10 +136,15 @@ def main():
 
     phase_acc: dict = {}
 
+    # at 8 ranks per node, per-rank thread pools must share the host:
+    # 16 fetch threads x 8 ranks oversubscribes the CPUs during the
+    # data phase (ROADMAP r1 flagged this for the SCALE run)
+    workers = max(4, 16 // max(world, 1))
+
     def one_step(timed: bool = False):
         builder = PackedFleetBuilder(
             my_machines, output_dir=out_dir, model_register_dir=None,
-            device=device, save_models=True,
+            device=device, save_models=True, data_workers=workers,
         )
         results = builder.build_all()
         failed = [n for n, r in results if isinstance(r, BaseException)]

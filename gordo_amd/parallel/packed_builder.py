@@ -132,9 +132,10 @@ class PackedFleetBuilder:
         self.replace_cache = replace_cache
         # async model-save pool: serialization of a finished group
         # overlaps the next group's GPU fits (joined in build_all)
-        self._save_pool = concurrent.futures.ThreadPoolExecutor(
-            max(8, min(16, (os.cpu_count() or 8)))
-        )
+        # share the host across ranks: WORLD_SIZE ranks each run a pool
+        world = int(os.environ.get("WORLD_SIZE", "1") or 1)
+        pool_n = max(4, min(16, (os.cpu_count() or 8) // max(world, 1)))
+        self._save_pool = concurrent.futures.ThreadPoolExecutor(pool_n)
         self._save_futures: List[Tuple[MachinePlan, Any]] = []
         # wall-clock phase ledger for the build-step budget table
         # (BASELINE.md): sequential wall segments; overlapped work
