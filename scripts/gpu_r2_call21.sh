@@ -1,12 +1,13 @@
 #!/bin/bash
+# Round-2 call 21 (final): full GPU suite + smoke + bench on the final
+# tree (defensive _pad_io, fold caps, init-cache widening, key hoisting).
 set -x
 cd "$GRAFT_REPO_ROOT" || cd /root/repo
 mkdir -p gpurun_out
-export TMPDIR=/tmp
-( cd /tmp && PROF_G=31 PROF_B=256 PROF_REPS=2 timeout 600 rocprofv3 \
-  --pmc SQ_WAIT_ANY SQ_WAIT_INST_ANY SQ_ACTIVE_INST_ANY SQ_WAVE_CYCLES \
-  -d "$GRAFT_REPO_ROOT/gpurun_out/pmc_r2b" -- \
-  python "$GRAFT_REPO_ROOT/scripts/prof_lstm.py" \
-  > "$GRAFT_REPO_ROOT/gpurun_out/pmc_r2b.log" 2>&1 )
-tail -2 gpurun_out/pmc_r2b.log
-find gpurun_out/pmc_r2b -name "*.db" | head -1
+exec > >(tee gpurun_out/r2_call21.log) 2>&1
+
+timeout 560 python -m pytest tests -m gpu -q 2>&1 | tail -3
+
+timeout 120 python -c "import __graft_entry__ as g; g.smoke(); print('smoke OK')"
+
+timeout 240 python bench.py --gpus 1 --steps 3 --warmup 1 2>&1 | tail -1
