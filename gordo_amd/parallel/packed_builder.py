@@ -153,27 +153,52 @@ class PackedFleetBuilder:
         t_all0 = time.time()
         plans = [MachinePlan(machine=m) for m in self.machines]
         t0 = time.time()
-        self._fetch_data(plans)
-        self._phase("fetch", time.time() - t0)
-        t0 = time.time()
         self._instantiate_models(plans)
         self._phase("instantiate", time.time() - t0)
 
+        # Fetches run on a thread pool in the BACKGROUND; each
+        # prospective group's futures are joined only when that group is
+        # about to build, so the data fetch of group g+1 overlaps the
+        # GPU fits of group g (pandas resample/join releases the GIL).
+        # GORDO_PREFETCH=0 restores the blocking up-front fetch.
+        prefetch = os.environ.get("GORDO_PREFETCH", "1") != "0"
         packable = [p for p in plans if p.packable and p.error is None]
+        # fallback plans go through ModelBuilder.build, which fetches its
+        # own data — no builder-side fetch for them.
         fallback = [p for p in plans if not p.packable and p.error is None]
 
-        # NOTE: running independent groups on concurrent threads/streams
-        # was measured SLOWER (28.3k vs 33.4k machines/hour): the extra
-        # python threads contend on the GIL with the fold-fit threads
-        # and starve kernel dispatch. Groups run sequentially; only the
-        # CV folds within a group overlap (streams, _fit_folds).
-        for group in self._group(packable):
-            try:
-                self._build_group(group)
-            except Exception as e:  # isolate group failures
-                logger.exception("Pack group build failed")
-                for p in group:
-                    p.error = e
+        fetch_pool = concurrent.futures.ThreadPoolExecutor(self.data_workers)
+        try:
+            futs = {
+                id(p): fetch_pool.submit(self._fetch_one, p)
+                for p in packable
+            }
+            if not prefetch:
+                t0 = time.time()
+                for f in futs.values():
+                    f.result()
+                self._phase("fetch", time.time() - t0)
+
+            # NOTE: running independent groups on concurrent threads/streams
+            # was measured SLOWER (28.3k vs 33.4k machines/hour): the extra
+            # python threads contend on the GIL with the fold-fit threads
+            # and starve kernel dispatch. Groups run sequentially; only the
+            # CV folds within a group overlap (streams, _fit_folds).
+            for pre_group in self._pre_group(packable):
+                t0 = time.time()
+                for p in pre_group:
+                    futs[id(p)].result()
+                self._phase("fetch", time.time() - t0)
+                ready = [p for p in pre_group if p.error is None]
+                for group in self._group(ready):
+                    try:
+                        self._build_group(group)
+                    except Exception as e:  # isolate group failures
+                        logger.exception("Pack group build failed")
+                        for p in group:
+                            p.error = e
+        finally:
+            fetch_pool.shutdown(wait=True)
 
         t0 = time.time()
         for p in fallback:
@@ -224,19 +249,44 @@ class PackedFleetBuilder:
         return results
 
     # ---- stages ----------------------------------------------------------
-    def _fetch_data(self, plans: List[MachinePlan]):
-        def fetch(p: MachinePlan):
-            try:
-                start = time.time()
-                dataset = GordoBaseDataset.from_dict(p.machine.dataset.to_dict())
-                p.X, p.y = dataset.get_data()
-                p.query_duration = time.time() - start
-                p.dataset_meta = dataset.get_metadata()
-            except BaseException as e:
-                p.error = e
+    @staticmethod
+    def _fetch_one(p: MachinePlan):
+        try:
+            start = time.time()
+            dataset = GordoBaseDataset.from_dict(p.machine.dataset.to_dict())
+            p.X, p.y = dataset.get_data()
+            p.query_duration = time.time() - start
+            p.dataset_meta = dataset.get_metadata()
+        except BaseException as e:
+            p.error = e
 
-        with concurrent.futures.ThreadPoolExecutor(self.data_workers) as ex:
-            list(ex.map(fetch, plans))
+    @staticmethod
+    def _pre_group(plans: List[MachinePlan]) -> List[List[MachinePlan]]:
+        """Config-level grouping done BEFORE any data is fetched, so each
+        prospective group's fetch futures can be joined lazily.
+
+        Key = model definition + evaluation + dataset shape-affecting
+        fields (tag COUNTS, date range, resolution, filters — not tag
+        names). Machines whose fetched data still differs in row count
+        or width are split afterwards by `_group`, which keys on the
+        actual arrays."""
+        groups: Dict[str, List[MachinePlan]] = {}
+        for p in plans:
+            d = dict(p.machine.dataset.to_dict())
+            for k in ("tags", "tag_list", "target_tag_list"):
+                if isinstance(d.get(k), (list, tuple)):
+                    d[k] = len(d[k])
+            key = json.dumps(
+                {
+                    "model": p.machine.model,
+                    "evaluation": p.machine.evaluation,
+                    "dataset": d,
+                },
+                sort_keys=True,
+                default=str,
+            )
+            groups.setdefault(key, []).append(p)
+        return list(groups.values())
 
     def _instantiate_models(self, plans: List[MachinePlan]):
         for p in plans:

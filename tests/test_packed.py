@@ -619,3 +619,64 @@ def test_pad8_direct_train_batch(monkeypatch):
     Xb = torch.rand(2, 32, 5)
     loss = dp.train_batch(Xb, Xb.clone())
     assert torch.isfinite(loss).all()
+
+
+def test_prefetch_pre_grouping_and_fetch_failure_isolation(tmp_path):
+    """The lazy-fetch path: machines pre-group by config (tag COUNTS,
+    not names), and a fetch failure in one machine surfaces as that
+    machine's error without breaking its group peers."""
+    from gordo_amd.parallel import PackedFleetBuilder
+    from gordo_amd.parallel.packed_builder import MachinePlan
+    from gordo_amd.workflow import NormalizedConfig
+
+    def machine(name, tags=4, hours=48):
+        return {
+            "name": name,
+            "dataset": {
+                "type": "SineWaveDataset",
+                "tag_list": [f"{name}-t{j}" for j in range(tags)],
+                "train_start_date": "2019-01-01T00:00:00+00:00",
+                "train_end_date": f"2019-01-01T00:00:00+00:00".replace(
+                    "01T00", f"0{1 + hours // 24}T00"
+                ),
+            },
+            "model": {
+                "gordo_amd.machine.model.models.KerasAutoEncoder": {
+                    "kind": "feedforward_hourglass",
+                    "epochs": 1,
+                }
+            },
+            "evaluation": {"cv_mode": "full_build"},
+        }
+
+    cfg = {
+        "machines": [machine(f"a-{i}") for i in range(3)]
+        + [machine("b-0", hours=24)]  # different date range → own pre-group
+    }
+    norm = NormalizedConfig(cfg, project_name="p")
+    fb = PackedFleetBuilder(norm.machines, save_models=False)
+    plans = [MachinePlan(machine=m) for m in norm.machines]
+    fb._instantiate_models(plans)
+    pre = fb._pre_group([p for p in plans if p.packable])
+    # per-machine tag NAMES must not split the a-* machines apart
+    assert sorted(len(g) for g in pre) == [1, 3]
+
+    # fetch failure isolation: poison one machine's fetch
+    orig = PackedFleetBuilder._fetch_one
+
+    def poisoned(p):
+        if p.machine.name == "a-1":
+            p.error = RuntimeError("store unreachable")
+            return
+        orig(p)
+
+    PackedFleetBuilder._fetch_one = staticmethod(poisoned)
+    try:
+        results = dict(fb.build_all())
+    finally:
+        PackedFleetBuilder._fetch_one = staticmethod(orig)
+    assert isinstance(results["a-1"], RuntimeError)
+    for name in ("a-0", "a-2", "b-0"):
+        assert not isinstance(results[name], BaseException), name
+        cv = results[name].metadata.build_metadata.model.cross_validation
+        assert cv.scores
