@@ -184,13 +184,21 @@ class PackedFleetBuilder:
             # python threads contend on the GIL with the fold-fit threads
             # and starve kernel dispatch. Groups run sequentially; only the
             # CV folds within a group overlap (streams, _fit_folds).
-            for pre_group in self._pre_group(packable):
+            # Largest pre-groups first: the final group's async
+            # adopt+save is the only one the end-of-build save_join
+            # cannot hide, so leave the smallest for last. Ties keep
+            # config order (sort is stable).
+            pre_groups = sorted(
+                self._pre_group(packable), key=len, reverse=True
+            )
+            for pre_group in pre_groups:
                 t0 = time.time()
                 for p in pre_group:
                     futs[id(p)].result()
                 self._phase("fetch", time.time() - t0)
                 ready = [p for p in pre_group if p.error is None]
-                for group in self._group(ready):
+                for group in sorted(self._group(ready), key=len,
+                                    reverse=True):
                     try:
                         self._build_group(group)
                     except Exception as e:  # isolate group failures
