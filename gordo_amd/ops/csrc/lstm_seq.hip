@@ -1399,7 +1399,15 @@ __global__ __launch_bounds__(256) void lstm_seq_fwd_v4_kernel(
   }
   for (int i = tid; i < ldg; i += 256)
     bS[i] = (i < H4) ? bias[(size_t)g * H4 + i] : lf2bf(0.f);
-  for (int i = tid; i < ROWS * LDKc; i += 256) hxS[i] = lf2bf(0.f);
+  // zero the h section and the K-tail padding ONLY: the x section
+  // [H, H+F) is written concurrently by the x_0 staging loop below
+  // (different threads own the same address in the two loops, so
+  // zeroing it here would race with that staging — seen as corrupted
+  // t=0 gates under co-residency before this guard)
+  for (int i = tid; i < ROWS * LDKc; i += 256) {
+    int k = i % LDKc;
+    if (k < H || k >= H + F) hxS[i] = lf2bf(0.f);
+  }
   for (int i = tid; i < ROWS * H; i += 256) cS[i] = 0.f;
   // x_0 into the K-buffer's x section (16-byte lanes: F % 8 == 0)
   for (int i = tid; i < ROWS * (F / 8); i += 256) {
