@@ -750,3 +750,41 @@ def test_lstm_seq_v4_fused_deterministic(rows, monkeypatch):
             assert torch.equal(a, c)
         (hi,) = ops.lstm_seq_fwd_fused(x, Wx, Wh, b, store_aux=False)
         assert torch.equal(hi, ref[0])
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize(
+    "name,env,G,B,H",
+    [
+        ("v1", {"GORDO_LSTM_V1": "1"}, 2, 64, 48),
+        ("v3", {}, 2, 64, 42),                  # default pipelined path
+        ("v2", {}, 40, 512, 48),                # H%16==0 + filled grid
+        ("big", {}, 2, 64, 128),                # streamed-Wh H>64 path
+    ],
+)
+def test_lstm_seq_scan_deterministic(name, env, G, B, H, monkeypatch):
+    """Every scan dispatch path (v1/v2/v3/big-H) is bitwise
+    deterministic across repeated forward AND backward launches —
+    companion to the v4 determinism regression."""
+    require_hip()
+    for k, v in env.items():
+        monkeypatch.setenv(k, v)
+    T, H4 = 16, 4 * H
+    g = torch.Generator().manual_seed(7)
+    xW = (torch.randn(G, B, T, H4, generator=g) * 0.3).to(
+        "cuda", torch.bfloat16
+    )
+    Wh = (torch.randn(G, H, H4, generator=g) * 0.2).to(
+        "cuda", torch.bfloat16
+    )
+    hs, cs, ga = ops.lstm_seq_fwd(xW, Wh)
+    dSeq = (torch.randn(G, B, T, H, generator=g) * 0.1).to(
+        "cuda", torch.bfloat16
+    )
+    dref = ops.lstm_seq_bwd(dSeq, ga, cs, Wh, last_only=False)
+    for _ in range(10):
+        out = ops.lstm_seq_fwd(xW, Wh)
+        for a, c in zip((hs, cs, ga), out):
+            assert torch.equal(a, c), name
+        d = ops.lstm_seq_bwd(dSeq, ga, cs, Wh, last_only=False)
+        assert torch.equal(d, dref), name
