@@ -680,3 +680,71 @@ def test_prefetch_pre_grouping_and_fetch_failure_isolation(tmp_path):
         assert not isinstance(results[name], BaseException), name
         cv = results[name].metadata.build_metadata.model.cross_validation
         assert cv.scores
+
+
+def test_packed_builder_cache_resume(tmp_path):
+    """Fleet crash-resume semantics (reference disk_registry cache,
+    build_model.py:634 check_cache): a second build with the same
+    register dir must LOAD every machine from cache — proven by
+    poisoning the training entry so any actual fit would raise."""
+    from gordo_amd.parallel import PackedFleetBuilder
+    from gordo_amd.parallel import packed_builder as pb
+    from gordo_amd.workflow import NormalizedConfig
+
+    def machine(name):
+        return {
+            "name": name,
+            "dataset": {
+                "type": "SineWaveDataset",
+                "tag_list": [f"{name}-t{j}" for j in range(4)],
+                "train_start_date": "2019-01-01T00:00:00+00:00",
+                "train_end_date": "2019-01-02T00:00:00+00:00",
+            },
+            "model": {
+                "gordo_amd.machine.model.models.KerasAutoEncoder": {
+                    "kind": "feedforward_hourglass",
+                    "epochs": 1,
+                }
+            },
+            "evaluation": {"cv_mode": "full_build"},
+        }
+
+    cfg = {"machines": [machine(f"m-{i}") for i in range(3)]}
+    out_dir = str(tmp_path / "models")
+    reg_dir = str(tmp_path / "registry")
+    norm = NormalizedConfig(cfg, project_name="p")
+    fb = PackedFleetBuilder(
+        norm.machines, output_dir=out_dir, model_register_dir=reg_dir
+    )
+    first = dict(fb.build_all())
+    assert all(not isinstance(v, BaseException) for v in first.values())
+
+    # rerun: every group must resolve from the registry without a
+    # single pack being constructed
+    def boom(*a, **k):
+        raise AssertionError("cache miss: pack constructed on resume")
+
+    norm2 = NormalizedConfig(cfg, project_name="p")
+    fb2 = PackedFleetBuilder(
+        norm2.machines, output_dir=out_dir, model_register_dir=reg_dir
+    )
+    orig = fb2._make_pack
+    fb2._make_pack = boom
+    try:
+        second = dict(fb2.build_all())
+    finally:
+        fb2._make_pack = orig
+    for name, m in second.items():
+        assert not isinstance(m, BaseException), (name, m)
+        cv1 = first[name].metadata.build_metadata.model.cross_validation
+        cv2 = m.metadata.build_metadata.model.cross_validation
+        assert cv2.scores == cv1.scores
+
+    # replace_cache forces a retrain (the poison must now trip)
+    fb3 = PackedFleetBuilder(
+        norm2.machines, output_dir=out_dir, model_register_dir=reg_dir,
+        replace_cache=True,
+    )
+    fb3._make_pack = boom
+    third = dict(fb3.build_all())
+    assert all(isinstance(v, AssertionError) for v in third.values())
